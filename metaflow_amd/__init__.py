@@ -1,0 +1,70 @@
+"""metaflow_amd — an MI355X-native ML workflow engine.
+
+Re-implements the capabilities of Netflix/metaflow (FlowSpec/@step API,
+AST-derived DAGs with foreach/join/switch, content-addressed artifact
+datastore, local subprocess runtime with gang-scheduled @parallel steps,
+resume/clone recovery, Flow/Run/Step/Task client) designed MI355X-first:
+the gang scheduler pins ranks onto GPUs with RCCL over xGMI, and the
+training hot path (metaflow_amd.ops) is hand-written CDNA4 HIP.
+"""
+
+from .flowspec import FlowSpec
+from .decorators import step
+from .parameters import Parameter, JSONType
+from .includefile import IncludeFile
+from .current import current
+from .unbounded_foreach import UnboundedForeachInput
+from .plugins.retry_decorator import retry
+from .plugins.catch_decorator import catch
+from .plugins.timeout_decorator import timeout
+from .plugins.resources_decorator import resources
+from .plugins.environment_decorator import environment
+from .plugins.parallel_decorator import parallel, torch_parallel
+from .plugins.checkpoint_decorator import checkpoint
+from .plugins.project_decorator import project
+from .plugins.schedule_decorator import schedule
+from .client import (
+    Metaflow,
+    Flow,
+    Run,
+    Step,
+    Task,
+    DataArtifact,
+    namespace,
+    get_namespace,
+)
+from .multicore_utils import parallel_map, parallel_imap_unordered
+
+__version__ = "0.1.0"
+
+S3 = None  # cloud datatools are not part of the single-node build (yet)
+
+__all__ = [
+    "FlowSpec",
+    "step",
+    "Parameter",
+    "JSONType",
+    "IncludeFile",
+    "current",
+    "UnboundedForeachInput",
+    "retry",
+    "catch",
+    "timeout",
+    "resources",
+    "environment",
+    "parallel",
+    "torch_parallel",
+    "checkpoint",
+    "project",
+    "schedule",
+    "Metaflow",
+    "Flow",
+    "Run",
+    "Step",
+    "Task",
+    "DataArtifact",
+    "namespace",
+    "get_namespace",
+    "parallel_map",
+    "parallel_imap_unordered",
+]

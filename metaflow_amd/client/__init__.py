@@ -1,0 +1,353 @@
+"""Client API: read-only object tree Metaflow -> Flow -> Run -> Step -> Task
+-> DataArtifact with namespace filtering.
+
+Parity target: /root/reference/metaflow/client/core.py:227-2765. Backed
+directly by the local datastore + metadata files.
+"""
+
+import os
+
+from ..config import DATASTORE_LOCAL_DIR
+from ..datastore import FlowDataStore
+from ..datastore.storage import LocalStorage
+from ..exceptions import NamespaceMismatchError, NotFoundError
+from ..metadata.local import LocalMetadataProvider, _username
+
+_current_namespace = "user:%s" % _username()
+
+
+def namespace(ns):
+    """Set the active namespace; None disables filtering. Returns it."""
+    global _current_namespace
+    _current_namespace = ns
+    return ns
+
+
+def get_namespace():
+    return _current_namespace
+
+
+def default_namespace():
+    return namespace("user:%s" % _username())
+
+
+def _datastore_root():
+    root = os.environ.get("MFX_DATASTORE_SYSROOT_LOCAL")
+    if root:
+        return root
+    return LocalStorage.get_datastore_root_from_config(create_on_absent=False)
+
+
+def _storage():
+    return LocalStorage(_datastore_root())
+
+
+def _check_namespace(tags):
+    if _current_namespace is None:
+        return True
+    return _current_namespace in (tags or [])
+
+
+class MetaflowObject(object):
+    def __repr__(self):
+        return "%s('%s')" % (type(self).__name__, self.pathspec)
+
+
+class Metaflow(MetaflowObject):
+    """Entry point: all flows in the datastore."""
+
+    @property
+    def flows(self):
+        return list(self)
+
+    def __iter__(self):
+        storage = _storage()
+        root = storage.root
+        if root is None or not os.path.isdir(root):
+            return
+        for name in sorted(os.listdir(root)):
+            if name.startswith(".") or name.startswith("_"):
+                continue
+            if os.path.isdir(os.path.join(root, name)):
+                yield Flow(name)
+
+    def __repr__(self):
+        return "Metaflow()"
+
+
+class Flow(MetaflowObject):
+    def __init__(self, name):
+        self.id = name
+        self.pathspec = name
+        self._storage = _storage()
+        self._fds = FlowDataStore(name, self._storage)
+        self._meta = LocalMetadataProvider(name, self._storage)
+
+    @property
+    def latest_run(self):
+        for run in self:
+            return run
+        return None
+
+    @property
+    def latest_successful_run(self):
+        for run in self:
+            if run.successful:
+                return run
+        return None
+
+    def runs(self, *tags):
+        for run in self:
+            if all(t in run.tags for t in tags):
+                yield run
+
+    def __iter__(self):
+        for info in self._meta.list_runs():
+            tags = _full_tags(info)
+            if _check_namespace(tags):
+                yield Run("%s/%s" % (self.id, info["run_id"]), _info=info)
+
+    def __getitem__(self, run_id):
+        return Run("%s/%s" % (self.id, run_id))
+
+
+def _full_tags(info):
+    tags = list(info.get("tags", []))
+    user = info.get("user")
+    if user:
+        tags.append("user:%s" % user)
+    return tags
+
+
+class Run(MetaflowObject):
+    def __init__(self, pathspec, _info=None):
+        parts = pathspec.split("/")
+        if len(parts) != 2:
+            raise NotFoundError("Run pathspec must be 'Flow/run_id'.")
+        self.flow_name, self.id = parts
+        self.pathspec = pathspec
+        self._storage = _storage()
+        self._fds = FlowDataStore(self.flow_name, self._storage)
+        self._meta = LocalMetadataProvider(self.flow_name, self._storage)
+        self._info = _info or self._meta.get_run(self.id)
+        if self._info is None:
+            raise NotFoundError("Run %s not found." % pathspec)
+        if not _check_namespace(_full_tags(self._info)):
+            raise NamespaceMismatchError(_current_namespace)
+
+    @property
+    def tags(self):
+        return _full_tags(self._info)
+
+    @property
+    def successful(self):
+        info = self._meta.get_run(self.id) or {}
+        if info.get("status") == "successful":
+            return True
+        # fall back to the end task's DONE marker
+        try:
+            return self["end"].task.successful
+        except Exception:
+            return False
+
+    @property
+    def finished(self):
+        info = self._meta.get_run(self.id) or {}
+        return info.get("status") in ("successful", "failed")
+
+    @property
+    def data(self):
+        end = self["end"].task
+        return end.data if end else None
+
+    def steps(self):
+        return list(self)
+
+    def __iter__(self):
+        for step_name in self._fds.list_steps(self.id):
+            if step_name.startswith("_"):
+                continue
+            yield Step("%s/%s" % (self.pathspec, step_name))
+
+    def __getitem__(self, step_name):
+        steps = self._fds.list_steps(self.id)
+        if step_name not in steps:
+            raise NotFoundError("Step %s/%s not found."
+                                % (self.pathspec, step_name))
+        return Step("%s/%s" % (self.pathspec, step_name))
+
+    def add_tags(self, tags):
+        self._meta.add_run_tags(self.id, tags)
+        self._info = self._meta.get_run(self.id)
+
+    def remove_tags(self, tags):
+        self._meta.remove_run_tags(self.id, tags)
+        self._info = self._meta.get_run(self.id)
+
+    @property
+    def end_task(self):
+        try:
+            return self["end"].task
+        except NotFoundError:
+            return None
+
+
+class Step(MetaflowObject):
+    def __init__(self, pathspec):
+        parts = pathspec.split("/")
+        if len(parts) != 3:
+            raise NotFoundError("Step pathspec must be 'Flow/run/step'.")
+        self.flow_name, self.run_id, self.id = parts
+        self.pathspec = pathspec
+        self._storage = _storage()
+        self._fds = FlowDataStore(self.flow_name, self._storage)
+
+    @property
+    def task(self):
+        """The first (often only) task."""
+        for t in self:
+            return t
+        return None
+
+    def tasks(self):
+        return list(self)
+
+    def __iter__(self):
+        task_ids = self._fds.list_tasks(self.run_id, self.id)
+
+        def sort_key(t):
+            main = t.split("_node_")[0]
+            node = t.split("_node_")[1] if "_node_" in t else "0"
+            try:
+                return (int(main), int(node))
+            except ValueError:
+                return (1 << 30, 0)
+
+        for task_id in sorted(task_ids, key=sort_key):
+            yield Task("%s/%s" % (self.pathspec, task_id))
+
+    def __getitem__(self, task_id):
+        return Task("%s/%s" % (self.pathspec, task_id))
+
+    @property
+    def parent(self):
+        return Run("%s/%s" % (self.flow_name, self.run_id))
+
+    @property
+    def control_task(self):
+        for t in self:
+            if "_node_" not in t.id:
+                return t
+        return None
+
+
+class Task(MetaflowObject):
+    def __init__(self, pathspec):
+        parts = pathspec.split("/")
+        if len(parts) != 4:
+            raise NotFoundError("Task pathspec must be 'Flow/run/step/task'.")
+        self.flow_name, self.run_id, self.step_name, self.id = parts
+        self.pathspec = pathspec
+        self._storage = _storage()
+        self._fds = FlowDataStore(self.flow_name, self._storage)
+        self._ds = self._fds.get_task_datastore(self.run_id, self.step_name,
+                                                self.id)
+
+    @property
+    def successful(self):
+        if self._ds.attempt is None:
+            return False
+        ok = self._ds.load_metadata("attempt_ok") or {}
+        return bool(ok.get("ok"))
+
+    @property
+    def finished(self):
+        return self._ds.attempt is not None
+
+    @property
+    def exception(self):
+        return self._ds.get("_exception")
+
+    @property
+    def data(self):
+        return MetaflowData(self._ds)
+
+    @property
+    def artifacts(self):
+        return MetaflowData(self._ds)
+
+    @property
+    def index(self):
+        frames = self._ds.load_metadata("foreach_stack") or []
+        return frames[-1][3] if frames else None
+
+    @property
+    def foreach_stack(self):
+        return self._ds.load_metadata("foreach_stack") or []
+
+    @property
+    def stdout(self):
+        return self._ds.load_logs("stdout")
+
+    @property
+    def stderr(self):
+        return self._ds.load_logs("stderr")
+
+    @property
+    def parent(self):
+        return Step("%s/%s/%s" % (self.flow_name, self.run_id,
+                                  self.step_name))
+
+    def __iter__(self):
+        for name in self._ds.artifact_names():
+            if not name.startswith("_"):
+                yield DataArtifact(self, name)
+
+    def __getitem__(self, name):
+        if name not in self._ds:
+            raise NotFoundError("Artifact %s/%s not found."
+                                % (self.pathspec, name))
+        return DataArtifact(self, name)
+
+
+class MetaflowData(object):
+    """Attribute access to a task's artifacts (task.data.model)."""
+
+    def __init__(self, task_ds):
+        object.__setattr__(self, "_ds", task_ds)
+
+    def __getattr__(self, name):
+        ds = object.__getattribute__(self, "_ds")
+        if name in ds:
+            return ds[name]
+        raise AttributeError(name)
+
+    def __contains__(self, name):
+        return name in object.__getattribute__(self, "_ds")
+
+    def _artifacts(self):
+        return object.__getattribute__(self, "_ds").artifact_names()
+
+    def __repr__(self):
+        names = [n for n in self._artifacts() if not n.startswith("_")]
+        return "<MetaflowData: %s>" % ", ".join(sorted(names))
+
+
+class DataArtifact(MetaflowObject):
+    def __init__(self, task, name):
+        self._task = task
+        self.id = name
+        self.pathspec = "%s/%s" % (task.pathspec, name)
+
+    @property
+    def data(self):
+        return self._task._ds[self.id]
+
+    @property
+    def sha(self):
+        return self._task._ds.artifact_sha(self.id)
+
+    @property
+    def size(self):
+        info = self._task._ds.artifact_info(self.id) or {}
+        return info.get("size")

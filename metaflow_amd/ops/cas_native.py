@@ -1,0 +1,24 @@
+"""Loader for the native C++ CAS engine (_mfx_cas extension).
+
+Returns None-ish (raises) when the extension isn't built; cas.py treats any
+failure here as "use the pure-Python path". Unlike the GPU kernels, the CAS
+engine is an optional accelerator — correctness is identical either way.
+"""
+
+_engine = None
+_tried = False
+
+
+def engine():
+    global _engine, _tried
+    if not _tried:
+        _tried = True
+        try:
+            from . import _mfx_cas  # built in-tree by setup.py
+
+            _engine = _mfx_cas.Engine()
+        except Exception:
+            _engine = None
+    if _engine is None:
+        raise ImportError("_mfx_cas extension not built")
+    return _engine

@@ -1,0 +1,34 @@
+"""Plugin registry.
+
+Parity target: /root/reference/metaflow/plugins/__init__.py:11-195 —
+declarative name -> class lists for step decorators so `--with name` and
+static `@name` both resolve here.
+"""
+
+from .retry_decorator import RetryDecorator
+from .catch_decorator import CatchDecorator
+from .timeout_decorator import TimeoutDecorator
+from .resources_decorator import ResourcesDecorator
+from .environment_decorator import EnvironmentDecorator
+from .parallel_decorator import ParallelDecorator, TorchParallelDecorator
+from .checkpoint_decorator import CheckpointDecorator
+from .project_decorator import ProjectDecorator
+from .schedule_decorator import ScheduleDecorator
+
+STEP_DECORATORS = {
+    cls.name: cls
+    for cls in (
+        RetryDecorator,
+        CatchDecorator,
+        TimeoutDecorator,
+        ResourcesDecorator,
+        EnvironmentDecorator,
+        ParallelDecorator,
+        TorchParallelDecorator,
+        CheckpointDecorator,
+    )
+}
+
+FLOW_DECORATORS = {
+    cls.name: cls for cls in (ProjectDecorator, ScheduleDecorator)
+}

@@ -1,0 +1,631 @@
+"""Local runtime: single-process event-loop scheduler.
+
+Parity target: /root/reference/metaflow/runtime.py (NativeRuntime :352,
+execute :794, _queue_tasks :1390, join gating :1163, foreach fan-out :1332,
+Worker :2238, CLIArgs :2094, TruncatedBuffer :2066). Fresh design:
+
+* every task is a subprocess re-invoking the flow file's CLI
+  (``python flow.py step <name> --run-id ... --task-id ...``);
+* the scheduler is a selectors-based poll loop over child stdout/stderr;
+* join readiness is tracked by (step, foreach-stack) keys where a stack is
+  a tuple of (split_step, index, num_splits) frames;
+* ``num_parallel`` gang steps are launched natively: N rank processes with
+  MFX_PARALLEL_* rendezvous env and per-rank GPU pinning
+  (HIP_VISIBLE_DEVICES) — the MI355X-first replacement for the reference's
+  control/worker UBF emulation;
+* resume = clone-at-queue-time: each task whose origin counterpart
+  succeeded is cloned (metadata-only, CAS dedup) instead of launched.
+"""
+
+import os
+import selectors
+import shlex
+import signal
+import socket
+import subprocess
+import sys
+import time
+from collections import deque
+
+from .config import (
+    EXIT_DISALLOW_RETRY,
+    MAX_ATTEMPTS,
+    MAX_LOG_SIZE,
+    MAX_NUM_SPLITS,
+    MAX_WORKERS,
+)
+from .exceptions import MFXException, TaskFailedException
+from .task import PARAMETERS_STEP, PARAMETERS_TASK_ID, dump_parameters
+from .unbounded_foreach import UBF_CONTROL
+
+
+class TaskSpec(object):
+    __slots__ = ("step", "input_paths", "split_index", "stack", "task_id",
+                 "retry_count", "gang", "env", "ubf_context")
+
+    def __init__(self, step, input_paths, split_index, stack, task_id=None,
+                 retry_count=0, gang=None, env=None, ubf_context=None):
+        self.step = step
+        self.input_paths = input_paths
+        self.split_index = split_index
+        self.stack = stack  # tuple of (split_step, index, num_splits)
+        self.task_id = task_id
+        self.retry_count = retry_count
+        self.gang = gang  # (num_nodes, node_index, control_task_id, port)
+        self.env = env or {}
+        self.ubf_context = ubf_context
+
+    @property
+    def key(self):
+        return (self.step, self.stack)
+
+
+class TruncatedBuffer(object):
+    def __init__(self, maxsize=MAX_LOG_SIZE):
+        self._buf = bytearray()
+        self._maxsize = maxsize
+        self._truncated = False
+
+    def write(self, data):
+        if len(self._buf) < self._maxsize:
+            self._buf.extend(data[: self._maxsize - len(self._buf)])
+        elif not self._truncated:
+            self._truncated = True
+            self._buf.extend(b"\n[log truncated]\n")
+
+    def get_bytes(self):
+        return bytes(self._buf)
+
+
+class Worker(object):
+    def __init__(self, spec, cmd, env, echo_prefix, quiet=False):
+        self.spec = spec
+        self.echo_prefix = echo_prefix
+        self.quiet = quiet
+        self.stdout_buf = TruncatedBuffer()
+        self.stderr_buf = TruncatedBuffer()
+        self._line_buf = {1: b"", 2: b""}
+        full_env = dict(os.environ)
+        full_env.update(env)
+        self.proc = subprocess.Popen(
+            cmd,
+            env=full_env,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
+            bufsize=0,
+        )
+        os.set_blocking(self.proc.stdout.fileno(), False)
+        os.set_blocking(self.proc.stderr.fileno(), False)
+
+    def fds(self):
+        return [self.proc.stdout, self.proc.stderr]
+
+    def read_available(self, fileobj):
+        is_err = fileobj is self.proc.stderr
+        buf = self.stderr_buf if is_err else self.stdout_buf
+        channel = 2 if is_err else 1
+        closed = False
+        try:
+            while True:
+                chunk = os.read(fileobj.fileno(), 65536)
+                if not chunk:
+                    closed = True
+                    break
+                buf.write(chunk)
+                if not self.quiet:
+                    self._echo_lines(channel, chunk)
+        except BlockingIOError:
+            pass
+        return closed
+
+    def _echo_lines(self, channel, chunk):
+        data = self._line_buf[channel] + chunk
+        *lines, rest = data.split(b"\n")
+        self._line_buf[channel] = rest
+        stream = sys.stderr if channel == 2 else sys.stdout
+        for line in lines:
+            stream.write("%s %s\n"
+                         % (self.echo_prefix,
+                            line.decode("utf-8", "replace")))
+        stream.flush()
+
+    def flush_echo(self):
+        for channel in (1, 2):
+            if self._line_buf[channel] and not self.quiet:
+                stream = sys.stderr if channel == 2 else sys.stdout
+                stream.write("%s %s\n" % (
+                    self.echo_prefix,
+                    self._line_buf[channel].decode("utf-8", "replace")))
+                self._line_buf[channel] = b""
+                stream.flush()
+
+    def poll(self):
+        return self.proc.poll()
+
+    def kill(self):
+        try:
+            self.proc.terminate()
+        except OSError:
+            pass
+
+
+class NativeRuntime(object):
+    def __init__(
+        self,
+        flow_cls,
+        graph,
+        flow_datastore,
+        metadata,
+        flow_file,
+        run_id=None,
+        param_values=None,
+        clone_run_id=None,
+        steps_to_rerun=None,
+        max_workers=MAX_WORKERS,
+        max_num_splits=MAX_NUM_SPLITS,
+        quiet=False,
+        top_level_args=None,
+        tags=None,
+    ):
+        self.flow_cls = flow_cls
+        self.graph = graph
+        self.flow_datastore = flow_datastore
+        self.metadata = metadata
+        self.flow_file = flow_file
+        self.param_values = param_values or {}
+        self.clone_run_id = clone_run_id
+        self.steps_to_rerun = set(steps_to_rerun or ())
+        self.max_workers = max_workers
+        self.max_num_splits = max_num_splits
+        self.quiet = quiet
+        self.top_level_args = top_level_args or []
+        self.run_id = run_id or metadata.new_run_id(tags=tags)
+        if run_id:
+            metadata.register_run(run_id, tags=tags,
+                                  origin_run_id=clone_run_id)
+
+        self._task_seq = 0
+        self._run_queue = deque()
+        self._active = {}  # fd -> worker ; plus worker set
+        self._workers = []
+        self._finished = {}  # (step, stack) -> pathspec
+        self._join_arrivals = {}  # (join_step, stack) -> list of pathspecs
+        self._failed = False
+        self._failure_msg = None
+        self._origin_index = None
+        self._selector = selectors.DefaultSelector()
+        self._params_pathspec = None
+
+    # ------------------------------------------------------------- utilities
+    def _new_task_id(self):
+        self._task_seq += 1
+        return str(self._task_seq)
+
+    def _echo(self, msg):
+        if not self.quiet:
+            sys.stdout.write("[mfx] %s\n" % msg)
+            sys.stdout.flush()
+
+    def _pathspec(self, step, task_id):
+        return "%s/%s/%s" % (self.run_id, step, task_id)
+
+    # ----------------------------------------------------------- constants
+    def persist_constants(self):
+        graph_info = self.graph.to_dict()
+        self._params_pathspec = dump_parameters(
+            self.flow_datastore, self.run_id, self.param_values, graph_info)
+
+    # -------------------------------------------------------------- resume
+    def _build_origin_index(self):
+        """(step, stack) -> origin TaskDataStore for all DONE origin tasks."""
+        index = {}
+        origin = self.clone_run_id
+        for step in self.flow_datastore.list_steps(origin):
+            if step == PARAMETERS_STEP:
+                continue
+            for ds in self.flow_datastore.get_done_task_datastores(
+                    origin, step):
+                ok = ds.load_metadata("attempt_ok") or {}
+                if not ok.get("ok"):
+                    continue
+                frames = ds.load_metadata("foreach_stack") or []
+                stack = tuple((f[0], f[3], f[2]) for f in frames)
+                index[(step, stack)] = ds
+        return index
+
+    def _maybe_clone(self, spec):
+        """If the origin run has a successful counterpart, clone it and
+        return its datastore; else None."""
+        if self._origin_index is None:
+            return None
+        if spec.step in self.steps_to_rerun:
+            return None
+        origin_ds = self._origin_index.get(spec.key)
+        if origin_ds is None:
+            return None
+        task_id = spec.task_id or self._new_task_id()
+        spec.task_id = task_id
+        new_ds = self.flow_datastore.get_task_datastore(
+            self.run_id, spec.step, task_id, attempt=0, mode="w")
+        new_ds.clone(origin_ds)
+        self.metadata.register_task(self.run_id, spec.step, task_id, 0,
+                                    {"cloned_from": origin_ds.pathspec})
+        self._echo("Cloned %s from %s" % (self._pathspec(spec.step, task_id),
+                                          origin_ds.pathspec))
+        return new_ds
+
+    # -------------------------------------------------------------- executing
+    def execute(self):
+        self.metadata.heartbeat(self.run_id)
+        if self.clone_run_id:
+            self._origin_index = self._build_origin_index()
+            # clone _parameters from origin (parameters are fixed on resume)
+            origin_params = self.flow_datastore.get_task_datastore(
+                self.clone_run_id, PARAMETERS_STEP, PARAMETERS_TASK_ID)
+            if origin_params.attempt is not None:
+                new_params = self.flow_datastore.get_task_datastore(
+                    self.run_id, PARAMETERS_STEP, PARAMETERS_TASK_ID,
+                    attempt=0, mode="w")
+                new_params.clone(origin_params)
+                self._params_pathspec = "%s/%s/%s" % (
+                    self.run_id, PARAMETERS_STEP, PARAMETERS_TASK_ID)
+        if self._params_pathspec is None:
+            self.persist_constants()
+
+        self._queue_spec(TaskSpec("start", [self._params_pathspec], None,
+                                  ()))
+
+        last_hb = time.time()
+        try:
+            while self._run_queue or self._workers:
+                if self._failed and not self._workers:
+                    break
+                self._launch_ready()
+                self._poll_workers()
+                if time.time() - last_hb > 10:
+                    self.metadata.heartbeat(self.run_id)
+                    last_hb = time.time()
+        except KeyboardInterrupt:
+            self._failed = True
+            self._failure_msg = "interrupted"
+            self._kill_all()
+            raise
+        finally:
+            self.metadata.register_run_done(self.run_id, not self._failed)
+
+        if self._failed:
+            raise TaskFailedException(
+                self._failure_msg or "Flow run failed.")
+        self._echo("Run %s done." % self.run_id)
+
+    # ---------------------------------------------------------------- queueing
+    def _queue_spec(self, spec):
+        if self._failed:
+            return
+        cloned = self._maybe_clone(spec)
+        if cloned is not None:
+            # treat as finished immediately
+            self._task_finished_bookkeeping(spec, cloned)
+            return
+        self._run_queue.append(spec)
+
+    def _queue_tasks_after(self, spec, task_ds):
+        """A task finished OK: schedule its successors."""
+        step = spec.step
+        node = self.graph[step]
+        if node.type == "end":
+            return
+        transition = task_ds.load_metadata("transition")
+        if transition is None:
+            self._fail("Step %s finished without a transition." % step)
+            return
+        pathspec = task_ds.pathspec.split("/", 1)[1]  # strip flow name
+
+        num_parallel = transition.get("num_parallel")
+        foreach = transition.get("foreach")
+        num_splits = transition.get("num_splits")
+        out_funcs = transition["out_funcs"]
+
+        # unbounded foreach: control task
+        if foreach is not None and num_splits is None:
+            target = out_funcs[0]
+            control_stack = spec.stack + ((step, 0, None),)
+            self._queue_spec(TaskSpec(
+                target, [pathspec], 0, control_stack,
+                ubf_context=UBF_CONTROL))
+            return
+
+        if num_parallel is not None:
+            self._queue_gang(spec, pathspec, out_funcs[0], num_parallel)
+            return
+
+        if foreach is not None:
+            if num_splits > self.max_num_splits:
+                self._fail(
+                    "Step %s fans out %d ways; --max-num-splits is %d."
+                    % (step, num_splits, self.max_num_splits))
+                return
+            target = out_funcs[0]
+            for i in range(num_splits):
+                child_stack = spec.stack + ((step, i, num_splits),)
+                self._queue_spec(TaskSpec(target, [pathspec], i,
+                                          child_stack))
+            return
+
+        for target in out_funcs:
+            tnode = self.graph[target]
+            if tnode.type == "join":
+                self._register_join_arrival(spec, pathspec, target, node)
+            else:
+                self._queue_spec(TaskSpec(target, [pathspec], None,
+                                          spec.stack))
+
+    def _register_join_arrival(self, spec, pathspec, join_step, from_node):
+        jnode = self.graph[join_step]
+        split_name = getattr(jnode, "matching_join_of", None)
+        if split_name is None and jnode.split_parents:
+            split_name = jnode.split_parents[-1]
+        split_node = self.graph[split_name] if split_name else None
+
+        if split_node is not None and split_node.type in ("foreach",
+                                                          "split-parallel"):
+            # foreach join: pop the frame; expected = num_splits
+            frame = spec.stack[-1]
+            join_stack = spec.stack[:-1]
+            expected = frame[2]
+            order_key = frame[1]  # index within the foreach
+        else:
+            join_stack = spec.stack
+            expected = len(jnode.in_funcs)
+            order_key = sorted(jnode.in_funcs).index(from_node.name)
+
+        key = (join_step, join_stack)
+        arrivals = self._join_arrivals.setdefault(key, [])
+        arrivals.append((order_key, pathspec))
+        if expected is not None and len(arrivals) >= expected:
+            arrivals.sort()
+            input_paths = [p for _k, p in arrivals]
+            del self._join_arrivals[key]
+            self._queue_spec(TaskSpec(join_step, input_paths, None,
+                                      join_stack))
+
+    def _register_ubf_join(self, spec, task_ds):
+        """Control task finished: gate the join on its mapper tasks."""
+        mapper_paths = task_ds.load_metadata("control_mapper_tasks") or []
+        node = self.graph[spec.step]
+        join_step = None
+        for out in node.out_funcs:
+            if self.graph[out].type == "join":
+                join_step = out
+        if join_step is None:
+            self._fail("UBF control step %s has no join successor."
+                       % spec.step)
+            return
+        join_stack = spec.stack[:-1]
+        input_paths = list(mapper_paths)
+        self._queue_spec(TaskSpec(join_step, input_paths, None, join_stack))
+
+    def _queue_gang(self, spec, pathspec, target, num_parallel):
+        """Native gang scheduling for @parallel steps: N rank processes with
+        rendezvous env + GPU pinning. Replaces the reference's control-task
+        subprocess emulation (parallel_decorator.py:175-246)."""
+        control_id = self._new_task_id()
+        port = _free_port()
+        n_gpus = _visible_gpu_count()
+        for rank in range(num_parallel):
+            task_id = control_id if rank == 0 else "%s_node_%d" % (
+                control_id, rank)
+            env = {
+                "MFX_PARALLEL_MAIN_IP": "127.0.0.1",
+                "MFX_PARALLEL_MAIN_PORT": str(port),
+                "MFX_PARALLEL_NUM_NODES": str(num_parallel),
+                "MFX_PARALLEL_NODE_INDEX": str(rank),
+                "MFX_PARALLEL_CONTROL_TASK_ID": control_id,
+                # torch.distributed env:// rendezvous, one rank per GPU
+                "MASTER_ADDR": "127.0.0.1",
+                "MASTER_PORT": str(port),
+                "RANK": str(rank),
+                "LOCAL_RANK": "0",
+                "WORLD_SIZE": str(num_parallel),
+            }
+            if n_gpus > 0:
+                env["HIP_VISIBLE_DEVICES"] = str(rank % n_gpus)
+                env["CUDA_VISIBLE_DEVICES"] = str(rank % n_gpus)
+            child_stack = spec.stack + ((spec.step, rank, num_parallel),)
+            self._queue_spec(TaskSpec(
+                target, [pathspec], rank, child_stack, task_id=task_id,
+                gang=(num_parallel, rank, control_id, port), env=env))
+
+    # ------------------------------------------------------------- launching
+    def _launch_ready(self):
+        if self._failed:
+            self._run_queue.clear()
+            return
+        while self._run_queue:
+            spec = self._run_queue[0]
+            if len(self._workers) >= self.max_workers and spec.gang is None:
+                break
+            # gang members bypass the cap: a partially-launched gang
+            # deadlocks in rendezvous
+            self._run_queue.popleft()
+            self._launch(spec)
+
+    def _max_retries_for(self, step):
+        func = getattr(self.flow_cls, step)
+        user_retries = 0
+        for deco in getattr(func, "decorators", []):
+            u, _e = deco.step_task_retry_count()
+            user_retries += u
+        return min(user_retries, MAX_ATTEMPTS - 1)
+
+    def _launch(self, spec):
+        if spec.task_id is None:
+            spec.task_id = self._new_task_id()
+        max_retries = self._max_retries_for(spec.step)
+        cmd = [
+            sys.executable, self.flow_file,
+        ] + list(self.top_level_args) + [
+            "step", spec.step,
+            "--run-id", self.run_id,
+            "--task-id", spec.task_id,
+            "--input-paths", ",".join(spec.input_paths),
+            "--retry-count", str(spec.retry_count),
+            "--max-user-code-retries", str(max_retries),
+        ]
+        if spec.split_index is not None:
+            cmd += ["--split-index", str(spec.split_index)]
+        if spec.ubf_context:
+            cmd += ["--ubf-context", spec.ubf_context]
+        if self.clone_run_id:
+            cmd += ["--origin-run-id", self.clone_run_id]
+
+        env = dict(spec.env)
+        # let decorators mutate args/env (e.g. @environment)
+        func = getattr(self.flow_cls, spec.step)
+        args_holder = {"cmd": cmd, "env": env}
+        for deco in getattr(func, "decorators", []):
+            deco.runtime_step_cli(args_holder, spec.retry_count, max_retries,
+                                  spec.ubf_context)
+
+        prefix = "[%s/%s]" % (spec.step, spec.task_id)
+        if os.environ.get("MFX_DEBUG_SUBCOMMAND"):
+            self._echo("exec: %s" % " ".join(shlex.quote(c)
+                                             for c in args_holder["cmd"]))
+        worker = Worker(spec, args_holder["cmd"], args_holder["env"], prefix,
+                        quiet=self.quiet)
+        self._workers.append(worker)
+        for f in worker.fds():
+            self._selector.register(f, selectors.EVENT_READ, worker)
+        self.metadata.register_task(self.run_id, spec.step, spec.task_id,
+                                    spec.retry_count)
+
+    # --------------------------------------------------------------- polling
+    def _poll_workers(self):
+        if not self._workers:
+            return
+        events = self._selector.select(timeout=1.0)
+        for key, _mask in events:
+            worker = key.data
+            worker.read_available(key.fileobj)
+        # reap exited workers
+        still = []
+        for worker in self._workers:
+            rc = worker.poll()
+            if rc is None:
+                still.append(worker)
+                continue
+            # drain remaining output
+            for f in worker.fds():
+                worker.read_available(f)
+                try:
+                    self._selector.unregister(f)
+                except KeyError:
+                    pass
+            worker.flush_echo()
+            self._worker_exited(worker, rc)
+        self._workers = still
+
+    def _worker_exited(self, worker, rc):
+        spec = worker.spec
+        # persist captured logs into the task's attempt
+        try:
+            log_ds = self.flow_datastore.get_task_datastore(
+                self.run_id, spec.step, spec.task_id,
+                attempt=spec.retry_count, mode="w")
+            log_ds.save_logs("stdout", worker.stdout_buf.get_bytes())
+            log_ds.save_logs("stderr", worker.stderr_buf.get_bytes())
+        except Exception:
+            pass
+
+        if rc == 0:
+            task_ds = self.flow_datastore.get_task_datastore(
+                self.run_id, spec.step, spec.task_id)
+            if task_ds.attempt is None:
+                self._fail("Task %s exited 0 but left no DONE marker."
+                           % self._pathspec(spec.step, spec.task_id))
+                return
+            self._task_finished_bookkeeping(spec, task_ds)
+            return
+
+        # failure path
+        if rc == -signal.SIGSEGV:
+            self._echo("Task %s segfaulted."
+                       % self._pathspec(spec.step, spec.task_id))
+        max_retries = self._max_retries_for(spec.step)
+        can_retry = (
+            spec.retry_count < max_retries
+            and rc != EXIT_DISALLOW_RETRY
+            and spec.gang is None
+        )
+        if can_retry:
+            self._echo("Task %s failed (rc=%d); retrying (%d/%d)."
+                       % (self._pathspec(spec.step, spec.task_id), rc,
+                          spec.retry_count + 1, max_retries))
+            self._run_queue.append(TaskSpec(
+                spec.step, spec.input_paths, spec.split_index, spec.stack,
+                task_id=spec.task_id, retry_count=spec.retry_count + 1,
+                gang=spec.gang, env=spec.env,
+                ubf_context=spec.ubf_context))
+        else:
+            self._fail("Task %s failed (rc=%d)."
+                       % (self._pathspec(spec.step, spec.task_id), rc))
+
+    def _task_finished_bookkeeping(self, spec, task_ds):
+        self._finished[spec.key] = task_ds.pathspec
+        if spec.gang is not None and spec.split_index != 0:
+            # non-control gang ranks don't drive transitions; the join
+            # arrival is still registered below
+            pass
+        # UBF control tasks gate joins through _control_mapper_tasks
+        if spec.ubf_context == UBF_CONTROL:
+            self._register_ubf_join(spec, task_ds)
+            return
+        node = self.graph[spec.step]
+        if spec.gang is not None:
+            # every rank's only successor is the gang join
+            transition = task_ds.load_metadata("transition")
+            if transition:
+                for target in transition["out_funcs"]:
+                    if self.graph[target].type == "join":
+                        self._register_join_arrival(
+                            spec, task_ds.pathspec.split("/", 1)[1],
+                            target, node)
+                    elif spec.split_index == 0:
+                        self._queue_spec(TaskSpec(
+                            target, [task_ds.pathspec.split("/", 1)[1]],
+                            None, spec.stack[:-1]))
+            return
+        self._queue_tasks_after(spec, task_ds)
+
+    def _fail(self, msg):
+        self._failed = True
+        if self._failure_msg is None:
+            self._failure_msg = msg
+        self._echo("FAIL: %s" % msg)
+        self._kill_all()
+
+    def _kill_all(self):
+        for worker in self._workers:
+            worker.kill()
+
+
+def _free_port():
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _visible_gpu_count():
+    env = os.environ.get("MFX_NUM_GPUS")
+    if env is not None:
+        return int(env)
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            return torch.cuda.device_count()
+    except Exception:
+        pass
+    return 0

@@ -1,0 +1,162 @@
+"""Datastore unit tests: storage atomicity, CAS dedup, task lifecycle."""
+
+import os
+
+from metaflow_amd.datastore import FlowDataStore
+from metaflow_amd.datastore.cas import (
+    CODEC_GZIP,
+    CODEC_RAW,
+    ContentAddressedStore,
+    pack,
+    unpack,
+)
+from metaflow_amd.datastore.storage import LocalStorage
+
+
+def make_fds(tmp_path, flow="TestFlow"):
+    return FlowDataStore(flow, LocalStorage(str(tmp_path)))
+
+
+def test_local_storage_roundtrip(tmp_path):
+    s = LocalStorage(str(tmp_path))
+    s.save_bytes(iter([("a/b/c", (b"hello", {"k": 1}))]))
+    assert s.is_file(["a/b/c"]) == [True]
+    assert s.is_file(["a/b/missing"]) == [False]
+    [(path, data, meta)] = list(s.load_bytes(["a/b/c"]))
+    assert data == b"hello"
+    assert meta == {"k": 1}
+    assert s.size_file("a/b/c") == 5
+
+
+def test_local_storage_no_overwrite(tmp_path):
+    s = LocalStorage(str(tmp_path))
+    s.save_bytes(iter([("x", (b"one", None))]))
+    s.save_bytes(iter([("x", (b"two", None))]), overwrite=False)
+    [(_, data, _)] = list(s.load_bytes(["x"]))
+    assert data == b"one"
+    s.save_bytes(iter([("x", (b"two", None))]), overwrite=True)
+    [(_, data, _)] = list(s.load_bytes(["x"]))
+    assert data == b"two"
+
+
+def test_pack_unpack():
+    data = b"payload" * 100
+    assert unpack(pack(data, CODEC_RAW)) == data
+    assert unpack(pack(data, CODEC_GZIP)) == data
+    assert len(pack(data, CODEC_GZIP)) < len(data)
+
+
+def test_cas_dedup(tmp_path):
+    s = LocalStorage(str(tmp_path))
+    cas = ContentAddressedStore("data", s)
+    blobs = [b"aaa", b"bbb", b"aaa"]
+    results = cas.save_blobs(blobs)
+    assert results[0][1] == results[2][1]  # same content -> same key
+    loaded = dict(cas.load_blobs([k for _u, k in results]))
+    assert loaded[results[0][1]] == b"aaa"
+    assert loaded[results[1][1]] == b"bbb"
+    # only two physical files
+    files = []
+    for dirpath, _dirs, names in os.walk(str(tmp_path)):
+        files += [n for n in names if not n.endswith("_meta")]
+    assert len(files) == 2
+
+
+def test_cas_large_blob_raw(tmp_path):
+    from metaflow_amd import config
+
+    s = LocalStorage(str(tmp_path))
+    cas = ContentAddressedStore("data", s)
+    big = os.urandom(config.CAS_COMPRESS_MAX_SIZE + 1)
+    [(_u, key)] = cas.save_blobs([big])
+    [(k, data)] = list(cas.load_blobs([key]))
+    assert data == big
+
+
+def test_task_datastore_lifecycle(tmp_path):
+    fds = make_fds(tmp_path)
+    ds = fds.get_task_datastore("1", "start", "1", attempt=0, mode="w")
+    ds.init_task()
+    ds.save_artifacts([("x", 42), ("y", [1, 2, 3])])
+    assert ds.latest_done_attempt() is None
+    ds.done()
+    assert ds.latest_done_attempt() == 0
+
+    rd = fds.get_task_datastore("1", "start", "1")
+    assert rd.attempt == 0
+    assert "x" in rd
+    assert rd["x"] == 42
+    assert rd["y"] == [1, 2, 3]
+    assert rd.artifact_sha("x") is not None
+
+
+def test_task_datastore_attempts(tmp_path):
+    fds = make_fds(tmp_path)
+    a0 = fds.get_task_datastore("1", "s", "1", attempt=0, mode="w")
+    a0.init_task()
+    a0.save_artifacts([("v", "first")])
+    a0.done()
+    a1 = fds.get_task_datastore("1", "s", "1", attempt=1, mode="w")
+    a1.init_task()
+    a1.save_artifacts([("v", "second")])
+    a1.done()
+    rd = fds.get_task_datastore("1", "s", "1")
+    assert rd.attempt == 1
+    assert rd["v"] == "second"
+
+
+def test_passdown(tmp_path):
+    fds = make_fds(tmp_path)
+    parent = fds.get_task_datastore("1", "a", "1", attempt=0, mode="w")
+    parent.init_task()
+    parent.save_artifacts([("big", list(range(100))), ("keep", "yes")])
+    parent.done()
+
+    child = fds.get_task_datastore("1", "b", "2", attempt=0, mode="w")
+    child.init_task()
+    rd_parent = fds.get_task_datastore("1", "a", "1")
+    child.passdown(rd_parent)
+    child.done()
+
+    rd = fds.get_task_datastore("1", "b", "2")
+    assert rd["keep"] == "yes"
+    assert rd.artifact_sha("big") == rd_parent.artifact_sha("big")
+
+
+def test_clone(tmp_path):
+    fds = make_fds(tmp_path)
+    orig = fds.get_task_datastore("1", "a", "1", attempt=0, mode="w")
+    orig.init_task()
+    orig.save_artifacts([("v", 99)])
+    orig.save_metadata("transition", {"out_funcs": ["end"]})
+    orig.save_metadata("foreach_stack", [])
+    orig.done()
+
+    new = fds.get_task_datastore("2", "a", "1", attempt=0, mode="w")
+    new.clone(fds.get_task_datastore("1", "a", "1"))
+    rd = fds.get_task_datastore("2", "a", "1")
+    assert rd["v"] == 99
+    ok = rd.load_metadata("attempt_ok")
+    assert ok["ok"] and ok["cloned"]
+
+
+def test_tensor_serializer_roundtrip():
+    import torch
+
+    from metaflow_amd.datastore import serializers
+
+    for dtype in (torch.float32, torch.bfloat16, torch.int64):
+        t = (torch.randn(3, 5) * 10).to(dtype)
+        blob, enc = serializers.serialize(t)
+        assert enc == serializers.ENC_TENSOR
+        back = serializers.deserialize(blob, enc)
+        assert back.dtype == dtype
+        assert back.shape == t.shape
+        assert torch.equal(back, t)
+
+
+def test_flow_datastore_raw_data(tmp_path):
+    fds = make_fds(tmp_path)
+    [(uri, key)] = fds.save_data([b"code package"])
+    [(k, data)] = fds.load_data([key])
+    assert data == b"code package"

@@ -1,0 +1,101 @@
+"""End-to-end runtime tests: run real flows as subprocesses (CPU)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+FLOWS = os.path.join(os.path.dirname(os.path.abspath(__file__)), "flows")
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_flow(flow_file, datastore_root, *args, check=True, timeout=180):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    cmd = [
+        sys.executable, os.path.join(FLOWS, flow_file),
+        "--datastore-root", datastore_root,
+    ] + list(args)
+    proc = subprocess.run(cmd, capture_output=True, text=True, env=env,
+                          timeout=timeout)
+    if check and proc.returncode != 0:
+        raise AssertionError(
+            "flow failed rc=%d\nstdout:\n%s\nstderr:\n%s"
+            % (proc.returncode, proc.stdout, proc.stderr))
+    return proc
+
+
+def latest_run_id(datastore_root, flow_name):
+    meta = os.path.join(datastore_root, flow_name, "_meta")
+    runs = sorted(os.listdir(meta))
+    return runs[-1]
+
+
+def read_artifact(datastore_root, flow_name, run_id, step, name):
+    """Read an artifact via the client API."""
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = datastore_root
+    import importlib
+
+    import metaflow_amd.client as client
+
+    importlib.reload(client)
+    client.namespace(None)
+    task = client.Task("%s/%s/%s/%s" % (
+        flow_name, run_id, step,
+        next(iter(client.Step("%s/%s/%s" % (flow_name, run_id, step)))).id))
+    return getattr(task.data, name)
+
+
+def test_linear_flow(tmp_datastore):
+    run_flow("linear_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "LinearFlow")
+    assert read_artifact(tmp_datastore, "LinearFlow", run_id, "end",
+                         "final") == 31
+
+
+def test_linear_flow_with_param(tmp_datastore):
+    run_flow("linear_flow.py", tmp_datastore, "run", "--alpha", "5")
+    run_id = latest_run_id(tmp_datastore, "LinearFlow")
+    assert read_artifact(tmp_datastore, "LinearFlow", run_id, "end",
+                         "final") == 51
+
+
+def test_branch_flow(tmp_datastore):
+    run_flow("branch_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "BranchFlow")
+    assert read_artifact(tmp_datastore, "BranchFlow", run_id, "join",
+                         "total") == 3
+
+
+def test_foreach_flow(tmp_datastore):
+    run_flow("foreach_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "ForeachFlow")
+    assert read_artifact(tmp_datastore, "ForeachFlow", run_id, "join",
+                         "total") == 30
+
+
+def test_parallel_flow(tmp_datastore):
+    run_flow("parallel_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "ParallelFlow")
+    assert read_artifact(tmp_datastore, "ParallelFlow", run_id, "join",
+                         "ok") is True
+
+
+def test_show_command(tmp_datastore):
+    proc = run_flow("linear_flow.py", tmp_datastore, "show")
+    assert "start" in proc.stdout
+    assert "middle" in proc.stdout
+
+
+def test_dump_command(tmp_datastore):
+    run_flow("linear_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "LinearFlow")
+    # find end task id via the datastore layout
+    end_dir = os.path.join(tmp_datastore, "LinearFlow", run_id, "end")
+    task_id = os.listdir(end_dir)[0]
+    proc = run_flow("linear_flow.py", tmp_datastore, "dump",
+                    "%s/end/%s" % (run_id, task_id))
+    assert "final" in proc.stdout
