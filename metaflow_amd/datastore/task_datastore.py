@@ -249,6 +249,11 @@ class TaskDataStore(object):
         assert self._mode == "w"
         self.init_task()
         self._objects = dict(origin._objects)
+        # the scheduler drives successors off these two metadata entries
+        for name in ("transition", "foreach_stack", "control_mapper_tasks"):
+            value = origin.load_metadata(name)
+            if value is not None:
+                self.save_metadata(name, value)
         self.save_metadata("attempt_ok", {"ok": True, "cloned": True,
                                           "origin": origin.pathspec})
         self.done()

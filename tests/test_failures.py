@@ -1,0 +1,56 @@
+"""Retry, catch, and resume recovery tests (CPU, subprocess-driven)."""
+
+import os
+
+from .test_runtime import latest_run_id, read_artifact, run_flow
+
+
+def test_retry(tmp_datastore, tmp_path, monkeypatch):
+    marker_dir = tmp_path / "markers"
+    marker_dir.mkdir()
+    monkeypatch.setenv("RETRY_MARKER_DIR", str(marker_dir))
+    run_flow("retry_flow.py", tmp_datastore, "run")
+    attempts = sorted(os.listdir(str(marker_dir)))
+    assert attempts == ["attempt_0", "attempt_1", "attempt_2"]
+    run_id = latest_run_id(tmp_datastore, "RetryFlow")
+    assert read_artifact(tmp_datastore, "RetryFlow", run_id, "flaky",
+                         "attempts_seen") == 2
+
+
+def test_catch(tmp_datastore):
+    run_flow("catch_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "CatchFlow")
+    failure = read_artifact(tmp_datastore, "CatchFlow", run_id, "will_fail",
+                            "failure")
+    assert "intentional" in failure.exception
+
+
+def test_resume(tmp_datastore, tmp_path, monkeypatch):
+    counter_dir = tmp_path / "counters"
+    counter_dir.mkdir()
+    monkeypatch.setenv("RESUME_COUNTER_DIR", str(counter_dir))
+
+    monkeypatch.setenv("RESUME_FAIL", "1")
+    proc = run_flow("resume_flow.py", tmp_datastore, "run", check=False)
+    assert proc.returncode != 0
+    orig_run = latest_run_id(tmp_datastore, "ResumeFlow")
+
+    monkeypatch.setenv("RESUME_FAIL", "0")
+    run_flow("resume_flow.py", tmp_datastore, "resume",
+             "--origin-run-id", orig_run)
+    new_run = latest_run_id(tmp_datastore, "ResumeFlow")
+    assert new_run != orig_run
+
+    # start was cloned (ran once), middle ran twice, end once
+    assert int(open(counter_dir / "start").read()) == 1
+    assert int(open(counter_dir / "middle").read()) == 2
+    assert int(open(counter_dir / "end").read()) == 1
+    assert read_artifact(tmp_datastore, "ResumeFlow", new_run, "end",
+                         "z") == 11
+
+
+def test_with_retry_attach(tmp_datastore, tmp_path, monkeypatch):
+    """--with retry attaches the decorator to all steps."""
+    proc = run_flow("linear_flow.py", tmp_datastore,
+                    "--with", "retry:times=1", "run")
+    assert proc.returncode == 0
