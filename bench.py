@@ -1,0 +1,157 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: @parallel Llama-3-8B bf16 train step (BASELINE
+config 3) — tokens/sec, whole-job aggregate.
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches this under torch.distributed.run (one rank per
+GPU, RCCL over xGMI); this script is exactly the body the @parallel gang
+step runs inside the workflow engine (see tests/flows/train_flow.py for the
+FlowSpec-wrapped version). Synthetic data (random tokens), random-init
+weights, bf16 compute, fp32 Adam moments.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--batch", type=int, default=2,
+                   help="per-GPU micro batch")
+    p.add_argument("--seq", type=int, default=4096)
+    p.add_argument("--model", type=str, default="llama3-8b",
+                   choices=["llama3-8b", "llama3-70b", "tiny"])
+    p.add_argument("--bucket-mb", type=int, default=64)
+    p.add_argument("--lr", type=float, default=3e-4)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = args.gpus if world == 1 else world
+
+    import torch.distributed as dist
+
+    from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from metaflow_amd.parallel.ddp import (
+        FlatParamModel,
+        FusedAdamW,
+        init_process_group_from_env,
+    )
+
+    use_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_gpu else \
+        torch.device("cpu")
+    if use_gpu:
+        torch.cuda.set_device(device)
+
+    distributed = world > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        init_process_group_from_env()
+
+    cfg = {
+        "llama3-8b": LlamaConfig.llama3_8b,
+        "llama3-70b": LlamaConfig.llama3_70b,
+        "tiny": LlamaConfig.tiny,
+    }[args.model]()
+    seq = min(args.seq, cfg.max_seq_len)
+
+    torch.manual_seed(1234 + rank)
+    t0 = time.time()
+    model = LlamaForCausalLM(cfg).to(device)
+    flat = FlatParamModel(model, bucket_mb=args.bucket_mb)
+    flat.install_overlap_hooks()
+    opt = FusedAdamW(flat, lr=args.lr)
+    if rank == 0:
+        print("# model %s: %.2fB params, init %.1fs"
+              % (args.model, model.num_params() / 1e9, time.time() - t0),
+              flush=True)
+
+    # synthetic batch (fixed per rank: loss must fall, proving a real
+    # fwd+bwd+optimizer step is in the timed region)
+    tokens = torch.randint(0, cfg.vocab_size, (args.batch, seq + 1),
+                           device=device)
+    inp, tgt = tokens[:, :-1], tokens[:, 1:].contiguous()
+
+    def one_step():
+        flat.zero_grad()
+        loss = model(inp, tgt)
+        loss.backward()
+        flat.finish_grad_sync()
+        opt.step()
+        return loss
+
+    losses = []
+    for _ in range(args.warmup):
+        losses.append(one_step())
+
+    if distributed:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t_start = time.time()
+    for _ in range(args.steps):
+        losses.append(one_step())
+    if use_gpu:
+        torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.time() - t_start
+
+    # max over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=device if use_gpu else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    first_loss = float(losses[0].item())
+    last_loss = float(losses[-1].item())
+
+    tokens_per_step = args.batch * seq * n_gpus
+    toks_per_sec = tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec",
+            "value": toks_per_sec,
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": seq,
+                "parallelism": "dp%d" % n_gpus,
+                "first_loss": round(first_loss, 4),
+                "last_loss": round(last_loss, 4),
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,156 @@
+"""Llama-3 architecture on the metaflow_amd HIP kernel library.
+
+BASELINE config 3: the @parallel Llama-3-8B bf16 train step. All hot ops
+are hand-written gfx950 HIP (ops.attention / rmsnorm / rope / swiglu /
+cross_entropy); plain GEMMs go through torch.matmul -> hipBLASLt. Weights
+random-init (no network for checkpoints), bf16.
+"""
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from ..ops import kernels as K
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    max_seq_len: int = 8192
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    tie_embeddings: bool = False
+
+    @classmethod
+    def llama3_8b(cls):
+        return cls()
+
+    @classmethod
+    def llama3_70b(cls):
+        return cls(hidden_size=8192, intermediate_size=28672, num_layers=80,
+                   num_heads=64, num_kv_heads=8)
+
+    @classmethod
+    def tiny(cls, vocab=1024, seq=256):
+        """CPU-testable config."""
+        return cls(vocab_size=vocab, hidden_size=256, intermediate_size=688,
+                   num_layers=2, num_heads=2, num_kv_heads=1, head_dim=128,
+                   max_seq_len=seq)
+
+
+class Linear(nn.Module):
+    """Bias-free bf16 linear -> hipBLASLt GEMM."""
+
+    def __init__(self, din, dout, dtype=torch.bfloat16):
+        super().__init__()
+        self.weight = nn.Parameter(
+            torch.empty(dout, din, dtype=dtype))
+
+    def forward(self, x):
+        return torch.nn.functional.linear(x, self.weight)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim, eps):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim, dtype=torch.bfloat16))
+        self.eps = eps
+
+    def forward(self, x):
+        return K.rmsnorm(x, self.weight, self.eps)
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        h, hd = cfg.hidden_size, cfg.head_dim
+        self.input_norm = RMSNorm(h, cfg.rms_eps)
+        self.q_proj = Linear(h, cfg.num_heads * hd)
+        self.k_proj = Linear(h, cfg.num_kv_heads * hd)
+        self.v_proj = Linear(h, cfg.num_kv_heads * hd)
+        self.o_proj = Linear(cfg.num_heads * hd, h)
+        self.post_norm = RMSNorm(h, cfg.rms_eps)
+        self.gate_proj = Linear(h, cfg.intermediate_size)
+        self.up_proj = Linear(h, cfg.intermediate_size)
+        self.down_proj = Linear(cfg.intermediate_size, h)
+
+    def forward(self, x, cos_t, sin_t):
+        cfg = self.cfg
+        B, S, h = x.shape
+        res = x
+        y = self.input_norm(x)
+        q = self.q_proj(y).view(B, S, cfg.num_heads, cfg.head_dim)
+        k = self.k_proj(y).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        v = self.v_proj(y).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        q = K.rope(q, cos_t, sin_t)
+        k = K.rope(k, cos_t, sin_t)
+        # [B,S,H,D] -> [B,H,S,D] for the attention kernel
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        o = K.attention(q, k, v)
+        o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
+        x = res + self.o_proj(o)
+
+        res = x
+        y = self.post_norm(x)
+        x = res + self.down_proj(K.swiglu(self.gate_proj(y),
+                                          self.up_proj(y)))
+        return x
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                  dtype=torch.bfloat16)
+        self.layers = nn.ModuleList(
+            DecoderLayer(cfg) for _ in range(cfg.num_layers))
+        self.final_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
+        self.lm_head = Linear(cfg.hidden_size, cfg.vocab_size)
+        cos_t, sin_t = K.rope_tables(cfg.max_seq_len, cfg.head_dim,
+                                     cfg.rope_theta)
+        self.register_buffer("cos_t", cos_t, persistent=False)
+        self.register_buffer("sin_t", sin_t, persistent=False)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        std = 0.02
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                nn.init.normal_(p, mean=0.0, std=std)
+            elif "norm" in name:
+                nn.init.ones_(p)
+        # scaled init for output projections (GPT-2 style)
+        scale = 1.0 / math.sqrt(2 * self.cfg.num_layers)
+        for layer in self.layers:
+            layer.o_proj.weight.data.mul_(scale)
+            layer.down_proj.weight.data.mul_(scale)
+
+    def forward(self, tokens, targets=None):
+        """tokens [B, S] int64; returns mean loss if targets given, else
+        logits."""
+        x = self.embed(tokens)
+        for layer in self.layers:
+            x = layer(x, self.cos_t, self.sin_t)
+        x = self.final_norm(x)
+        if targets is None:
+            return self.lm_head(x)
+        logits = self.lm_head(x)
+        B, S, V = logits.shape
+        loss = K.cross_entropy(logits.reshape(B * S, V),
+                               targets.reshape(B * S))
+        return loss.mean()
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())

@@ -14,6 +14,8 @@ def engine():
     if not _tried:
         _tried = True
         try:
+            import torch  # noqa: F401  (loads libc10.so for the extension)
+
             from . import _mfx_cas  # built in-tree by setup.py
 
             _engine = _mfx_cas.Engine()

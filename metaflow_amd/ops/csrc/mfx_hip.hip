@@ -1,0 +1,245 @@
+// PyTorch bindings for the metaflow_amd gfx950 kernel library (_mfx_hip).
+// All tensors bf16 unless stated; shapes validated here so kernels stay lean.
+
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+#include "elementwise.hip"
+#include "adam.hip"
+#include "cross_entropy.hip"
+#include "attention.hip"
+
+namespace {
+
+constexpr int kBlock = 256;
+// memory-bound grid cap (guide G11): 256 CUs x 8 blocks
+constexpr int kGridCap = 2048;
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+inline const short* bf(const torch::Tensor& t) {
+  return reinterpret_cast<const short*>(t.data_ptr());
+}
+inline short* bfm(torch::Tensor& t) {
+  return reinterpret_cast<short*>(t.data_ptr());
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+inline int grid_for(long long work_items) {
+  long long blocks = (work_items + kBlock - 1) / kBlock;
+  return (int)std::min<long long>(blocks, kGridCap);
+}
+
+// ------------------------------------------------------------- rmsnorm
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
+  const long long rows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto inv = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  rmsnorm_fwd_kernel<kBlock><<<(int)rows, kBlock, 0, cur_stream()>>>(
+      bf(x), bf(w), bfm(y), inv.data_ptr<float>(), H, (float)eps);
+  return {y, inv};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor dy,
+                                       torch::Tensor inv_rms) {
+  check_bf16(x, "x");
+  check_bf16(dy, "dy");
+  const int H = x.size(-1);
+  const long long rows = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  int grid = (int)std::min<long long>(rows, 1024);
+  size_t lds = (H + kBlock / 64) * sizeof(float);
+  rmsnorm_bwd_kernel<kBlock><<<grid, kBlock, lds, cur_stream()>>>(
+      bf(x), bf(w), bf(dy), inv_rms.data_ptr<float>(), bfm(dx),
+      dw.data_ptr<float>(), (int)rows, H);
+  return {dx, dw};
+}
+
+// ---------------------------------------------------------------- rope
+torch::Tensor rope(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
+                   long rows_per_pos, long seqlen, long pos0, bool backward) {
+  check_bf16(x, "x");
+  const int D = x.size(-1);
+  const long long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  const long long total = rows * (D / 2);
+  rope_kernel<<<grid_for(total), kBlock, 0, cur_stream()>>>(
+      bf(x), bfm(y), cos_t.data_ptr<float>(), sin_t.data_ptr<float>(), rows,
+      D, (int)rows_per_pos, (int)seqlen, (int)pos0, backward ? -1.f : 1.f);
+  return y;
+}
+
+// -------------------------------------------------------------- swiglu
+torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u) {
+  check_bf16(g, "g");
+  check_bf16(u, "u");
+  TORCH_CHECK(g.numel() % 8 == 0);
+  auto y = torch::empty_like(g);
+  long long n8 = g.numel() / 8;
+  swiglu_fwd_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
+      bf(g), bf(u), bfm(y), n8);
+  return y;
+}
+
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor g, torch::Tensor u,
+                                      torch::Tensor dy) {
+  auto dg = torch::empty_like(g);
+  auto du = torch::empty_like(u);
+  long long n8 = g.numel() / 8;
+  swiglu_bwd_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
+      bf(g), bf(u), bf(dy), bfm(dg), bfm(du), n8);
+  return {dg, du};
+}
+
+torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a");
+  TORCH_CHECK(a.numel() % 8 == 0);
+  auto y = torch::empty_like(a);
+  long long n8 = a.numel() / 8;
+  add_bf16_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
+      bf(a), bf(b), bfm(y), n8);
+  return y;
+}
+
+// ---------------------------------------------------------------- adam
+void adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
+           c10::optional<torch::Tensor> master, double lr, double beta1,
+           double beta2, double eps, double weight_decay, long step,
+           double grad_scale) {
+  const long long n = p.numel();
+  float bias_c1 = 1.f / (1.f - powf((float)beta1, (float)step));
+  float bias_c2 = 1.f / (1.f - powf((float)beta2, (float)step));
+  if (p.scalar_type() == torch::kBFloat16) {
+    TORCH_CHECK(n % 4 == 0, "bf16 adamw requires numel % 4 == 0");
+    float* master_ptr =
+        master.has_value() ? master->data_ptr<float>() : nullptr;
+    adamw_kernel<<<grid_for(n / 4), kBlock, 0, cur_stream()>>>(
+        bfm(p), bf(g), m.data_ptr<float>(), v.data_ptr<float>(), master_ptr,
+        n, (float)lr, (float)beta1, (float)beta2, (float)eps,
+        (float)weight_decay, bias_c1, bias_c2, (float)grad_scale);
+  } else {
+    adamw_f32_kernel<<<grid_for(n), kBlock, 0, cur_stream()>>>(
+        p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+        v.data_ptr<float>(), n, (float)lr, (float)beta1, (float)beta2,
+        (float)eps, (float)weight_decay, bias_c1, bias_c2,
+        (float)grad_scale);
+  }
+}
+
+// ------------------------------------------------------- cross entropy
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
+                                             torch::Tensor targets,
+                                             long ignore_index) {
+  check_bf16(logits, "logits");
+  const int V = logits.size(-1);
+  const long long N = logits.numel() / V;
+  auto loss = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+  auto rmax = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+  auto rlse = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+  int grid = (int)std::min<long long>(N, kGridCap);
+  cross_entropy_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
+      bf(logits), targets.data_ptr<long long>(), loss.data_ptr<float>(),
+      rmax.data_ptr<float>(), rlse.data_ptr<float>(), N, V,
+      (long long)ignore_index);
+  return {loss, rlse};
+}
+
+torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
+                                torch::Tensor row_lse, torch::Tensor dloss,
+                                long ignore_index) {
+  const int V = logits.size(-1);
+  const long long N = logits.numel() / V;
+  auto dlogits = torch::empty_like(logits);
+  int grid = (int)std::min<long long>(N, kGridCap);
+  cross_entropy_bwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
+      bf(logits), targets.data_ptr<long long>(), row_lse.data_ptr<float>(),
+      dloss.data_ptr<float>(), bfm(dlogits), N, V, (long long)ignore_index);
+  return dlogits;
+}
+
+// ----------------------------------------------------------- attention
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, double scale) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(v, "v");
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(D == 128, "attention requires head dim 128");
+  TORCH_CHECK(S % 64 == 0, "attention requires seqlen % 64 == 0");
+  TORCH_CHECK(H % Hkv == 0, "GQA requires H % Hkv == 0");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
+  dim3 grid(S / 64, H, B);
+  attn_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
+      bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
+      (float)scale);
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dout, torch::Tensor lse,
+                                    double scale) {
+  const int B = q.size(0), H = q.size(1), S = q.size(2);
+  const int Hkv = k.size(1);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
+
+  const long long rows = (long long)B * H * S;
+  {
+    int rows_per_block = kBlock / 64;
+    long long blocks = (rows + rows_per_block - 1) / rows_per_block;
+    attn_bwd_delta_kernel<<<(int)blocks, kBlock, 0, cur_stream()>>>(
+        bf(dout), bf(o), delta.data_ptr<float>(), rows);
+  }
+  {
+    dim3 grid(S / 64, Hkv, B);
+    attn_bwd_dkdv_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+        delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
+        (float)scale);
+  }
+  {
+    dim3 grid(S / 64, H, B);
+    attn_bwd_dq_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+        delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale);
+  }
+  return {dq, dk, dv};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");
+  m.def("rope", &rope, "RoPE (half-rotation), fwd or bwd via sign");
+  m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
+  m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
+  m.def("add_bf16", &add_bf16, "fused bf16 add");
+  m.def("adamw", &adamw, "fused AdamW (bf16 p/g, fp32 m/v[, master])");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward (causal, GQA)");
+  m.def("attn_bwd", &attn_bwd, "flash attention backward");
+}

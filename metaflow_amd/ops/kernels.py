@@ -1,0 +1,248 @@
+"""Python surface for the gfx950 kernel library.
+
+Every op has two implementations:
+  * the HIP kernel (used on GPU — REQUIRED there: if the extension is
+    missing on a GPU box we raise KernelExtensionMissing rather than
+    silently fall back to eager, so GPU runs always exercise native code);
+  * a plain PyTorch fp32 reference (`*_ref`), used on CPU and by the
+    numerics tests as ground truth.
+"""
+
+import math
+
+import torch
+
+from ..exceptions import KernelExtensionMissing
+
+_ext = None
+_ext_err = None
+
+
+def hip_ext():
+    """The _mfx_hip extension; raises loudly if unavailable on GPU."""
+    global _ext, _ext_err
+    if _ext is None and _ext_err is None:
+        try:
+            from . import _mfx_hip
+
+            _ext = _mfx_hip
+        except Exception as e:  # noqa: BLE001
+            _ext_err = e
+    if _ext is None:
+        raise KernelExtensionMissing(
+            "metaflow_amd._mfx_hip is not built (%s). Run "
+            "`python setup.py build_ext --inplace` (hipcc, gfx950)."
+            % _ext_err)
+    return _ext
+
+
+def extension_loaded():
+    try:
+        hip_ext()
+        return True
+    except KernelExtensionMissing:
+        return False
+
+
+# ============================== RMSNorm ====================================
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        y, inv = hip_ext().rmsnorm_fwd(x, w, eps)
+        ctx.save_for_backward(x, w, inv)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, inv = ctx.saved_tensors
+        dx, dw = hip_ext().rmsnorm_bwd(x, w, dy.contiguous(), inv)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x, w, eps=1e-5):
+    if x.is_cuda:
+        return _RMSNorm.apply(x.contiguous(), w.contiguous(), eps)
+    return rmsnorm_ref(x, w, eps).to(x.dtype)
+
+
+def rmsnorm_ref(x, w, eps=1e-5):
+    xf = x.float()
+    inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * inv * w.float()).to(x.dtype)
+
+
+# ================================ RoPE =====================================
+def rope_tables(seqlen, dim, theta=500000.0, device="cpu"):
+    """Host-precomputed fp32 cos/sin tables [seqlen, dim/2] (guide App B:
+    on-device trig turns a memory-bound op VALU-bound)."""
+    half = dim // 2
+    freqs = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64)
+                             / half))
+    pos = torch.arange(seqlen, dtype=torch.float64)
+    ang = torch.outer(pos, freqs)
+    return (ang.cos().float().to(device).contiguous(),
+            ang.sin().float().to(device).contiguous())
+
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos_t, sin_t, rows_per_pos, seqlen, pos0):
+        ctx.tables = (cos_t, sin_t)
+        ctx.meta = (rows_per_pos, seqlen, pos0)
+        return hip_ext().rope(x, cos_t, sin_t, rows_per_pos, seqlen, pos0,
+                              False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos_t, sin_t = ctx.tables
+        rows_per_pos, seqlen, pos0 = ctx.meta
+        dx = hip_ext().rope(dy.contiguous(), cos_t, sin_t, rows_per_pos,
+                            seqlen, pos0, True)
+        return dx, None, None, None, None, None
+
+
+def rope(x, cos_t, sin_t, pos0=0):
+    """x: [B, S, Hh, D] (positions along dim 1). Half-rotation (Llama-HF)."""
+    B, S, Hh, D = x.shape
+    if x.is_cuda:
+        return _Rope.apply(x.contiguous(), cos_t, sin_t, Hh, S, pos0)
+    return rope_ref(x, cos_t, sin_t, pos0)
+
+
+def rope_ref(x, cos_t, sin_t, pos0=0):
+    B, S, Hh, D = x.shape
+    half = D // 2
+    c = cos_t[pos0:pos0 + S].view(1, S, 1, half).to(torch.float32)
+    s = sin_t[pos0:pos0 + S].view(1, S, 1, half).to(torch.float32)
+    xf = x.float()
+    x1, x2 = xf[..., :half], xf[..., half:]
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1).to(x.dtype)
+
+
+# =============================== SwiGLU ====================================
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g, u):
+        ctx.save_for_backward(g, u)
+        return hip_ext().swiglu_fwd(g, u)
+
+    @staticmethod
+    def backward(ctx, dy):
+        g, u = ctx.saved_tensors
+        dg, du = hip_ext().swiglu_bwd(g, u, dy.contiguous())
+        return dg, du
+
+
+def swiglu(g, u):
+    if g.is_cuda:
+        return _SwiGLU.apply(g.contiguous(), u.contiguous())
+    return swiglu_ref(g, u)
+
+
+def swiglu_ref(g, u):
+    gf = g.float()
+    return (torch.nn.functional.silu(gf) * u.float()).to(g.dtype)
+
+
+# ============================ cross entropy ================================
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        loss, lse = hip_ext().cross_entropy_fwd(logits, targets,
+                                                ignore_index)
+        ctx.save_for_backward(logits, targets, lse)
+        ctx.ignore_index = ignore_index
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse = ctx.saved_tensors
+        dlogits = hip_ext().cross_entropy_bwd(
+            logits, targets, lse, dloss.contiguous().float(),
+            ctx.ignore_index)
+        return dlogits, None, None
+
+
+def cross_entropy(logits, targets, ignore_index=-100):
+    """Per-token loss [N] fp32 from bf16 logits [N, V]."""
+    if logits.is_cuda:
+        return _CrossEntropy.apply(logits.contiguous(),
+                                   targets.contiguous(), ignore_index)
+    return cross_entropy_ref(logits, targets, ignore_index)
+
+
+def cross_entropy_ref(logits, targets, ignore_index=-100):
+    return torch.nn.functional.cross_entropy(
+        logits.float(), targets, ignore_index=ignore_index,
+        reduction="none")
+
+
+# ============================== attention ==================================
+class _Attention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        o, lse = hip_ext().attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = hip_ext().attn_bwd(q, k, v, o, dout.contiguous(), lse,
+                                        ctx.scale)
+        return dq, dk, dv, None
+
+
+def attention(q, k, v, scale=None):
+    """Causal GQA flash attention. q: [B,H,S,128], k/v: [B,Hkv,S,128]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.size(-1))
+    if q.is_cuda:
+        return _Attention.apply(q.contiguous(), k.contiguous(),
+                                v.contiguous(), scale)
+    return attention_ref(q, k, v, scale)
+
+
+def attention_ref(q, k, v, scale=None):
+    """fp32 eager reference (causal, GQA)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.size(-1))
+    B, H, S, D = q.shape
+    Hkv = k.size(1)
+    G = H // Hkv
+    kf = k.float().repeat_interleave(G, dim=1)
+    vf = v.float().repeat_interleave(G, dim=1)
+    s = torch.matmul(q.float(), kf.transpose(-1, -2)) * scale
+    mask = torch.ones(S, S, dtype=torch.bool, device=q.device).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.matmul(p, vf).to(q.dtype)
+
+
+# ================================ adam =====================================
+def adamw_step(p, g, m, v, step, lr, beta1=0.9, beta2=0.95, eps=1e-8,
+               weight_decay=0.1, master=None, grad_scale=1.0):
+    """Fused AdamW on one tensor (bf16 or fp32 params, fp32 m/v)."""
+    if p.is_cuda:
+        hip_ext().adamw(p, g, m, v, master, lr, beta1, beta2, eps,
+                        weight_decay, step, grad_scale)
+        return
+    # CPU reference
+    gf = g.float() * grad_scale
+    m.mul_(beta1).add_(gf, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+    c1 = 1 / (1 - beta1 ** step)
+    c2 = 1 / (1 - beta2 ** step)
+    src = master if master is not None else p.float()
+    upd = src - lr * ((m * c1) / ((v * c2).sqrt() + eps)
+                      + weight_decay * src)
+    if master is not None:
+        master.copy_(upd)
+    p.copy_(upd.to(p.dtype))
+
+
+def add_bf16(a, b):
+    if a.is_cuda:
+        return hip_ext().add_bf16(a.contiguous(), b.contiguous())
+    return a + b

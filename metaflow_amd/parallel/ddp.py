@@ -1,0 +1,173 @@
+"""Flat-buffer data parallelism over RCCL/xGMI + fused AdamW.
+
+MI355X-first design (SURVEY §5 'distributed communication backend'):
+
+* all bf16 params live in ONE contiguous buffer (params become views), and
+  all grads in a matching flat buffer — autograd accumulates straight into
+  the flat slices (no per-tensor .grad allocations, no copy into buckets);
+* the flat grad buffer is divided into contiguous buckets (~64 MiB: sized
+  for xGMI per-link ring bandwidth, not NVSwitch); as backward finishes the
+  last param of a bucket, an async all-reduce(AVG) is launched so
+  communication overlaps the rest of backward;
+* the optimizer is ONE fused AdamW kernel call over the whole flat buffer
+  (ops/csrc/adam.hip) — fp32 m/v (+optional fp32 master weights), so a
+  full 8B update is a handful of kernel launches instead of ~300.
+"""
+
+import torch
+import torch.distributed as dist
+
+from ..ops import kernels as K
+
+
+class FlatParamModel(object):
+    """Flattens a module's parameters into one bf16 buffer + flat grads."""
+
+    def __init__(self, module, bucket_mb=64):
+        self.module = module
+        params = [p for p in module.parameters() if p.requires_grad]
+        # stable ordering: reverse autograd-completion order roughly matches
+        # registration order reversed; buckets are launched as they fill
+        self.params = params
+        total = sum(self._padded(p.numel()) for p in params)
+        device = params[0].device
+        dtype = params[0].dtype
+        self.flat_param = torch.empty(total, dtype=dtype, device=device)
+        self.flat_grad = torch.zeros(total, dtype=dtype, device=device)
+
+        offset = 0
+        self.offsets = []
+        for p in params:
+            n = p.numel()
+            self.flat_param[offset:offset + n].copy_(p.detach().reshape(-1))
+            # re-point the parameter at the flat storage
+            p.data = self.flat_param[offset:offset + n].view_as(p)
+            p.grad = self.flat_grad[offset:offset + n].view_as(p)
+            self.offsets.append(offset)
+            offset += self._padded(n)
+
+        # buckets: contiguous ranges of the flat buffer
+        bucket_elems = (bucket_mb << 20) // self.flat_param.element_size()
+        self.buckets = []  # (start, end, last_param_index)
+        start = 0
+        for i, p in enumerate(params):
+            end = self.offsets[i] + self._padded(p.numel())
+            if end - start >= bucket_elems or i == len(params) - 1:
+                self.buckets.append([start, end, i])
+                start = end
+        self._pending = []
+        self._hooks = []
+
+    @staticmethod
+    def _padded(n):
+        return (n + 3) & ~3  # fused adam wants numel % 4 == 0
+
+    def zero_grad(self):
+        self.flat_grad.zero_()
+
+    # ----------------------------------------------------------- allreduce
+    def install_overlap_hooks(self):
+        """Launch each bucket's all-reduce as soon as its last param's grad
+        is accumulated (params complete roughly in reverse order, so buckets
+        are checked by completion count)."""
+        if not (dist.is_available() and dist.is_initialized()
+                and dist.get_world_size() > 1):
+            return
+        self._done = set()
+        bucket_last_param = {}
+        prev = -1
+        for bi, (_s, _e, last) in enumerate(self.buckets):
+            for pi in range(prev + 1, last + 1):
+                bucket_last_param[pi] = None
+            bucket_last_param[last] = bi
+            prev = last
+        bucket_sizes = {}
+        prev = -1
+        for bi, (_s, _e, last) in enumerate(self.buckets):
+            bucket_sizes[bi] = set(range(prev + 1, last + 1))
+            prev = last
+        self._bucket_param_sets = bucket_sizes
+        self._bucket_done_count = {bi: 0 for bi in range(len(self.buckets))}
+        param_to_bucket = {}
+        prev = -1
+        for bi, (_s, _e, last) in enumerate(self.buckets):
+            for pi in range(prev + 1, last + 1):
+                param_to_bucket[pi] = bi
+            prev = last
+
+        for pi, p in enumerate(self.params):
+            bi = param_to_bucket[pi]
+
+            def hook(_param, bi=bi):
+                self._bucket_done_count[bi] += 1
+                if self._bucket_done_count[bi] == len(
+                        self._bucket_param_sets[bi]):
+                    s, e, _ = self.buckets[bi]
+                    work = dist.all_reduce(self.flat_grad[s:e],
+                                           op=dist.ReduceOp.AVG,
+                                           async_op=True)
+                    self._pending.append(work)
+
+            self._hooks.append(p.register_post_accumulate_grad_hook(hook))
+
+    def finish_grad_sync(self):
+        """Wait for overlapped all-reduces (or do one synchronous pass if
+        hooks are not installed)."""
+        if self._pending:
+            for work in self._pending:
+                work.wait()
+            self._pending.clear()
+            for bi in self._bucket_done_count:
+                self._bucket_done_count[bi] = 0
+        elif (dist.is_available() and dist.is_initialized()
+              and dist.get_world_size() > 1):
+            dist.all_reduce(self.flat_grad, op=dist.ReduceOp.AVG)
+
+
+class FusedAdamW(object):
+    """AdamW over a FlatParamModel: one kernel call per step."""
+
+    def __init__(self, flat_model, lr=3e-4, betas=(0.9, 0.95), eps=1e-8,
+                 weight_decay=0.1, master_weights=False):
+        fp = flat_model.flat_param
+        self.flat = flat_model
+        self.lr = lr
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.step_count = 0
+        self.m = torch.zeros(fp.numel(), dtype=torch.float32,
+                             device=fp.device)
+        self.v = torch.zeros_like(self.m)
+        self.master = fp.float() if master_weights else None
+
+    def step(self, grad_scale=1.0):
+        self.step_count += 1
+        K.adamw_step(self.flat.flat_param, self.flat.flat_grad, self.m,
+                     self.v, self.step_count, self.lr, self.beta1,
+                     self.beta2, self.eps, self.weight_decay,
+                     master=self.master, grad_scale=grad_scale)
+
+    def state_dict_tensors(self):
+        """Shard tensors for @checkpoint (per-rank)."""
+        out = {"flat_param": self.flat.flat_param, "adam_m": self.m,
+               "adam_v": self.v}
+        if self.master is not None:
+            out["master"] = self.master
+        return out
+
+
+def init_process_group_from_env(backend=None):
+    """env:// rendezvous (MASTER_ADDR/PORT, RANK, WORLD_SIZE) — set either
+    by torchrun or by the gang scheduler (runtime._queue_gang)."""
+    import os
+
+    if dist.is_initialized():
+        return
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    dist.init_process_group(backend=backend)

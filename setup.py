@@ -1,0 +1,45 @@
+"""Build the native extensions in-tree:
+
+  python setup.py build_ext --inplace
+
+* metaflow_amd/ops/_mfx_hip.so  — gfx950 HIP kernel library (torch ext)
+* metaflow_amd/ops/_mfx_cas.so  — C++ CAS engine (threaded SHA-256 + IO)
+
+gfx950-only by design (PYTORCH_ROCM_ARCH=gfx950); the .so files travel with
+the repo snapshot to the GPU box.
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+os.environ.setdefault("MAX_JOBS", str(os.cpu_count() or 8))
+
+from setuptools import setup  # noqa: E402
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension, CppExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "metaflow_amd", "ops", "csrc")
+
+ext_modules = [
+    CUDAExtension(
+        name="metaflow_amd.ops._mfx_hip",
+        sources=[os.path.join(CSRC, "mfx_hip.hip")],
+        extra_compile_args={
+            "cxx": ["-O3", "-std=c++17"],
+            "nvcc": ["-O3", "-std=c++17"],
+        },
+    ),
+    CppExtension(
+        name="metaflow_amd.ops._mfx_cas",
+        sources=[os.path.join(CSRC, "cas_engine.cpp")],
+        extra_compile_args={"cxx": ["-O3", "-std=c++17", "-pthread"]},
+    ),
+]
+
+setup(
+    name="metaflow_amd",
+    version="0.1.0",
+    packages=["metaflow_amd"],
+    ext_modules=ext_modules,
+    cmdclass={"build_ext": BuildExtension},
+)
