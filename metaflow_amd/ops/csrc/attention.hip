@@ -26,8 +26,11 @@ DEV f4 mfma16(bf16x8 a, bf16x8 b, f4 c) {
 }
 
 // byte offset of element (r, c) in a swizzled LDS tile with ROWB bytes/row
+// (rowb must be a power of 2; the XOR is masked so it stays inside the row
+// — rowb=64 tiles get a 4-row spread instead of 8, still conflict-free
+// enough at 2 lanes/bank)
 DEV int swz(int r, int c, int rowb) {
-  return r * rowb + (((c) * 2) ^ ((r & 7) << 4));
+  return r * rowb + (((c) * 2) ^ (((r & 7) << 4) & (rowb - 1)));
 }
 
 // load an 8-element fragment from a swizzled tile row (c must be mult of 8)

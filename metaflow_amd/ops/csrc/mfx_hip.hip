@@ -155,7 +155,7 @@ std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
   auto rlse = torch::empty({N}, logits.options().dtype(torch::kFloat32));
   int grid = (int)std::min<long long>(N, kGridCap);
   cross_entropy_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
-      bf(logits), targets.data_ptr<long long>(), loss.data_ptr<float>(),
+      bf(logits), reinterpret_cast<const long long*>(targets.data_ptr<int64_t>()), loss.data_ptr<float>(),
       rmax.data_ptr<float>(), rlse.data_ptr<float>(), N, V,
       (long long)ignore_index);
   return {loss, rlse};
@@ -169,7 +169,7 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
   auto dlogits = torch::empty_like(logits);
   int grid = (int)std::min<long long>(N, kGridCap);
   cross_entropy_bwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
-      bf(logits), targets.data_ptr<long long>(), row_lse.data_ptr<float>(),
+      bf(logits), reinterpret_cast<const long long*>(targets.data_ptr<int64_t>()), row_lse.data_ptr<float>(),
       dloss.data_ptr<float>(), bfm(dlogits), N, V, (long long)ignore_index);
   return dlogits;
 }

@@ -1,0 +1,91 @@
+"""Nested foreach, switch, Runner API, client API, and the CPU gang train
+flow (gloo world=2)."""
+
+import os
+
+from .test_runtime import FLOWS, REPO, latest_run_id, read_artifact, run_flow
+
+
+def test_nested_foreach(tmp_datastore):
+    run_flow("nested_foreach_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "NestedForeachFlow")
+    assert read_artifact(tmp_datastore, "NestedForeachFlow", run_id,
+                         "outer_join", "total") == 180
+
+
+def test_switch_fast(tmp_datastore):
+    run_flow("switch_flow.py", tmp_datastore, "run", "--route", "fast")
+    run_id = latest_run_id(tmp_datastore, "SwitchFlow")
+    assert read_artifact(tmp_datastore, "SwitchFlow", run_id, "finish",
+                         "final") == "fast"
+
+
+def test_switch_slow(tmp_datastore):
+    run_flow("switch_flow.py", tmp_datastore, "run", "--route", "slow")
+    run_id = latest_run_id(tmp_datastore, "SwitchFlow")
+    assert read_artifact(tmp_datastore, "SwitchFlow", run_id, "finish",
+                         "final") == "slow"
+
+
+def test_runner_api(tmp_datastore, monkeypatch):
+    monkeypatch.setenv("MFX_NUM_GPUS", "0")
+    monkeypatch.setenv(
+        "PYTHONPATH", REPO + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    from metaflow_amd.runner import Runner
+
+    r = Runner(os.path.join(FLOWS, "linear_flow.py"),
+               datastore_root=tmp_datastore)
+    ex = r.run(alpha=7)
+    assert ex.status == "successful"
+    run = ex.run
+    assert run is not None
+    task = run["end"].task
+    assert task.data.final == 71
+    assert task.successful
+
+
+def test_client_api(tmp_datastore):
+    run_flow("branch_flow.py", tmp_datastore, "run")
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    import importlib
+
+    import metaflow_amd.client as client
+
+    importlib.reload(client)
+    client.namespace(None)
+
+    mf = client.Metaflow()
+    flows = [f.id for f in mf]
+    assert "BranchFlow" in flows
+    flow = client.Flow("BranchFlow")
+    run = flow.latest_run
+    assert run.successful
+    steps = {s.id for s in run}
+    assert {"start", "a", "b", "join", "end"} <= steps
+    join_task = run["join"].task
+    assert join_task.data.total == 3
+    art = join_task["total"]
+    assert art.data == 3 and art.sha and art.size > 0
+    # namespace filtering
+    client.namespace("user:nonexistent-user")
+    try:
+        client.Flow("BranchFlow").latest_run
+        got = client.Flow("BranchFlow").latest_run
+        assert got is None
+    finally:
+        client.namespace(None)
+
+
+def test_gang_train_flow_cpu(tmp_datastore):
+    """The full @parallel + @torch_parallel + @checkpoint path on CPU:
+    gang of 2 over gloo through the real scheduler."""
+    run_flow("train_flow.py", tmp_datastore, "run", "--num-nodes", "2",
+             "--steps-n", "2", timeout=420)
+    run_id = latest_run_id(tmp_datastore, "TrainFlow")
+    all_losses = read_artifact(tmp_datastore, "TrainFlow", run_id, "join",
+                               "all_losses")
+    assert set(all_losses) == {0, 1}
+    # checkpoint artifact exists on the control task
+    idx = read_artifact(tmp_datastore, "TrainFlow", run_id, "train",
+                        "_checkpoint_final_index")
+    assert "flat_param" in idx and "adam_m" in idx
