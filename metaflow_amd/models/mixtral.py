@@ -1,0 +1,275 @@
+"""Mixtral-style MoE on the metaflow_amd kernel library.
+
+BASELINE config 5: Mixtral 8x7B @parallel train step with expert
+all-to-all. Expert parallelism over RCCL/xGMI: experts are sharded across
+ranks; tokens are routed top-k, exchanged with all_to_all_single (variable
+splits, autograd-wrapped so backward is the transposed all-to-all), run
+through the owner's experts (ops.swiglu + hipBLASLt GEMMs), exchanged back
+and combined with routing weights. xGMI note: all-to-all is
+point-to-point and uses all 7 links concurrently — better suited to the
+topology than ring collectives.
+"""
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from ..ops import kernels as K
+from .llama import Linear, RMSNorm
+
+
+@dataclass
+class MixtralConfig:
+    vocab_size: int = 32000
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    num_experts: int = 8
+    top_k: int = 2
+    max_seq_len: int = 8192
+    rope_theta: float = 1000000.0
+    rms_eps: float = 1e-5
+
+    @classmethod
+    def mixtral_8x7b(cls):
+        return cls()
+
+    @classmethod
+    def tiny(cls, vocab=1024, seq=256):
+        return cls(vocab_size=vocab, hidden_size=256, intermediate_size=512,
+                   num_layers=2, num_heads=2, num_kv_heads=1, head_dim=128,
+                   num_experts=4, top_k=2, max_seq_len=seq)
+
+
+class _AllToAll(torch.autograd.Function):
+    """all_to_all_single with explicit splits; backward = transposed."""
+
+    @staticmethod
+    def forward(ctx, x, out_splits, in_splits, group):
+        import torch.distributed as dist
+
+        ctx.splits = (out_splits, in_splits)
+        ctx.group = group
+        out = x.new_empty((sum(out_splits),) + tuple(x.shape[1:]))
+        dist.all_to_all_single(out, x.contiguous(),
+                               output_split_sizes=out_splits,
+                               input_split_sizes=in_splits, group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        out_splits, in_splits = ctx.splits
+        import torch.distributed as dist
+
+        gin = grad.new_empty((sum(in_splits),) + tuple(grad.shape[1:]))
+        dist.all_to_all_single(gin, grad.contiguous(),
+                               output_split_sizes=in_splits,
+                               input_split_sizes=out_splits,
+                               group=ctx.group)
+        return gin, None, None, None
+
+
+def all_to_all(x, out_splits, in_splits, group=None):
+    return _AllToAll.apply(x, out_splits, in_splits, group)
+
+
+class Expert(nn.Module):
+    def __init__(self, hidden, inter):
+        super().__init__()
+        self.gate_proj = Linear(hidden, inter)
+        self.up_proj = Linear(hidden, inter)
+        self.down_proj = Linear(inter, hidden)
+
+    def forward(self, x):
+        return self.down_proj(K.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class MoELayer(nn.Module):
+    """Top-k router + (optionally expert-parallel) experts.
+
+    With EP (world > 1): rank r owns experts [r*E/W, (r+1)*E/W); the
+    dispatch is a single all-to-all of the routed token copies, grouped by
+    destination rank.
+    """
+
+    def __init__(self, cfg: MixtralConfig, ep_group=None):
+        super().__init__()
+        self.cfg = cfg
+        self.router = Linear(cfg.hidden_size, cfg.num_experts)
+        self.ep_group = ep_group
+        import torch.distributed as dist
+
+        if (dist.is_available() and dist.is_initialized()
+                and dist.get_world_size(ep_group) > 1
+                and cfg.num_experts % dist.get_world_size(ep_group) == 0):
+            self.ep_world = dist.get_world_size(ep_group)
+            self.ep_rank = dist.get_rank(ep_group)
+        else:
+            self.ep_world = 1
+            self.ep_rank = 0
+        self.experts_per_rank = cfg.num_experts // self.ep_world
+        self.local_experts = nn.ModuleList(
+            Expert(cfg.hidden_size, cfg.intermediate_size)
+            for _ in range(self.experts_per_rank))
+        if self.ep_world > 1:
+            # grads are complete after the token all-to-all: exclude from
+            # the DP all-reduce (see FlatParamModel no-sync region)
+            for p in self.local_experts.parameters():
+                p._mfx_no_sync = True
+
+    def forward(self, x):
+        cfg = self.cfg
+        B, S, h = x.shape
+        flat = x.reshape(B * S, h)
+        logits = self.router(flat).float()
+        probs = torch.softmax(logits, dim=-1)
+        weights, experts = probs.topk(cfg.top_k, dim=-1)  # [N, k]
+        weights = weights / weights.sum(-1, keepdim=True)
+
+        N = flat.size(0)
+        k = cfg.top_k
+        # one entry per (token, choice): sort by destination expert
+        flat_expert = experts.reshape(-1)              # [N*k]
+        order = torch.argsort(flat_expert, stable=True)
+        token_idx = order // k                         # source token
+        sorted_expert = flat_expert[order]
+        counts = torch.bincount(sorted_expert, minlength=cfg.num_experts)
+
+        dispatched = flat[token_idx]                   # [N*k, h]
+
+        if self.ep_world > 1:
+            import torch.distributed as dist
+
+            # counts per destination rank
+            per_rank = counts.reshape(self.ep_world,
+                                      self.experts_per_rank).sum(-1)
+            in_splits = per_rank.tolist()
+            recv_counts = torch.empty(
+                self.ep_world * self.experts_per_rank, dtype=counts.dtype,
+                device=counts.device)
+            # exchange per-expert counts so the owner can group its input
+            dist.all_to_all_single(
+                recv_counts, counts,
+                output_split_sizes=[self.experts_per_rank] * self.ep_world,
+                input_split_sizes=[self.experts_per_rank] * self.ep_world,
+                group=self.ep_group)
+            out_splits = recv_counts.reshape(
+                self.ep_world, self.experts_per_rank).sum(-1).tolist()
+            incoming = all_to_all(dispatched, out_splits, in_splits,
+                                  self.ep_group)
+            # incoming is grouped by (source rank, local expert); regroup
+            # by local expert
+            blocks = []
+            offsets = []
+            off = 0
+            rc = recv_counts.reshape(self.ep_world, self.experts_per_rank)
+            for sr in range(self.ep_world):
+                for le in range(self.experts_per_rank):
+                    n = int(rc[sr, le])
+                    offsets.append((le, off, n))
+                    off += n
+            outs = incoming.new_empty(incoming.shape)
+            for le in range(self.experts_per_rank):
+                segs = [incoming[o:o + n] for (l, o, n) in offsets
+                        if l == le and n > 0]
+                if segs:
+                    y = self.local_experts[le](torch.cat(segs))
+                    # scatter back into position
+                    pos = 0
+                    for (l, o, n) in offsets:
+                        if l == le and n > 0:
+                            outs[o:o + n] = y[pos:pos + n]
+                            pos += n
+            returned = all_to_all(outs, in_splits, out_splits,
+                                  self.ep_group)
+        else:
+            returned = dispatched.new_empty(dispatched.shape)
+            off = 0
+            for e in range(cfg.num_experts):
+                n = int(counts[e])
+                if n > 0:
+                    returned[off:off + n] = self.local_experts[e](
+                        dispatched[off:off + n])
+                off += n
+
+        # combine: weight each routed copy and scatter-add to source token
+        w_sorted = weights.reshape(-1)[order].to(returned.dtype)
+        contrib = returned * w_sorted.unsqueeze(-1)
+        out = torch.zeros_like(flat)
+        out.index_add_(0, token_idx, contrib)
+        return out.reshape(B, S, h)
+
+
+class MixtralDecoderLayer(nn.Module):
+    def __init__(self, cfg: MixtralConfig, ep_group=None):
+        super().__init__()
+        self.cfg = cfg
+        h, hd = cfg.hidden_size, cfg.head_dim
+        self.input_norm = RMSNorm(h, cfg.rms_eps)
+        self.q_proj = Linear(h, cfg.num_heads * hd)
+        self.k_proj = Linear(h, cfg.num_kv_heads * hd)
+        self.v_proj = Linear(h, cfg.num_kv_heads * hd)
+        self.o_proj = Linear(cfg.num_heads * hd, h)
+        self.post_norm = RMSNorm(h, cfg.rms_eps)
+        self.moe = MoELayer(cfg, ep_group)
+
+    def forward(self, x, cos_t, sin_t):
+        cfg = self.cfg
+        B, S, _ = x.shape
+        res = x
+        y = self.input_norm(x)
+        q = K.rope(self.q_proj(y).view(B, S, cfg.num_heads, cfg.head_dim),
+                   cos_t, sin_t).transpose(1, 2)
+        kk = K.rope(self.k_proj(y).view(B, S, cfg.num_kv_heads,
+                                        cfg.head_dim),
+                    cos_t, sin_t).transpose(1, 2)
+        v = self.v_proj(y).view(B, S, cfg.num_kv_heads,
+                                cfg.head_dim).transpose(1, 2)
+        o = K.attention(q, kk, v).transpose(1, 2).reshape(
+            B, S, cfg.num_heads * cfg.head_dim)
+        x = res + self.o_proj(o)
+        x = x + self.moe(self.post_norm(x))
+        return x
+
+
+class MixtralForCausalLM(nn.Module):
+    def __init__(self, cfg: MixtralConfig, ep_group=None):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                  dtype=torch.bfloat16)
+        self.layers = nn.ModuleList(
+            MixtralDecoderLayer(cfg, ep_group)
+            for _ in range(cfg.num_layers))
+        self.final_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
+        self.lm_head = Linear(cfg.hidden_size, cfg.vocab_size)
+        cos_t, sin_t = K.rope_tables(cfg.max_seq_len, cfg.head_dim,
+                                     cfg.rope_theta)
+        self.register_buffer("cos_t", cos_t, persistent=False)
+        self.register_buffer("sin_t", sin_t, persistent=False)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        for p in self.parameters():
+            if p.dim() >= 2:
+                nn.init.normal_(p, mean=0.0, std=0.02)
+
+    def forward(self, tokens, targets=None):
+        x = self.embed(tokens)
+        for layer in self.layers:
+            x = layer(x, self.cos_t, self.sin_t)
+        x = self.final_norm(x)
+        logits = self.lm_head(x)
+        if targets is None:
+            return logits
+        B, S, V = logits.shape
+        loss = K.cross_entropy(logits.reshape(B * S, V),
+                               targets.reshape(B * S))
+        return loss.mean()
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())

@@ -11,6 +11,7 @@
 #include "adam.hip"
 #include "cross_entropy.hip"
 #include "attention.hip"
+#include "debug_kernels.hip"
 
 namespace {
 
@@ -230,6 +231,24 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
 
 }  // namespace
 
+
+namespace {
+torch::Tensor dbg_st(torch::Tensor kmat, torch::Tensor qmat) {
+  auto out = torch::empty({64, 32},
+                          kmat.options().dtype(torch::kFloat32));
+  dbg_st_kernel<<<1, 256, 0, cur_stream()>>>(bf(kmat), bf(qmat),
+                                             out.data_ptr<float>());
+  return out;
+}
+torch::Tensor dbg_dv(torch::Tensor pmat, torch::Tensor bmat) {
+  auto out = torch::empty({64, 128},
+                          pmat.options().dtype(torch::kFloat32));
+  dbg_dv_kernel<<<1, 256, 0, cur_stream()>>>(bf(pmat), bf(bmat),
+                                             out.data_ptr<float>());
+  return out;
+}
+}  // namespace
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");
@@ -242,4 +261,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (causal, GQA)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward");
+  m.def("dbg_st", &dbg_st, "debug S^T path");
+  m.def("dbg_dv", &dbg_dv, "debug dV path");
 }

@@ -25,9 +25,16 @@ class FlatParamModel(object):
 
     def __init__(self, module, bucket_mb=64):
         self.module = module
-        params = [p for p in module.parameters() if p.requires_grad]
-        # stable ordering: reverse autograd-completion order roughly matches
-        # registration order reversed; buckets are launched as they fill
+        all_params = [p for p in module.parameters() if p.requires_grad]
+        # params marked _mfx_no_sync (e.g. expert-parallel weights whose
+        # grads are already complete after the token all-to-all) are placed
+        # AFTER the sync region so buckets never cover them
+        sync = [p for p in all_params
+                if not getattr(p, "_mfx_no_sync", False)]
+        nosync = [p for p in all_params
+                  if getattr(p, "_mfx_no_sync", False)]
+        params = sync + nosync
+        self.n_sync_params = len(sync)
         self.params = params
         total = sum(self._padded(p.numel()) for p in params)
         device = params[0].device
@@ -46,15 +53,16 @@ class FlatParamModel(object):
             self.offsets.append(offset)
             offset += self._padded(n)
 
-        # buckets: contiguous ranges of the flat buffer
+        # buckets: contiguous ranges of the SYNC region of the flat buffer
         bucket_elems = (bucket_mb << 20) // self.flat_param.element_size()
         self.buckets = []  # (start, end, last_param_index)
         start = 0
-        for i, p in enumerate(params):
+        for i, p in enumerate(params[:self.n_sync_params]):
             end = self.offsets[i] + self._padded(p.numel())
-            if end - start >= bucket_elems or i == len(params) - 1:
+            if end - start >= bucket_elems or i == self.n_sync_params - 1:
                 self.buckets.append([start, end, i])
                 start = end
+        self.sync_end = self.buckets[-1][1] if self.buckets else 0
         self._pending = []
         self._hooks = []
 
@@ -95,7 +103,7 @@ class FlatParamModel(object):
                 param_to_bucket[pi] = bi
             prev = last
 
-        for pi, p in enumerate(self.params):
+        for pi, p in enumerate(self.params[:self.n_sync_params]):
             bi = param_to_bucket[pi]
 
             def hook(_param, bi=bi):
@@ -121,7 +129,8 @@ class FlatParamModel(object):
                 self._bucket_done_count[bi] = 0
         elif (dist.is_available() and dist.is_initialized()
               and dist.get_world_size() > 1):
-            dist.all_reduce(self.flat_grad, op=dist.ReduceOp.AVG)
+            dist.all_reduce(self.flat_grad[:self.sync_end],
+                            op=dist.ReduceOp.AVG)
 
 
 class FusedAdamW(object):

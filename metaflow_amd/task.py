@@ -237,6 +237,15 @@ class MFXTask(object):
         if node.type == "join":
             inputs = Inputs([TaskInput(ds) for ds in input_dss])
 
+        # ---- unbounded-foreach control task ---------------------------------
+        # (reference protocol: plugins/test_unbounded_foreach_decorator.py:99
+        # + runtime.py:1178-1264 — the control task spawns/waits mappers and
+        # persists _control_mapper_tasks for the join)
+        if ubf_context == UBF_CONTROL:
+            return self._run_ubf_control(
+                flow, node, output, input_dss, run_id, step_name, task_id,
+                retry_count)
+
         # ---- decorator hooks + user code ------------------------------------
         task_ok = True
         error = None
@@ -327,6 +336,83 @@ class MFXTask(object):
             output.done()
         else:
             raise TaskFailed(error)
+
+    def _run_ubf_control(self, flow, node, output, input_dss, run_id,
+                         step_name, task_id, retry_count):
+        """Control side of an unbounded foreach: enumerate the UBF input,
+        spawn one mapper subprocess per item (same step, task ids
+        <ctrl>_mapper_<i>), wait, persist _control_mapper_tasks."""
+        import subprocess
+
+        from .config import MAX_WORKERS
+
+        parent = input_dss[0]
+        ptrans = parent.load_metadata("transition") or {}
+        var = ptrans.get("foreach")
+        seq = getattr(flow, var)
+        try:
+            num = len(seq)
+        except TypeError:
+            num = len(list(iter(seq)))
+
+        def mapper_cmd(i, mapper_id):
+            cmd = list(sys.argv)
+            cmd.insert(0, sys.executable)
+
+            def replace(flag, value):
+                if flag in cmd:
+                    cmd[cmd.index(flag) + 1] = value
+                else:
+                    cmd.extend([flag, value])
+
+            replace("--task-id", mapper_id)
+            replace("--split-index", str(i))
+            replace("--ubf-context", UBF_TASK)
+            return cmd
+
+        mapper_ids = ["%s_mapper_%d" % (task_id, i) for i in range(num)]
+        procs = {}
+        failed = []
+        i = 0
+        while i < num or procs:
+            while i < num and len(procs) < MAX_WORKERS:
+                procs[i] = subprocess.Popen(mapper_cmd(i, mapper_ids[i]))
+                i += 1
+            done = []
+            for idx, proc in procs.items():
+                rc = proc.poll()
+                if rc is not None:
+                    done.append(idx)
+                    if rc != 0:
+                        failed.append(idx)
+            for idx in done:
+                del procs[idx]
+            if not done:
+                import time
+
+                time.sleep(0.05)
+        if failed:
+            raise TaskFailed(MFXException(
+                "UBF mappers %s failed." % failed))
+
+        mapper_paths = ["%s/%s/%s" % (run_id, step_name, mid)
+                        for mid in mapper_ids]
+        flow._control_mapper_tasks = mapper_paths
+        flow._transition = {
+            "out_funcs": list(node.out_funcs),
+            "foreach": None,
+            "condition": None,
+            "num_parallel": None,
+            "num_splits": None,
+        }
+        flow._task_ok = True
+        output.persist(flow)
+        output.save_metadata("transition", flow._transition)
+        output.save_metadata("control_mapper_tasks", mapper_paths)
+        output.save_metadata("attempt_ok", {"ok": True, "control": True})
+        self.metadata.register_metadata(run_id, step_name, task_id,
+                                        retry_count, {"attempt_ok": True})
+        output.done()
 
     @staticmethod
     def _exec_step_function(func, flow, inputs=None):

@@ -89,3 +89,17 @@ def test_gang_train_flow_cpu(tmp_datastore):
     idx = read_artifact(tmp_datastore, "TrainFlow", run_id, "train",
                         "_checkpoint_final_index")
     assert "flat_param" in idx and "adam_m" in idx
+
+
+def test_unbounded_foreach(tmp_datastore):
+    run_flow("ubf_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "UBFFlow")
+    assert read_artifact(tmp_datastore, "UBFFlow", run_id, "join",
+                         "total") == 30
+    # control + 3 mappers exist under the work step
+    work_dir = os.path.join(tmp_datastore, "UBFFlow", run_id, "work")
+    tasks = os.listdir(work_dir)
+    assert len([t for t in tasks if "_mapper_" in t]) == 3
+    ctl = read_artifact(tmp_datastore, "UBFFlow", run_id, "work",
+                        "_control_mapper_tasks")
+    assert len(ctl) == 3
