@@ -28,7 +28,8 @@ def parse_args():
                    help="per-GPU micro batch")
     p.add_argument("--seq", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3-8b",
-                   choices=["llama3-8b", "llama3-70b", "tiny"])
+                   choices=["llama3-8b", "llama3-70b", "tiny",
+                            "mixtral-8x7b", "mixtral-tiny"])
     p.add_argument("--bucket-mb", type=int, default=64)
     p.add_argument("--lr", type=float, default=3e-4)
     return p.parse_args()
@@ -61,16 +62,25 @@ def main():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         init_process_group_from_env()
 
-    cfg = {
-        "llama3-8b": LlamaConfig.llama3_8b,
-        "llama3-70b": LlamaConfig.llama3_70b,
-        "tiny": LlamaConfig.tiny,
-    }[args.model]()
+    from metaflow_amd.models.mixtral import (
+        MixtralConfig,
+        MixtralForCausalLM,
+    )
+
+    model_factories = {
+        "llama3-8b": (LlamaConfig.llama3_8b, LlamaForCausalLM),
+        "llama3-70b": (LlamaConfig.llama3_70b, LlamaForCausalLM),
+        "tiny": (LlamaConfig.tiny, LlamaForCausalLM),
+        "mixtral-8x7b": (MixtralConfig.mixtral_8x7b, MixtralForCausalLM),
+        "mixtral-tiny": (MixtralConfig.tiny, MixtralForCausalLM),
+    }
+    cfg_fn, model_cls = model_factories[args.model]
+    cfg = cfg_fn()
     seq = min(args.seq, cfg.max_seq_len)
 
-    torch.manual_seed(1234 + rank)
+    torch.manual_seed(1234)  # same init on all ranks (DP)
     t0 = time.time()
-    model = LlamaForCausalLM(cfg).to(device)
+    model = model_cls(cfg).to(device)
     flat = FlatParamModel(model, bucket_mb=args.bucket_mb)
     flat.install_overlap_hooks()
     opt = FusedAdamW(flat, lr=args.lr)
@@ -79,8 +89,9 @@ def main():
               % (args.model, model.num_params() / 1e9, time.time() - t0),
               flush=True)
 
-    # synthetic batch (fixed per rank: loss must fall, proving a real
-    # fwd+bwd+optimizer step is in the timed region)
+    # synthetic batch (fixed per rank but distinct across ranks: loss must
+    # fall, proving a real fwd+bwd+optimizer step is in the timed region)
+    torch.manual_seed(5678 + rank)
     tokens = torch.randint(0, cfg.vocab_size, (args.batch, seq + 1),
                            device=device)
     inp, tgt = tokens[:, :-1], tokens[:, 1:].contiguous()

@@ -204,6 +204,46 @@ def main(flow):
             traceback.print_exc()
             sys.exit(1)
 
+    @cli.command(help="Re-run ONE step of a finished run against its "
+                      "original input artifacts (debugging).")
+    @click.argument("step_name")
+    @click.option("--run-id", "origin_run_id", required=True,
+                  help="The finished run to spin against.")
+    def spin(step_name, origin_run_id):
+        from .task import PARAMETERS_STEP as PSTEP
+
+        node = state.graph[step_name]
+        if node.type == "join" and len(node.in_funcs) > 1:
+            in_steps = sorted(node.in_funcs)
+        else:
+            in_steps = sorted(node.in_funcs) or [PSTEP]
+        input_paths = []
+        for s in in_steps:
+            tasks = state.flow_datastore.list_tasks(origin_run_id, s)
+            if not tasks:
+                raise MFXException(
+                    "Origin run %s has no tasks for step %s"
+                    % (origin_run_id, s))
+            input_paths.append("%s/%s/%s" % (origin_run_id, s,
+                                             sorted(tasks)[0]))
+        spin_run_id = "spin%s" % origin_run_id
+        task = MFXTask(state.flow, state.graph, state.flow_datastore,
+                       state.metadata)
+        task.run_step(step_name, spin_run_id, "1", input_paths, None, 0, 0)
+        click.echo("spin task done: %s/%s/1" % (spin_run_id, step_name))
+
+    @cli.command(name="tag", help="Mutate run tags: tag add/remove RUN_ID "
+                                  "TAG [TAG...]")
+    @click.argument("action", type=click.Choice(["add", "remove"]))
+    @click.argument("run_id")
+    @click.argument("tags", nargs=-1, required=True)
+    def tag_cmd(action, run_id, tags):
+        if action == "add":
+            state.metadata.add_run_tags(run_id, list(tags))
+        else:
+            state.metadata.remove_run_tags(run_id, list(tags))
+        click.echo("tags updated")
+
     @cli.command(help="Show the flow structure.")
     def show():
         click.echo("Flow: %s" % flow_cls.__name__)

@@ -191,6 +191,19 @@ class Run(MetaflowObject):
         except NotFoundError:
             return None
 
+    @property
+    def code_package_key(self):
+        return (self._meta.get_run(self.id) or {}).get("code_package_key")
+
+    def extract_code(self, dest):
+        """Extract this run's code snapshot into dest (reproducibility)."""
+        from ..package import CodePackage
+
+        key = self.code_package_key
+        if key is None:
+            raise NotFoundError("Run %s has no code package." % self.pathspec)
+        return CodePackage.extract(self._fds, key, dest)
+
 
 class Step(MetaflowObject):
     def __init__(self, pathspec):

@@ -70,6 +70,11 @@ class LocalMetadataProvider(object):
             "status": "running",
         })
 
+    def update_run_info(self, run_id, extra):
+        info = self._load(self._run_path(run_id)) or {}
+        info.update(extra)
+        self._save(self._run_path(run_id), info)
+
     def register_run_done(self, run_id, success):
         info = self._load(self._run_path(run_id)) or {}
         info["status"] = "successful" if success else "failed"

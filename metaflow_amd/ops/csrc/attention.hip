@@ -209,7 +209,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int r = 0; r < 4; ++r)
         *(short*)((char*)pt + swz(wid * 16 + lgrp * 4 + r, n * 16 + l16,
                                   BKV * 2)) = f2bf(s[n][r]);
-    __builtin_amdgcn_s_waitcnt(0);  // lgkm drain before own-wave reads
+    __syncthreads();  // real barrier: per-lane AA would otherwise allow
+    // the compiler to reorder the cross-lane LDS exchange
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 pa = frag8(pt, wid * 16 + l16, kk * 32 + lgrp * 8, BKV * 2);
@@ -268,8 +269,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     short* __restrict__ dk, short* __restrict__ dv,
     int B, int H, int Hkv, int S, float scale) {
   constexpr int BKV = 64, BQ2 = 32;
-  __shared__ short kt[BKV * ATT_D];       // K rm (A for S^T)
-  __shared__ short vt[BKV * ATT_D];       // V rm (A for dP^T)
   __shared__ short qtile[BQ2 * ATT_D];    // Q rm (B for S^T)
   __shared__ short qtt[ATT_D * BQ2];      // Q^T (B for dK)
   __shared__ short dot[BQ2 * ATT_D];      // dO rm (B for dP^T)
@@ -289,8 +288,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 
   const long long kvoff =
       (((long long)b * Hkv + hk) * S + kvb * BKV) * ATT_D;
-  stage_rm<BKV, BLOCK>(kt, k + kvoff, ATT_D);
-  stage_rm<BKV, BLOCK>(vt, v + kvoff, ATT_D);
 
   f4 acc_dk[8], acc_dv[8];
 #pragma unroll
@@ -298,14 +295,16 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     acc_dk[n] = (f4){0, 0, 0, 0};
     acc_dv[n] = (f4){0, 0, 0, 0};
   }
-  __syncthreads();
 
-  // preload K/V fragments (A operands, rows wid*16 + l16)
+  // preload K/V A-fragments straight from global (each lane 16 B of its
+  // strip row; one-time cost, saves 32 KiB of LDS)
   bf16x8 kf[4], vf[4];
 #pragma unroll
   for (int kk = 0; kk < 4; ++kk) {
-    kf[kk] = frag8(kt, wid * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
-    vf[kk] = frag8(vt, wid * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
+    const long long row_off =
+        kvoff + (long long)(wid * 16 + l16) * ATT_D + kk * 32 + lgrp * 8;
+    kf[kk] = *(const bf16x8*)(k + row_off);
+    vf[kk] = *(const bf16x8*)(v + row_off);
   }
 
   const int jq0 = (kvb * BKV) / BQ2;
@@ -366,7 +365,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
           *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
                                        n * 16 + l16, BQ2 * 2)) =
               f2bf(st[n][r]);
-      __builtin_amdgcn_s_waitcnt(0);
+      __syncthreads();
       {
         bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BQ2 * 2);
 #pragma unroll
@@ -391,7 +390,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
           *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
                                        n * 16 + l16, BQ2 * 2)) =
               f2bf(st[n][r]);
-      __builtin_amdgcn_s_waitcnt(0);
+      __syncthreads();
       {
         bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BQ2 * 2);
 #pragma unroll
@@ -517,7 +516,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
                                      n * 16 + l16, BKV2 * 2)) =
             f2bf(s[n][r]);
-    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
     {
       bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BKV2 * 2);
 #pragma unroll

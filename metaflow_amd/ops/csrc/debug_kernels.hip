@@ -75,7 +75,7 @@ __global__ __launch_bounds__(256) void dbg_dv_kernel(
     for (int r = 0; r < 4; ++r)
       *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r, n * 16 + l16,
                                    64)) = f2bf(st[n][r]);
-  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
 
   f4_ acc[8];
 #pragma unroll

@@ -53,6 +53,7 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
   auto inv = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   rmsnorm_fwd_kernel<kBlock><<<(int)rows, kBlock, 0, cur_stream()>>>(
       bf(x), bf(w), bfm(y), inv.data_ptr<float>(), H, (float)eps);
+  HIP_CHECK_KERNEL();
   return {y, inv};
 }
 
@@ -70,6 +71,7 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
   rmsnorm_bwd_kernel<kBlock><<<grid, kBlock, lds, cur_stream()>>>(
       bf(x), bf(w), bf(dy), inv_rms.data_ptr<float>(), bfm(dx),
       dw.data_ptr<float>(), (int)rows, H);
+  HIP_CHECK_KERNEL();
   return {dx, dw};
 }
 
@@ -84,6 +86,7 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
   rope_kernel<<<grid_for(total), kBlock, 0, cur_stream()>>>(
       bf(x), bfm(y), cos_t.data_ptr<float>(), sin_t.data_ptr<float>(), rows,
       D, (int)rows_per_pos, (int)seqlen, (int)pos0, backward ? -1.f : 1.f);
+  HIP_CHECK_KERNEL();
   return y;
 }
 
@@ -96,6 +99,7 @@ torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u) {
   long long n8 = g.numel() / 8;
   swiglu_fwd_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
       bf(g), bf(u), bfm(y), n8);
+  HIP_CHECK_KERNEL();
   return y;
 }
 
@@ -106,6 +110,7 @@ std::vector<torch::Tensor> swiglu_bwd(torch::Tensor g, torch::Tensor u,
   long long n8 = g.numel() / 8;
   swiglu_bwd_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
       bf(g), bf(u), bf(dy), bfm(dg), bfm(du), n8);
+  HIP_CHECK_KERNEL();
   return {dg, du};
 }
 
@@ -116,6 +121,7 @@ torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
   long long n8 = a.numel() / 8;
   add_bf16_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
       bf(a), bf(b), bfm(y), n8);
+  HIP_CHECK_KERNEL();
   return y;
 }
 
@@ -135,12 +141,14 @@ void adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
         bfm(p), bf(g), m.data_ptr<float>(), v.data_ptr<float>(), master_ptr,
         n, (float)lr, (float)beta1, (float)beta2, (float)eps,
         (float)weight_decay, bias_c1, bias_c2, (float)grad_scale);
+  HIP_CHECK_KERNEL();
   } else {
     adamw_f32_kernel<<<grid_for(n), kBlock, 0, cur_stream()>>>(
         p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
         v.data_ptr<float>(), n, (float)lr, (float)beta1, (float)beta2,
         (float)eps, (float)weight_decay, bias_c1, bias_c2,
         (float)grad_scale);
+  HIP_CHECK_KERNEL();
   }
 }
 
@@ -159,6 +167,7 @@ std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
       bf(logits), reinterpret_cast<const long long*>(targets.data_ptr<int64_t>()), loss.data_ptr<float>(),
       rmax.data_ptr<float>(), rlse.data_ptr<float>(), N, V,
       (long long)ignore_index);
+  HIP_CHECK_KERNEL();
   return {loss, rlse};
 }
 
@@ -172,6 +181,7 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
   cross_entropy_bwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
       bf(logits), reinterpret_cast<const long long*>(targets.data_ptr<int64_t>()), row_lse.data_ptr<float>(),
       dloss.data_ptr<float>(), bfm(dlogits), N, V, (long long)ignore_index);
+  HIP_CHECK_KERNEL();
   return dlogits;
 }
 
@@ -192,6 +202,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   attn_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
       bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
       (float)scale);
+  HIP_CHECK_KERNEL();
   return {o, lse};
 }
 
@@ -212,6 +223,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
     long long blocks = (rows + rows_per_block - 1) / rows_per_block;
     attn_bwd_delta_kernel<<<(int)blocks, kBlock, 0, cur_stream()>>>(
         bf(dout), bf(o), delta.data_ptr<float>(), rows);
+  HIP_CHECK_KERNEL();
   }
   {
     dim3 grid(S / 64, Hkv, B);
@@ -219,12 +231,14 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
         bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
         delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
         (float)scale);
+  HIP_CHECK_KERNEL();
   }
   {
     dim3 grid(S / 64, H, B);
     attn_bwd_dq_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
         bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
         delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale);
+  HIP_CHECK_KERNEL();
   }
   return {dq, dk, dv};
 }
@@ -238,6 +252,7 @@ torch::Tensor dbg_st(torch::Tensor kmat, torch::Tensor qmat) {
                           kmat.options().dtype(torch::kFloat32));
   dbg_st_kernel<<<1, 256, 0, cur_stream()>>>(bf(kmat), bf(qmat),
                                              out.data_ptr<float>());
+  HIP_CHECK_KERNEL();
   return out;
 }
 torch::Tensor dbg_dv(torch::Tensor pmat, torch::Tensor bmat) {
@@ -245,6 +260,7 @@ torch::Tensor dbg_dv(torch::Tensor pmat, torch::Tensor bmat) {
                           pmat.options().dtype(torch::kFloat32));
   dbg_dv_kernel<<<1, 256, 0, cur_stream()>>>(bf(pmat), bf(bmat),
                                              out.data_ptr<float>());
+  HIP_CHECK_KERNEL();
   return out;
 }
 }  // namespace

@@ -255,8 +255,23 @@ class NativeRuntime(object):
         return new_ds
 
     # -------------------------------------------------------------- executing
+    def _save_code_package(self):
+        """Content-addressed code snapshot for reproducibility (reference
+        MetaflowPackage, package/__init__.py:43)."""
+        try:
+            from .package import CodePackage
+
+            pkg = CodePackage(os.path.dirname(os.path.abspath(
+                self.flow_file)) or ".")
+            _uri, key = pkg.save(self.flow_datastore)
+            self.metadata.update_run_info(self.run_id,
+                                          {"code_package_key": key})
+        except Exception as e:  # packaging must never fail the run
+            self._echo("code packaging skipped: %s" % e)
+
     def execute(self):
         self.metadata.heartbeat(self.run_id)
+        self._save_code_package()
         if self.clone_run_id:
             self._origin_index = self._build_origin_index()
             # clone _parameters from origin (parameters are fixed on resume)
