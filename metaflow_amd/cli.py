@@ -286,10 +286,21 @@ def main(flow):
     @cli.command(help="Show logs of a task: PATHSPEC = run/step/task")
     @click.argument("pathspec")
     @click.option("--stderr", is_flag=True, default=False)
-    def logs(pathspec, stderr):
+    @click.option("--timestamps", is_flag=True, default=False)
+    def logs(pathspec, stderr, timestamps):
+        from .mflog import parse
+
         parts = pathspec.split("/")
         ds = state.flow_datastore.get_task_datastore(parts[0], parts[1],
                                                      parts[2])
-        click.echo(ds.load_logs("stderr" if stderr else "stdout"))
+        raw = ds.load_logs("stderr" if stderr else "stdout")
+        for line in raw.splitlines():
+            p = parse(line)
+            if p is None:
+                click.echo(line)
+            elif timestamps:
+                click.echo("%s %s" % (p.ts, p.msg))
+            else:
+                click.echo(p.msg)
 
     cli(standalone_mode=True)

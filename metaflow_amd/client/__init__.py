@@ -298,13 +298,30 @@ class Task(MetaflowObject):
     def foreach_stack(self):
         return self._ds.load_metadata("foreach_stack") or []
 
+    @staticmethod
+    def _plain(raw):
+        from ..mflog import parse
+
+        out = []
+        for line in raw.splitlines():
+            p = parse(line)
+            out.append(p.msg if p else line)
+        return "\n".join(out) + ("\n" if out else "")
+
     @property
     def stdout(self):
-        return self._ds.load_logs("stdout")
+        return self._plain(self._ds.load_logs("stdout"))
 
     @property
     def stderr(self):
-        return self._ds.load_logs("stderr")
+        return self._plain(self._ds.load_logs("stderr"))
+
+    def loglines(self, stream="stdout"):
+        """Structured MFLogline records for this task."""
+        from ..mflog import merge_logs
+
+        raw = self._ds.load_logs(stream).encode()
+        return list(merge_logs([raw]))
 
     @property
     def parent(self):

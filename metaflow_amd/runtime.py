@@ -270,7 +270,14 @@ class NativeRuntime(object):
             self._echo("code packaging skipped: %s" % e)
 
     def execute(self):
+        from .sidecar import SidecarSubProcess
+
         self.metadata.heartbeat(self.run_id)
+        self._heartbeat_sidecar = SidecarSubProcess("heartbeat", {
+            "flow_name": self.flow_cls.__name__,
+            "run_id": self.run_id,
+            "datastore_root": self.flow_datastore.datastore_root,
+        })
         self._save_code_package()
         if self.clone_run_id:
             self._origin_index = self._build_origin_index()
@@ -307,6 +314,10 @@ class NativeRuntime(object):
             raise
         finally:
             self.metadata.register_run_done(self.run_id, not self._failed)
+            try:
+                self._heartbeat_sidecar.terminate()
+            except Exception:
+                pass
 
         if self._failed:
             raise TaskFailedException(
@@ -542,13 +553,18 @@ class NativeRuntime(object):
 
     def _worker_exited(self, worker, rc):
         spec = worker.spec
-        # persist captured logs into the task's attempt
+        # persist captured logs (mflog-structured) into the task's attempt
         try:
+            from . import mflog
+
+            source = "%s/%s" % (spec.step, spec.task_id)
             log_ds = self.flow_datastore.get_task_datastore(
                 self.run_id, spec.step, spec.task_id,
                 attempt=spec.retry_count, mode="w")
-            log_ds.save_logs("stdout", worker.stdout_buf.get_bytes())
-            log_ds.save_logs("stderr", worker.stderr_buf.get_bytes())
+            log_ds.save_logs("stdout", mflog.decorate_stream(
+                source, worker.stdout_buf.get_bytes()))
+            log_ds.save_logs("stderr", mflog.decorate_stream(
+                source, worker.stderr_buf.get_bytes()))
         except Exception:
             pass
 
