@@ -80,7 +80,8 @@ def main():
 
     torch.manual_seed(1234)  # same init on all ranks (DP)
     t0 = time.time()
-    model = model_cls(cfg).to(device)
+    with torch.device(device):  # construct + random-init directly on GPU
+        model = model_cls(cfg)
     flat = FlatParamModel(model, bucket_mb=args.bucket_mb)
     flat.install_overlap_hooks()
     opt = FusedAdamW(flat, lr=args.lr)
