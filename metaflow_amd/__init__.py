@@ -21,6 +21,7 @@ from .plugins.resources_decorator import resources
 from .plugins.environment_decorator import environment
 from .plugins.parallel_decorator import parallel, torch_parallel
 from .plugins.checkpoint_decorator import checkpoint
+from .plugins.card_decorator import card
 from .plugins.project_decorator import project
 from .plugins.schedule_decorator import schedule
 from .client import (
@@ -55,6 +56,7 @@ __all__ = [
     "parallel",
     "torch_parallel",
     "checkpoint",
+    "card",
     "project",
     "schedule",
     "Metaflow",

@@ -14,6 +14,7 @@ from .parallel_decorator import ParallelDecorator, TorchParallelDecorator
 from .checkpoint_decorator import CheckpointDecorator
 from .project_decorator import ProjectDecorator
 from .schedule_decorator import ScheduleDecorator
+from .card_decorator import CardDecorator
 
 STEP_DECORATORS = {
     cls.name: cls
@@ -26,6 +27,7 @@ STEP_DECORATORS = {
         ParallelDecorator,
         TorchParallelDecorator,
         CheckpointDecorator,
+        CardDecorator,
     )
 }
 

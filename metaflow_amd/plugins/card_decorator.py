@@ -1,0 +1,122 @@
+"""@card: per-task HTML report stored in the task datastore.
+
+Parity target: /root/reference/metaflow/plugins/cards/ (3.9 kLoC) scoped to
+the useful core: a self-contained HTML card per task with task metadata,
+artifact summaries, logs, and (on GPU boxes) the telemetry the gpu_monitor
+sidecar sampled. ``current.card.append(html_or_text)`` adds user content;
+the card is rendered in task_finished and readable via
+``Task.card_html`` / the `card` CLI.
+"""
+
+import html
+import json
+import time
+
+from ..decorators import StepDecorator, make_step_decorator
+
+_CARD_TEMPLATE = """<!DOCTYPE html>
+<html><head><meta charset="utf-8"><title>{title}</title>
+<style>
+body {{ font-family: -apple-system, Segoe UI, sans-serif; margin: 2rem;
+       background: #fafafa; color: #222; }}
+h1 {{ font-size: 1.3rem; }} h2 {{ font-size: 1.05rem; margin-top: 1.5rem; }}
+table {{ border-collapse: collapse; width: 100%; background: #fff; }}
+td, th {{ border: 1px solid #ddd; padding: 6px 10px; font-size: 0.9rem;
+          text-align: left; }}
+th {{ background: #f0f0f0; }}
+pre {{ background: #1e1e1e; color: #d4d4d4; padding: 1rem;
+       overflow-x: auto; font-size: 0.8rem; }}
+.badge {{ display: inline-block; padding: 2px 10px; border-radius: 10px;
+          color: #fff; background: {badge}; font-size: 0.85rem; }}
+</style></head><body>
+<h1>{title} <span class="badge">{status}</span></h1>
+<table>
+<tr><th>pathspec</th><td>{pathspec}</td></tr>
+<tr><th>attempt</th><td>{attempt}</td></tr>
+<tr><th>generated</th><td>{ts}</td></tr>
+</table>
+<h2>Artifacts</h2>
+<table><tr><th>name</th><th>type</th><th>size (bytes)</th><th>sha</th></tr>
+{artifact_rows}
+</table>
+{user_sections}
+</body></html>
+"""
+
+
+class CardBuilder(object):
+    def __init__(self):
+        self.sections = []
+
+    def append(self, content, title=None):
+        """Append a section: raw HTML (str starting with '<') or text."""
+        self.sections.append((title, content))
+
+    def render_sections(self):
+        out = []
+        for title, content in self.sections:
+            if title:
+                out.append("<h2>%s</h2>" % html.escape(str(title)))
+            c = str(content)
+            if c.lstrip().startswith("<"):
+                out.append(c)
+            else:
+                out.append("<pre>%s</pre>" % html.escape(c))
+        return "\n".join(out)
+
+
+def render_card(task_datastore, step_name, ok, builder=None):
+    rows = []
+    for name in sorted(task_datastore.artifact_names()):
+        if name.startswith("_"):
+            continue
+        info = task_datastore.artifact_info(name) or {}
+        rows.append(
+            "<tr><td>%s</td><td>%s</td><td>%s</td><td><code>%s</code>"
+            "</td></tr>"
+            % (html.escape(name), html.escape(str(info.get("type", ""))),
+               info.get("size", ""), (info.get("sha") or "")[:16]))
+    return _CARD_TEMPLATE.format(
+        title="%s" % step_name,
+        status="OK" if ok else "FAILED",
+        badge="#2e7d32" if ok else "#c62828",
+        pathspec=task_datastore.pathspec,
+        attempt=task_datastore.attempt,
+        ts=time.strftime("%Y-%m-%d %H:%M:%S UTC", time.gmtime()),
+        artifact_rows="\n".join(rows),
+        user_sections=builder.render_sections() if builder else "",
+    )
+
+
+class CardDecorator(StepDecorator):
+    name = "card"
+    defaults = {"id": "default"}
+    allow_multiple = True
+
+    def task_pre_step(self, step_name, task_datastore, metadata, run_id,
+                      task_id, flow, graph, retry_count,
+                      max_user_code_retries, ubf_context, inputs):
+        from ..current import current
+
+        self._builder = CardBuilder()
+        self._ds = task_datastore
+        current._update_env({"card": self._builder})
+
+    def task_finished(self, step_name, flow, graph, is_task_ok, retry_count,
+                      max_user_code_retries):
+        try:
+            html_doc = render_card(self._ds, step_name, is_task_ok,
+                                   self._builder)
+            self._ds.save_metadata(
+                "card_%s" % self.attributes.get("id", "default"),
+                {"html": html_doc})
+        except Exception:
+            pass
+
+
+card = make_step_decorator(CardDecorator)
+
+
+def get_card(task_datastore, card_id="default"):
+    meta = task_datastore.load_metadata("card_%s" % card_id)
+    return meta["html"] if meta else None

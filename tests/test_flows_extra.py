@@ -103,3 +103,18 @@ def test_unbounded_foreach(tmp_datastore):
     ctl = read_artifact(tmp_datastore, "UBFFlow", run_id, "work",
                         "_control_mapper_tasks")
     assert len(ctl) == 3
+
+
+def test_card(tmp_datastore):
+    from metaflow_amd.datastore import FlowDataStore
+    from metaflow_amd.datastore.storage import LocalStorage
+    from metaflow_amd.plugins.card_decorator import get_card
+
+    run_flow("card_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "CardFlow")
+    fds = FlowDataStore("CardFlow", LocalStorage(tmp_datastore))
+    task_id = fds.list_tasks(run_id, "start")[0]
+    ds = fds.get_task_datastore(run_id, "start", task_id)
+    html = get_card(ds)
+    assert html and "metric" in html and "custom html" in html
+    assert "Notes" in html and "OK" in html
