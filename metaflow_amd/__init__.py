@@ -22,6 +22,8 @@ from .plugins.environment_decorator import environment
 from .plugins.parallel_decorator import parallel, torch_parallel
 from .plugins.checkpoint_decorator import checkpoint
 from .plugins.card_decorator import card
+from .plugins.secrets_decorator import secrets
+from .plugins.exit_hook_decorator import exit_hook
 from .plugins.project_decorator import project
 from .plugins.schedule_decorator import schedule
 from .client import (
@@ -57,6 +59,8 @@ __all__ = [
     "torch_parallel",
     "checkpoint",
     "card",
+    "secrets",
+    "exit_hook",
     "project",
     "schedule",
     "Metaflow",

@@ -153,7 +153,18 @@ def main(flow):
                            "run_id": runtime.run_id,
                            "metadata": "local@%s"
                            % state.flow_datastore.datastore_root}, f)
-        runtime.execute()
+        success = False
+        try:
+            runtime.execute()
+            success = True
+        finally:
+            for deco in getattr(flow_cls, "_flow_decorators", []):
+                if deco.name == "exit_hook":
+                    try:
+                        deco.run_hooks(sys.argv[0], success,
+                                       runtime.run_id)
+                    except Exception:
+                        traceback.print_exc()
         return runtime
 
     @cli.command(help="Run the flow locally.")

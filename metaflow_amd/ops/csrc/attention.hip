@@ -91,6 +91,17 @@ DEV void stage_tr(short* dst, const short* src, long long src_row_stride) {
   }
 }
 
+// intra-wave LDS exchange fence: the P/dS round-trips move data between
+// lanes of the SAME wave; DS ops issue in order per wave, so a full block
+// barrier is overkill — we only need the writes drained and the compiler
+// forbidden from reordering the reads above the writes (per-lane alias
+// analysis would otherwise allow it).
+DEV void wave_lds_fence() {
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_wave_barrier();
+  __builtin_amdgcn_sched_barrier(0);
+}
+
 // ============================ FORWARD =====================================
 // BQ=64, BKV=64, 256 threads (4 waves), each wave owns 16 q rows.
 
@@ -209,8 +220,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int r = 0; r < 4; ++r)
         *(short*)((char*)pt + swz(wid * 16 + lgrp * 4 + r, n * 16 + l16,
                                   BKV * 2)) = f2bf(s[n][r]);
-    __syncthreads();  // real barrier: per-lane AA would otherwise allow
-    // the compiler to reorder the cross-lane LDS exchange
+    wave_lds_fence();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 pa = frag8(pt, wid * 16 + l16, kk * 32 + lgrp * 8, BKV * 2);
@@ -365,7 +375,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
           *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
                                        n * 16 + l16, BQ2 * 2)) =
               f2bf(st[n][r]);
-      __syncthreads();
+      wave_lds_fence();
       {
         bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BQ2 * 2);
 #pragma unroll
@@ -390,7 +400,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
           *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
                                        n * 16 + l16, BQ2 * 2)) =
               f2bf(st[n][r]);
-      __syncthreads();
+      wave_lds_fence();
       {
         bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BQ2 * 2);
 #pragma unroll
@@ -516,7 +526,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
                                      n * 16 + l16, BKV2 * 2)) =
             f2bf(s[n][r]);
-    __syncthreads();
+    wave_lds_fence();
     {
       bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BKV2 * 2);
 #pragma unroll

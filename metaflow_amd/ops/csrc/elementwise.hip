@@ -112,27 +112,45 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ x,
 // pos_of_row: row -> position index (seq pos), stride trick avoids a
 // lookup table: pos = (row / heads) % seqlen handled by caller via
 // rows_per_pos.
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+
 __global__ void rope_kernel(const short* __restrict__ x,
                             short* __restrict__ y,
                             const float* __restrict__ cos_t,
                             const float* __restrict__ sin_t,
                             long long rows, int D, int rows_per_pos,
                             int seqlen, int pos0, float sin_sign) {
+  // vectorized: each thread rotates 8 (x1,x2) pairs -> 2x bf16x8 loads,
+  // 2x f32x4 table loads, 2x bf16x8 stores (guide G13)
   const int half = D / 2;
-  const long long total = rows * (long long)half;
+  const int hv = half / 8;  // bf16x8 chunks per half
+  const long long total = rows * (long long)hv;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        idx < total; idx += stride) {
-    const long long row = idx / half;
-    const int d = (int)(idx % half);
+    const long long row = idx / hv;
+    const int d8 = (int)(idx % hv);
     const int pos = pos0 + (int)((row / rows_per_pos) % seqlen);
-    const float c = cos_t[(long long)pos * half + d];
-    const float s = sin_t[(long long)pos * half + d] * sin_sign;
-    const long long base = row * (long long)D;
-    const float x1 = bf2f(x[base + d]);
-    const float x2 = bf2f(x[base + half + d]);
-    y[base + d] = f2bf(x1 * c - x2 * s);
-    y[base + half + d] = f2bf(x2 * c + x1 * s);
+    const long long tbase = (long long)pos * half + d8 * 8;
+    const long long base = row * (long long)D + d8 * 8;
+    bf16x8 v1 = *(const bf16x8*)(x + base);
+    bf16x8 v2 = *(const bf16x8*)(x + base + half);
+    f32x4v c0 = *(const f32x4v*)(cos_t + tbase);
+    f32x4v c1 = *(const f32x4v*)(cos_t + tbase + 4);
+    f32x4v s0 = *(const f32x4v*)(sin_t + tbase);
+    f32x4v s1 = *(const f32x4v*)(sin_t + tbase + 4);
+    bf16x8 o1, o2;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float c = (j < 4 ? c0[j] : c1[j - 4]);
+      const float s = (j < 4 ? s0[j] : s1[j - 4]) * sin_sign;
+      const float x1 = bf2f(v1[j]);
+      const float x2 = bf2f(v2[j]);
+      o1[j] = f2bf(x1 * c - x2 * s);
+      o2[j] = f2bf(x2 * c + x1 * s);
+    }
+    *(bf16x8*)(y + base) = o1;
+    *(bf16x8*)(y + base + half) = o2;
   }
 }
 

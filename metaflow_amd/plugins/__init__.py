@@ -15,6 +15,8 @@ from .checkpoint_decorator import CheckpointDecorator
 from .project_decorator import ProjectDecorator
 from .schedule_decorator import ScheduleDecorator
 from .card_decorator import CardDecorator
+from .secrets_decorator import SecretsDecorator
+from .exit_hook_decorator import ExitHookDecorator
 
 STEP_DECORATORS = {
     cls.name: cls
@@ -28,9 +30,11 @@ STEP_DECORATORS = {
         TorchParallelDecorator,
         CheckpointDecorator,
         CardDecorator,
+        SecretsDecorator,
     )
 }
 
 FLOW_DECORATORS = {
-    cls.name: cls for cls in (ProjectDecorator, ScheduleDecorator)
+    cls.name: cls
+    for cls in (ProjectDecorator, ScheduleDecorator, ExitHookDecorator)
 }
