@@ -111,7 +111,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse, int B, int H, int Hkv, int S, float scale) {
   constexpr int BQ = 64, BKV = 64;
-  __shared__ short qt[BQ * ATT_D];      // swizzled [64][128]
   __shared__ short kt[BKV * ATT_D];     // swizzled [64][128] (N-major for B)
   __shared__ short vtt[ATT_D * BKV];    // V^T swizzled [128][64]
   __shared__ short pt[BQ * BKV];        // P swizzled [64][64]
@@ -128,8 +127,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const long long qoff = (((long long)b * H + h) * S + qb * BQ) * ATT_D;
   const long long kvoff0 = ((long long)b * Hkv + hk) * S * ATT_D;
 
-  stage_rm<BQ, BLOCK>(qt, q + qoff, ATT_D);
-
   // per-wave per-lane state: 4 rows (reg r), replicated across 16 lanes
   float m_run[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
   float l_run[4] = {0.f, 0.f, 0.f, 0.f};
@@ -137,12 +134,14 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
   for (int n = 0; n < 8; ++n) acc_o[n] = (f4){0, 0, 0, 0};
 
-  // preload Q fragments for this wave (rows wid*16 + l16)
-  __syncthreads();
+  // preload Q A-fragments straight from global (one 16 B read per lane per
+  // k-step; keeping Q out of LDS doubles occupancy)
   bf16x8 qf[4];
 #pragma unroll
   for (int kk = 0; kk < 4; ++kk)
-    qf[kk] = frag8(qt, wid * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
+    qf[kk] = *(const bf16x8*)(
+        q + qoff + (long long)(wid * 16 + l16) * ATT_D + kk * 32
+        + lgrp * 8);
 
   const int kv_tiles = (qb * BQ) / BKV + 1;  // causal bound
   for (int j = 0; j < kv_tiles; ++j) {
@@ -439,9 +438,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     short* __restrict__ dq,
     int B, int H, int Hkv, int S, float scale) {
   constexpr int BQ = 64, BKV2 = 32;
-  __shared__ short qt[BQ * ATT_D];       // Q rm (A for S)
-  __shared__ short dot[BQ * ATT_D];      // dO rm (A for dP)
-  __shared__ short kt[BKV2 * ATT_D];     // K rm (B for S + B for dP? no: V)
+  __shared__ short kt[BKV2 * ATT_D];     // K rm (B for S)
   __shared__ short ktt[ATT_D * BKV2];    // K^T (B for dQ)
   __shared__ short vt[BKV2 * ATT_D];     // V rm (B for dP)
   __shared__ short ptile[BQ * BKV2];     // dS round-trip
@@ -461,8 +458,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const long long qoff = (hoff + qb * BQ) * ATT_D;
   const long long kvoff0 = ((long long)b * Hkv + hk) * S * ATT_D;
 
-  stage_rm<BQ, BLOCK>(qt, q + qoff, ATT_D);
-  stage_rm<BQ, BLOCK>(dot, dout + qoff, ATT_D);
   if (threadIdx.x < BQ) {
     lse_s[threadIdx.x] = lse[hoff + qb * BQ + threadIdx.x];
     del_s[threadIdx.x] = delta[hoff + qb * BQ + threadIdx.x];
@@ -473,11 +468,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   for (int n = 0; n < 8; ++n) acc_dq[n] = (f4){0, 0, 0, 0};
   __syncthreads();
 
+  // preload Q/dO A-fragments straight from global (keeps 32 KiB of LDS
+  // free; one-time cost per block)
   bf16x8 qf[4], df[4];
 #pragma unroll
   for (int kk = 0; kk < 4; ++kk) {
-    qf[kk] = frag8(qt, wid * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
-    df[kk] = frag8(dot, wid * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
+    const long long row_off =
+        qoff + (long long)(wid * 16 + l16) * ATT_D + kk * 32 + lgrp * 8;
+    qf[kk] = *(const bf16x8*)(q + row_off);
+    df[kk] = *(const bf16x8*)(dout + row_off);
   }
 
   const int nkv = (qb + 1) * BQ / BKV2;  // causal bound

@@ -1,0 +1,28 @@
+"""Matrix harness: DAG shapes x behaviors x dual checkers."""
+
+import pytest
+
+from .matrix.harness import (
+    GRAPHS,
+    ArtifactFlow,
+    CurrentInfoFlow,
+    ForeachContextFlow,
+    StepCounterFlow,
+    run_matrix_case,
+)
+
+BEHAVIORS = {
+    "artifact": ArtifactFlow,
+    "counter": StepCounterFlow,
+    "current": CurrentInfoFlow,
+    "foreach_ctx": ForeachContextFlow,
+}
+
+CASES = [(g, b) for g in GRAPHS for b in BEHAVIORS]
+
+
+@pytest.mark.parametrize("graph_name,behavior", CASES,
+                         ids=["%s-%s" % c for c in CASES])
+def test_matrix(graph_name, behavior, tmp_path, tmp_datastore):
+    test = BEHAVIORS[behavior]()
+    run_matrix_case(graph_name, test, str(tmp_path), tmp_datastore)
