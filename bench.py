@@ -32,6 +32,9 @@ def parse_args():
                             "mixtral-8x7b", "mixtral-tiny"])
     p.add_argument("--bucket-mb", type=int, default=64)
     p.add_argument("--lr", type=float, default=3e-4)
+    p.add_argument("--tunableop", action="store_true",
+                   help="enable PyTorch TunableOp (hipBLASLt algorithm "
+                        "autotuning) during warmup")
     return p.parse_args()
 
 
@@ -50,6 +53,12 @@ def main():
         FusedAdamW,
         init_process_group_from_env,
     )
+
+    if args.tunableop:
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                              "/tmp/mfx_tunableop_%d.csv" % rank)
 
     use_gpu = torch.cuda.is_available()
     device = torch.device("cuda", local_rank) if use_gpu else \

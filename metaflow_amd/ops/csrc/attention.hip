@@ -99,7 +99,6 @@ DEV void stage_tr(short* dst, const short* src, long long src_row_stride) {
 DEV void wave_lds_fence() {
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
-  __builtin_amdgcn_sched_barrier(0);
 }
 
 // ============================ FORWARD =====================================
