@@ -20,6 +20,33 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension, CppExtensio
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "metaflow_amd", "ops", "csrc")
 
+# torch's hipcc path emits no depfiles, so edits to #included .hip files
+# never trigger a rebuild of the single TU. Track deps ourselves: if any
+# csrc file is newer than the built object, drop the object + the hipified
+# intermediate so ninja recompiles.
+def _force_rebuild_if_stale():
+    import glob
+
+    objs = glob.glob(os.path.join(ROOT, "build", "temp*", "metaflow_amd",
+                                  "ops", "csrc", "mfx_hip*.o"))
+    if not objs:
+        return
+    newest_src = max(
+        os.path.getmtime(p)
+        for p in glob.glob(os.path.join(CSRC, "*"))
+        if not p.endswith("_hip.hip"))
+    for obj in objs:
+        if os.path.getmtime(obj) < newest_src:
+            os.unlink(obj)
+            gen = os.path.join(CSRC, "mfx_hip_hip.hip")
+            if os.path.exists(gen):
+                os.unlink(gen)
+            # force re-hipify by bumping the source mtime
+            os.utime(os.path.join(CSRC, "mfx_hip.hip"))
+
+
+_force_rebuild_if_stale()
+
 ext_modules = [
     CUDAExtension(
         name="metaflow_amd.ops._mfx_hip",
