@@ -92,3 +92,28 @@ __global__ __launch_bounds__(256) void dbg_dv_kernel(
     for (int r = 0; r < 4; ++r)
       out[(wid * 16 + lgrp * 4 + r) * 128 + n * 16 + l16] = acc[n][r];
 }
+
+// probe: mfma_f32_32x32x16_bf16 layout — C = A(32x16) @ B(16x32) with
+// assumed layouts: A[lane&31][(lane>>5)*8+i], B[(lane>>5)*8+i][lane&31],
+// C: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
+typedef __attribute__((ext_vector_type(16))) float f16x_;
+__global__ void dbg_mfma32_kernel(const short* __restrict__ a,
+                                  const short* __restrict__ b,
+                                  float* __restrict__ c /* [32][32] */) {
+  const int lane = threadIdx.x & 63;
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+  bf16x8 av, bv;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    av[i] = a[l32 * 16 + hi * 8 + i];       // A row-major [32][16]
+    bv[i] = b[(hi * 8 + i) * 32 + l32];     // B row-major [16][32]
+  }
+  f16x_ acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, bv, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    c[row * 32 + l32] = acc[r];
+  }
+}

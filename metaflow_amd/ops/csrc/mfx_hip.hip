@@ -265,6 +265,16 @@ torch::Tensor dbg_dv(torch::Tensor pmat, torch::Tensor bmat) {
 }
 }  // namespace
 
+namespace {
+torch::Tensor dbg_mfma32(torch::Tensor a, torch::Tensor b) {
+  auto out = torch::empty({32, 32}, a.options().dtype(torch::kFloat32));
+  dbg_mfma32_kernel<<<1, 64, 0, cur_stream()>>>(bf(a), bf(b),
+                                                out.data_ptr<float>());
+  HIP_CHECK_KERNEL();
+  return out;
+}
+}  // namespace
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");
@@ -279,4 +289,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, "flash attention backward");
   m.def("dbg_st", &dbg_st, "debug S^T path");
   m.def("dbg_dv", &dbg_dv, "debug dV path");
+  m.def("dbg_mfma32", &dbg_mfma32, "mfma 32x32x16 layout probe");
 }
