@@ -11,6 +11,7 @@
 #include "adam.hip"
 #include "cross_entropy.hip"
 #include "attention.hip"
+#include "attention_v2.hip"
 #include "debug_kernels.hip"
 
 namespace {
@@ -198,10 +199,18 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(H % Hkv == 0, "GQA requires H % Hkv == 0");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
-  dim3 grid(S / 64, H, B);
-  attn_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
-      bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
-      (float)scale);
+  if (S % 256 == 0) {
+    // v2: 8-wave 32x32 swapped-QK^T structure
+    dim3 grid(S / 256, H, B);
+    attn_fwd_v2_kernel<512><<<grid, 512, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
+        (float)scale);
+  } else {
+    dim3 grid(S / 64, H, B);
+    attn_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
+        (float)scale);
+  }
   HIP_CHECK_KERNEL();
   return {o, lse};
 }
