@@ -14,6 +14,7 @@ from .parameters import Parameter, JSONType
 from .includefile import IncludeFile
 from .current import current
 from .unbounded_foreach import UnboundedForeachInput
+from .user_config import Config, ConfigValue, FlowMutator, MutableFlow
 from .plugins.retry_decorator import retry
 from .plugins.catch_decorator import catch
 from .plugins.timeout_decorator import timeout
@@ -50,6 +51,10 @@ __all__ = [
     "IncludeFile",
     "current",
     "UnboundedForeachInput",
+    "Config",
+    "ConfigValue",
+    "FlowMutator",
+    "MutableFlow",
     "retry",
     "catch",
     "timeout",

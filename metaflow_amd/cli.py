@@ -51,9 +51,16 @@ def _make_storage(datastore, datastore_root):
     return impl(root)
 
 
-def _init_state(state, datastore, datastore_root, metadata, quiet, with_):
+def _init_state(state, datastore, datastore_root, metadata, quiet, with_,
+                config=()):
     from .decorators import attach_decorators
+    from .user_config import resolve_configs
 
+    overrides = {}
+    for spec in config:
+        name, _, path = spec.partition("=")
+        overrides[name] = path
+    resolve_configs(state.flow_cls, overrides)
     if with_:
         attach_decorators(state.flow_cls, with_)
     state.graph = FlowGraph(state.flow_cls)
@@ -73,6 +80,8 @@ def _init_state(state, datastore, datastore_root, metadata, quiet, with_):
     ]
     if quiet:
         state.top_level_args.append("--quiet")
+    for spec in config:
+        state.top_level_args.extend(["--config", spec])
     # run step_init hooks
     for step_name in state.flow_cls._steps:
         func = getattr(state.flow_cls, step_name)
@@ -121,8 +130,11 @@ def main(flow):
     @click.option("--quiet", is_flag=True, default=False)
     @click.option("--with", "with_", multiple=True,
                   help="Attach a decorator to all steps, e.g. retry:times=2")
-    def cli(datastore, datastore_root, metadata, quiet, with_):
-        _init_state(state, datastore, datastore_root, metadata, quiet, with_)
+    @click.option("--config", multiple=True,
+                  help="Override a Config file: name=path.json")
+    def cli(datastore, datastore_root, metadata, quiet, with_, config):
+        _init_state(state, datastore, datastore_root, metadata, quiet,
+                    with_, config)
 
     def _run_common(kwargs, clone_run_id=None, steps_to_rerun=None):
         from .runtime import NativeRuntime

@@ -118,3 +118,20 @@ def test_card(tmp_datastore):
     html = get_card(ds)
     assert html and "metric" in html and "custom html" in html
     assert "Notes" in html and "OK" in html
+
+
+def test_config_and_mutator(tmp_datastore, tmp_path):
+    run_flow("config_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "ConfigFlow")
+    assert read_artifact(tmp_datastore, "ConfigFlow", run_id, "end",
+                         "value") == 30
+    assert read_artifact(tmp_datastore, "ConfigFlow", run_id, "end",
+                         "retry_attached") is True
+    # config file override via --config
+    cfgf = tmp_path / "override.json"
+    cfgf.write_text('{"scale": 7, "retries": 0}')
+    run_flow("config_flow.py", tmp_datastore, "--config",
+             "cfg=%s" % cfgf, "run")
+    run_id = latest_run_id(tmp_datastore, "ConfigFlow")
+    assert read_artifact(tmp_datastore, "ConfigFlow", run_id, "end",
+                         "value") == 70
