@@ -32,6 +32,68 @@ DEV unsigned int pack_bf2(float lo, float hi) {
          | ((unsigned int)(unsigned short)f2bf(hi) << 16);
 }
 
+// Turn a per-lane C-tile column (st[2][16]: lane holds values X[row][mycol]
+// for rows (r&3)+8*(r>>2)+4*hi of two 32-row sub-tiles) into MFMA
+// A-fragments with m = mycol and k = the row dimension: pack bf16 pairs,
+// exchange with the partner half-wave (shfl_xor 32), assemble 4 k-steps.
+DEV void col_to_afrags(const f16f st[2], bf16x8 pa[4], int hi) {
+  unsigned int pk[16], xp[16];
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int j2 = 0; j2 < 8; ++j2)
+      pk[t * 8 + j2] = pack_bf2(st[t][j2 * 2], st[t][j2 * 2 + 1]);
+#pragma unroll
+  for (int i = 0; i < 16; ++i) xp[i] = __shfl_xor(pk[i], 32, WAVE);
+#pragma unroll
+  for (int ks = 0; ks < 4; ++ks) {
+    union {
+      unsigned int w[4];
+      bf16x8 v;
+    } u;
+    if (hi == 0) {
+      u.w[0] = pk[4 * ks];
+      u.w[1] = pk[4 * ks + 1];
+      u.w[2] = xp[4 * ks];
+      u.w[3] = xp[4 * ks + 1];
+    } else {
+      u.w[0] = xp[4 * ks + 2];
+      u.w[1] = xp[4 * ks + 3];
+      u.w[2] = pk[4 * ks + 2];
+      u.w[3] = pk[4 * ks + 3];
+    }
+    pa[ks] = u.v;
+  }
+}
+
+
+// single-subtile variant of col_to_afrags: one 32-row C sub-tile -> 2
+// A-fragment k-steps (k = 0..31 of the row dimension)
+DEV void col_to_afrags1(const f16f& st, bf16x8 pa[2], int hi) {
+  unsigned int pk[8], xp[8];
+#pragma unroll
+  for (int j2 = 0; j2 < 8; ++j2)
+    pk[j2] = pack_bf2(st[j2 * 2], st[j2 * 2 + 1]);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) xp[i] = __shfl_xor(pk[i], 32, WAVE);
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    union { unsigned int w[4]; bf16x8 v; } u;
+    if (hi == 0) {
+      u.w[0] = pk[4 * ks];
+      u.w[1] = pk[4 * ks + 1];
+      u.w[2] = xp[4 * ks];
+      u.w[3] = xp[4 * ks + 1];
+    } else {
+      u.w[0] = xp[4 * ks + 2];
+      u.w[1] = xp[4 * ks + 3];
+      u.w[2] = pk[4 * ks + 2];
+      u.w[3] = pk[4 * ks + 3];
+    }
+    pa[ks] = u.v;
+  }
+}
+
 template <int BLOCK>  // BLOCK = 512 (8 waves)
 __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
@@ -143,40 +205,8 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     }
 
     // ---- P -> bf16 A-fragments via packed pairs + partner exchange ----
-    // own pairs: pk[t*8 + j2] packs kv offsets {t*32 + pairbase(j2) + 4hi}
-    unsigned int pk[16], xp[16];
-#pragma unroll
-    for (int t = 0; t < 2; ++t)
-#pragma unroll
-      for (int j2 = 0; j2 < 8; ++j2) {
-        pk[t * 8 + j2] = pack_bf2(st[t][j2 * 2], st[t][j2 * 2 + 1]);
-      }
-#pragma unroll
-    for (int i = 0; i < 16; ++i) xp[i] = __shfl_xor(pk[i], 32, WAVE);
-
-    // A-frag for PV k-step ks (kv = j*64 + ks*16 + hi*8 + i):
-    //   hi=0: [pk[4ks], pk[4ks+1], xp[4ks], xp[4ks+1]]   (kv ..0-3,4-7)
-    //   hi=1: [xp[4ks+2], xp[4ks+3], pk[4ks+2], pk[4ks+3]] (kv 8-11,12-15)
     bf16x8 pa[4];
-#pragma unroll
-    for (int ks = 0; ks < 4; ++ks) {
-      union {
-        unsigned int w[4];
-        bf16x8 v;
-      } u;
-      if (hi == 0) {
-        u.w[0] = pk[4 * ks];
-        u.w[1] = pk[4 * ks + 1];
-        u.w[2] = xp[4 * ks];
-        u.w[3] = xp[4 * ks + 1];
-      } else {
-        u.w[0] = xp[4 * ks + 2];
-        u.w[1] = xp[4 * ks + 3];
-        u.w[2] = pk[4 * ks + 2];
-        u.w[3] = pk[4 * ks + 3];
-      }
-      pa[ks] = u.v;
-    }
+    col_to_afrags(st, pa, hi);
 
     // ---- PV: O(32q x 128d) += P(32q x 64kv) @ V(64kv x 128d) ----
 #pragma unroll
@@ -203,4 +233,240 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
   }
   if (hi == 0)
     lse[hoff + my_qrow] = m_run + __logf(l_run);
+}
+
+// ===================== BACKWARD v2: dQ ====================================
+// Grid (S/256, H, B), 8 waves x 32 q rows. Swapped structure: lane's
+// C-column is its OWN q row, so lse/delta are per-lane scalars and
+// dS^T -> dS A-fragments use the same col_to_afrags dance. LDS: K rm +
+// K^T + V rm (48 KiB).
+
+template <int BLOCK>
+__global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    short* __restrict__ dq, int B, int H, int Hkv, int S, float scale) {
+  constexpr int BQ = 256, BKV = 64;
+  __shared__ short kt[BKV * ATT_D];    // K rm swizzled (B for S^T)
+  __shared__ short ktt[ATT_D * BKV];   // K^T swizzled (B for dQ)
+  __shared__ short vt[BKV * ATT_D];    // V rm swizzled (B for dP^T)
+
+  const int qb = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hk = h / (H / Hkv);
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int my_qrow = qb * BQ + wid * 32 + l32;
+  const long long hoff = ((long long)b * H + h) * S;
+  const long long qoff = (hoff + qb * BQ) * ATT_D;
+  const long long kvoff0 = ((long long)b * Hkv + hk) * S * ATT_D;
+
+  bf16x8 q_reg[8], do_reg[8];
+  {
+    const short* qrow = q + (hoff + my_qrow) * ATT_D;
+    const short* drow = dout + (hoff + my_qrow) * ATT_D;
+#pragma unroll
+    for (int s = 0; s < 8; ++s) {
+      q_reg[s] = *(const bf16x8*)(qrow + s * 16 + hi * 8);
+      do_reg[s] = *(const bf16x8*)(drow + s * 16 + hi * 8);
+    }
+  }
+  const float my_lse = lse[hoff + my_qrow];
+  const float my_del = delta[hoff + my_qrow];
+
+  f16f acc_dq[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) acc_dq[n] = (f16f){};
+
+  const int kv_tiles = (qb * BQ + BQ) / BKV;
+  for (int j = 0; j < kv_tiles; ++j) {
+    __syncthreads();
+    const short* ksrc = k + kvoff0 + (long long)j * BKV * ATT_D;
+    stage_rm<BKV, BLOCK>(kt, ksrc, ATT_D);
+    stage_tr<BKV, BLOCK>(ktt, ksrc, ATT_D);
+    stage_rm<BKV, BLOCK>(vt, v + kvoff0 + (long long)j * BKV * ATT_D,
+                         ATT_D);
+    __syncthreads();
+    if (j * BKV > qb * BQ + wid * 32 + 31) continue;
+
+    // per 32-kv sub-tile: S^T, dP^T, dS, dQ — keeps live regs low
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f16f st = (f16f){}, dpt = (f16f){};
+#pragma unroll
+      for (int s = 0; s < 8; ++s) {
+        bf16x8 kf = frag8(kt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
+        bf16x8 vf = frag8(vt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
+        st = mfma32(kf, q_reg[s], st);
+        dpt = mfma32(vf, do_reg[s], dpt);
+      }
+      // dS = P * (dP - delta) * scale, P = exp(S*scale - lse), causal
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvg = j * BKV + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float p = 0.f;
+        if (kvg <= my_qrow)
+          p = __expf(st[r] * scale - my_lse);
+        st[r] = p * (dpt[r] - my_del) * scale;
+      }
+      // dQ(32q x 128d) += dS(32q x 32kv) @ K(32kv x 128d)
+      bf16x8 pa[2];
+      col_to_afrags1(st, pa, hi);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          bf16x8 kf = frag8(ktt, n * 32 + l32,
+                            t * 32 + ks * 16 + hi * 8, BKV * 2);
+          acc_dq[n] = mfma32(pa[ks], kf, acc_dq[n]);
+        }
+    }
+  }
+
+  // epilogue: C row = q local, col = d
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const long long obase =
+        qoff + (long long)(wid * 32 + row_local) * ATT_D;
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      dq[obase + n * 32 + l32] = f2bf(acc_dq[n][r]);
+  }
+}
+
+// ===================== BACKWARD v2: dK/dV =================================
+// Grid (S/256, Hkv, B), 8 waves x 32 kv rows; K/V rows live in registers
+// (B-operands); loops over the G q-heads and q tiles of 64. C col = own
+// kv row, so P/dS columns feed col_to_afrags for the dV/dK MFMAs.
+// LDS: Q rm + Q^T + dO rm + dO^T (64 KiB) + lse/delta rows.
+
+template <int BLOCK>
+__global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    short* __restrict__ dk, short* __restrict__ dv,
+    int B, int H, int Hkv, int S, float scale) {
+  constexpr int BKVB = 256, BQ2 = 64;
+  __shared__ short qt[BQ2 * ATT_D];    // Q rm swizzled (A for S^T)
+  __shared__ short qtt[ATT_D * BQ2];   // Q^T swizzled (B for dK)
+  __shared__ short dot[BQ2 * ATT_D];   // dO rm swizzled (A for dP^T)
+  __shared__ short dott[ATT_D * BQ2];  // dO^T swizzled (B for dV)
+  __shared__ float lse_s[BQ2];
+  __shared__ float del_s[BQ2];
+
+  const int kvb = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int b = blockIdx.z;
+  const int G = H / Hkv;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int my_kvrow = kvb * BKVB + wid * 32 + l32;
+  const long long kvoff =
+      (((long long)b * Hkv + hk) * S + kvb * BKVB) * ATT_D;
+
+  // B-operand K/V rows are re-read from global per q-tile (L2-hot after
+  // the first pass; keeping them in registers costs 64 VGPR and spills)
+  const short* krow = k + kvoff + (long long)(wid * 32 + l32) * ATT_D;
+  const short* vrow = v + kvoff + (long long)(wid * 32 + l32) * ATT_D;
+
+  f16f acc_dk[4], acc_dv[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    acc_dk[n] = (f16f){};
+    acc_dv[n] = (f16f){};
+  }
+
+  const int jq0 = (kvb * BKVB) / BQ2;
+  const int nq = S / BQ2;
+  for (int g = 0; g < G; ++g) {
+    const int h = hk * G + g;
+    const long long hoff = ((long long)b * H + h) * S;
+    for (int jq = jq0; jq < nq; ++jq) {
+      __syncthreads();
+      const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
+      const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
+      stage_rm<BQ2, BLOCK>(qt, qsrc, ATT_D);
+      stage_tr<BQ2, BLOCK>(qtt, qsrc, ATT_D);
+      stage_rm<BQ2, BLOCK>(dot, dsrc, ATT_D);
+      stage_tr<BQ2, BLOCK>(dott, dsrc, ATT_D);
+      if (threadIdx.x < BQ2) {
+        lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
+        del_s[threadIdx.x] = delta[hoff + jq * BQ2 + threadIdx.x];
+      }
+      __syncthreads();
+      // wave-uniform skip: all of this wave's kv rows above every q row
+      if (jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
+
+      // per 32-q sub-tile: S, dP, P/dS, dV, dK — keeps live regs low
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        f16f st = (f16f){}, dpt = (f16f){};
+#pragma unroll
+        for (int s = 0; s < 8; ++s) {
+          bf16x8 qf = frag8(qt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
+          bf16x8 df = frag8(dot, t * 32 + l32, s * 16 + hi * 8,
+                            ATT_D * 2);
+          bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
+          bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
+          st = mfma32(qf, kr, st);
+          dpt = mfma32(df, vr, dpt);
+        }
+        // P (into st) with causal mask q >= kv; dS (into dpt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const int qrg = jq * BQ2 + qrl;
+          float p = 0.f;
+          if (qrg >= my_kvrow)
+            p = __expf(st[r] * scale - lse_s[qrl]);
+          st[r] = p;
+          dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
+        }
+        // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d)
+        bf16x8 pa[2];
+        col_to_afrags1(st, pa, hi);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            bf16x8 df = frag8(dott, n * 32 + l32,
+                              t * 32 + ks * 16 + hi * 8, BQ2 * 2);
+            acc_dv[n] = mfma32(pa[ks], df, acc_dv[n]);
+          }
+        // dK += dS^T @ Q
+        col_to_afrags1(dpt, pa, hi);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            bf16x8 qf = frag8(qtt, n * 32 + l32,
+                              t * 32 + ks * 16 + hi * 8, BQ2 * 2);
+            acc_dk[n] = mfma32(pa[ks], qf, acc_dk[n]);
+          }
+      }
+    }
+  }
+
+  // epilogue: C row = kv local, col = d
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const long long obase =
+        kvoff + (long long)(wid * 32 + row_local) * ATT_D;
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      dk[obase + n * 32 + l32] = f2bf(acc_dk[n][r]);
+      dv[obase + n * 32 + l32] = f2bf(acc_dv[n][r]);
+    }
+  }
 }
