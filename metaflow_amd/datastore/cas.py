@@ -109,7 +109,6 @@ class ContentAddressedStore(object):
         """
         results = []
         to_save = []
-        engine = _native_engine()
         for blob in blob_iter:
             # keys are opaque; loads always use the key stored in the
             # artifact index, so the Merkle fast path is safe
@@ -122,19 +121,16 @@ class ContentAddressedStore(object):
         exists = self._storage.is_file([p for _, p, _ in to_save])
 
         def _packed():
+            raw_header = MAGIC + bytes([1, CODEC_RAW, 0, 0])
             for (key, path, blob), present in zip(to_save, exists):
                 if present:
                     continue
                 big = raw or len(blob) >= CAS_COMPRESS_MAX_SIZE
-                if (engine is not None and big
-                        and hasattr(self._storage, "_abs")):
-                    # native path: multithreaded write with header, atomic
-                    engine.save_blob(self._storage._abs(path),
-                                     MAGIC + bytes([1, CODEC_RAW, 0, 0]),
-                                     blob)
-                    continue
-                codec = CODEC_RAW if big else CODEC_GZIP
-                yield path, (pack(blob, codec), None)
+                if big:
+                    # zero-copy: header + payload written separately
+                    yield path, ([raw_header, blob], None)
+                else:
+                    yield path, (pack(bytes(blob), CODEC_GZIP), None)
 
         self._storage.save_bytes(_packed(), overwrite=False)
         return results

@@ -133,6 +133,11 @@ class LocalStorage(DataStoreStorage):
                             if not chunk:
                                 break
                             f.write(chunk)
+                    elif isinstance(data, (list, tuple)):
+                        # scatter write: avoids concatenating header +
+                        # payload (zero-copy path for large blobs)
+                        for part in data:
+                            f.write(part)
                     else:
                         f.write(data)
                 os.replace(tmp, ap)

@@ -3,19 +3,21 @@
 BASELINE config 4: @checkpoint save/load of 70B random-init shards across
 288 GB HBM x8. Design:
 
-* one CAS blob per tensor (content-hashed: an unchanged shard between two
-  checkpoints dedups to a pure existence check — no bytes move);
-* GPU tensors are staged D2H through a reusable pinned buffer with
-  hipMemcpyAsync on a side stream (chunked, so HBM-sized tensors never need
-  a full host-sized intermediate);
+* one CAS blob per tensor holding the RAW storage bytes (no per-blob
+  framing: dtype/shape live in the index artifact), so the save path is
+  zero-copy from (pinned) host memory to the write syscall;
+* content-hashed keys: an unchanged shard between two checkpoints dedups
+  to a pure existence check — no bytes move;
+* GPU tensors stage D2H through a reusable pinned buffer with
+  non-blocking copies on a dedicated side stream, chunked so HBM-sized
+  tensors never need a second host-sized intermediate;
 * the index artifact (name -> {sha, dtype, shape}) is saved like any
   artifact, so `resume` and the Client API see checkpoints natively.
 """
 
-import struct
-
-_PIN_BUF_BYTES = 256 << 20  # 256 MiB staging buffer
+_PIN_BUF_BYTES = 512 << 20  # pinned staging buffer for D2H
 _pin_buf = None
+_side_stream = None
 
 
 def _get_pin_buf(torch):
@@ -29,45 +31,40 @@ def _get_pin_buf(torch):
     return _pin_buf
 
 
-def _tensor_to_bytes(t):
-    """Serialize one tensor to raw bytes; GPU tensors stream through the
-    pinned staging buffer on a dedicated side stream."""
+def _tensor_to_buffer(t):
+    """(memoryview_of_raw_bytes, dtype_str, shape). GPU tensors stream
+    through the pinned staging buffer on a side stream."""
+    import numpy as np
     import torch
 
-    from ..datastore.serializers import serialize_tensor
-
-    if not t.is_cuda:
-        return serialize_tensor(t)
-
+    global _side_stream
     t = t.detach().contiguous()
+    shape = tuple(t.shape)
+    dtype = str(t.dtype).replace("torch.", "")
+    if not t.is_cuda:
+        arr = t.reshape(-1).view(torch.uint8).numpy()
+        return memoryview(arr), dtype, shape
+
     nbytes = t.element_size() * t.numel()
     flat = t.reshape(-1).view(torch.uint8)
-    out = bytearray()
-    # header identical to serializers.serialize_tensor
-    from ..datastore import serializers as S
-
-    if not S._DTYPE_IDS:
-        S._init_dtype_table()
-    shape = tuple(t.shape)
-    out += S._TENSOR_MAGIC + struct.pack("<BB", S._DTYPE_IDS[t.dtype],
-                                         len(shape))
-    out += struct.pack("<%dq" % len(shape), *shape)
-
+    host = np.empty(nbytes, dtype=np.uint8)
     pin = _get_pin_buf(torch)
-    stream = torch.cuda.Stream()
+    pin_np = pin.numpy()
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
     offset = 0
-    with torch.cuda.stream(stream):
+    with torch.cuda.stream(_side_stream):
         while offset < nbytes:
             n = min(_PIN_BUF_BYTES, nbytes - offset)
             pin[:n].copy_(flat[offset:offset + n], non_blocking=True)
-            stream.synchronize()
-            out += pin[:n].numpy().tobytes()
+            _side_stream.synchronize()
+            host[offset:offset + n] = pin_np[:n]
             offset += n
-    return bytes(out)
+    return memoryview(host), dtype, shape
 
 
 def save_state_dict(task_datastore, state_dict, name="checkpoint"):
-    """Persist a state dict; returns {tensor_name: sha}."""
+    """Persist a state dict; returns {tensor_name: {sha, dtype, shape}}."""
     import torch
 
     cas = task_datastore._ca_store
@@ -75,9 +72,10 @@ def save_state_dict(task_datastore, state_dict, name="checkpoint"):
     other = {}
     for key, value in state_dict.items():
         if isinstance(value, torch.Tensor):
-            blob = _tensor_to_bytes(value)
-            (_uri, sha), = cas.save_blobs([blob], raw=True)
-            index[key] = {"sha": sha, "nbytes": len(blob)}
+            buf, dtype, shape = _tensor_to_buffer(value)
+            (_uri, sha), = cas.save_blobs([buf], raw=True)
+            index[key] = {"sha": sha, "dtype": dtype, "shape": list(shape),
+                          "nbytes": len(buf)}
         else:
             other[key] = value
     task_datastore.save_artifacts([
@@ -90,7 +88,7 @@ def save_state_dict(task_datastore, state_dict, name="checkpoint"):
 
 
 def load_state_dict(task_datastore, name="checkpoint", map_location="cpu"):
-    from ..datastore.serializers import deserialize_tensor
+    import torch
 
     index = task_datastore["_checkpoint_%s_index" % name]
     meta = task_datastore.get("_checkpoint_%s_meta" % name, {})
@@ -100,7 +98,12 @@ def load_state_dict(task_datastore, name="checkpoint", map_location="cpu"):
     for key, info in index.items():
         sha_to_names.setdefault(info["sha"], []).append(key)
     for sha, blob in cas.load_blobs(list(sha_to_names)):
-        t = deserialize_tensor(blob)
         for key in sha_to_names[sha]:
-            out[key] = t.to(map_location) if map_location != "cpu" else t
+            info = index[key]
+            dtype = getattr(torch, info["dtype"])
+            t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+            t = t.view(dtype).reshape(info["shape"])
+            if map_location != "cpu":
+                t = t.to(map_location)
+            out[key] = t
     return out
