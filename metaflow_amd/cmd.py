@@ -1,0 +1,102 @@
+"""Top-level `mfx` CLI: inspect flows/runs without a flow file.
+
+Parity target: /root/reference/metaflow/cmd/ (`metaflow status`, etc.).
+
+    python -m metaflow_amd status
+    python -m metaflow_amd runs <FlowName>
+    python -m metaflow_amd logs <Flow/run/step/task>
+    python -m metaflow_amd card <Flow/run/step/task>
+    python -m metaflow_amd gpus
+"""
+
+import click
+
+
+@click.group()
+def cli():
+    pass
+
+
+@cli.command(help="List flows in the local datastore.")
+def status():
+    from .client import Metaflow, namespace
+
+    namespace(None)
+    flows = list(Metaflow())
+    if not flows:
+        click.echo("No flows found (datastore root empty).")
+        return
+    for flow in flows:
+        run = flow.latest_run
+        click.echo("%-30s latest run: %s  (%s)"
+                   % (flow.id, run.id if run else "-",
+                      "ok" if run and run.successful else "…"))
+
+
+@cli.command(help="List runs of a flow.")
+@click.argument("flow_name")
+@click.option("--limit", default=10)
+def runs(flow_name, limit):
+    from .client import Flow, namespace
+
+    namespace(None)
+    for i, run in enumerate(Flow(flow_name)):
+        if i >= limit:
+            break
+        click.echo("%-22s %-10s tags=%s"
+                   % (run.id,
+                      "ok" if run.successful else "failed/running",
+                      ",".join(t for t in run.tags
+                               if not t.startswith("user:"))))
+
+
+@cli.command(help="Show merged logs of a task pathspec.")
+@click.argument("pathspec")
+@click.option("--stderr", is_flag=True)
+def logs(pathspec, stderr):
+    from .client import Task, namespace
+
+    namespace(None)
+    task = Task(pathspec)
+    click.echo(task.stderr if stderr else task.stdout)
+
+
+@cli.command(help="Write a task's HTML card to a file.")
+@click.argument("pathspec")
+@click.option("--out", default="card.html")
+def card(pathspec, out):
+    from .client import Task, namespace
+    from .plugins.card_decorator import get_card
+
+    namespace(None)
+    task = Task(pathspec)
+    html = get_card(task._ds)
+    if not html:
+        raise click.ClickException("No card for %s" % pathspec)
+    with open(out, "w") as f:
+        f.write(html)
+    click.echo("wrote %s" % out)
+
+
+@cli.command(help="Show visible GPUs (rocm-smi summary).")
+def gpus():
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            for i in range(torch.cuda.device_count()):
+                p = torch.cuda.get_device_properties(i)
+                click.echo("%d: %s  %.0f GB" % (
+                    i, p.name, p.total_memory / 1e9))
+        else:
+            click.echo("No GPUs visible.")
+    except Exception as e:  # noqa: BLE001
+        click.echo("torch unavailable: %s" % e)
+
+
+def main():
+    cli(standalone_mode=True)
+
+
+if __name__ == "__main__":
+    main()
