@@ -234,6 +234,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
         bf(dout), bf(o), delta.data_ptr<float>(), rows);
   HIP_CHECK_KERNEL();
   }
+  if (S % 256 == 0) {
+    dim3 gkv(S / 256, Hkv, B);
+    attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+        delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
+        (float)scale);
+    HIP_CHECK_KERNEL();
+    dim3 gq(S / 256, H, B);
+    attn_bwd_dq_v2_kernel<512><<<gq, 512, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+        delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale);
+    HIP_CHECK_KERNEL();
+    return {dq, dk, dv};
+  }
   {
     dim3 grid(S / 64, Hkv, B);
     attn_bwd_dkdv_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
