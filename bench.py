@@ -57,6 +57,10 @@ def main():
     if args.tunableop:
         os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
         os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS",
+                              "100")
+        os.environ.setdefault(
+            "PYTORCH_TUNABLEOP_MAX_TUNING_ITERATIONS", "30")
         os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
                               "/tmp/mfx_tunableop_%d.csv" % rank)
 

@@ -30,7 +30,10 @@ DEV f4 mfma16(bf16x8 a, bf16x8 b, f4 c) {
 // — rowb=64 tiles get a 4-row spread instead of 8, still conflict-free
 // enough at 2 lanes/bank)
 DEV int swz(int r, int c, int rowb) {
-  return r * rowb + (((c) * 2) ^ (((r & 7) << 4) & (rowb - 1)));
+  // spread up to 16 rows across 16-byte slots (256B rows get the full
+  // 16-way spread -> 2 lanes/slot for 32-row reads; narrower rows are
+  // masked down automatically)
+  return r * rowb + (((c) * 2) ^ (((r & 15) << 4) & (rowb - 1)));
 }
 
 // load an 8-element fragment from a swizzled tile row (c must be mult of 8)
