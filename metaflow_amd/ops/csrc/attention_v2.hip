@@ -100,8 +100,11 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse, int B, int H, int Hkv, int S, float scale) {
   constexpr int BQ = 256, BKV = 64;  // 8 waves x 32 q rows
-  __shared__ short kt[BKV * ATT_D];   // K row-major swizzled [64][128]
-  __shared__ short vtt[ATT_D * BKV];  // V^T swizzled [128][64]
+  // double-buffered K/V tiles: stage j+1 overlaps compute on j (T14
+  // async-stage split: global loads issue BEFORE compute, LDS writes
+  // happen after — HBM latency hides under the MFMA phase)
+  __shared__ short kt[2][BKV * ATT_D];   // K rm swizzled [64][128]
+  __shared__ short vtt[2][ATT_D * BKV];  // V^T swizzled [128][64]
 
   const int qb = blockIdx.x;
   const int h = blockIdx.y;
@@ -132,18 +135,51 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
 #pragma unroll
   for (int n = 0; n < 4; ++n) acc_o[n] = (f16f){};
 
+  // per-thread staging slices: 2 x bf16x8 of K and of V per tile
+  const int tid = threadIdx.x;
+  bf16x8 stg_k[2], stg_v[2];
+  int stg_r[2], stg_c8[2];
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = u * BLOCK + tid;
+    stg_r[u] = idx / 16;
+    stg_c8[u] = idx % 16;
+  }
+
+#define MFX_STAGE_LOAD(jj)                                               \
+  {                                                                      \
+    const short* kb_ = k + kvoff0 + (long long)(jj) * BKV * ATT_D;       \
+    const short* vb_ = v + kvoff0 + (long long)(jj) * BKV * ATT_D;       \
+    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                      \
+      stg_k[u] = *(const bf16x8*)(kb_ + stg_r[u] * ATT_D + stg_c8[u] * 8);\
+      stg_v[u] = *(const bf16x8*)(vb_ + stg_r[u] * ATT_D + stg_c8[u] * 8);\
+    }                                                                    \
+  }
+
+#define MFX_STAGE_WRITE(buf)                                             \
+  {                                                                      \
+    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                      \
+      *(bf16x8*)((char*)kt[buf]                                          \
+                 + swz(stg_r[u], stg_c8[u] * 8, ATT_D * 2)) = stg_k[u];  \
+      _Pragma("unroll") for (int jj = 0; jj < 8; ++jj)                   \
+        *(short*)((char*)vtt[buf]                                        \
+                  + swz(stg_c8[u] * 8 + jj, stg_r[u], BKV * 2)) =        \
+            stg_v[u][jj];                                                \
+    }                                                                    \
+  }
+
   const int kv_tiles = (qb * BQ + BQ) / BKV;  // causal bound (diag incl.)
+  MFX_STAGE_LOAD(0);
+  MFX_STAGE_WRITE(0);
+  __syncthreads();
   for (int j = 0; j < kv_tiles; ++j) {
-    __syncthreads();
-    stage_rm<BKV, BLOCK>(kt, k + kvoff0 + (long long)j * BKV * ATT_D,
-                         ATT_D);
-    stage_tr<BKV, BLOCK>(vtt, v + kvoff0 + (long long)j * BKV * ATT_D,
-                         ATT_D);
-    __syncthreads();
+    const int cur = j & 1;
+    if (j + 1 < kv_tiles) MFX_STAGE_LOAD(j + 1);  // issue loads early
 
     // wave-uniform skip: this wave's rows are all below the tile's kv
-    // range (fully masked) — barriers above/below still run
-    if (j * BKV > qb * BQ + wid * 32 + 31) continue;
+    // range (fully masked) — staging + barrier still run below
+    const bool active = (j * BKV <= qb * BQ + wid * 32 + 31);
+    if (active) {
 
     // ---- S^T = K (64x128) @ Q^T: two 32-kv sub-tiles ----
     f16f st[2];
@@ -152,7 +188,8 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
       st[t] = (f16f){};
 #pragma unroll
       for (int s = 0; s < 8; ++s) {
-        bf16x8 kf = frag8(kt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
+        bf16x8 kf = frag8(kt[cur], t * 32 + l32, s * 16 + hi * 8,
+                          ATT_D * 2);
         st[t] = mfma32(kf, q_reg[s], st[t]);
       }
     }
@@ -213,11 +250,18 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     for (int ks = 0; ks < 4; ++ks) {
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
-        bf16x8 vf = frag8(vtt, n * 32 + l32, ks * 16 + hi * 8, BKV * 2);
+        bf16x8 vf = frag8(vtt[cur], n * 32 + l32, ks * 16 + hi * 8,
+                          BKV * 2);
         acc_o[n] = mfma32(pa[ks], vf, acc_o[n]);
       }
     }
+    }  // active
+
+    if (j + 1 < kv_tiles) MFX_STAGE_WRITE(cur ^ 1);
+    __syncthreads();
   }
+#undef MFX_STAGE_LOAD
+#undef MFX_STAGE_WRITE
 
   // ---- epilogue: O /= l (row-matched), write bf16 + lse ----
   float inv_l = 1.f / l_run;
