@@ -45,31 +45,59 @@ class MixtralConfig:
                    num_experts=4, top_k=2, max_seq_len=seq)
 
 
+def _a2a_single(out, inp, out_splits, in_splits, group):
+    """all_to_all_single, with a send/recv emulation for gloo (which has no
+    all-to-all) so the EP path is CPU-testable with world_size > 1."""
+    import torch.distributed as dist
+
+    if dist.get_backend(group) != "gloo":
+        dist.all_to_all_single(out, inp, output_split_sizes=out_splits,
+                               input_split_sizes=in_splits, group=group)
+        return
+    rank = dist.get_rank(group)
+    world = dist.get_world_size(group)
+    in_off = [0]
+    for s in in_splits:
+        in_off.append(in_off[-1] + s)
+    out_off = [0]
+    for s in out_splits:
+        out_off.append(out_off[-1] + s)
+    # local copy + pairwise exchanges
+    out[out_off[rank]:out_off[rank + 1]] = inp[in_off[rank]:
+                                               in_off[rank + 1]]
+    reqs = []
+    for peer in range(world):
+        if peer == rank:
+            continue
+        send = inp[in_off[peer]:in_off[peer + 1]].contiguous()
+        reqs.append(dist.isend(send, peer, group=group))
+    for peer in range(world):
+        if peer == rank:
+            continue
+        recv = out.new_empty((out_splits[peer],) + tuple(out.shape[1:]))
+        dist.recv(recv, peer, group=group)
+        out[out_off[peer]:out_off[peer + 1]] = recv
+    for r in reqs:
+        r.wait()
+
+
 class _AllToAll(torch.autograd.Function):
     """all_to_all_single with explicit splits; backward = transposed."""
 
     @staticmethod
     def forward(ctx, x, out_splits, in_splits, group):
-        import torch.distributed as dist
-
         ctx.splits = (out_splits, in_splits)
         ctx.group = group
         out = x.new_empty((sum(out_splits),) + tuple(x.shape[1:]))
-        dist.all_to_all_single(out, x.contiguous(),
-                               output_split_sizes=out_splits,
-                               input_split_sizes=in_splits, group=group)
+        _a2a_single(out, x.contiguous(), out_splits, in_splits, group)
         return out
 
     @staticmethod
     def backward(ctx, grad):
         out_splits, in_splits = ctx.splits
-        import torch.distributed as dist
-
         gin = grad.new_empty((sum(in_splits),) + tuple(grad.shape[1:]))
-        dist.all_to_all_single(gin, grad.contiguous(),
-                               output_split_sizes=in_splits,
-                               input_split_sizes=out_splits,
-                               group=ctx.group)
+        _a2a_single(gin, grad.contiguous(), in_splits, out_splits,
+                    ctx.group)
         return gin, None, None, None
 
 
@@ -152,11 +180,10 @@ class MoELayer(nn.Module):
                 self.ep_world * self.experts_per_rank, dtype=counts.dtype,
                 device=counts.device)
             # exchange per-expert counts so the owner can group its input
-            dist.all_to_all_single(
-                recv_counts, counts,
-                output_split_sizes=[self.experts_per_rank] * self.ep_world,
-                input_split_sizes=[self.experts_per_rank] * self.ep_world,
-                group=self.ep_group)
+            _a2a_single(recv_counts, counts.contiguous(),
+                        [self.experts_per_rank] * self.ep_world,
+                        [self.experts_per_rank] * self.ep_world,
+                        self.ep_group)
             out_splits = recv_counts.reshape(
                 self.ep_world, self.experts_per_rank).sum(-1).tolist()
             incoming = all_to_all(dispatched, out_splits, in_splits,
