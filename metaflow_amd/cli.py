@@ -198,10 +198,42 @@ def main(flow):
     @click.option("--max-num-splits", default=MAX_NUM_SPLITS, type=int)
     @click.option("--tag", multiple=True)
     @click.option("--runner-attribute-file", default=None)
-    def resume(origin_run_id, step_to_rerun, **kwargs):
+    @click.option("--reentrant", is_flag=True, default=False,
+                  help="Safe under concurrent invocation: one caller wins "
+                       "leader election and resumes; the rest wait.")
+    def resume(origin_run_id, step_to_rerun, reentrant, **kwargs):
         steps = [step_to_rerun] if step_to_rerun else []
-        _run_common(kwargs, clone_run_id=origin_run_id,
-                    steps_to_rerun=steps)
+        if reentrant:
+            # leader election via exclusive-create in the datastore
+            # (reference runtime.py:1660-1786 uses metadata task
+            # registration as the lock; the primitive is the same)
+            import time as _time
+            import uuid
+
+            run_id = kwargs.get("run_id") or "resume%s" % origin_run_id
+            kwargs["run_id"] = run_id
+            lock_path = state.flow_datastore.storage.path_join(
+                flow_cls.__name__, run_id, "_resume_leader.lock")
+            token = uuid.uuid4().hex.encode()
+            if state.flow_datastore.storage.create_exclusive(lock_path,
+                                                             token):
+                _run_common(kwargs, clone_run_id=origin_run_id,
+                            steps_to_rerun=steps)
+            else:
+                click.echo("[mfx] another resume holds the leader lock; "
+                           "waiting for run %s" % run_id)
+                deadline = _time.time() + 24 * 3600
+                while _time.time() < deadline:
+                    info = state.metadata.get_run(run_id) or {}
+                    if info.get("status") == "successful":
+                        return
+                    if info.get("status") == "failed":
+                        raise MFXException("Leader's resume failed.")
+                    _time.sleep(2)
+                raise MFXException("Timed out waiting for the leader.")
+        else:
+            _run_common(kwargs, clone_run_id=origin_run_id,
+                        steps_to_rerun=steps)
 
     @cli.command(help="[internal] Execute one task of one step.")
     @click.argument("step_name")

@@ -168,6 +168,19 @@ class LocalStorage(DataStoreStorage):
             else:
                 yield path, None, None
 
+    def create_exclusive(self, path, data):
+        """Atomically create a file; returns True iff WE created it (the
+        leader-election primitive: O_CREAT|O_EXCL)."""
+        ap = self._abs(path)
+        os.makedirs(os.path.dirname(ap), exist_ok=True)
+        try:
+            fd = os.open(ap, os.O_CREAT | os.O_EXCL | os.O_WRONLY, 0o644)
+        except FileExistsError:
+            return False
+        with os.fdopen(fd, "wb") as f:
+            f.write(data)
+        return True
+
     def load_file_path(self, path):
         """Local fast path: return the filesystem path for a key (lets the
         native CAS engine mmap/pread instead of round-tripping bytes)."""

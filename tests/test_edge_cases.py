@@ -1,0 +1,72 @@
+"""Segfault detection, large artifacts, log truncation, IncludeFile."""
+
+import os
+
+from .test_runtime import latest_run_id, read_artifact, run_flow
+
+
+def test_segfault_detected(tmp_datastore):
+    proc = run_flow("segfault_flow.py", tmp_datastore, "run", check=False)
+    assert proc.returncode != 0
+    assert "segfault" in (proc.stdout + proc.stderr).lower()
+
+
+def test_large_artifact_roundtrip(tmp_datastore):
+    run_flow("bigdata_flow.py", tmp_datastore, "run", timeout=300)
+    run_id = latest_run_id(tmp_datastore, "BigDataFlow")
+    big = read_artifact(tmp_datastore, "BigDataFlow", run_id, "start", "big")
+    assert big.nbytes == 64 << 20
+    assert int(big[999999]) == 999999
+    # stored raw (uncompressed codec) exactly once in the CAS
+    data_dir = os.path.join(tmp_datastore, "BigDataFlow", "data")
+    sizes = []
+    for dirpath, _d, names in os.walk(data_dir):
+        for n in names:
+            if not n.endswith("_meta"):
+                sizes.append(os.path.getsize(os.path.join(dirpath, n)))
+    big_blobs = [s for s in sizes if s >= 64 << 20]
+    assert len(big_blobs) == 1, sizes
+
+
+def test_include_file(tmp_datastore, tmp_path):
+    f = tmp_path / "input.txt"
+    f.write_text("a\nb\nc\n")
+    run_flow("include_flow.py", tmp_datastore, "run", "--data-file", str(f))
+    run_id = latest_run_id(tmp_datastore, "IncludeFlow")
+    assert read_artifact(tmp_datastore, "IncludeFlow", run_id, "start",
+                         "n_lines") == 3
+
+
+def test_log_truncation(tmp_datastore, tmp_path):
+    flow = tmp_path / "loud_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, step\n"
+        "class LoudFlow(FlowSpec):\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        for i in range(40000):\n"
+        "            print('x' * 64, i)\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n"
+        "if __name__ == '__main__':\n"
+        "    LoudFlow()\n")
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))) + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    run_id = latest_run_id(tmp_datastore, "LoudFlow")
+    start_dir = os.path.join(tmp_datastore, "LoudFlow", run_id, "start")
+    task_id = os.listdir(start_dir)[0]
+    log = os.path.join(start_dir, task_id, "0.stdout.log")
+    # capped at ~1 MB + truncation marker + mflog framing
+    assert os.path.getsize(log) < 3 << 20
+    assert b"[log truncated]" in open(log, "rb").read()

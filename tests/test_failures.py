@@ -54,3 +54,38 @@ def test_with_retry_attach(tmp_datastore, tmp_path, monkeypatch):
     proc = run_flow("linear_flow.py", tmp_datastore,
                     "--with", "retry:times=1", "run")
     assert proc.returncode == 0
+
+
+def test_reentrant_resume(tmp_datastore, tmp_path, monkeypatch):
+    """Two concurrent `resume --reentrant` calls: exactly one executes,
+    the other waits and exits cleanly; tasks ran once."""
+    import subprocess
+    import sys
+
+    counter_dir = tmp_path / "counters"
+    counter_dir.mkdir()
+    monkeypatch.setenv("RESUME_COUNTER_DIR", str(counter_dir))
+    monkeypatch.setenv("RESUME_FAIL", "1")
+    proc = run_flow("resume_flow.py", tmp_datastore, "run", check=False)
+    assert proc.returncode != 0
+    orig_run = latest_run_id(tmp_datastore, "ResumeFlow")
+
+    monkeypatch.setenv("RESUME_FAIL", "0")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))) + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    flows_dir = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                             "flows")
+    cmd = [sys.executable, os.path.join(flows_dir, "resume_flow.py"),
+           "--datastore-root", tmp_datastore, "resume",
+           "--origin-run-id", orig_run, "--reentrant"]
+    procs = [subprocess.Popen(cmd, env=env, stdout=subprocess.PIPE,
+                              stderr=subprocess.PIPE, text=True)
+             for _ in range(2)]
+    rcs = [p.wait(timeout=300) for p in procs]
+    assert rcs == [0, 0], [(p.stdout.read(), p.stderr.read())
+                           for p in procs]
+    # middle ran exactly twice total (once in failed run + once in resume)
+    assert int(open(counter_dir / "middle").read()) == 2
+    assert int(open(counter_dir / "end").read()) == 1
