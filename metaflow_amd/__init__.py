@@ -25,6 +25,7 @@ from .plugins.checkpoint_decorator import checkpoint
 from .plugins.card_decorator import card
 from .plugins.secrets_decorator import secrets
 from .plugins.exit_hook_decorator import exit_hook
+from .plugins.trigger_decorator import trigger, trigger_on_finish
 from .plugins.project_decorator import project
 from .plugins.schedule_decorator import schedule
 from .client import (
@@ -66,6 +67,8 @@ __all__ = [
     "card",
     "secrets",
     "exit_hook",
+    "trigger",
+    "trigger_on_finish",
     "project",
     "schedule",
     "Metaflow",

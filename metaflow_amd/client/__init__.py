@@ -156,6 +156,16 @@ class Run(MetaflowObject):
         return info.get("status") in ("successful", "failed")
 
     @property
+    def is_alive(self):
+        """True if the run is unfinished and heartbeat is fresh (<60 s)."""
+        import time
+
+        if self.finished:
+            return False
+        hb = self._meta._load(self._meta._heartbeat_path(self.id)) or {}
+        return time.time() - hb.get("ts", 0) < 60
+
+    @property
     def data(self):
         end = self["end"].task
         return end.data if end else None
@@ -327,6 +337,52 @@ class Task(MetaflowObject):
     def parent(self):
         return Step("%s/%s/%s" % (self.flow_name, self.run_id,
                                   self.step_name))
+
+    @property
+    def parent_tasks(self):
+        """Tasks whose outputs fed this task (foreach-aware: matched by
+        graph in_funcs + foreach-stack prefix)."""
+        graph = self._graph_info()
+        if graph is None:
+            return []
+        node = graph.get(self.step_name, {})
+        my_stack = self.foreach_stack
+        out = []
+        for in_step in node.get("in_funcs", []):
+            for t in Step("%s/%s/%s" % (self.flow_name, self.run_id,
+                                        in_step)):
+                ts = t.foreach_stack
+                if ts == my_stack or ts == my_stack[:-1] \
+                        or my_stack == ts[:-1] or ts[:len(my_stack)] \
+                        == my_stack:
+                    out.append(t)
+        return out
+
+    @property
+    def child_tasks(self):
+        graph = self._graph_info()
+        if graph is None:
+            return []
+        node = graph.get(self.step_name, {})
+        my_stack = self.foreach_stack
+        out = []
+        for out_step in node.get("out_funcs", []):
+            for t in Step("%s/%s/%s" % (self.flow_name, self.run_id,
+                                        out_step)):
+                ts = t.foreach_stack
+                if ts == my_stack or my_stack == ts[:-1] \
+                        or ts == my_stack[:-1] \
+                        or my_stack[:len(ts)] == ts:
+                    out.append(t)
+        return out
+
+    def _graph_info(self):
+        try:
+            params = self._fds.get_task_datastore(self.run_id,
+                                                  "_parameters", "0")
+            return params.get("_graph_info")
+        except Exception:
+            return None
 
     def __iter__(self):
         for name in self._ds.artifact_names():

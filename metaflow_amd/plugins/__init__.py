@@ -17,6 +17,7 @@ from .schedule_decorator import ScheduleDecorator
 from .card_decorator import CardDecorator
 from .secrets_decorator import SecretsDecorator
 from .exit_hook_decorator import ExitHookDecorator
+from .trigger_decorator import TriggerDecorator, TriggerOnFinishDecorator
 
 STEP_DECORATORS = {
     cls.name: cls
@@ -36,5 +37,6 @@ STEP_DECORATORS = {
 
 FLOW_DECORATORS = {
     cls.name: cls
-    for cls in (ProjectDecorator, ScheduleDecorator, ExitHookDecorator)
+    for cls in (ProjectDecorator, ScheduleDecorator, ExitHookDecorator,
+                TriggerDecorator, TriggerOnFinishDecorator)
 }

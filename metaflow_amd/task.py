@@ -231,6 +231,11 @@ class MFXTask(object):
         )
         if parallel_ctx is not None:
             current._update_env({"parallel": parallel_ctx})
+        from .plugins.trigger_decorator import TriggerInfo
+
+        trig = TriggerInfo.from_env()
+        if trig is not None:
+            current._update_env({"trigger": trig})
 
         # ---- join inputs ----------------------------------------------------
         inputs = None
