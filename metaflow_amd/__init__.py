@@ -39,6 +39,7 @@ from .client import (
     get_namespace,
 )
 from .multicore_utils import parallel_map, parallel_imap_unordered
+from .datatools import ObjectStore
 
 __version__ = "0.1.0"
 
@@ -79,6 +80,7 @@ __all__ = [
     "DataArtifact",
     "namespace",
     "get_namespace",
+    "ObjectStore",
     "parallel_map",
     "parallel_imap_unordered",
 ]
