@@ -24,7 +24,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--batch", type=int, default=4,
+    p.add_argument("--batch", type=int, default=8,
                    help="per-GPU micro batch")
     p.add_argument("--seq", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3-8b",
