@@ -69,42 +69,45 @@ class RMSNorm(nn.Module):
 
 
 class DecoderLayer(nn.Module):
+    """Fused projections: one QKV GEMM (the narrow separate K/V GEMMs are
+    inefficient on MFMA) and one gate|up GEMM feeding the fused-layout
+    SwiGLU kernel (no slice copies; its backward emits d_gateup directly).
+    RoPE runs on the [B,H,S,D] layout the attention kernel consumes, so
+    the transpose copy after the QKV split is the only data movement."""
+
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         self.cfg = cfg
         h, hd = cfg.hidden_size, cfg.head_dim
+        self.qkv_proj = Linear(
+            h, (cfg.num_heads + 2 * cfg.num_kv_heads) * hd)
         self.input_norm = RMSNorm(h, cfg.rms_eps)
-        self.q_proj = Linear(h, cfg.num_heads * hd)
-        self.k_proj = Linear(h, cfg.num_kv_heads * hd)
-        self.v_proj = Linear(h, cfg.num_kv_heads * hd)
         self.o_proj = Linear(cfg.num_heads * hd, h)
         self.post_norm = RMSNorm(h, cfg.rms_eps)
-        self.gate_proj = Linear(h, cfg.intermediate_size)
-        self.up_proj = Linear(h, cfg.intermediate_size)
+        self.gate_up_proj = Linear(h, 2 * cfg.intermediate_size)
         self.down_proj = Linear(cfg.intermediate_size, h)
 
     def forward(self, x, cos_t, sin_t):
         cfg = self.cfg
         B, S, h = x.shape
+        hd = cfg.head_dim
+        nq, nkv = cfg.num_heads, cfg.num_kv_heads
         res = x
         y = self.input_norm(x)
-        q = self.q_proj(y).view(B, S, cfg.num_heads, cfg.head_dim)
-        k = self.k_proj(y).view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        v = self.v_proj(y).view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        q = K.rope(q, cos_t, sin_t)
-        k = K.rope(k, cos_t, sin_t)
-        # [B,S,H,D] -> [B,H,S,D] for the attention kernel
-        q = q.transpose(1, 2)
-        k = k.transpose(1, 2)
-        v = v.transpose(1, 2)
+        qkv = self.qkv_proj(y)
+        q, k, v = qkv.split([nq * hd, nkv * hd, nkv * hd], dim=-1)
+        q = q.view(B, S, nq, hd).transpose(1, 2).contiguous()
+        k = k.view(B, S, nkv, hd).transpose(1, 2).contiguous()
+        v = v.view(B, S, nkv, hd).transpose(1, 2).contiguous()
+        q = K.rope(q, cos_t, sin_t, layout="bhsd")
+        k = K.rope(k, cos_t, sin_t, layout="bhsd")
         o = K.attention(q, k, v)
-        o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
+        o = o.transpose(1, 2).reshape(B, S, nq * hd)
         x = res + self.o_proj(o)
 
         res = x
         y = self.post_norm(x)
-        x = res + self.down_proj(K.swiglu(self.gate_proj(y),
-                                          self.up_proj(y)))
+        x = res + self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
         return x
 
 

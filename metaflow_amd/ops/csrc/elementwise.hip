@@ -221,3 +221,60 @@ __global__ void add_bf16_kernel(const short* __restrict__ a,
     yv[i] = o;
   }
 }
+
+// ------------------------------------------------- fused-layout SwiGLU
+// gateup: [N, 2I] rows holding [gate | up] halves (the fused MLP GEMM
+// output); avoids slice copies entirely — bwd writes d_gateup in the
+// same fused layout, feeding the fused GEMM backward directly.
+__global__ void swiglu_gu_fwd_kernel(const short* __restrict__ gu,
+                                     short* __restrict__ y,
+                                     long long rows, int I) {
+  const int iv = I / 8;
+  const long long total = rows * iv;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const long long row = idx / iv;
+    const int c8 = (int)(idx % iv);
+    const long long base = row * (long long)(2 * I) + c8 * 8;
+    bf16x8 g = *(const bf16x8*)(gu + base);
+    bf16x8 u = *(const bf16x8*)(gu + base + I);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      o[j] = f2bf(gf * sig * bf2f(u[j]));
+    }
+    *(bf16x8*)(y + row * (long long)I + c8 * 8) = o;
+  }
+}
+
+__global__ void swiglu_gu_bwd_kernel(const short* __restrict__ gu,
+                                     const short* __restrict__ dy,
+                                     short* __restrict__ dgu,
+                                     long long rows, int I) {
+  const int iv = I / 8;
+  const long long total = rows * iv;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const long long row = idx / iv;
+    const int c8 = (int)(idx % iv);
+    const long long base = row * (long long)(2 * I) + c8 * 8;
+    bf16x8 g = *(const bf16x8*)(gu + base);
+    bf16x8 u = *(const bf16x8*)(gu + base + I);
+    bf16x8 d = *(const bf16x8*)(dy + row * (long long)I + c8 * 8);
+    bf16x8 og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]), uf = bf2f(u[j]), df = bf2f(d[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      og[j] = f2bf(df * uf * (sig + silu * (1.f - sig)));
+      ou[j] = f2bf(df * silu);
+    }
+    *(bf16x8*)(dgu + base) = og;
+    *(bf16x8*)(dgu + base + I) = ou;
+  }
+}

@@ -115,6 +115,33 @@ std::vector<torch::Tensor> swiglu_bwd(torch::Tensor g, torch::Tensor u,
   return {dg, du};
 }
 
+std::vector<torch::Tensor> swiglu_gu_fwd(torch::Tensor gu) {
+  check_bf16(gu, "gu");
+  const int I2 = gu.size(-1);
+  TORCH_CHECK(I2 % 16 == 0, "fused swiglu needs 2I % 16 == 0");
+  const int I = I2 / 2;
+  const long long rows = gu.numel() / I2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto y = torch::empty(sizes, gu.options());
+  long long total = rows * (I / 8);
+  swiglu_gu_fwd_kernel<<<grid_for(total), kBlock, 0, cur_stream()>>>(
+      bf(gu), bfm(y), rows, I);
+  HIP_CHECK_KERNEL();
+  return {y};
+}
+
+torch::Tensor swiglu_gu_bwd(torch::Tensor gu, torch::Tensor dy) {
+  const int I = gu.size(-1) / 2;
+  const long long rows = gu.numel() / (2 * I);
+  auto dgu = torch::empty_like(gu);
+  long long total = rows * (I / 8);
+  swiglu_gu_bwd_kernel<<<grid_for(total), kBlock, 0, cur_stream()>>>(
+      bf(gu), bf(dy.contiguous()), bfm(dgu), rows, I);
+  HIP_CHECK_KERNEL();
+  return dgu;
+}
+
 torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a");
   TORCH_CHECK(a.numel() % 8 == 0);
@@ -304,6 +331,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope", &rope, "RoPE (half-rotation), fwd or bwd via sign");
   m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
   m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
+  m.def("swiglu_gu_fwd", [](torch::Tensor gu){ return swiglu_gu_fwd(gu)[0]; }, "fused-layout SwiGLU fwd");
+  m.def("swiglu_gu_bwd", &swiglu_gu_bwd, "fused-layout SwiGLU bwd");
   m.def("add_bf16", &add_bf16, "fused bf16 add");
   m.def("adamw", &adamw, "fused AdamW (bf16 p/g, fp32 m/v[, master])");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward");

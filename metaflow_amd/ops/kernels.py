@@ -101,19 +101,33 @@ class _Rope(torch.autograd.Function):
         return dx, None, None, None, None, None
 
 
-def rope(x, cos_t, sin_t, pos0=0):
-    """x: [B, S, Hh, D] (positions along dim 1). Half-rotation (Llama-HF)."""
-    B, S, Hh, D = x.shape
+def rope(x, cos_t, sin_t, pos0=0, layout="bshd"):
+    """Half-rotation RoPE. layout 'bshd': x [B,S,Hh,D] (positions on dim
+    1); layout 'bhsd': x [B,Hh,S,D] (positions on dim 2 — what the
+    attention kernel consumes, so fused-QKV slices need no extra copy)."""
+    if layout == "bshd":
+        B, S, Hh, D = x.shape
+        rows_per_pos = Hh
+    else:
+        B, Hh, S, D = x.shape
+        rows_per_pos = 1
     if x.is_cuda:
-        return _Rope.apply(x.contiguous(), cos_t, sin_t, Hh, S, pos0)
-    return rope_ref(x, cos_t, sin_t, pos0)
+        return _Rope.apply(x.contiguous(), cos_t, sin_t, rows_per_pos, S,
+                           pos0)
+    return rope_ref(x, cos_t, sin_t, pos0, layout)
 
 
-def rope_ref(x, cos_t, sin_t, pos0=0):
-    B, S, Hh, D = x.shape
+def rope_ref(x, cos_t, sin_t, pos0=0, layout="bshd"):
+    D = x.shape[-1]
     half = D // 2
-    c = cos_t[pos0:pos0 + S].view(1, S, 1, half).to(torch.float32)
-    s = sin_t[pos0:pos0 + S].view(1, S, 1, half).to(torch.float32)
+    if layout == "bshd":
+        S = x.shape[1]
+        c = cos_t[pos0:pos0 + S].view(1, S, 1, half).to(torch.float32)
+        s = sin_t[pos0:pos0 + S].view(1, S, 1, half).to(torch.float32)
+    else:
+        S = x.shape[2]
+        c = cos_t[pos0:pos0 + S].view(1, 1, S, half).to(torch.float32)
+        s = sin_t[pos0:pos0 + S].view(1, 1, S, half).to(torch.float32)
     xf = x.float()
     x1, x2 = xf[..., :half], xf[..., half:]
     return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1).to(x.dtype)
@@ -142,6 +156,28 @@ def swiglu(g, u):
 def swiglu_ref(g, u):
     gf = g.float()
     return (torch.nn.functional.silu(gf) * u.float()).to(g.dtype)
+
+
+class _SwiGLUFused(torch.autograd.Function):
+    """SwiGLU over the fused [.., 2I] gate|up GEMM output — no slice
+    copies; backward emits d_gateup in the same fused layout."""
+
+    @staticmethod
+    def forward(ctx, gu):
+        ctx.save_for_backward(gu)
+        return hip_ext().swiglu_gu_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (gu,) = ctx.saved_tensors
+        return hip_ext().swiglu_gu_bwd(gu, dy)
+
+
+def swiglu_fused(gu):
+    if gu.is_cuda:
+        return _SwiGLUFused.apply(gu.contiguous())
+    I = gu.shape[-1] // 2
+    return swiglu_ref(gu[..., :I], gu[..., I:])
 
 
 # ============================ cross entropy ================================

@@ -195,3 +195,20 @@ def test_tiny_model_step(dev):
         losses.append(float(loss.item()))
     assert all(l == l for l in losses), "NaN loss"
     assert losses[-1] < losses[0], "loss did not decrease: %s" % losses
+
+
+def test_swiglu_fused(dev):
+    from metaflow_amd.ops import kernels as K
+
+    torch.manual_seed(0)
+    gu = torch.randn(512, 2048, dtype=torch.bfloat16, device=dev,
+                     requires_grad=True)
+    y = K.swiglu_fused(gu)
+    gr = gu.detach().clone().float().requires_grad_(True)
+    I = 1024
+    yr = torch.nn.functional.silu(gr[..., :I]) * gr[..., I:]
+    assert rel_err(y, yr) < 2e-2
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    assert rel_err(gu.grad, gr.grad) < 2e-2
