@@ -44,7 +44,18 @@ class TorchParallelDecorator(ParallelDecorator):
 
             backend = self.attributes.get("backend")
             if backend is None:
-                backend = "nccl" if torch.cuda.is_available() else "gloo"
+                num_nodes = int(os.environ.get("MFX_PARALLEL_NUM_NODES",
+                                               "1"))
+                n_gpus = (torch.cuda.device_count()
+                          if torch.cuda.is_available() else 0)
+                # RCCL needs one DISTINCT device per rank; oversubscribed
+                # gangs (e.g. CPU tests or a 2-rank gang on a 1-GPU box)
+                # fall back to gloo. The scheduler pins one visible device
+                # per rank, so n_gpus is what THIS rank can see.
+                visible = int(os.environ.get("MFX_PARALLEL_TOTAL_GPUS",
+                                             str(n_gpus)))
+                backend = "nccl" if (n_gpus > 0 and num_nodes <= visible) \
+                    else "gloo"
             if not dist.is_initialized():
                 dist.init_process_group(backend=backend)
             try:

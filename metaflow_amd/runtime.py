@@ -454,6 +454,7 @@ class NativeRuntime(object):
                 "LOCAL_RANK": "0",
                 "WORLD_SIZE": str(num_parallel),
             }
+            env["MFX_PARALLEL_TOTAL_GPUS"] = str(n_gpus)
             if n_gpus > 0:
                 env["HIP_VISIBLE_DEVICES"] = str(rank % n_gpus)
                 env["CUDA_VISIBLE_DEVICES"] = str(rank % n_gpus)
