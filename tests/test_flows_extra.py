@@ -135,3 +135,23 @@ def test_config_and_mutator(tmp_datastore, tmp_path):
     run_id = latest_run_id(tmp_datastore, "ConfigFlow")
     assert read_artifact(tmp_datastore, "ConfigFlow", run_id, "end",
                          "value") == 70
+
+
+def test_client_lineage_and_liveness(tmp_datastore):
+    run_flow("foreach_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "ForeachFlow")
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    import importlib
+
+    import metaflow_amd.client as client
+
+    importlib.reload(client)
+    client.namespace(None)
+    run = client.Run("ForeachFlow/%s" % run_id)
+    assert not run.is_alive  # finished
+    join_task = run["join"].task
+    parents = join_task.parent_tasks
+    assert len(parents) == 4  # the four foreach work tasks
+    work0 = run["work"].tasks()[0]
+    kids = work0.child_tasks
+    assert any(t.step_name == "join" for t in kids)
