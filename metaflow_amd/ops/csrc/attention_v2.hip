@@ -418,10 +418,14 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
   const long long kvoff =
       (((long long)b * Hkv + hk) * S + kvb * BKVB) * ATT_D;
 
-  // B-operand K/V rows are re-read from global per q-tile (L2-hot after
-  // the first pass; keeping them in registers costs 64 VGPR and spills)
+  // K rows live in registers (hot in every S^T mfma); V rows are
+  // re-read from L2 per q-tile — both in registers spills ~500 B/lane
   const short* krow = k + kvoff + (long long)(wid * 32 + l32) * ATT_D;
   const short* vrow = v + kvoff + (long long)(wid * 32 + l32) * ATT_D;
+  bf16x8 k_reg[8];
+#pragma unroll
+  for (int s = 0; s < 8; ++s)
+    k_reg[s] = *(const bf16x8*)(krow + s * 16 + hi * 8);
 
   f16f acc_dk[4], acc_dv[4];
 #pragma unroll
@@ -460,9 +464,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           bf16x8 qf = frag8(qt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
           bf16x8 df = frag8(dot, t * 32 + l32, s * 16 + hi * 8,
                             ATT_D * 2);
-          bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
           bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
-          st = mfma32(qf, kr, st);
+          st = mfma32(qf, k_reg[s], st);
           dpt = mfma32(df, vr, dpt);
         }
         // P (into st) with causal mask q >= kv; dS (into dpt)

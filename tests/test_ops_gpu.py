@@ -212,3 +212,17 @@ def test_swiglu_fused(dev):
     y.backward(dy)
     yr.backward(dy.float())
     assert rel_err(gu.grad, gr.grad) < 2e-2
+
+
+def test_mfma32_layout_probe(dev):
+    """Regression guard for the 32x32x16 fragment layout every v2 kernel
+    assumes (asymmetric inputs per guide G9: transpose-detecting)."""
+    from metaflow_amd.ops import kernels as K
+
+    ext = K.hip_ext()
+    torch.manual_seed(3)
+    a = torch.randn(32, 16, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(16, 32, dtype=torch.bfloat16, device=dev)
+    c = ext.dbg_mfma32(a, b)
+    ref = a.float() @ b.float()
+    assert rel_err(c, ref) < 1e-3
