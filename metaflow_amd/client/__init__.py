@@ -275,6 +275,10 @@ class Task(MetaflowObject):
         self._fds = FlowDataStore(self.flow_name, self._storage)
         self._ds = self._fds.get_task_datastore(self.run_id, self.step_name,
                                                 self.id)
+        if os.environ.get("MFX_CLIENT_CACHE", "0") == "1":
+            from .filecache import FileBlobCache
+
+            self._fds.ca_store.set_blob_cache(FileBlobCache())
 
     @property
     def successful(self):

@@ -1,0 +1,81 @@
+"""HTTP metadata service + provider round-trip, and a full flow run with
+--metadata service."""
+
+import os
+import socket
+import subprocess
+import sys
+import time
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.fixture
+def service(tmp_path):
+    port = _free_port()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "metaflow_amd.metadata.service",
+         "--root", str(tmp_path / "meta_root"), "--port", str(port)],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    url = "http://127.0.0.1:%d" % port
+    # wait for readiness
+    import urllib.request
+
+    for _ in range(100):
+        try:
+            urllib.request.urlopen(url + "/version", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.1)
+    else:
+        proc.kill()
+        raise RuntimeError("service did not start")
+    yield url
+    proc.terminate()
+    proc.wait(timeout=5)
+
+
+def test_provider_roundtrip(service):
+    from metaflow_amd.metadata.service import ServiceMetadataProvider
+
+    p = ServiceMetadataProvider("TestFlow", url=service)
+    run_id = p.new_run_id(tags=["exp:1"])
+    assert run_id
+    p.register_task(run_id, "start", "1", 0, {"hello": 1})
+    p.heartbeat(run_id)
+    p.add_run_tags(run_id, ["extra"])
+    info = p.get_run(run_id)
+    assert "exp:1" in info["tags"] and "extra" in info["tags"]
+    tasks = p.list_tasks(run_id)
+    assert tasks and tasks[0]["step_name"] == "start"
+    p.register_run_done(run_id, True)
+    assert p.get_run(run_id)["status"] == "successful"
+    assert p.get_task(run_id, "start", "1")["task_id"] == "1"
+
+
+def test_flow_run_with_service_metadata(service, tmp_datastore,
+                                        monkeypatch):
+    from .test_runtime import latest_run_id, run_flow
+
+    monkeypatch.setenv("MFX_SERVICE_URL", service)
+    run_flow("linear_flow.py", tmp_datastore, "--metadata", "service",
+             "run")
+    run_id = latest_run_id(tmp_datastore, "LinearFlow")
+    from metaflow_amd.metadata.service import ServiceMetadataProvider
+
+    p = ServiceMetadataProvider("LinearFlow", url=service)
+    info = p.get_run(run_id)
+    assert info and info["status"] == "successful"
+    assert len(p.list_tasks(run_id)) >= 3
