@@ -78,6 +78,57 @@ def card(pathspec, out):
     click.echo("wrote %s" % out)
 
 
+@cli.command(name="card-server",
+             help="Serve task cards over HTTP for browsing.")
+@click.option("--port", default=8324)
+def card_server(port):
+    import http.server
+
+    from .client import Metaflow, Task, namespace
+    from .plugins.card_decorator import get_card
+
+    namespace(None)
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def log_message(self, *a):
+            pass
+
+        def _send(self, body, ctype="text/html"):
+            self.send_response(200)
+            self.send_header("Content-Type", ctype)
+            self.end_headers()
+            self.wfile.write(body.encode())
+
+        def do_GET(self):
+            path = self.path.strip("/")
+            if not path:
+                rows = []
+                for flow in Metaflow():
+                    run = flow.latest_run
+                    if not run:
+                        continue
+                    for step in run:
+                        for task in step:
+                            html = get_card(task._ds)
+                            if html:
+                                rows.append(
+                                    '<li><a href="/%s">%s</a></li>'
+                                    % (task.pathspec, task.pathspec))
+                self._send("<h1>mfx cards</h1><ul>%s</ul>"
+                           % "".join(rows))
+                return
+            try:
+                task = Task(path)
+                html = get_card(task._ds)
+                self._send(html or "<p>no card</p>")
+            except Exception as e:
+                self._send("<p>error: %s</p>" % e)
+
+    server = http.server.ThreadingHTTPServer(("127.0.0.1", port), Handler)
+    click.echo("serving cards at http://127.0.0.1:%d" % port)
+    server.serve_forever()
+
+
 @cli.command(help="Show visible GPUs (rocm-smi summary).")
 def gpus():
     try:

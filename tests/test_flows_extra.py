@@ -155,3 +155,40 @@ def test_client_lineage_and_liveness(tmp_datastore):
     work0 = run["work"].tasks()[0]
     kids = work0.child_tasks
     assert any(t.step_name == "join" for t in kids)
+
+
+def test_card_server(tmp_datastore):
+    import subprocess
+    import sys
+    import time
+    import urllib.request
+
+    run_flow("card_flow.py", tmp_datastore, "run")
+    import socket
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "metaflow_amd", "card-server",
+         "--port", str(port)], env=env,
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        body = None
+        for _ in range(100):
+            try:
+                body = urllib.request.urlopen(
+                    "http://127.0.0.1:%d/" % port, timeout=1).read()
+                break
+            except Exception:
+                time.sleep(0.1)
+        assert body and b"CardFlow" in body
+        link = body.decode().split('href="/')[1].split('"')[0]
+        card = urllib.request.urlopen(
+            "http://127.0.0.1:%d/%s" % (port, link), timeout=5).read()
+        assert b"custom html" in card
+    finally:
+        proc.terminate()
+        proc.wait(timeout=5)
