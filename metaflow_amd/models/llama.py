@@ -87,13 +87,16 @@ class DecoderLayer(nn.Module):
         self.gate_up_proj = Linear(h, 2 * cfg.intermediate_size)
         self.down_proj = Linear(cfg.intermediate_size, h)
 
-    def forward(self, x, cos_t, sin_t):
+    def forward(self, res, pending, cos_t, sin_t):
+        """Carries (residual_stream, pending_branch): every residual add
+        fuses with the next RMSNorm (K.add_rmsnorm). `pending` is the
+        previous layer's un-added MLP output (None for layer 0)."""
         cfg = self.cfg
-        B, S, h = x.shape
+        B, S, _ = res.shape
         hd = cfg.head_dim
         nq, nkv = cfg.num_heads, cfg.num_kv_heads
-        res = x
-        y = self.input_norm(x)
+        res, y = K.add_rmsnorm(res, pending, self.input_norm.weight,
+                               self.input_norm.eps)
         qkv = self.qkv_proj(y)
         q, k, v = qkv.split([nq * hd, nkv * hd, nkv * hd], dim=-1)
         q = q.view(B, S, nq, hd).transpose(1, 2).contiguous()
@@ -103,12 +106,11 @@ class DecoderLayer(nn.Module):
         k = K.rope(k, cos_t, sin_t, layout="bhsd")
         o = K.attention(q, k, v)
         o = o.transpose(1, 2).reshape(B, S, nq * hd)
-        x = res + self.o_proj(o)
 
-        res = x
-        y = self.post_norm(x)
-        x = res + self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
-        return x
+        res, y = K.add_rmsnorm(res, self.o_proj(o), self.post_norm.weight,
+                               self.post_norm.eps)
+        pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
+        return res, pending
 
 
 class LlamaForCausalLM(nn.Module):
@@ -143,10 +145,12 @@ class LlamaForCausalLM(nn.Module):
     def forward(self, tokens, targets=None):
         """tokens [B, S] int64; returns mean loss if targets given, else
         logits."""
-        x = self.embed(tokens)
+        res = self.embed(tokens)
+        pending = None
         for layer in self.layers:
-            x = layer(x, self.cos_t, self.sin_t)
-        x = self.final_norm(x)
+            res, pending = layer(res, pending, self.cos_t, self.sin_t)
+        _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
+                             self.final_norm.eps)
         if targets is None:
             return self.lm_head(x)
         logits = self.lm_head(x)

@@ -12,8 +12,13 @@
 // y[n][h] = x[n][h] * w[h] * rsqrt(mean_h(x^2) + eps); saves inv_rms[n].
 // One block (256 threads) per row; rows = B*S, H up to 16384 with bf16x8.
 
+// optional fused residual add: when res != null, operates on s = x + res
+// and writes s to sum_out (the new residual stream) — one launch instead
+// of add + norm, and s is read back from LDS-fresh L2 instead of recompute.
 template <int BLOCK>
 __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
+                                   const short* __restrict__ res,
+                                   short* __restrict__ sum_out,
                                    const short* __restrict__ w,
                                    short* __restrict__ y,
                                    float* __restrict__ inv_rms,
@@ -22,12 +27,21 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
   const long long row = blockIdx.x;
   const short* xr = x + row * (long long)H;
   short* yr = y + row * (long long)H;
+  const bf16x8* rv = res ? (const bf16x8*)(res + row * (long long)H)
+                         : nullptr;
+  bf16x8* sv = sum_out ? (bf16x8*)(sum_out + row * (long long)H) : nullptr;
 
   float ssq = 0.f;
   const int HV = H / 8;
   const bf16x8* xv = (const bf16x8*)xr;
   for (int i = threadIdx.x; i < HV; i += BLOCK) {
     bf16x8 v = xv[i];
+    if (rv) {
+      bf16x8 r = rv[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = f2bf(bf2f(v[j]) + bf2f(r[j]));
+      sv[i] = v;
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = bf2f(v[j]);
@@ -39,9 +53,10 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
   if (threadIdx.x == 0 && inv_rms) inv_rms[row] = inv;
 
   const bf16x8* wv = (const bf16x8*)w;
+  const bf16x8* src = rv ? (const bf16x8*)sv : xv;
   bf16x8* yv = (bf16x8*)yr;
   for (int i = threadIdx.x; i < HV; i += BLOCK) {
-    bf16x8 v = xv[i];
+    bf16x8 v = src[i];
     bf16x8 wk = wv[i];
     bf16x8 o;
 #pragma unroll
@@ -54,10 +69,13 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
 // dx[n] = w*dy*inv - x*inv^3/H * sum_h(dy*w*x); dw += sum_n(dy*x*inv).
 // dw accumulated per-block in LDS (fp32 H <= 8192 -> 32 KiB), one
 // atomicAdd per element at the end (guide G12).
+// extra_dsum: optional gradient flowing directly into the (fused) sum
+// output — added to dx so add+norm backward is also one launch.
 template <int BLOCK>
 __global__ void rmsnorm_bwd_kernel(const short* __restrict__ x,
                                    const short* __restrict__ w,
                                    const short* __restrict__ dy,
+                                   const short* __restrict__ extra_dsum,
                                    const float* __restrict__ inv_rms,
                                    short* __restrict__ dx,
                                    float* __restrict__ dw,  // fp32 accum
@@ -87,13 +105,17 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ x,
     dot = block_sum<BLOCK>(dot, scratch);
     const float c = dot * inv * inv * inv / (float)H;
 
+    const bf16x8* ev = extra_dsum
+        ? (const bf16x8*)(extra_dsum + row * (long long)H) : nullptr;
     for (int i = threadIdx.x; i < HV; i += BLOCK) {
       bf16x8 xi = xv[i], di = dyv[i], wi = wv[i];
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xf = bf2f(xi[j]), df = bf2f(di[j]);
-        o[j] = f2bf(df * bf2f(wi[j]) * inv - xf * c);
+        float g = df * bf2f(wi[j]) * inv - xf * c;
+        if (ev) g += bf2f(ev[i][j]);
+        o[j] = f2bf(g);
         dw_part[i * 8 + j] += df * xf * inv;
       }
       dxv[i] = o;

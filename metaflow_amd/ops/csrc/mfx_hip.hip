@@ -53,14 +53,33 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
   auto y = torch::empty_like(x);
   auto inv = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   rmsnorm_fwd_kernel<kBlock><<<(int)rows, kBlock, 0, cur_stream()>>>(
-      bf(x), bf(w), bfm(y), inv.data_ptr<float>(), H, (float)eps);
+      bf(x), nullptr, nullptr, bf(w), bfm(y), inv.data_ptr<float>(), H,
+      (float)eps);
   HIP_CHECK_KERNEL();
   return {y, inv};
 }
 
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x,
+                                           torch::Tensor res,
+                                           torch::Tensor w, double eps) {
+  check_bf16(x, "x");
+  check_bf16(res, "res");
+  const int H = x.size(-1);
+  const long long rows = x.numel() / H;
+  auto s = torch::empty_like(x);
+  auto y = torch::empty_like(x);
+  auto inv = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  rmsnorm_fwd_kernel<kBlock><<<(int)rows, kBlock, 0, cur_stream()>>>(
+      bf(x), bf(res), bfm(s), bf(w), bfm(y), inv.data_ptr<float>(), H,
+      (float)eps);
+  HIP_CHECK_KERNEL();
+  return {s, y, inv};
+}
+
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
                                        torch::Tensor dy,
-                                       torch::Tensor inv_rms) {
+                                       torch::Tensor inv_rms,
+                                       c10::optional<torch::Tensor> dsum) {
   check_bf16(x, "x");
   check_bf16(dy, "dy");
   const int H = x.size(-1);
@@ -69,8 +88,9 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
   auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   int grid = (int)std::min<long long>(rows, 1024);
   size_t lds = (H + kBlock / 64) * sizeof(float);
+  const short* extra = dsum.has_value() ? bf(*dsum) : nullptr;
   rmsnorm_bwd_kernel<kBlock><<<grid, kBlock, lds, cur_stream()>>>(
-      bf(x), bf(w), bf(dy), inv_rms.data_ptr<float>(), bfm(dx),
+      bf(x), bf(w), bf(dy), extra, inv_rms.data_ptr<float>(), bfm(dx),
       dw.data_ptr<float>(), (int)rows, H);
   HIP_CHECK_KERNEL();
   return {dx, dw};
@@ -328,6 +348,7 @@ torch::Tensor dbg_mfma32(torch::Tensor a, torch::Tensor b) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd, "fused residual add + RMSNorm");
   m.def("rope", &rope, "RoPE (half-rotation), fwd or bwd via sign");
   m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
   m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");

@@ -55,7 +55,7 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w, inv = ctx.saved_tensors
-        dx, dw = hip_ext().rmsnorm_bwd(x, w, dy.contiguous(), inv)
+        dx, dw = hip_ext().rmsnorm_bwd(x, w, dy.contiguous(), inv, None)
         return dx, dw.to(w.dtype), None
 
 
@@ -63,6 +63,38 @@ def rmsnorm(x, w, eps=1e-5):
     if x.is_cuda:
         return _RMSNorm.apply(x.contiguous(), w.contiguous(), eps)
     return rmsnorm_ref(x, w, eps).to(x.dtype)
+
+
+class _AddRMSNorm(torch.autograd.Function):
+    """(s, y) = (x + res, rmsnorm(x + res) * w) in one kernel; backward
+    folds the gradient arriving at s into the norm backward (one launch)."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps):
+        s, y, inv = hip_ext().add_rmsnorm_fwd(x, res, w, eps)
+        ctx.save_for_backward(s, w, inv)
+        return s, y
+
+    @staticmethod
+    def backward(ctx, ds, dy):
+        s, w, inv = ctx.saved_tensors
+        if ds is not None:
+            ds = ds.contiguous()
+        dx, dw = hip_ext().rmsnorm_bwd(s, w, dy.contiguous(), inv, ds)
+        # d/dx and d/dres are identical (s = x + res)
+        return dx, dx, dw.to(w.dtype), None
+
+
+def add_rmsnorm(x, res, w, eps=1e-5):
+    """Fused residual add + RMSNorm: returns (sum, normed). res may be
+    None (falls back to plain rmsnorm, sum = x)."""
+    if res is None:
+        return x, rmsnorm(x, w, eps)
+    if x.is_cuda:
+        return _AddRMSNorm.apply(x.contiguous(), res.contiguous(),
+                                 w.contiguous(), eps)
+    s = x + res
+    return s, rmsnorm_ref(s, w, eps).to(x.dtype)
 
 
 def rmsnorm_ref(x, w, eps=1e-5):

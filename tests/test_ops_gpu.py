@@ -226,3 +226,30 @@ def test_mfma32_layout_probe(dev):
     c = ext.dbg_mfma32(a, b)
     ref = a.float() @ b.float()
     assert rel_err(c, ref) < 1e-3
+
+
+def test_add_rmsnorm_fused(dev):
+    from metaflow_amd.ops import kernels as K
+
+    torch.manual_seed(0)
+    x = torch.randn(256, 4096, dtype=torch.bfloat16, device=dev,
+                    requires_grad=True)
+    r = torch.randn_like(x, requires_grad=True)
+    w = torch.randn(4096, dtype=torch.bfloat16, device=dev,
+                    requires_grad=True)
+    s, y = K.add_rmsnorm(x, r, w)
+    xr = x.detach().clone().float().requires_grad_(True)
+    rr = r.detach().clone().float().requires_grad_(True)
+    wr = w.detach().clone().float().requires_grad_(True)
+    sr = xr + rr
+    yr = K.rmsnorm_ref(sr, wr)
+    assert rel_err(s, sr) < 2e-2
+    assert rel_err(y, yr) < 2e-2
+    # both outputs used downstream (like the model: s -> residual, y -> op)
+    ds = torch.randn_like(s)
+    dy = torch.randn_like(y)
+    (s * ds + y * dy).sum().backward()
+    (sr * ds.float() + yr * dy.float()).sum().backward()
+    assert rel_err(x.grad, xr.grad) < 2e-2
+    assert rel_err(r.grad, rr.grad) < 2e-2
+    assert rel_err(w.grad, wr.grad) < 2e-2
