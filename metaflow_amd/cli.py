@@ -7,7 +7,6 @@ inter-process API: the scheduler launches every task as
 """
 
 import json
-import os
 import sys
 import traceback
 
@@ -16,7 +15,6 @@ import click
 from .config import (
     DEFAULT_DATASTORE,
     DEFAULT_METADATA,
-    EXIT_DISALLOW_RETRY,
     MAX_NUM_SPLITS,
     MAX_WORKERS,
 )
@@ -26,7 +24,7 @@ from .exceptions import MFXException
 from .graph import FlowGraph
 from .lint import lint
 from .metadata import METADATA_PROVIDERS
-from .task import MFXTask, PARAMETERS_STEP, PARAMETERS_TASK_ID
+from .task import MFXTask
 
 
 class CLIState(object):

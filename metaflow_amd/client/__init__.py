@@ -7,7 +7,6 @@ directly by the local datastore + metadata files.
 
 import os
 
-from ..config import DATASTORE_LOCAL_DIR
 from ..datastore import FlowDataStore
 from ..datastore.storage import LocalStorage
 from ..exceptions import NamespaceMismatchError, NotFoundError

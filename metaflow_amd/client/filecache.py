@@ -7,7 +7,6 @@ set_blob_cache so repeated client reads skip the datastore.
 """
 
 import os
-import time
 
 DEFAULT_MAX_BYTES = 2 << 30  # 2 GiB
 

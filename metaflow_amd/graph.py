@@ -20,9 +20,6 @@ import ast
 import inspect
 import textwrap
 
-from .exceptions import GraphException
-
-
 def _deindent(src):
     return textwrap.dedent(src)
 

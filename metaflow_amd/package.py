@@ -10,7 +10,6 @@ import io
 import json
 import os
 import tarfile
-import time
 
 DEFAULT_SUFFIXES = (".py", ".txt", ".yaml", ".yml", ".json", ".hip", ".cpp",
                     ".h", ".sh", ".toml", ".cfg")

@@ -9,7 +9,6 @@ the card is rendered in task_finished and readable via
 """
 
 import html
-import json
 import time
 
 from ..decorators import StepDecorator, make_step_decorator

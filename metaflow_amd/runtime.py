@@ -34,7 +34,7 @@ from .config import (
     MAX_NUM_SPLITS,
     MAX_WORKERS,
 )
-from .exceptions import MFXException, TaskFailedException
+from .exceptions import TaskFailedException
 from .task import PARAMETERS_STEP, PARAMETERS_TASK_ID, dump_parameters
 from .unbounded_foreach import UBF_CONTROL
 

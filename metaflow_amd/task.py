@@ -14,13 +14,11 @@ The scheduler only ever learns about the task through its exit code and the
 datastore/metadata state it leaves behind.
 """
 
-import json
 import os
 import sys
 import traceback
 
 from .current import current, Parallel
-from .datastore.task_datastore import TaskDataStore  # noqa: F401 (doc)
 from .exceptions import MFXException
 from .flowspec import ForeachFrame
 from .unbounded_foreach import UBF_CONTROL, UBF_TASK
