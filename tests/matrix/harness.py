@@ -70,6 +70,19 @@ GRAPHS = {
         ("join_p", "join", ["end"], {}),
         ("end", "end", [], {}),
     ],
+    "switch": [
+        ("start", "switch", ["fast", "slow"], {"var": "route"}),
+        ("fast", "linear", ["merge"], {}),
+        ("slow", "linear", ["merge"], {}),
+        ("merge", "linear", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
+    "ubf": [
+        ("start", "ubf", ["inner"], {"var": "payload", "n": 3}),
+        ("inner", "linear", ["join_u"], {}),
+        ("join_u", "join", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
 }
 
 
@@ -91,6 +104,17 @@ def generate_flow(graph_name, test, class_name):
     lines = [
         "import os",
         "from metaflow_amd import FlowSpec, step, current, parallel",
+        "from metaflow_amd import UnboundedForeachInput",
+        "",
+        "class MXListUBF(UnboundedForeachInput):",
+        "    def __init__(self, items):",
+        "        self.items = list(items)",
+        "    def __iter__(self):",
+        "        return iter(self.items)",
+        "    def __len__(self):",
+        "        return len(self.items)",
+        "    def __getitem__(self, i):",
+        "        return self if i is None else self.items[i]",
         "",
         "class %s(FlowSpec):" % class_name,
     ]
@@ -102,14 +126,24 @@ def generate_flow(graph_name, test, class_name):
         if kind == "foreach":
             lines.append("        self.%s = list(range(%d))"
                          % (extra["var"], extra["n"]))
+        elif kind == "ubf":
+            lines.append("        self.%s = MXListUBF(range(%d))"
+                         % (extra["var"], extra["n"]))
+        elif kind == "switch":
+            lines.append("        self.%s = '%s'"
+                         % (extra["var"], "fast"))
         for b in body:
             lines.append("        " + b)
         # transition
         if kind == "end":
             pass
-        elif kind == "foreach":
+        elif kind in ("foreach", "ubf"):
             lines.append("        self.next(self.%s, foreach='%s')"
                          % (targets[0], extra["var"]))
+        elif kind == "switch":
+            lines.append("        self.next(%s, condition='%s')"
+                         % (", ".join("self.%s" % t for t in targets),
+                            extra["var"]))
         elif kind == "parallel":
             lines.append("        self.next(self.%s, num_parallel=%d)"
                          % (targets[0], extra["n"]))
@@ -200,9 +234,12 @@ class StepCounterFlow(MatrixTest):
 
     def check(self, run, graph):
         data = run["end"].task.data
-        # every non-join step's mark must have propagated to the end
+        # every non-join step's mark must have propagated to the end;
+        # switch graphs only run the chosen branch ('fast' here)
+        skipped = {t for n, k, ts, _e in graph if k == "switch"
+                   for t in ts if t != "fast"}
         for name, kind, _t, _e in graph:
-            if kind != "join":
+            if kind != "join" and name not in skipped:
                 assert getattr(data, "mark_%s" % name) == name
 
 
