@@ -262,7 +262,9 @@ def main(flow):
     @click.argument("step_name")
     @click.option("--run-id", "origin_run_id", required=True,
                   help="The finished run to spin against.")
-    def spin(step_name, origin_run_id):
+    @click.option("--split-index", default=None, type=int,
+                  help="Foreach index when spinning a foreach-child step.")
+    def spin(step_name, origin_run_id, split_index):
         from .task import PARAMETERS_STEP as PSTEP
 
         node = state.graph[step_name]
@@ -282,7 +284,8 @@ def main(flow):
         spin_run_id = "spin%s" % origin_run_id
         task = MFXTask(state.flow, state.graph, state.flow_datastore,
                        state.metadata)
-        task.run_step(step_name, spin_run_id, "1", input_paths, None, 0, 0)
+        task.run_step(step_name, spin_run_id, "1", input_paths,
+                      split_index, 0, 0)
         click.echo("spin task done: %s/%s/1" % (spin_run_id, step_name))
 
     @cli.command(name="tag", help="Mutate run tags: tag add/remove RUN_ID "

@@ -97,13 +97,22 @@ def load_state_dict(task_datastore, name="checkpoint", map_location="cpu"):
     sha_to_names = {}
     for key, info in index.items():
         sha_to_names.setdefault(info["sha"], []).append(key)
+    import warnings
+
     for sha, blob in cas.load_blobs(list(sha_to_names)):
         for key in sha_to_names[sha]:
             info = index[key]
             dtype = getattr(torch, info["dtype"])
-            t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
-            t = t.view(dtype).reshape(info["shape"])
-            if map_location != "cpu":
-                t = t.to(map_location)
+            if map_location == "cpu":
+                # writable copy (restored tensors get trained on)
+                t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+                t = t.view(dtype).reshape(info["shape"])
+            else:
+                # zero host copies: view the blob read-only, H2D is the
+                # only copy
+                with warnings.catch_warnings():
+                    warnings.simplefilter("ignore")
+                    t = torch.frombuffer(blob, dtype=torch.uint8)
+                t = t.view(dtype).reshape(info["shape"]).to(map_location)
             out[key] = t
     return out
