@@ -100,6 +100,8 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
 torch::Tensor rope(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
                    long rows_per_pos, long seqlen, long pos0, bool backward) {
   check_bf16(x, "x");
+  TORCH_CHECK(cos_t.is_cuda() && sin_t.is_cuda(),
+              "rope tables must be on the GPU (host pointers fault)");
   const int D = x.size(-1);
   const long long rows = x.numel() / D;
   auto y = torch::empty_like(x);

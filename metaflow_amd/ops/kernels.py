@@ -104,16 +104,22 @@ def rmsnorm_ref(x, w, eps=1e-5):
 
 
 # ================================ RoPE =====================================
-def rope_tables(seqlen, dim, theta=500000.0, device="cpu"):
-    """Host-precomputed fp32 cos/sin tables [seqlen, dim/2] (guide App B:
-    on-device trig turns a memory-bound op VALU-bound)."""
+def rope_tables(seqlen, dim, theta=500000.0, device=None):
+    """Precomputed fp32 cos/sin tables [seqlen, dim/2] (guide App B:
+    on-device trig turns a memory-bound op VALU-bound). device=None keeps
+    the active default device (so models built under `with torch.device`
+    get device-resident tables — the kernel must never see host pointers)."""
     half = dim // 2
     freqs = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64)
                              / half))
     pos = torch.arange(seqlen, dtype=torch.float64)
     ang = torch.outer(pos, freqs)
-    return (ang.cos().float().to(device).contiguous(),
-            ang.sin().float().to(device).contiguous())
+    cos_t = ang.cos().float().contiguous()
+    sin_t = ang.sin().float().contiguous()
+    if device is not None:
+        cos_t = cos_t.to(device)
+        sin_t = sin_t.to(device)
+    return cos_t, sin_t
 
 
 class _Rope(torch.autograd.Function):
