@@ -13,6 +13,7 @@ steps can move many objects at device-filling throughput.
 """
 
 import os
+import random
 import shutil
 import time
 from collections import namedtuple
@@ -21,6 +22,15 @@ from concurrent.futures import ThreadPoolExecutor
 from .exceptions import MFXException
 
 StoredObject = namedtuple("StoredObject", ["key", "path", "size"])
+
+
+def _maybe_inject_failure():
+    """Test hook (reference parity: s3op --inject-failure): when
+    MFX_INJECT_IO_FAILURES=P (0..100), IO ops raise a transient OSError
+    with probability P% — exercised by the retry machinery tests."""
+    p = float(os.environ.get("MFX_INJECT_IO_FAILURES", "0"))
+    if p > 0 and random.random() * 100 < p:
+        raise OSError("injected transient IO failure")
 
 
 class ObjectStore(object):
@@ -48,6 +58,7 @@ class ObjectStore(object):
         err = None
         for attempt in range(self._retries):
             try:
+                _maybe_inject_failure()
                 return fn(*args)
             except OSError as e:
                 err = e

@@ -70,3 +70,14 @@ def test_log_truncation(tmp_datastore, tmp_path):
     # capped at ~1 MB + truncation marker + mflog framing
     assert os.path.getsize(log) < 3 << 20
     assert b"[log truncated]" in open(log, "rb").read()
+
+
+def test_scheduler_stress(tmp_datastore):
+    """48-way foreach with 25% transient failures under the 16-worker cap:
+    every task recovers via @retry."""
+    proc = run_flow("stress_flow.py", tmp_datastore, "run",
+                    "--max-num-splits", "64", timeout=420)
+    run_id = latest_run_id(tmp_datastore, "StressFlow")
+    retried = read_artifact(tmp_datastore, "StressFlow", run_id, "join",
+                            "retried")
+    assert retried > 0
