@@ -94,6 +94,52 @@ DEV void col_to_afrags1(const f16f& st, bf16x8 pa[2], int hi) {
   }
 }
 
+// ================= panel layout + hardware transpose reads =================
+// A [ROWS][128] bf16 tile stored as 8 panels of [ROWS][16]: element
+// (row, d) lives at (d>>4)*ROWS*16 + row*16 + (d&15). Properties:
+//  * staging from row-major global is pure vec8 -> vec8 (no scalar scatter);
+//  * A-fragment reads (row fixed per lane, 8 contiguous d) are one
+//    16-byte read: panel (d0>>4), offset row*16 + (d0&8);
+//  * B-fragments with K = the row dimension come from ds_read_b64_tr_b16:
+//    the HW pattern elem = base + (l&15) + j*16 + (l>>4)*64 walks column
+//    (l&15) of a [4][16] subtile, which in panel storage is exactly
+//    4 consecutive rows at one d-column — solve base per lane so lane
+//    l = 32*hi + l32 receives rows q0+hi*8+j at column n*32+l32.
+
+template <int ROWS, int BLOCK>
+DEV void stage_panel(short* dst, const short* src,
+                     long long src_row_stride) {
+  constexpr int TOT = ROWS * 16;  // vec8 slots
+#pragma unroll
+  for (int it = 0; it < (TOT + BLOCK - 1) / BLOCK; ++it) {
+    const int idx = it * BLOCK + threadIdx.x;
+    if (idx < TOT) {
+      const int r = idx % ROWS, c8 = idx / ROWS;
+      bf16x8 v = *(const bf16x8*)(src + r * src_row_stride + c8 * 8);
+      *(bf16x8*)(dst + (c8 >> 1) * (ROWS * 16) + r * 16 + (c8 & 1) * 8) =
+          v;
+    }
+  }
+}
+
+template <int ROWS>
+DEV bf16x8 frag8_panel(const short* tile, int row, int d0) {
+  return *(const bf16x8*)(tile + (d0 >> 4) * (ROWS * 16) + row * 16
+                          + (d0 & 15));
+}
+
+// issue one tr-read: returns 2 VGPRs holding rows (base_q .. base_q+3) at
+// this lane's column; caller must s_waitcnt lgkmcnt + sched_barrier before
+// consuming (rule #18).
+DEV unsigned long long tr_read(const short* lds_ptr) {
+  // generic pointers to __shared__ carry the LDS offset in the low 32
+  // bits; ds_* instructions take that 32-bit address
+  unsigned int addr = (unsigned int)(unsigned long long)lds_ptr;
+  unsigned long long out;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(out) : "v"(addr));
+  return out;
+}
+
 template <int BLOCK>  // BLOCK = 512 (8 waves)
 __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
@@ -398,10 +444,11 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
     short* __restrict__ dk, short* __restrict__ dv,
     int B, int H, int Hkv, int S, float scale) {
   constexpr int BKVB = 256, BQ2 = 64;
-  __shared__ short qt[BQ2 * ATT_D];    // Q rm swizzled (A for S^T)
-  __shared__ short qtt[ATT_D * BQ2];   // Q^T swizzled (B for dK)
-  __shared__ short dot[BQ2 * ATT_D];   // dO rm swizzled (A for dP^T)
-  __shared__ short dott[ATT_D * BQ2];  // dO^T swizzled (B for dV)
+  // panel layout [8][BQ2][16] serves BOTH the A-fragment reads (contiguous
+  // within a panel row) and the B-fragments via ds_read_b64_tr_b16 — no
+  // transposed copies, staging is pure vec8 (see panel helpers above)
+  __shared__ short qp[BQ2 * ATT_D];    // Q panels
+  __shared__ short dop[BQ2 * ATT_D];   // dO panels
   __shared__ float lse_s[BQ2];
   __shared__ float del_s[BQ2];
 
@@ -443,10 +490,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
       __syncthreads();
       const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
       const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
-      stage_rm<BQ2, BLOCK>(qt, qsrc, ATT_D);
-      stage_tr<BQ2, BLOCK>(qtt, qsrc, ATT_D);
-      stage_rm<BQ2, BLOCK>(dot, dsrc, ATT_D);
-      stage_tr<BQ2, BLOCK>(dott, dsrc, ATT_D);
+      stage_panel<BQ2, BLOCK>(qp, qsrc, ATT_D);
+      stage_panel<BQ2, BLOCK>(dop, dsrc, ATT_D);
       if (threadIdx.x < BQ2) {
         lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
         del_s[threadIdx.x] = delta[hoff + jq * BQ2 + threadIdx.x];
@@ -461,9 +506,10 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
         f16f st = (f16f){}, dpt = (f16f){};
 #pragma unroll
         for (int s = 0; s < 8; ++s) {
-          bf16x8 qf = frag8(qt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
-          bf16x8 df = frag8(dot, t * 32 + l32, s * 16 + hi * 8,
-                            ATT_D * 2);
+          bf16x8 qf = frag8_panel<BQ2>(qp, t * 32 + l32,
+                                       s * 16 + hi * 8);
+          bf16x8 df = frag8_panel<BQ2>(dop, t * 32 + l32,
+                                       s * 16 + hi * 8);
           bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
           st = mfma32(qf, k_reg[s], st);
           dpt = mfma32(df, vr, dpt);
@@ -479,27 +525,49 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           st[r] = p;
           dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
         }
+        // per-lane tr_read base (elements): delivers rows q0+hi*8+j at
+        // column n*32+l32 of the panel tile (derivation at panel helpers)
+        const int lane_col_panel = (((lane >> 4) & 1)) * (BQ2 * 16)
+                                   - (lane >> 4) * 64 + hi * 8 * 16;
         // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d)
         bf16x8 pa[2];
         col_to_afrags1(st, pa, hi);
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
+        for (int ks = 0; ks < 2; ++ks) {
+          const int q0 = t * 32 + ks * 16;
+          union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
-            bf16x8 df = frag8(dott, n * 32 + l32,
-                              t * 32 + ks * 16 + hi * 8, BQ2 * 2);
-            acc_dv[n] = mfma32(pa[ks], df, acc_dv[n]);
+            const int base = n * 2 * (BQ2 * 16) + q0 * 16
+                             + lane_col_panel;
+            bfr[n].u[0] = tr_read(dop + base);
+            bfr[n].u[1] = tr_read(dop + base + 4 * 16);
           }
-        // dK += dS^T @ Q
+          asm volatile("s_waitcnt lgkmcnt(0)");
+          __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc_dv[n] = mfma32(pa[ks], bfr[n].v, acc_dv[n]);
+        }
+        // dK(32kv x 128d) += dS^T(32kv x 32q) @ Q(32q x 128d)
         col_to_afrags1(dpt, pa, hi);
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
+        for (int ks = 0; ks < 2; ++ks) {
+          const int q0 = t * 32 + ks * 16;
+          union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
-            bf16x8 qf = frag8(qtt, n * 32 + l32,
-                              t * 32 + ks * 16 + hi * 8, BQ2 * 2);
-            acc_dk[n] = mfma32(pa[ks], qf, acc_dk[n]);
+            const int base = n * 2 * (BQ2 * 16) + q0 * 16
+                             + lane_col_panel;
+            bfr[n].u[0] = tr_read(qp + base);
+            bfr[n].u[1] = tr_read(qp + base + 4 * 16);
           }
+          asm volatile("s_waitcnt lgkmcnt(0)");
+          __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc_dk[n] = mfma32(pa[ks], bfr[n].v, acc_dk[n]);
+        }
       }
     }
   }
