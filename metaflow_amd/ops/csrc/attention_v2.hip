@@ -525,10 +525,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           st[r] = p;
           dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
         }
-        // per-lane tr_read base (elements): delivers rows q0+hi*8+j at
-        // column n*32+l32 of the panel tile (derivation at panel helpers)
-        const int lane_col_panel = (((lane >> 4) & 1)) * (BQ2 * 16)
-                                   - (lane >> 4) * 64 + hi * 8 * 16;
+        // ds_read_b64_tr_b16 semantics: each lane loads 64b at its OWN
+        // address; the HW transposes 16-bit elements within each 16-lane
+        // group (a [4][16] tile). To make lane l=32*hi+l32 RECEIVE rows
+        // q0+hi*8+j at panel column n*32+l32, lane l must LOAD the 4
+        // elements of row q0+hi*8+((l>>2)&3) at col offset 4*(l&3):
+        const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
+        const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
         // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d)
         bf16x8 pa[2];
         col_to_afrags1(st, pa, hi);
@@ -538,8 +541,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
-            const int base = n * 2 * (BQ2 * 16) + q0 * 16
-                             + lane_col_panel;
+            const int base = n * 2 * (BQ2 * 16) + tr_panel
+                             + (q0 + hi * 8) * 16 + tr_lane_off;
             bfr[n].u[0] = tr_read(dop + base);
             bfr[n].u[1] = tr_read(dop + base + 4 * 16);
           }
@@ -557,8 +560,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
-            const int base = n * 2 * (BQ2 * 16) + q0 * 16
-                             + lane_col_panel;
+            const int base = n * 2 * (BQ2 * 16) + tr_panel
+                             + (q0 + hi * 8) * 16 + tr_lane_off;
             bfr[n].u[0] = tr_read(qp + base);
             bfr[n].u[1] = tr_read(qp + base + 4 * 16);
           }
