@@ -146,11 +146,12 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse, int B, int H, int Hkv, int S, float scale) {
   constexpr int BQ = 256, BKV = 64;  // 8 waves x 32 q rows
-  // double-buffered K/V tiles: stage j+1 overlaps compute on j (T14
-  // async-stage split: global loads issue BEFORE compute, LDS writes
-  // happen after — HBM latency hides under the MFMA phase)
-  __shared__ short kt[2][BKV * ATT_D];   // K rm swizzled [64][128]
-  __shared__ short vtt[2][ATT_D * BKV];  // V^T swizzled [128][64]
+  // double-buffered K/V tiles in the 16-wide PANEL layout [8][64][16]:
+  // staging is pure vec8 writes, K A-fragments are contiguous panel
+  // reads, V B-fragments come from ds_read_b64_tr_b16 (no transposed
+  // copy). Stage j+1 overlaps compute on j (T14 async-stage split).
+  __shared__ short kp[2][BKV * ATT_D];
+  __shared__ short vp[2][BKV * ATT_D];
 
   const int qb = blockIdx.x;
   const int h = blockIdx.y;
@@ -205,12 +206,10 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
 #define MFX_STAGE_WRITE(buf)                                             \
   {                                                                      \
     _Pragma("unroll") for (int u = 0; u < 2; ++u) {                      \
-      *(bf16x8*)((char*)kt[buf]                                          \
-                 + swz(stg_r[u], stg_c8[u] * 8, ATT_D * 2)) = stg_k[u];  \
-      _Pragma("unroll") for (int jj = 0; jj < 8; ++jj)                   \
-        *(short*)((char*)vtt[buf]                                        \
-                  + swz(stg_c8[u] * 8 + jj, stg_r[u], BKV * 2)) =        \
-            stg_v[u][jj];                                                \
+      const int pel_ = (stg_c8[u] >> 1) * (BKV * 16) + stg_r[u] * 16     \
+                       + (stg_c8[u] & 1) * 8;                            \
+      *(bf16x8*)(kp[buf] + pel_) = stg_k[u];                             \
+      *(bf16x8*)(vp[buf] + pel_) = stg_v[u];                             \
     }                                                                    \
   }
 
@@ -234,8 +233,8 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
       st[t] = (f16f){};
 #pragma unroll
       for (int s = 0; s < 8; ++s) {
-        bf16x8 kf = frag8(kt[cur], t * 32 + l32, s * 16 + hi * 8,
-                          ATT_D * 2);
+        bf16x8 kf = frag8_panel<BKV>(kp[cur], t * 32 + l32,
+                                     s * 16 + hi * 8);
         st[t] = mfma32(kf, q_reg[s], st[t]);
       }
     }
@@ -292,13 +291,25 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     col_to_afrags(st, pa, hi);
 
     // ---- PV: O(32q x 128d) += P(32q x 64kv) @ V(64kv x 128d) ----
+    // B-fragments via hardware transpose (see tr_read/panel derivation)
+    {
+      const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
+      const int tr_panel = ((lane >> 4) & 1) * (BKV * 16);
 #pragma unroll
-    for (int ks = 0; ks < 4; ++ks) {
+      for (int ks = 0; ks < 4; ++ks) {
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        bf16x8 vf = frag8(vtt[cur], n * 32 + l32, ks * 16 + hi * 8,
-                          BKV * 2);
-        acc_o[n] = mfma32(pa[ks], vf, acc_o[n]);
+        for (int n = 0; n < 4; ++n) {
+          const int base = n * 2 * (BKV * 16) + tr_panel
+                           + (ks * 16 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(vp[cur] + base);
+          bfr[n].u[1] = tr_read(vp[cur] + base + 4 * 16);
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc_o[n] = mfma32(pa[ks], bfr[n].v, acc_o[n]);
       }
     }
     }  // active
@@ -338,9 +349,10 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int B, int H, int Hkv, int S, float scale) {
   constexpr int BQ = 256, BKV = 64;
-  __shared__ short kt[BKV * ATT_D];    // K rm swizzled (B for S^T)
-  __shared__ short ktt[ATT_D * BKV];   // K^T swizzled (B for dQ)
-  __shared__ short vt[BKV * ATT_D];    // V rm swizzled (B for dP^T)
+  // panel layout serves both A-fragments (contiguous) and the dQ
+  // B-fragments via ds_read_b64_tr_b16 — K^T copy eliminated
+  __shared__ short kp[BKV * ATT_D];    // K panels
+  __shared__ short vp[BKV * ATT_D];    // V panels
 
   const int qb = blockIdx.x;
   const int h = blockIdx.y;
@@ -376,11 +388,10 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
   const int kv_tiles = (qb * BQ + BQ) / BKV;
   for (int j = 0; j < kv_tiles; ++j) {
     __syncthreads();
-    const short* ksrc = k + kvoff0 + (long long)j * BKV * ATT_D;
-    stage_rm<BKV, BLOCK>(kt, ksrc, ATT_D);
-    stage_tr<BKV, BLOCK>(ktt, ksrc, ATT_D);
-    stage_rm<BKV, BLOCK>(vt, v + kvoff0 + (long long)j * BKV * ATT_D,
-                         ATT_D);
+    stage_panel<BKV, BLOCK>(kp, k + kvoff0 + (long long)j * BKV * ATT_D,
+                            ATT_D);
+    stage_panel<BKV, BLOCK>(vp, v + kvoff0 + (long long)j * BKV * ATT_D,
+                            ATT_D);
     __syncthreads();
     if (j * BKV > qb * BQ + wid * 32 + 31) continue;
 
@@ -390,8 +401,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
       f16f st = (f16f){}, dpt = (f16f){};
 #pragma unroll
       for (int s = 0; s < 8; ++s) {
-        bf16x8 kf = frag8(kt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
-        bf16x8 vf = frag8(vt, t * 32 + l32, s * 16 + hi * 8, ATT_D * 2);
+        bf16x8 kf = frag8_panel<BKV>(kp, t * 32 + l32, s * 16 + hi * 8);
+        bf16x8 vf = frag8_panel<BKV>(vp, t * 32 + l32, s * 16 + hi * 8);
         st = mfma32(kf, q_reg[s], st);
         dpt = mfma32(vf, do_reg[s], dpt);
       }
@@ -407,14 +418,25 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
       // dQ(32q x 128d) += dS(32q x 32kv) @ K(32kv x 128d)
       bf16x8 pa[2];
       col_to_afrags1(st, pa, hi);
+      const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
+      const int tr_panel = ((lane >> 4) & 1) * (BKV * 16);
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kv0 = t * 32 + ks * 16;
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
         for (int n = 0; n < 4; ++n) {
-          bf16x8 kf = frag8(ktt, n * 32 + l32,
-                            t * 32 + ks * 16 + hi * 8, BKV * 2);
-          acc_dq[n] = mfma32(pa[ks], kf, acc_dq[n]);
+          const int base = n * 2 * (BKV * 16) + tr_panel
+                           + (kv0 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(kp + base);
+          bfr[n].u[1] = tr_read(kp + base + 4 * 16);
         }
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc_dq[n] = mfma32(pa[ks], bfr[n].v, acc_dq[n]);
+      }
     }
   }
 
