@@ -277,6 +277,7 @@ class NativeRuntime(object):
             "flow_name": self.flow_cls.__name__,
             "run_id": self.run_id,
             "datastore_root": self.flow_datastore.datastore_root,
+            "provider": getattr(self.metadata, "TYPE", "local"),
         })
         self._save_code_package()
         if self.clone_run_id:
@@ -431,13 +432,23 @@ class NativeRuntime(object):
         input_paths = list(mapper_paths)
         self._queue_spec(TaskSpec(join_step, input_paths, None, join_stack))
 
+    def _gpus_per_rank(self, step):
+        """@resources(gpu=N) on the gang step -> N devices per rank."""
+        func = getattr(self.flow_cls, step)
+        for deco in getattr(func, "decorators", []):
+            if deco.name == "resources":
+                return max(1, int(deco.attributes.get("gpu") or 1))
+        return 1
+
     def _queue_gang(self, spec, pathspec, target, num_parallel):
         """Native gang scheduling for @parallel steps: N rank processes with
-        rendezvous env + GPU pinning. Replaces the reference's control-task
-        subprocess emulation (parallel_decorator.py:175-246)."""
+        rendezvous env + GPU pinning (@resources(gpu=k) gives each rank k
+        devices). Replaces the reference's control-task subprocess
+        emulation (parallel_decorator.py:175-246)."""
         control_id = self._new_task_id()
         port = _free_port()
         n_gpus = _visible_gpu_count()
+        gpr = self._gpus_per_rank(target)
         for rank in range(num_parallel):
             task_id = control_id if rank == 0 else "%s_node_%d" % (
                 control_id, rank)
@@ -456,8 +467,10 @@ class NativeRuntime(object):
             }
             env["MFX_PARALLEL_TOTAL_GPUS"] = str(n_gpus)
             if n_gpus > 0:
-                env["HIP_VISIBLE_DEVICES"] = str(rank % n_gpus)
-                env["CUDA_VISIBLE_DEVICES"] = str(rank % n_gpus)
+                devs = ",".join(str((rank * gpr + g) % n_gpus)
+                                for g in range(gpr))
+                env["HIP_VISIBLE_DEVICES"] = devs
+                env["CUDA_VISIBLE_DEVICES"] = devs
             child_stack = spec.stack + ((spec.step, rank, num_parallel),)
             self._queue_spec(TaskSpec(
                 target, [pathspec], rank, child_stack, task_id=task_id,

@@ -77,12 +77,18 @@ class SidecarSubProcess(object):
 def _heartbeat_worker(context):
     """Write run heartbeats every ~10 s until stdin closes."""
     from .datastore.storage import LocalStorage
-    from .metadata.local import LocalMetadataProvider
 
     flow = context["flow_name"]
     run_id = context["run_id"]
     root = context["datastore_root"]
-    meta = LocalMetadataProvider(flow, LocalStorage(root))
+    if context.get("provider") == "service":
+        from .metadata.service import ServiceMetadataProvider
+
+        meta = ServiceMetadataProvider(flow)
+    else:
+        from .metadata.local import LocalMetadataProvider
+
+        meta = LocalMetadataProvider(flow, LocalStorage(root))
     meta.heartbeat(run_id)
     import select
 
