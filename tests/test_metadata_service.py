@@ -67,15 +67,19 @@ def test_provider_roundtrip(service):
 
 def test_flow_run_with_service_metadata(service, tmp_datastore,
                                         monkeypatch):
-    from .test_runtime import latest_run_id, run_flow
+    from .test_runtime import run_flow
 
     monkeypatch.setenv("MFX_SERVICE_URL", service)
     run_flow("linear_flow.py", tmp_datastore, "--metadata", "service",
              "run")
-    run_id = latest_run_id(tmp_datastore, "LinearFlow")
     from metaflow_amd.metadata.service import ServiceMetadataProvider
 
     p = ServiceMetadataProvider("LinearFlow", url=service)
+    runs = p.list_runs()
+    assert runs, "service has no runs registered"
+    run_id = runs[0]["run_id"]
     info = p.get_run(run_id)
     assert info and info["status"] == "successful"
     assert len(p.list_tasks(run_id)) >= 3
+    # heartbeat sidecar went through the service provider too
+    assert p._request("GET", "/version")["api_version"] == 1
