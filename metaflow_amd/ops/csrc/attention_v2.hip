@@ -491,10 +491,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
   // re-read from L2 per q-tile — both in registers spills ~500 B/lane
   const short* krow = k + kvoff + (long long)(wid * 32 + l32) * ATT_D;
   const short* vrow = v + kvoff + (long long)(wid * 32 + l32) * ATT_D;
-  bf16x8 k_reg[8];
-#pragma unroll
-  for (int s = 0; s < 8; ++s)
-    k_reg[s] = *(const bf16x8*)(krow + s * 16 + hi * 8);
 
   f16f acc_dk[4], acc_dv[4];
 #pragma unroll
@@ -532,8 +528,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
                                        s * 16 + hi * 8);
           bf16x8 df = frag8_panel<BQ2>(dop, t * 32 + l32,
                                        s * 16 + hi * 8);
+          bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
           bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
-          st = mfma32(qf, k_reg[s], st);
+          st = mfma32(qf, kr, st);
           dpt = mfma32(df, vr, dpt);
         }
         // P (into st) with causal mask q >= kv; dS (into dpt)
