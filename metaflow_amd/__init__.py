@@ -40,6 +40,7 @@ from .client import (
 )
 from .multicore_utils import parallel_map, parallel_imap_unordered
 from .datatools import ObjectStore
+from .runner import Runner
 
 __version__ = "0.1.0"
 
@@ -81,6 +82,7 @@ __all__ = [
     "namespace",
     "get_namespace",
     "ObjectStore",
+    "Runner",
     "parallel_map",
     "parallel_imap_unordered",
 ]
