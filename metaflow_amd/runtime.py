@@ -217,20 +217,25 @@ class NativeRuntime(object):
 
     # -------------------------------------------------------------- resume
     def _build_origin_index(self):
-        """(step, stack) -> origin TaskDataStore for all DONE origin tasks."""
+        """(step, stack) -> origin TaskDataStore for all DONE origin
+        tasks. Datastores init concurrently (reference TaskDataStoreSet
+        prefetch, runtime.py:445)."""
+        from .datastore.datastore_set import TaskDataStoreSet
+
         index = {}
         origin = self.clone_run_id
-        for step in self.flow_datastore.list_steps(origin):
-            if step == PARAMETERS_STEP:
+        steps = [s for s in self.flow_datastore.list_steps(origin)
+                 if s != PARAMETERS_STEP]
+        for ds in TaskDataStoreSet(self.flow_datastore, origin,
+                                   steps=steps):
+            if ds.attempt is None:
                 continue
-            for ds in self.flow_datastore.get_done_task_datastores(
-                    origin, step):
-                ok = ds.load_metadata("attempt_ok") or {}
-                if not ok.get("ok"):
-                    continue
-                frames = ds.load_metadata("foreach_stack") or []
-                stack = tuple((f[0], f[3], f[2]) for f in frames)
-                index[(step, stack)] = ds
+            ok = ds.load_metadata("attempt_ok") or {}
+            if not ok.get("ok"):
+                continue
+            frames = ds.load_metadata("foreach_stack") or []
+            stack = tuple((f[0], f[3], f[2]) for f in frames)
+            index[(ds.step_name, stack)] = ds
         return index
 
     def _maybe_clone(self, spec):

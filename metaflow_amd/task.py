@@ -111,12 +111,18 @@ class MFXTask(object):
         raise MFXException("Bad input pathspec %r" % pathspec)
 
     def _input_datastores(self, input_paths):
-        out = []
-        for p in input_paths:
-            run_id, step, task = self._parse_pathspec(p)
-            ds = self.flow_datastore.get_task_datastore(run_id, step, task)
-            out.append(ds)
-        return out
+        """Joins with many inputs init their datastores concurrently
+        (reference task.py:252: parallel prefetch if >4 inputs)."""
+        specs = [self._parse_pathspec(p) for p in input_paths]
+        if len(specs) <= 4:
+            return [self.flow_datastore.get_task_datastore(r, s, t)
+                    for r, s, t in specs]
+        from concurrent.futures import ThreadPoolExecutor
+
+        with ThreadPoolExecutor(max_workers=8) as pool:
+            return list(pool.map(
+                lambda spec: self.flow_datastore.get_task_datastore(*spec),
+                specs))
 
     def _reconstruct_foreach_stack(self, node, input_dss, split_index):
         """Build this task's foreach frame stack from its first input's
