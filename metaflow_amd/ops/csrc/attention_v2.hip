@@ -144,7 +144,8 @@ template <int BLOCK>  // BLOCK = 512 (8 waves)
 __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
-    float* __restrict__ lse, int B, int H, int Hkv, int S, float scale) {
+    float* __restrict__ lse, int B, int H, int Hkv, int S, float scale,
+    int causal) {
   constexpr int BQ = 256, BKV = 64;  // 8 waves x 32 q rows
   // double-buffered K/V tiles in the 16-wide PANEL layout [8][64][16]:
   // staging is pure vec8 writes, K A-fragments are contiguous panel
@@ -213,7 +214,8 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
     }                                                                    \
   }
 
-  const int kv_tiles = (qb * BQ + BQ) / BKV;  // causal bound (diag incl.)
+  const int kv_tiles =
+      causal ? (qb * BQ + BQ) / BKV : S / BKV;  // causal bound incl diag
   MFX_STAGE_LOAD(0);
   MFX_STAGE_WRITE(0);
   __syncthreads();
@@ -223,7 +225,7 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
 
     // wave-uniform skip: this wave's rows are all below the tile's kv
     // range (fully masked) — staging + barrier still run below
-    const bool active = (j * BKV <= qb * BQ + wid * 32 + 31);
+    const bool active = !causal || (j * BKV <= qb * BQ + wid * 32 + 31);
     if (active) {
 
     // ---- S^T = K (64x128) @ Q^T: two 32-kv sub-tiles ----
@@ -241,7 +243,7 @@ __global__ __launch_bounds__(512) void attn_fwd_v2_kernel(
 
     // ---- scale + causal mask + in-register online softmax ----
     // lane's value (t, r) is S[my_qrow][kv = j*64 + t*32 + krow(r,hi)]
-    const bool diag = (j * BKV + BKV > qb * BQ);
+    const bool diag = causal && (j * BKV + BKV > qb * BQ);
     float pmax = -INFINITY;
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
@@ -347,7 +349,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq, int B, int H, int Hkv, int S, float scale) {
+    short* __restrict__ dq, int B, int H, int Hkv, int S, float scale,
+    int causal) {
   constexpr int BQ = 256, BKV = 64;
   // panel layout serves both A-fragments (contiguous) and the dQ
   // B-fragments via ds_read_b64_tr_b16 — K^T copy eliminated
@@ -385,7 +388,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
 #pragma unroll
   for (int n = 0; n < 4; ++n) acc_dq[n] = (f16f){};
 
-  const int kv_tiles = (qb * BQ + BQ) / BKV;
+  const int kv_tiles = causal ? (qb * BQ + BQ) / BKV : S / BKV;
   for (int j = 0; j < kv_tiles; ++j) {
     __syncthreads();
     stage_panel<BKV, BLOCK>(kp, k + kvoff0 + (long long)j * BKV * ATT_D,
@@ -393,7 +396,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
     stage_panel<BKV, BLOCK>(vp, v + kvoff0 + (long long)j * BKV * ATT_D,
                             ATT_D);
     __syncthreads();
-    if (j * BKV > qb * BQ + wid * 32 + 31) continue;
+    if (causal && j * BKV > qb * BQ + wid * 32 + 31) continue;
 
     // per 32-kv sub-tile: S^T, dP^T, dS, dQ — keeps live regs low
 #pragma unroll
@@ -411,7 +414,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
       for (int r = 0; r < 16; ++r) {
         const int kvg = j * BKV + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
         float p = 0.f;
-        if (kvg <= my_qrow)
+        if (!causal || kvg <= my_qrow)
           p = __expf(st[r] * scale - my_lse);
         st[r] = p * (dpt[r] - my_del) * scale;
       }
@@ -464,7 +467,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv,
-    int B, int H, int Hkv, int S, float scale) {
+    int B, int H, int Hkv, int S, float scale, int causal) {
   constexpr int BKVB = 256, BQ2 = 64;
   // panel layout [8][BQ2][16] serves BOTH the A-fragment reads (contiguous
   // within a panel row) and the B-fragments via ds_read_b64_tr_b16 — no
@@ -499,7 +502,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
     acc_dv[n] = (f16f){};
   }
 
-  const int jq0 = (kvb * BKVB) / BQ2;
+  const int jq0 = causal ? (kvb * BKVB) / BQ2 : 0;
   const int nq = S / BQ2;
   for (int g = 0; g < G; ++g) {
     const int h = hk * G + g;
@@ -516,7 +519,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
       }
       __syncthreads();
       // wave-uniform skip: all of this wave's kv rows above every q row
-      if (jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
+      if (causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
 
       // per 32-q sub-tile: S, dP, P/dS, dV, dK — keeps live regs low
 #pragma unroll
@@ -539,7 +542,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
           const int qrg = jq * BQ2 + qrl;
           float p = 0.f;
-          if (qrg >= my_kvrow)
+          if (!causal || qrg >= my_kvrow)
             p = __expf(st[r] * scale - lse_s[qrl]);
           st[r] = p;
           dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;

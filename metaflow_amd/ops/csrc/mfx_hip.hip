@@ -237,7 +237,8 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
 
 // ----------------------------------------------------------- attention
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, double scale) {
+                                    torch::Tensor v, double scale,
+                                    bool causal) {
   check_bf16(q, "q");
   check_bf16(k, "k");
   check_bf16(v, "v");
@@ -246,6 +247,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(D == 128, "attention requires head dim 128");
   TORCH_CHECK(S % 64 == 0, "attention requires seqlen % 64 == 0");
   TORCH_CHECK(H % Hkv == 0, "GQA requires H % Hkv == 0");
+  TORCH_CHECK(causal || S % 256 == 0,
+              "non-causal attention requires seqlen % 256 == 0");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   if (S % 256 == 0) {
@@ -253,7 +256,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     dim3 grid(S / 256, H, B);
     attn_fwd_v2_kernel<512><<<grid, 512, 0, cur_stream()>>>(
         bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
-        (float)scale);
+        (float)scale, causal ? 1 : 0);
   } else {
     dim3 grid(S / 64, H, B);
     attn_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
@@ -267,7 +270,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     torch::Tensor dout, torch::Tensor lse,
-                                    double scale) {
+                                    double scale, bool causal) {
   const int B = q.size(0), H = q.size(1), S = q.size(2);
   const int Hkv = k.size(1);
   auto dq = torch::empty_like(q);
@@ -283,17 +286,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
         bf(dout), bf(o), delta.data_ptr<float>(), rows);
   HIP_CHECK_KERNEL();
   }
+  TORCH_CHECK(causal || S % 256 == 0,
+              "non-causal attention requires seqlen % 256 == 0");
   if (S % 256 == 0) {
     dim3 gkv(S / 256, Hkv, B);
     attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
         bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
         delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
-        (float)scale);
+        (float)scale, causal ? 1 : 0);
     HIP_CHECK_KERNEL();
     dim3 gq(S / 256, H, B);
     attn_bwd_dq_v2_kernel<512><<<gq, 512, 0, cur_stream()>>>(
         bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-        delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale);
+        delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale,
+        causal ? 1 : 0);
     HIP_CHECK_KERNEL();
     return {dq, dk, dv};
   }

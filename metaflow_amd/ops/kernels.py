@@ -254,44 +254,107 @@ def cross_entropy_ref(logits, targets, ignore_index=-100):
 # ============================== attention ==================================
 class _Attention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale):
-        o, lse = hip_ext().attn_fwd(q, k, v, scale)
+    def forward(ctx, q, k, v, scale, causal):
+        o, lse = hip_ext().attn_fwd(q, k, v, scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
+        ctx.causal = causal
         return o
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, o, lse = ctx.saved_tensors
         dq, dk, dv = hip_ext().attn_bwd(q, k, v, o, dout.contiguous(), lse,
-                                        ctx.scale)
-        return dq, dk, dv, None
+                                        ctx.scale, ctx.causal)
+        return dq, dk, dv, None, None
 
 
-def attention(q, k, v, scale=None):
-    """Causal GQA flash attention. q: [B,H,S,128], k/v: [B,Hkv,S,128]."""
+def attention(q, k, v, scale=None, causal=True):
+    """GQA flash attention. q: [B,H,S,128], k/v: [B,Hkv,S,128]."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.size(-1))
     if q.is_cuda:
         return _Attention.apply(q.contiguous(), k.contiguous(),
-                                v.contiguous(), scale)
-    return attention_ref(q, k, v, scale)
+                                v.contiguous(), scale, causal)
+    return attention_ref(q, k, v, scale, causal)
 
 
-def attention_ref(q, k, v, scale=None):
-    """fp32 eager reference (causal, GQA)."""
+def _ref_scores(q, k, scale, causal):
+    """Scaled (and causally masked) fp32 score matrix with GQA expand."""
+    G = q.size(1) // k.size(1)
+    kf = k.float().repeat_interleave(G, dim=1)
+    s = torch.matmul(q.float(), kf.transpose(-1, -2)) * scale
+    if causal:
+        S, Skv = q.size(2), k.size(2)
+        mask = torch.ones(S, Skv, dtype=torch.bool, device=q.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    return s
+
+
+def attention_ref(q, k, v, scale=None, causal=True):
+    """fp32 eager reference (GQA)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.size(-1))
-    B, H, S, D = q.shape
+    G = q.size(1) // k.size(1)
+    s = _ref_scores(q, k, scale, causal)
+    p = torch.softmax(s, dim=-1)
+    vf = v.float().repeat_interleave(G, dim=1)
+    return torch.matmul(p, vf).to(q.dtype)
+
+
+def attn_fwd_raw(q, k, v, scale, causal=True):
+    """Non-autograd attention forward returning (o, lse).
+
+    ``lse[b,h,s] = log sum_kv exp(score*scale)`` — the merge statistic for
+    context-parallel (ring) attention. HIP kernel on GPU; fp32 reference
+    on CPU.
+    """
+    if q.is_cuda:
+        o, lse = hip_ext().attn_fwd(q.contiguous(), k.contiguous(),
+                                    v.contiguous(), scale, causal)
+        return o, lse
+    s = _ref_scores(q, k, scale, causal)
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.exp(s - lse.unsqueeze(-1))
+    G = q.size(1) // k.size(1)
+    vf = v.float().repeat_interleave(G, dim=1)
+    return torch.matmul(p, vf).to(q.dtype), lse
+
+
+def attn_bwd_raw(q, k, v, o, dout, lse, scale, causal=True):
+    """Non-autograd flash backward against a GLOBAL (o, lse).
+
+    The softmax statistics may cover MORE kv than (k, v) — exactly the
+    per-chunk backward a ring-attention pass needs: p is formed from the
+    global lse and delta from the global o, so each chunk's (dq, dk, dv)
+    contribution sums to the true gradient. Mirrors the HIP kernels'
+    formula (attention_v2.hip: p = exp(s*scale - lse),
+    ds = p * (dp - delta) * scale).
+    """
+    if q.is_cuda:
+        return hip_ext().attn_bwd(q.contiguous(), k.contiguous(),
+                                  v.contiguous(), o.contiguous(),
+                                  dout.contiguous(), lse, scale, causal)
+    G = q.size(1) // k.size(1)
     Hkv = k.size(1)
-    G = H // Hkv
+    s = _ref_scores(q, k, scale, causal)
+    p = torch.exp(s - lse.unsqueeze(-1))       # global-softmax probs
     kf = k.float().repeat_interleave(G, dim=1)
     vf = v.float().repeat_interleave(G, dim=1)
-    s = torch.matmul(q.float(), kf.transpose(-1, -2)) * scale
-    mask = torch.ones(S, S, dtype=torch.bool, device=q.device).tril()
-    s = s.masked_fill(~mask, float("-inf"))
-    p = torch.softmax(s, dim=-1)
-    return torch.matmul(p, vf).to(q.dtype)
+    dof = dout.float()
+    delta = (dof * o.float()).sum(-1)          # [B,H,S]
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    ds = p * (dp - delta.unsqueeze(-1)) * scale
+    dq = torch.matmul(ds, kf).to(q.dtype)
+    B, H, S, D = q.shape
+    Skv = k.size(2)
+    dsg = ds.view(B, Hkv, G, S, Skv)
+    qg = q.float().view(B, Hkv, G, S, D)
+    dog = dof.view(B, Hkv, G, S, D)
+    pg = p.view(B, Hkv, G, S, Skv)
+    dk = torch.einsum("bhgsk,bhgsd->bhkd", dsg, qg).to(k.dtype)
+    dv = torch.einsum("bhgsk,bhgsd->bhkd", pg, dog).to(v.dtype)
+    return dq, dk, dv
 
 
 # ================================ adam =====================================
