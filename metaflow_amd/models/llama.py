@@ -87,10 +87,12 @@ class DecoderLayer(nn.Module):
         self.gate_up_proj = Linear(h, 2 * cfg.intermediate_size)
         self.down_proj = Linear(cfg.intermediate_size, h)
 
-    def forward(self, res, pending, cos_t, sin_t):
+    def forward(self, res, pending, cos_t, sin_t, cp_group=None):
         """Carries (residual_stream, pending_branch): every residual add
         fuses with the next RMSNorm (K.add_rmsnorm). `pending` is the
-        previous layer's un-added MLP output (None for layer 0)."""
+        previous layer's un-added MLP output (None for layer 0). With
+        cp_group set, S is the LOCAL sequence shard and attention runs
+        the xGMI ring (parallel/ring_attention.py)."""
         cfg = self.cfg
         B, S, _ = res.shape
         hd = cfg.head_dim
@@ -104,7 +106,12 @@ class DecoderLayer(nn.Module):
         v = v.view(B, S, nkv, hd).transpose(1, 2).contiguous()
         q = K.rope(q, cos_t, sin_t, layout="bhsd")
         k = K.rope(k, cos_t, sin_t, layout="bhsd")
-        o = K.attention(q, k, v)
+        if cp_group is not None:
+            from ..parallel.ring_attention import ring_attention
+
+            o = ring_attention(q, k, v, group=cp_group)
+        else:
+            o = K.attention(q, k, v)
         o = o.transpose(1, 2).reshape(B, S, nq * hd)
 
         res, y = K.add_rmsnorm(res, self.o_proj(o), self.post_norm.weight,
@@ -114,9 +121,19 @@ class DecoderLayer(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    """Set ``cp_group`` (a torch.distributed group) for context-parallel
+    training: every rank in the group holds the SAME parameters but a
+    different contiguous sequence shard of each batch (rank r gets token
+    rows r*Sc..(r+1)*Sc); attention runs over the xGMI ring. Parameters
+    being replicated, the per-rank local-mean losses backprop to grads
+    whose cp-group AVERAGE is the true global-mean-loss gradient — i.e.
+    cp ranks join the flat-buffer DDP all-reduce exactly like extra data-
+    parallel ranks (parallel/ddp.py needs no changes)."""
+
+    def __init__(self, cfg: LlamaConfig, cp_group=None):
         super().__init__()
         self.cfg = cfg
+        self.cp_group = cp_group
         self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                   dtype=torch.bfloat16)
         self.layers = nn.ModuleList(
@@ -143,12 +160,22 @@ class LlamaForCausalLM(nn.Module):
             layer.down_proj.weight.data.mul_(scale)
 
     def forward(self, tokens, targets=None):
-        """tokens [B, S] int64; returns mean loss if targets given, else
+        """tokens [B, S] int64 (the LOCAL shard when cp_group is set);
+        returns mean loss over the local tokens if targets given, else
         logits."""
+        S = tokens.size(1)
+        off = 0
+        if self.cp_group is not None:
+            import torch.distributed as dist
+
+            off = dist.get_rank(self.cp_group) * S
+        cos_t = self.cos_t[off:off + S].contiguous() if off else self.cos_t
+        sin_t = self.sin_t[off:off + S].contiguous() if off else self.sin_t
         res = self.embed(tokens)
         pending = None
         for layer in self.layers:
-            res, pending = layer(res, pending, self.cos_t, self.sin_t)
+            res, pending = layer(res, pending, cos_t, sin_t,
+                                 self.cp_group)
         _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
                              self.final_norm.eps)
         if targets is None:
