@@ -100,12 +100,8 @@ class DecoderLayer(nn.Module):
         res, y = K.add_rmsnorm(res, pending, self.input_norm.weight,
                                self.input_norm.eps)
         qkv = self.qkv_proj(y)
-        q, k, v = qkv.split([nq * hd, nkv * hd, nkv * hd], dim=-1)
-        q = q.view(B, S, nq, hd).transpose(1, 2).contiguous()
-        k = k.view(B, S, nkv, hd).transpose(1, 2).contiguous()
-        v = v.view(B, S, nkv, hd).transpose(1, 2).contiguous()
-        q = K.rope(q, cos_t, sin_t, layout="bhsd")
-        k = K.rope(k, cos_t, sin_t, layout="bhsd")
+        # one fused kernel: QKV split + bhsd transpose + RoPE
+        q, k, v = K.rope_qkv(qkv, cos_t, sin_t, nq, nkv)
         if cp_group is not None:
             from ..parallel.ring_attention import ring_attention
 

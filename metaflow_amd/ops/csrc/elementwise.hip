@@ -176,6 +176,121 @@ __global__ void rope_kernel(const short* __restrict__ x,
   }
 }
 
+// --------------------------------------------- fused QKV split + RoPE
+// Consumes the fused QKV GEMM output [B, S, (NQ+2*NKV)*128] directly and
+// emits rope'd q [B,NQ,S,128], k [B,NKV,S,128] and copied v [B,NKV,S,128]
+// in the attention kernel's bhsd layout — replaces three slice+transpose
+// contiguous() copies and two rope launches per decoder layer with ONE
+// pass over the bytes. Reads and writes are both bf16x8 along D.
+__global__ void rope_qkv_fwd_kernel(
+    const short* __restrict__ qkv, short* __restrict__ qo,
+    short* __restrict__ ko, short* __restrict__ vo,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    int B, int S, int NQ, int NKV) {
+  constexpr int D = 128, half = 64, hv = half / 8;
+  const int HT = NQ + 2 * NKV;
+  const long long total = (long long)B * S * HT * hv;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const long long row = idx / hv;    // (b*S + s)*HT + h
+    const int d8 = (int)(idx % hv);
+    const int h = (int)(row % HT);
+    const long long bs = row / HT;
+    const int s = (int)(bs % S);
+    const int b = (int)(bs / S);
+    const long long src = row * D + d8 * 8;
+    short* dst_base;
+    if (h < NQ) {
+      dst_base = qo + (((long long)b * NQ + h) * S + s) * D;
+    } else if (h < NQ + NKV) {
+      dst_base = ko + (((long long)b * NKV + (h - NQ)) * S + s) * D;
+    } else {
+      dst_base = vo + (((long long)b * NKV + (h - NQ - NKV)) * S + s) * D;
+    }
+    bf16x8 v1 = *(const bf16x8*)(qkv + src);
+    bf16x8 v2 = *(const bf16x8*)(qkv + src + half);
+    if (h < NQ + NKV) {
+      const long long tbase = (long long)s * half + d8 * 8;
+      f32x4v c0 = *(const f32x4v*)(cos_t + tbase);
+      f32x4v c1 = *(const f32x4v*)(cos_t + tbase + 4);
+      f32x4v s0 = *(const f32x4v*)(sin_t + tbase);
+      f32x4v s1 = *(const f32x4v*)(sin_t + tbase + 4);
+      bf16x8 o1, o2;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float c = (j < 4 ? c0[j] : c1[j - 4]);
+        const float sn = (j < 4 ? s0[j] : s1[j - 4]);
+        const float x1 = bf2f(v1[j]);
+        const float x2 = bf2f(v2[j]);
+        o1[j] = f2bf(x1 * c - x2 * sn);
+        o2[j] = f2bf(x2 * c + x1 * sn);
+      }
+      *(bf16x8*)(dst_base + d8 * 8) = o1;
+      *(bf16x8*)(dst_base + d8 * 8 + half) = o2;
+    } else {
+      *(bf16x8*)(dst_base + d8 * 8) = v1;
+      *(bf16x8*)(dst_base + d8 * 8 + half) = v2;
+    }
+  }
+}
+
+// Backward: gathers dq/dk/dv [B,H,S,128] back into the fused GEMM-grad
+// layout [B, S, (NQ+2*NKV)*128], applying the inverse rotation to the
+// q/k parts (rope with -sin).
+__global__ void rope_qkv_bwd_kernel(
+    const short* __restrict__ dq, const short* __restrict__ dk,
+    const short* __restrict__ dv, short* __restrict__ dqkv,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    int B, int S, int NQ, int NKV) {
+  constexpr int D = 128, half = 64, hv = half / 8;
+  const int HT = NQ + 2 * NKV;
+  const long long total = (long long)B * S * HT * hv;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const long long row = idx / hv;
+    const int d8 = (int)(idx % hv);
+    const int h = (int)(row % HT);
+    const long long bs = row / HT;
+    const int s = (int)(bs % S);
+    const int b = (int)(bs / S);
+    const long long dst = row * D + d8 * 8;
+    const short* src_base;
+    if (h < NQ) {
+      src_base = dq + (((long long)b * NQ + h) * S + s) * D;
+    } else if (h < NQ + NKV) {
+      src_base = dk + (((long long)b * NKV + (h - NQ)) * S + s) * D;
+    } else {
+      src_base = dv + (((long long)b * NKV + (h - NQ - NKV)) * S + s) * D;
+    }
+    bf16x8 v1 = *(const bf16x8*)(src_base + d8 * 8);
+    bf16x8 v2 = *(const bf16x8*)(src_base + d8 * 8 + half);
+    if (h < NQ + NKV) {
+      const long long tbase = (long long)s * half + d8 * 8;
+      f32x4v c0 = *(const f32x4v*)(cos_t + tbase);
+      f32x4v c1 = *(const f32x4v*)(cos_t + tbase + 4);
+      f32x4v s0 = *(const f32x4v*)(sin_t + tbase);
+      f32x4v s1 = *(const f32x4v*)(sin_t + tbase + 4);
+      bf16x8 o1, o2;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float c = (j < 4 ? c0[j] : c1[j - 4]);
+        const float sn = -(j < 4 ? s0[j] : s1[j - 4]);
+        const float x1 = bf2f(v1[j]);
+        const float x2 = bf2f(v2[j]);
+        o1[j] = f2bf(x1 * c - x2 * sn);
+        o2[j] = f2bf(x2 * c + x1 * sn);
+      }
+      *(bf16x8*)(dqkv + dst) = o1;
+      *(bf16x8*)(dqkv + dst + half) = o2;
+    } else {
+      *(bf16x8*)(dqkv + dst) = v1;
+      *(bf16x8*)(dqkv + dst + half) = v2;
+    }
+  }
+}
+
 // ---------------------------------------------------------------- SwiGLU
 // y = silu(g) * u ; dg = dy * u * silu'(g) ; du = dy * silu(g)
 __global__ void swiglu_fwd_kernel(const short* __restrict__ g,

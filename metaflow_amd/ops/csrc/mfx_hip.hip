@@ -113,6 +113,48 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
   return y;
 }
 
+// ----------------------------------------------------- fused qkv rope
+std::vector<torch::Tensor> rope_qkv_fwd(torch::Tensor qkv,
+                                        torch::Tensor cos_t,
+                                        torch::Tensor sin_t, long nq,
+                                        long nkv) {
+  check_bf16(qkv, "qkv");
+  TORCH_CHECK(cos_t.is_cuda() && sin_t.is_cuda(),
+              "rope tables must be on the GPU (host pointers fault)");
+  const int B = qkv.size(0), S = qkv.size(1);
+  const int NQ = (int)nq, NKV = (int)nkv;
+  TORCH_CHECK(qkv.size(2) == (long)(NQ + 2 * NKV) * 128,
+              "qkv last dim must be (nq + 2*nkv) * 128");
+  auto opts = qkv.options();
+  auto q = torch::empty({B, NQ, S, 128}, opts);
+  auto k = torch::empty({B, NKV, S, 128}, opts);
+  auto v = torch::empty({B, NKV, S, 128}, opts);
+  const long long total = (long long)B * S * (NQ + 2 * NKV) * 8;
+  rope_qkv_fwd_kernel<<<grid_for(total), kBlock, 0, cur_stream()>>>(
+      bf(qkv), bfm(q), bfm(k), bfm(v), cos_t.data_ptr<float>(),
+      sin_t.data_ptr<float>(), B, S, NQ, NKV);
+  HIP_CHECK_KERNEL();
+  return {q, k, v};
+}
+
+torch::Tensor rope_qkv_bwd(torch::Tensor dq, torch::Tensor dk,
+                           torch::Tensor dv, torch::Tensor cos_t,
+                           torch::Tensor sin_t) {
+  check_bf16(dq, "dq");
+  check_bf16(dk, "dk");
+  check_bf16(dv, "dv");
+  const int B = dq.size(0), NQ = dq.size(1), S = dq.size(2);
+  const int NKV = dk.size(1);
+  auto dqkv = torch::empty({B, S, (long)(NQ + 2 * NKV) * 128},
+                           dq.options());
+  const long long total = (long long)B * S * (NQ + 2 * NKV) * 8;
+  rope_qkv_bwd_kernel<<<grid_for(total), kBlock, 0, cur_stream()>>>(
+      bf(dq), bf(dk), bf(dv), bfm(dqkv), cos_t.data_ptr<float>(),
+      sin_t.data_ptr<float>(), B, S, NQ, NKV);
+  HIP_CHECK_KERNEL();
+  return dqkv;
+}
+
 // -------------------------------------------------------------- swiglu
 torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u) {
   check_bf16(g, "g");
@@ -366,6 +408,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw", &adamw, "fused AdamW (bf16 p/g, fp32 m/v[, master])");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward");
+  m.def("rope_qkv_fwd", &rope_qkv_fwd,
+        "fused QKV split + transpose + RoPE");
+  m.def("rope_qkv_bwd", &rope_qkv_bwd,
+        "fused QKV rope backward -> GEMM-grad layout");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (causal, GQA)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward");
   m.def("dbg_st", &dbg_st, "debug S^T path");

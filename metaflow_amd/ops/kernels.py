@@ -251,6 +251,42 @@ def cross_entropy_ref(logits, targets, ignore_index=-100):
         reduction="none")
 
 
+# ============================ fused qkv rope ===============================
+class _RopeQKV(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, cos_t, sin_t, nq, nkv):
+        q, k, v = hip_ext().rope_qkv_fwd(qkv, cos_t, sin_t, nq, nkv)
+        ctx.save_for_backward(cos_t, sin_t)
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        cos_t, sin_t = ctx.saved_tensors
+        dqkv = hip_ext().rope_qkv_bwd(dq.contiguous(), dk.contiguous(),
+                                      dv.contiguous(), cos_t, sin_t)
+        return dqkv, None, None, None, None
+
+
+def rope_qkv(qkv, cos_t, sin_t, nq, nkv):
+    """Fused QKV split + bhsd transpose + RoPE.
+
+    qkv: the fused projection output [B, S, (nq+2*nkv)*128]; returns
+    rope'd q [B,nq,S,128], k [B,nkv,S,128] and plain v [B,nkv,S,128] in
+    one pass (elementwise.hip: rope_qkv_fwd_kernel) — replaces three
+    transpose-contiguous copies and two rope launches per decoder layer.
+    """
+    if qkv.is_cuda:
+        return _RopeQKV.apply(qkv.contiguous(), cos_t, sin_t, nq, nkv)
+    B, S, _ = qkv.shape
+    q, k, v = qkv.split([nq * 128, nkv * 128, nkv * 128], dim=-1)
+    q = q.reshape(B, S, nq, 128).transpose(1, 2).contiguous()
+    k = k.reshape(B, S, nkv, 128).transpose(1, 2).contiguous()
+    v = v.reshape(B, S, nkv, 128).transpose(1, 2).contiguous()
+    q = rope(q, cos_t, sin_t, layout="bhsd")
+    k = rope(k, cos_t, sin_t, layout="bhsd")
+    return q, k, v
+
+
 # ============================== attention ==================================
 class _Attention(torch.autograd.Function):
     @staticmethod

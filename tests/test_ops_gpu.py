@@ -171,6 +171,31 @@ def test_attention_bwd(dev):
     assert rel_err(v.grad, vr.grad) < 3e-2, "dv"
 
 
+def test_rope_qkv_fused(dev):
+    """Fused QKV split+transpose+RoPE vs the torch composition (which the
+    CPU fallback implements), forward and backward."""
+    from metaflow_amd.ops import kernels as K
+
+    torch.manual_seed(5)
+    B, S, nq, nkv = 2, 256, 4, 2
+    qkv = torch.randn(B, S, (nq + 2 * nkv) * 128, dtype=torch.bfloat16,
+                      device=dev, requires_grad=True)
+    cos_t, sin_t = K.rope_tables(S, 128, 500000.0, device=dev)
+    q, k, v = K.rope_qkv(qkv, cos_t, sin_t, nq, nkv)
+    dq, dk, dv = (torch.randn_like(q), torch.randn_like(k),
+                  torch.randn_like(v))
+    torch.autograd.backward([q, k, v], [dq, dk, dv])
+    g_fused = qkv.grad.clone()
+
+    qkv2 = qkv.detach().cpu().requires_grad_(True)
+    q2, k2, v2 = K.rope_qkv(qkv2, cos_t.cpu(), sin_t.cpu(), nq, nkv)
+    torch.autograd.backward([q2, k2, v2],
+                            [dq.cpu(), dk.cpu(), dv.cpu()])
+    for name, got, want in (("q", q, q2), ("k", k, k2), ("v", v, v2),
+                            ("dqkv", g_fused, qkv2.grad)):
+        assert rel_err(got, want.to(dev)) < 2e-2, name
+
+
 def test_attention_noncausal(dev):
     """Non-causal kernel path (causal=0 in attention_v2.hip) — the
     off-diagonal chunk op of ring attention (parallel/ring_attention.py):
