@@ -135,6 +135,23 @@ class ContentAddressedStore(object):
         self._storage.save_bytes(_packed(), overwrite=False)
         return results
 
+    def blob_file(self, key):
+        """Local raw-blob fast path: (filesystem_path, payload_offset) if
+        this key is stored as an uncompressed MFXB file on local disk,
+        else None. Lets large-tensor consumers (parallel/checkpoint.py)
+        ``readinto`` pinned memory instead of round-tripping bytes."""
+        path = self._storage.load_file_path(self._key_path(key))
+        if path is None:
+            return None
+        try:
+            with open(path, "rb") as f:
+                hdr = f.read(HEADER_LEN)
+        except OSError:
+            return None
+        if len(hdr) < HEADER_LEN or hdr[:4] != MAGIC or hdr[5] != CODEC_RAW:
+            return None
+        return path, HEADER_LEN
+
     def load_blobs(self, keys, force_raw=False):
         """Yield (key, bytes) for each key."""
         missing = []

@@ -87,6 +87,54 @@ def save_state_dict(task_datastore, state_dict, name="checkpoint"):
     return index
 
 
+def _load_file_to_device(path, off, nbytes, dtype, shape, device, torch):
+    """Pipelined file -> pinned -> HBM load: readinto one pinned half
+    while the other half's async H2D is in flight on the side stream."""
+    global _side_stream
+    dest = torch.empty(shape, dtype=dtype, device=device)
+    flat = dest.reshape(-1).view(torch.uint8)
+    pin = _get_pin_buf(torch)
+    half = _PIN_BUF_BYTES // 2
+    views = [pin[:half], pin[half:]]
+    np_views = [memoryview(v.numpy()) for v in views]
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
+    events = [torch.cuda.Event(), torch.cuda.Event()]
+    recorded = [False, False]
+    with open(path, "rb") as f:
+        f.seek(off)
+        offset, i = 0, 0
+        with torch.cuda.stream(_side_stream):
+            while offset < nbytes:
+                n = min(half, nbytes - offset)
+                if recorded[i]:
+                    events[i].synchronize()  # pin half i free to reuse?
+                got = f.readinto(np_views[i][:n])
+                if got != n:
+                    raise IOError("short read from %s" % path)
+                flat[offset:offset + n].copy_(views[i][:n],
+                                              non_blocking=True)
+                events[i].record(_side_stream)
+                recorded[i] = True
+                offset += n
+                i ^= 1
+        _side_stream.synchronize()
+    return dest
+
+
+def _load_file_to_cpu(path, off, nbytes, dtype, shape, torch):
+    """Single-copy CPU load: readinto a fresh writable buffer (the old
+    bytes -> bytearray path copied twice)."""
+    import numpy as np
+
+    arr = np.empty(nbytes, dtype=np.uint8)
+    with open(path, "rb") as f:
+        f.seek(off)
+        if f.readinto(memoryview(arr)) != nbytes:
+            raise IOError("short read from %s" % path)
+    return torch.from_numpy(arr).view(dtype).reshape(shape)
+
+
 def load_state_dict(task_datastore, name="checkpoint", map_location="cpu"):
     import torch
 
@@ -99,11 +147,30 @@ def load_state_dict(task_datastore, name="checkpoint", map_location="cpu"):
         sha_to_names.setdefault(info["sha"], []).append(key)
     import warnings
 
-    for sha, blob in cas.load_blobs(list(sha_to_names)):
+    to_cuda = str(map_location) != "cpu"
+    slow_shas = []
+    for sha, names in sha_to_names.items():
+        loc = cas.blob_file(sha)
+        if loc is None:
+            slow_shas.append(sha)
+            continue
+        path, off = loc
+        for key in names:
+            info = index[key]
+            dtype = getattr(torch, info["dtype"])
+            nbytes = info["nbytes"]
+            if to_cuda:
+                out[key] = _load_file_to_device(
+                    path, off, nbytes, dtype, info["shape"], map_location,
+                    torch)
+            else:
+                out[key] = _load_file_to_cpu(path, off, nbytes, dtype,
+                                             info["shape"], torch)
+    for sha, blob in cas.load_blobs(slow_shas):
         for key in sha_to_names[sha]:
             info = index[key]
             dtype = getattr(torch, info["dtype"])
-            if map_location == "cpu":
+            if not to_cuda:
                 # writable copy (restored tensors get trained on)
                 t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
                 t = t.view(dtype).reshape(info["shape"])

@@ -160,3 +160,40 @@ def test_flow_datastore_raw_data(tmp_path):
     [(uri, key)] = fds.save_data([b"code package"])
     [(k, data)] = fds.load_data([key])
     assert data == b"code package"
+
+
+def test_checkpoint_save_load_roundtrip(tmp_path):
+    """save_state_dict -> load_state_dict through the CAS raw fast path
+    (cas.blob_file + single-copy readinto), mixed dtypes + non-tensor
+    metadata; loaded tensors must be writable."""
+    import torch
+
+    from metaflow_amd.parallel.checkpoint import (
+        load_state_dict,
+        save_state_dict,
+    )
+
+    fds = make_fds(tmp_path)
+    ds = fds.get_task_datastore("9", "train", "t1", attempt=0, mode="w")
+    ds.init_task()
+    torch.manual_seed(0)
+    state = {
+        "w": torch.randn(1000, 33, dtype=torch.bfloat16),
+        "m": torch.randn(513, dtype=torch.float32),
+        "step": 7,
+    }
+    index = save_state_dict(ds, state, name="u")
+    assert set(index) == {"w", "m"}
+    ds.done()
+
+    rd = fds.get_task_datastore("9", "train", "t1")
+    # the blobs must be locally resolvable as raw files (fast path)
+    for info in index.values():
+        loc = rd._ca_store.blob_file(info["sha"])
+        assert loc is not None, "raw fast path not taken"
+    out = load_state_dict(rd, name="u")
+    assert out["step"] == 7
+    for k in ("w", "m"):
+        assert out[k].dtype == state[k].dtype
+        assert torch.equal(out[k], state[k]), k
+    out["w"] += 1  # writable (trained on after resume)
