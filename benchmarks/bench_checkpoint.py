@@ -68,6 +68,20 @@ def main():
     back = load_state_dict(ds, name="bench", map_location="cpu")
     load_s = time.time() - t
     assert len([k for k in back if k.startswith("shard_")]) == n_shards
+    assert torch.equal(back["shard_0"],
+                       state["shard_0"].cpu()), "cpu load corrupt"
+
+    load_dev_s = None
+    if device.startswith("cuda"):
+        # the real restore path: CAS file -> pinned -> HBM, overlapped
+        del back
+        t = time.time()
+        back = load_state_dict(ds, name="bench", map_location=device)
+        torch.cuda.synchronize()
+        load_dev_s = time.time() - t
+        assert back["shard_0"].is_cuda
+        assert torch.equal(back["shard_0"], state["shard_0"]), \
+            "device load corrupt"
 
     ds.done()
     import subprocess
@@ -81,6 +95,8 @@ def main():
         "save_gbps": total_gb / save_s,
         "dedup_save_gbps": total_gb / dedup_s,
         "load_gbps": total_gb / load_s,
+        "load_to_gpu_gbps":
+            (total_gb / load_dev_s) if load_dev_s else None,
         "config": {"gb": total_gb, "shards": n_shards, "device": device},
     }), flush=True)
 
