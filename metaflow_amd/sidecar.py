@@ -147,9 +147,27 @@ def _gpu_monitor_worker(context):
         sample()
 
 
+def _monitor_worker(context):
+    """Append monitor measure/count/gauge messages to a JSONL file."""
+    out_path = context.get("out_path", "/tmp/mfx_monitor.jsonl")
+    while True:
+        line = sys.stdin.readline()
+        if not line:
+            break
+        try:
+            msg = json.loads(line)
+        except ValueError:
+            continue
+        if msg.get("type") == "shutdown":
+            break
+        with open(out_path, "a") as f:
+            f.write(json.dumps(msg) + "\n")
+
+
 WORKERS = {
     "heartbeat": _heartbeat_worker,
     "gpu_monitor": _gpu_monitor_worker,
+    "monitor": _monitor_worker,
 }
 
 
