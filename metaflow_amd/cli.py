@@ -80,6 +80,10 @@ def _init_state(state, datastore, datastore_root, metadata, quiet, with_,
         state.top_level_args.append("--quiet")
     for spec in config:
         state.top_level_args.extend(["--config", spec])
+    # replay --with so task subprocesses attach the same decorators
+    # (reference: runtime replays top_level_options incl. --with)
+    for spec in with_ or ():
+        state.top_level_args.extend(["--with", spec])
     # run step_init hooks
     for step_name in state.flow_cls._steps:
         func = getattr(state.flow_cls, step_name)

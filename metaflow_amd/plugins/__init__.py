@@ -40,3 +40,8 @@ FLOW_DECORATORS = {
     for cls in (ProjectDecorator, ScheduleDecorator, ExitHookDecorator,
                 TriggerDecorator, TriggerOnFinishDecorator)
 }
+
+# merge metaflow_amd_extensions.* contributions (extension_support.py)
+from ..extension_support import load_extensions as _load_ext
+
+_load_ext(STEP_DECORATORS, FLOW_DECORATORS)

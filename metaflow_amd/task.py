@@ -256,6 +256,12 @@ class MFXTask(object):
                 retry_count)
 
         # ---- decorator hooks + user code ------------------------------------
+        from .monitor import get_system_logger, get_system_monitor
+
+        monitor = get_system_monitor()
+        get_system_logger().log({
+            "event": "task_start", "flow": flow.name, "run_id": run_id,
+            "step": step_name, "task_id": task_id, "attempt": retry_count})
         task_ok = True
         error = None
         try:
@@ -271,10 +277,12 @@ class MFXTask(object):
                     func, flow, self.graph, retry_count,
                     max_user_code_retries, ubf_context)
 
-            if node.type == "join":
-                self._exec_step_function(func, flow, inputs)
-            else:
-                self._exec_step_function(func, flow)
+            with monitor.measure("mfx.task.user_code"), \
+                    monitor.count("mfx.task.runs"):
+                if node.type == "join":
+                    self._exec_step_function(func, flow, inputs)
+                else:
+                    self._exec_step_function(func, flow)
 
             for deco in decorators:
                 deco.task_post_step(step_name, flow, self.graph, retry_count,

@@ -11,10 +11,13 @@ FLOWS = os.path.join(os.path.dirname(os.path.abspath(__file__)), "flows")
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def run_flow(flow_file, datastore_root, *args, check=True, timeout=180):
+def run_flow(flow_file, datastore_root, *args, check=True, timeout=180,
+             env_extra=None):
     env = dict(os.environ)
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     env["MFX_NUM_GPUS"] = "0"
+    if env_extra:
+        env.update(env_extra)
     cmd = [
         sys.executable, os.path.join(FLOWS, flow_file),
         "--datastore-root", datastore_root,
