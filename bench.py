@@ -31,6 +31,10 @@ def parse_args():
                    choices=["llama3-8b", "llama3-70b", "tiny",
                             "mixtral-8x7b", "mixtral-tiny"])
     p.add_argument("--bucket-mb", type=int, default=64)
+    p.add_argument("--cp", action="store_true",
+                   help="context parallelism: all ranks form one ring-"
+                        "attention group over a single shared batch "
+                        "(seq sharded across ranks)")
     p.add_argument("--lr", type=float, default=3e-4)
     p.add_argument("--tunableop", action="store_true",
                    help="enable PyTorch TunableOp (hipBLASLt algorithm "
@@ -91,10 +95,22 @@ def main():
     cfg = cfg_fn()
     seq = min(args.seq, cfg.max_seq_len)
 
+    use_cp = args.cp and distributed
+    if use_cp:
+        assert args.model in ("llama3-8b", "llama3-70b", "tiny"), \
+            "--cp supports the Llama family"
+        assert seq % world == 0, "--cp needs seq %% world == 0"
+        if use_gpu:
+            assert (seq // world) % 256 == 0, \
+                "--cp on GPU needs per-rank seq %% 256 == 0"
+
     torch.manual_seed(1234)  # same init on all ranks (DP)
     t0 = time.time()
     with torch.device(device):  # construct + random-init directly on GPU
-        model = model_cls(cfg)
+        if use_cp:
+            model = model_cls(cfg, cp_group=dist.group.WORLD)
+        else:
+            model = model_cls(cfg)
     flat = FlatParamModel(model, bucket_mb=args.bucket_mb)
     flat.install_overlap_hooks()
     opt = FusedAdamW(flat, lr=args.lr)
@@ -104,11 +120,17 @@ def main():
               flush=True)
 
     # synthetic batch (fixed per rank but distinct across ranks: loss must
-    # fall, proving a real fwd+bwd+optimizer step is in the timed region)
-    torch.manual_seed(5678 + rank)
+    # fall, proving a real fwd+bwd+optimizer step is in the timed region).
+    # Under --cp every rank sees the SAME batch and takes its seq shard.
+    torch.manual_seed(5678 + (0 if use_cp else rank))
     tokens = torch.randint(0, cfg.vocab_size, (args.batch, seq + 1),
                            device=device)
     inp, tgt = tokens[:, :-1], tokens[:, 1:].contiguous()
+    if use_cp:
+        sc = seq // world
+        sl = slice(rank * sc, (rank + 1) * sc)
+        inp = inp[:, sl].contiguous()
+        tgt = tgt[:, sl].contiguous()
 
     def one_step():
         flat.zero_grad()
@@ -145,7 +167,8 @@ def main():
     first_loss = float(losses[0].item())
     last_loss = float(losses[-1].item())
 
-    tokens_per_step = args.batch * seq * n_gpus
+    # cp: one shared batch across the group; dp: one batch per rank
+    tokens_per_step = args.batch * seq * (1 if use_cp else n_gpus)
     toks_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -159,15 +182,15 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if use_cp else "weak",
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch * n_gpus,
+                "global_batch": args.batch * (1 if use_cp else n_gpus),
                 "seq_len": seq,
-                "parallelism": "dp%d" % n_gpus,
+                "parallelism": ("cp%d" if use_cp else "dp%d") % n_gpus,
                 "first_loss": round(first_loss, 4),
                 "last_loss": round(last_loss, 4),
             },

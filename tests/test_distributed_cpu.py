@@ -56,6 +56,43 @@ def test_bench_tiny_world2(tmp_path):
     assert not any(l.startswith("{") for l in outs[1].splitlines())
 
 
+def test_bench_cp_tiny_world2(tmp_path):
+    """bench.py --cp over 2 gloo ranks: ring-attention context parallelism
+    through the flagship path; JSON reports cp2 + strong scaling."""
+    port = _free_port()
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({
+            "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": "2",
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(REPO, "bench.py"),
+             "--model", "tiny", "--steps", "2", "--warmup", "1",
+             "--batch", "1", "--seq", "256", "--cp"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, "rank failed:\n%s\n%s" % (out, err)
+        outs.append(out)
+    import json
+
+    json_lines = [l for l in outs[0].splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1
+    rec = json.loads(json_lines[0])
+    assert rec["config"]["parallelism"] == "cp2"
+    assert rec["scaling"] == "strong"
+    assert rec["config"]["global_batch"] == 1
+    assert rec["value"] > 0
+
+
 def test_flat_ddp_grad_sync():
     """Grad averaging across 2 gloo ranks through FlatParamModel hooks."""
     import torch.multiprocessing as mp
