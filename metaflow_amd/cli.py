@@ -192,7 +192,9 @@ def main(flow):
         _run_common(kwargs)
 
     @cli.command(help="Resume a failed run from where it left off.")
-    @click.option("--origin-run-id", required=True)
+    @click.option("--origin-run-id", default=None,
+                  help="Run to clone from (default: this flow's most "
+                       "recent run, reference semantics)")
     @click.option("--run-id", default=None)
     @click.option("--step-to-rerun", default=None,
                   help="Force this step (and everything after) to rerun.")
@@ -204,6 +206,12 @@ def main(flow):
                   help="Safe under concurrent invocation: one caller wins "
                        "leader election and resumes; the rest wait.")
     def resume(origin_run_id, step_to_rerun, reentrant, **kwargs):
+        if origin_run_id is None:
+            runs = state.metadata.list_runs()
+            if not runs:
+                raise MFXException(
+                    "No previous run of %s to resume." % flow_cls.__name__)
+            origin_run_id = sorted(r["run_id"] for r in runs)[-1]
         steps = [step_to_rerun] if step_to_rerun else []
         if reentrant:
             # leader election via exclusive-create in the datastore

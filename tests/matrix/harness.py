@@ -98,6 +98,20 @@ class MatrixTest(object):
     def check(self, run, graph):
         pass
 
+    def execute(self, flow_file, env, datastore_root):
+        """Run the generated flow; behaviors may override (e.g. resume)."""
+        return cli(flow_file, env, datastore_root, "run")
+
+
+def cli(flow_file, env, datastore_root, *args, extra_env=None):
+    e = dict(env)
+    if extra_env:
+        e.update(extra_env)
+    return subprocess.run(
+        [sys.executable, flow_file, "--quiet", "--datastore-root",
+         datastore_root] + list(args),
+        capture_output=True, text=True, env=e, timeout=300)
+
 
 def generate_flow(graph_name, test, class_name):
     graph = GRAPHS[graph_name]
@@ -168,10 +182,7 @@ def run_matrix_case(graph_name, test, tmp_dir, datastore_root):
     env = dict(os.environ)
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     env["MFX_NUM_GPUS"] = "0"
-    proc = subprocess.run(
-        [sys.executable, flow_file, "--quiet", "--datastore-root",
-         datastore_root, "run"],
-        capture_output=True, text=True, env=env, timeout=300)
+    proc = test.execute(flow_file, env, datastore_root)
     if proc.returncode != 0:
         raise AssertionError(
             "matrix flow %s x %s failed:\n%s\n%s"
@@ -286,3 +297,43 @@ class ForeachContextFlow(MatrixTest):
             t = run["join_f"].task
             if "mid" not in names:  # plain foreach x3
                 assert t.data.all_idx == [0, 1, 2], t.data.all_idx
+
+
+class ResumeFlow(MatrixTest):
+    """Clone-based resume across every DAG shape: the first run fails at
+    `end` (env MX_FAIL=1); `resume` must clone the whole successful
+    prefix (foreach fan-outs, joins, gangs included) and rerun only the
+    failed step."""
+
+    def body(self, name, kind, graph_name):
+        if kind == "join":
+            return ["self.merge_artifacts(inputs)"]
+        lines = ["self.mark_%s = '%s'" % (name, name)]
+        if kind == "end":
+            lines += [
+                "if os.environ.get('MX_FAIL') == '1':",
+                "    raise Exception('planned failure')",
+                "self.finished = 1",
+            ]
+        return lines
+
+    def execute(self, flow_file, env, datastore_root):
+        first = cli(flow_file, env, datastore_root, "run",
+                    extra_env={"MX_FAIL": "1"})
+        assert first.returncode != 0, "first run should fail at end"
+        return cli(flow_file, env, datastore_root, "resume",
+                   extra_env={"MX_FAIL": "0"})
+
+    def check(self, run, graph):
+        data = run["end"].task.data
+        assert data.finished == 1
+        skipped = {t for n, k, ts, _e in graph if k == "switch"
+                   for t in ts if t != "fast"}
+        for name, kind, _t, _e in graph:
+            if kind != "join" and name not in skipped:
+                assert getattr(data, "mark_%s" % name) == name
+        # the resumed run must contain tasks for every cloned step too
+        for name, _k, _t, _e in graph:
+            if name in skipped:
+                continue
+            assert len(list(run[name])) >= 1, "no tasks for %s" % name

@@ -7,6 +7,7 @@ from .matrix.harness import (
     ArtifactFlow,
     CurrentInfoFlow,
     ForeachContextFlow,
+    ResumeFlow,
     StepCounterFlow,
     run_matrix_case,
 )
@@ -16,6 +17,7 @@ BEHAVIORS = {
     "counter": StepCounterFlow,
     "current": CurrentInfoFlow,
     "foreach_ctx": ForeachContextFlow,
+    "resume": ResumeFlow,
 }
 
 CASES = [(g, b) for g in GRAPHS for b in BEHAVIORS]
