@@ -189,9 +189,28 @@ class Run(MetaflowObject):
         self._meta.add_run_tags(self.id, tags)
         self._info = self._meta.get_run(self.id)
 
+    def add_tag(self, tag):
+        self.add_tags([tag])
+
     def remove_tags(self, tags):
         self._meta.remove_run_tags(self.id, tags)
         self._info = self._meta.get_run(self.id)
+
+    def remove_tag(self, tag):
+        self.remove_tags([tag])
+
+    def replace_tags(self, removals, additions):
+        """Atomically remove and add tags (reference Run.replace_tags)."""
+        replace = getattr(self._meta, "replace_run_tags", None)
+        if replace is not None:
+            replace(self.id, removals, additions)
+            self._info = self._meta.get_run(self.id)
+        else:
+            self.remove_tags(removals)
+            self.add_tags(additions)
+
+    def replace_tag(self, removal, addition):
+        self.replace_tags([removal], [addition])
 
     @property
     def end_task(self):

@@ -145,6 +145,15 @@ class LocalMetadataProvider(object):
         info["tags"] = sorted(set(info.get("tags", [])) - set(tags))
         self._save(self._run_path(run_id), info)
 
+    def replace_run_tags(self, run_id, removals, additions):
+        """Atomic remove+add (reference client/core.py replace_tag): one
+        read-modify-write so a concurrent reader never sees the
+        intermediate state."""
+        info = self._load(self._run_path(run_id)) or {}
+        info["tags"] = sorted(
+            (set(info.get("tags", [])) - set(removals)) | set(additions))
+        self._save(self._run_path(run_id), info)
+
 
 def _username():
     try:

@@ -132,6 +132,11 @@ class ServiceMetadataProvider(object):
                       % (self.flow_name, run_id),
                       {"remove": list(tags)})
 
+    def replace_run_tags(self, run_id, removals, additions):
+        self._request("PATCH", "/flows/%s/runs/%s/tags"
+                      % (self.flow_name, run_id),
+                      {"remove": list(removals), "add": list(additions)})
+
 
 # ---------------------------------------------------------------- service
 def build_app(root):
@@ -218,9 +223,11 @@ def build_app(root):
     async def patch_tags(flow: str, run_id: str, request: Request):
         body = await request.json()
         p = provider(flow)
-        if body.get("add"):
+        if body.get("add") and body.get("remove"):
+            p.replace_run_tags(run_id, body["remove"], body["add"])
+        elif body.get("add"):
             p.add_run_tags(run_id, body["add"])
-        if body.get("remove"):
+        elif body.get("remove"):
             p.remove_run_tags(run_id, body["remove"])
         return {"ok": True}
 

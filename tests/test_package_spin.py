@@ -62,3 +62,11 @@ def test_tag_cli(tmp_datastore):
     assert "experiment:x1" in run.tags
     run.remove_tags(["experiment:x1"])
     assert "experiment:x1" not in client.Run("LinearFlow/%s" % run_id).tags
+    # singular + atomic replace (reference Run.add_tag/replace_tag)
+    run.add_tag("phase:a")
+    assert "phase:a" in client.Run("LinearFlow/%s" % run_id).tags
+    run.replace_tag("phase:a", "phase:b")
+    tags = client.Run("LinearFlow/%s" % run_id).tags
+    assert "phase:b" in tags and "phase:a" not in tags
+    run.remove_tag("phase:b")
+    assert "phase:b" not in client.Run("LinearFlow/%s" % run_id).tags
