@@ -48,14 +48,27 @@ class CardBuilder(object):
         self.sections = []
 
     def append(self, content, title=None):
-        """Append a section: raw HTML (str starting with '<') or text."""
+        """Append a section: a card component (card_components.Markdown/
+        Table/Image/Artifact), raw HTML (str starting with '<'), or
+        plain text."""
         self.sections.append((title, content))
+
+    def extend(self, contents):
+        for c in contents:
+            self.append(c)
 
     def render_sections(self):
         out = []
         for title, content in self.sections:
             if title:
                 out.append("<h2>%s</h2>" % html.escape(str(title)))
+            if hasattr(content, "render"):
+                try:
+                    out.append(content.render())
+                except Exception as ex:  # a bad component can't kill a card
+                    out.append("<pre>[component error: %s]</pre>"
+                               % html.escape(repr(ex)))
+                continue
             c = str(content)
             if c.lstrip().startswith("<"):
                 out.append(c)

@@ -118,6 +118,13 @@ def test_card(tmp_datastore):
     html = get_card(ds)
     assert html and "metric" in html and "custom html" in html
     assert "Notes" in html and "OK" in html
+    # typed components (card_components.py)
+    assert "<h3>Results</h3>" in html          # Markdown heading
+    assert "<b>0.125</b>" in html              # Markdown bold
+    assert "<th>a</th>" in html and "<td>4</td>" in html   # Table
+    assert "data:image/png;base64," in html    # Image data URI
+    assert "config" in html and "3e-04" in html.replace(
+        "0.0003", "3e-04")                     # Artifact pprint
 
 
 def test_config_and_mutator(tmp_datastore, tmp_path):
