@@ -31,7 +31,7 @@ class TorchParallelDecorator(ParallelDecorator):
     are visible, gloo otherwise (CPU tests)."""
 
     name = "torch_parallel"
-    defaults = {"backend": None}
+    defaults = {"backend": None, "context_parallel": 1}
 
     def task_decorate(self, step_func, flow, graph, retry_count,
                       max_user_code_retries, ubf_context):
@@ -58,6 +58,9 @@ class TorchParallelDecorator(ParallelDecorator):
                     else "gloo"
             if not dist.is_initialized():
                 dist.init_process_group(backend=backend)
+            cp = int(self.attributes.get("context_parallel") or 1)
+            if cp > 1:
+                self._make_grid(dist, cp)
             try:
                 return step_func(*args, **kwargs)
             finally:
@@ -66,6 +69,47 @@ class TorchParallelDecorator(ParallelDecorator):
 
         wrapped.__name__ = getattr(step_func, "__name__", "step")
         return wrapped
+
+    def _make_grid(self, dist, cp):
+        """Partition the gang into a dp x cp grid (SURVEY §5:
+        `@parallel(context_parallel=k)`): consecutive ranks share a
+        ring-attention group; same-index ranks across rings form the
+        data-parallel gradient group. Exposed as
+        ``current.parallel.cp_group`` / ``dp_group`` / ``cp_rank`` /
+        ``dp_rank`` (pass cp_group to LlamaForCausalLM /
+        ring_attention; FlatParamModel's all-reduce over dp_group x cp
+        averaging stays correct because cp shards see the same batch).
+        """
+        from ..current import current
+        from ..exceptions import MFXException
+
+        world = dist.get_world_size()
+        rank = dist.get_rank()
+        if world % cp != 0:
+            raise MFXException(
+                "context_parallel=%d must divide num_parallel=%d"
+                % (cp, world))
+        cp_group = dp_group = None
+        # new_group must be called by ALL ranks for EVERY group
+        for start in range(0, world, cp):
+            g = dist.new_group(list(range(start, start + cp)))
+            if start <= rank < start + cp:
+                cp_group = g
+        for idx in range(cp):
+            g = dist.new_group(list(range(idx, world, cp)))
+            if rank % cp == idx:
+                dp_group = g
+        par = getattr(current, "parallel", None)
+        info = dict(par._asdict()) if par is not None else {}
+        info.update({
+            "cp_group": cp_group, "dp_group": dp_group,
+            "cp_rank": rank % cp, "cp_degree": cp,
+            "dp_rank": rank // cp, "dp_degree": world // cp,
+        })
+        from collections import namedtuple
+
+        current._update_env(
+            {"parallel": namedtuple("Parallel", info)(**info)})
 
 
 parallel = make_step_decorator(ParallelDecorator)

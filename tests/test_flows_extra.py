@@ -199,3 +199,9 @@ def test_card_server(tmp_datastore):
     finally:
         proc.terminate()
         proc.wait(timeout=5)
+
+
+def test_torch_parallel_cp_grid(tmp_datastore):
+    """@torch_parallel(context_parallel=2): dp x cp grid groups exposed on
+    current.parallel, collective + ring attention run over cp_group."""
+    run_flow("cp_grid_flow.py", tmp_datastore, "run", timeout=300)
