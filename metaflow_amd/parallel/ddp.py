@@ -120,17 +120,30 @@ class FlatParamModel(object):
 
     def finish_grad_sync(self):
         """Wait for overlapped all-reduces (or do one synchronous pass if
-        hooks are not installed)."""
+        hooks are not installed). The exposed wait time is the NON-hidden
+        part of gradient communication — the number that explains the
+        1/2/4/8-GPU scaling curve (collective-time breakdown
+        through the monitor; enable with MFX_MONITOR=debug|sidecar)."""
+        import time
+
+        from ..monitor import get_system_monitor
+
+        t0 = time.time()
         if self._pending:
             for work in self._pending:
                 work.wait()
             self._pending.clear()
             for bi in self._bucket_done_count:
                 self._bucket_done_count[bi] = 0
+            get_system_monitor().gauge(
+                "mfx.ddp.allreduce_wait_ms", (time.time() - t0) * 1000)
         elif (dist.is_available() and dist.is_initialized()
               and dist.get_world_size() > 1):
             dist.all_reduce(self.flat_grad[:self.sync_end],
                             op=dist.ReduceOp.AVG)
+            get_system_monitor().gauge(
+                "mfx.ddp.allreduce_sync_ms", (time.time() - t0) * 1000)
+        self.last_comm_wait_ms = (time.time() - t0) * 1000
 
 
 class FusedAdamW(object):
