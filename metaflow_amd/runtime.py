@@ -525,6 +525,11 @@ class NativeRuntime(object):
             cmd += ["--origin-run-id", self.clone_run_id]
 
         env = dict(spec.env)
+        # propagate the active trace context into the child task
+        # (reference runtime.py:2337 inject_tracing_vars)
+        from .tracing import inject_tracing_vars
+
+        inject_tracing_vars(env)
         # let decorators mutate args/env (e.g. @environment)
         func = getattr(self.flow_cls, spec.step)
         args_holder = {"cmd": cmd, "env": env}
