@@ -58,7 +58,9 @@ class TorchParallelDecorator(ParallelDecorator):
                     else "gloo"
             if not dist.is_initialized():
                 dist.init_process_group(backend=backend)
-            cp = int(self.attributes.get("context_parallel") or 1)
+            cp_attr = self.attributes.get("context_parallel") or 1
+            cp = (dist.get_world_size() if cp_attr == "all"
+                  else int(cp_attr))
             if cp > 1:
                 self._make_grid(dist, cp)
             try:

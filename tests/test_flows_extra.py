@@ -205,3 +205,23 @@ def test_torch_parallel_cp_grid(tmp_datastore):
     """@torch_parallel(context_parallel=2): dp x cp grid groups exposed on
     current.parallel, collective + ring attention run over cp_group."""
     run_flow("cp_grid_flow.py", tmp_datastore, "run", timeout=300)
+
+
+def test_example_train_llama_cp(tmp_datastore):
+    """The long-context example flow end-to-end on CPU (2-rank ring)."""
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "train_llama_cp.py"),
+         "--quiet", "--datastore-root", tmp_datastore, "run",
+         "--num-gpus", "2", "--train-steps", "2", "--seq", "256"],
+        capture_output=True, text=True, env=env, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    run_id = latest_run_id(tmp_datastore, "TrainLlamaCP")
+    losses = read_artifact(tmp_datastore, "TrainLlamaCP", run_id, "join",
+                           "losses")
+    assert len(losses) == 2 and all(l == l for l in losses)
