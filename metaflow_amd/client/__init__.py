@@ -310,6 +310,20 @@ class Task(MetaflowObject):
         return self._ds.attempt is not None
 
     @property
+    def is_alive(self):
+        """True if this task is mid-execution: not finished and its
+        task-level heartbeat is fresh (<60 s)."""
+        import time
+
+        if self.finished:
+            return False
+        from ..metadata.local import LocalMetadataProvider
+
+        meta = LocalMetadataProvider(self.flow_name, self._storage)
+        ts = meta.task_heartbeat_ts(self.run_id, self.step_name, self.id)
+        return ts is not None and time.time() - ts < 60
+
+    @property
     def exception(self):
         return self._ds.get("_exception")
 

@@ -103,8 +103,24 @@ class LocalMetadataProvider(object):
                           metadata):
         self.register_task(run_id, step_name, task_id, attempt, metadata)
 
-    def heartbeat(self, run_id):
+    def heartbeat(self, run_id, step_name=None, task_id=None):
+        """Run-level liveness; with step_name/task_id, task-level too
+        (reference: heartbeats for task+run, heartbeat.py:21)."""
         self._save(self._heartbeat_path(run_id), {"ts": time.time()})
+        if step_name and task_id:
+            self._save(self._task_heartbeat_path(run_id, step_name,
+                                                 task_id),
+                       {"ts": time.time()})
+
+    def _task_heartbeat_path(self, run_id, step_name, task_id):
+        return self._storage.path_join(
+            self._meta_root(), str(run_id),
+            "task_heartbeat_%s_%s.json" % (step_name, task_id))
+
+    def task_heartbeat_ts(self, run_id, step_name, task_id):
+        hb = self._load(self._task_heartbeat_path(run_id, step_name,
+                                                  task_id)) or {}
+        return hb.get("ts")
 
     # ----------------------------------------------------------------- query
     def list_runs(self):

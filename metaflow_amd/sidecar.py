@@ -89,7 +89,16 @@ def _heartbeat_worker(context):
         from .metadata.local import LocalMetadataProvider
 
         meta = LocalMetadataProvider(flow, LocalStorage(root))
-    meta.heartbeat(run_id)
+    step_name = context.get("step_name")
+    task_id = context.get("task_id")
+
+    def beat():
+        try:
+            meta.heartbeat(run_id, step_name, task_id)
+        except TypeError:  # provider without task-level support
+            meta.heartbeat(run_id)
+
+    beat()
     import select
 
     while True:
@@ -103,7 +112,7 @@ def _heartbeat_worker(context):
                     break
             except ValueError:
                 pass
-        meta.heartbeat(run_id)
+        beat()
 
 
 def _gpu_monitor_worker(context):

@@ -173,6 +173,23 @@ class MFXTask(object):
         self.metadata.register_task(run_id, step_name, task_id, retry_count,
                                     metadata={"attempt_started": True})
 
+        # task-level liveness (reference task.py:797: heartbeats for
+        # task+run); lossy sidecar, never blocks the task
+        hb_sidecar = None
+        try:
+            from .sidecar import SidecarSubProcess
+
+            hb_sidecar = SidecarSubProcess("heartbeat", {
+                "flow_name": flow.name,
+                "run_id": run_id,
+                "datastore_root": self.flow_datastore.datastore_root,
+                "provider": getattr(self.metadata, "TYPE", "local"),
+                "step_name": step_name,
+                "task_id": str(task_id),
+            })
+        except Exception:
+            pass
+
         input_dss = self._input_datastores(input_paths)
         output = self.flow_datastore.get_task_datastore(
             run_id, step_name, task_id, attempt=retry_count, mode="w")
@@ -349,6 +366,8 @@ class MFXTask(object):
             except Exception:
                 traceback.print_exc()
 
+        if hb_sidecar is not None:
+            hb_sidecar.terminate()
         if task_ok:
             output.done()
         else:

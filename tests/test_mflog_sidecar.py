@@ -56,3 +56,29 @@ def test_heartbeat_sidecar(tmp_path):
         assert hb and hb.get("ts"), "no heartbeat written"
     finally:
         sc.terminate()
+
+
+def test_task_heartbeat_written(tmp_datastore):
+    """Every task writes task-level heartbeats (reference: task+run
+    liveness); a finished task reports is_alive False via the client."""
+    import os
+
+    from .test_runtime import latest_run_id, run_flow
+
+    run_flow("linear_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "LinearFlow")
+    meta_dir = os.path.join(tmp_datastore, "LinearFlow", "_meta", run_id)
+    hb = [f for f in os.listdir(meta_dir)
+          if f.startswith("task_heartbeat_")]
+    # one per task attempt: start, middle..., end (>= 2 at minimum)
+    assert len(hb) >= 2, hb
+
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    import importlib
+
+    import metaflow_amd.client as client
+
+    importlib.reload(client)
+    client.namespace(None)
+    task = client.Run("LinearFlow/%s" % run_id)["start"].task
+    assert task.finished and not task.is_alive
