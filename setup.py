@@ -29,6 +29,8 @@ def _force_rebuild_if_stale():
 
     objs = glob.glob(os.path.join(ROOT, "build", "temp*", "metaflow_amd",
                                   "ops", "csrc", "mfx_hip*.o"))
+    objs += glob.glob(os.path.join(ROOT, "build", "temp*", "metaflow_amd",
+                                   "ops", "csrc", "gemm_lt*.o"))
     if not objs:
         return
     newest_src = max(
@@ -38,11 +40,13 @@ def _force_rebuild_if_stale():
     for obj in objs:
         if os.path.getmtime(obj) < newest_src:
             os.unlink(obj)
-            gen = os.path.join(CSRC, "mfx_hip_hip.hip")
-            if os.path.exists(gen):
-                os.unlink(gen)
-            # force re-hipify by bumping the source mtime
+            for gen_name in ("mfx_hip_hip.hip", "gemm_lt_hip.hip"):
+                gen = os.path.join(CSRC, gen_name)
+                if os.path.exists(gen):
+                    os.unlink(gen)
+            # force re-hipify by bumping the source mtimes
             os.utime(os.path.join(CSRC, "mfx_hip.hip"))
+            os.utime(os.path.join(CSRC, "gemm_lt.hip"))
 
 
 _force_rebuild_if_stale()
@@ -51,6 +55,15 @@ ext_modules = [
     CUDAExtension(
         name="metaflow_amd.ops._mfx_hip",
         sources=[os.path.join(CSRC, "mfx_hip.hip")],
+        extra_compile_args={
+            "cxx": ["-O3", "-std=c++17"],
+            "nvcc": ["-O3", "-std=c++17"],
+        },
+    ),
+    CUDAExtension(
+        name="metaflow_amd.ops._mfx_gemm",
+        sources=[os.path.join(CSRC, "gemm_lt.hip")],
+        libraries=["hipblaslt"],
         extra_compile_args={
             "cxx": ["-O3", "-std=c++17"],
             "nvcc": ["-O3", "-std=c++17"],
