@@ -17,7 +17,12 @@ end-to-end.
 
 import argparse
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 
 # llama3-8b, B=8 S=4096 -> M = 32768 (fwd); K/N per projection
 SHAPES = {
