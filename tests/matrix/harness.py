@@ -83,6 +83,26 @@ GRAPHS = {
         ("join_u", "join", ["end"], {}),
         ("end", "end", [], {}),
     ],
+    # scheduler stress: a long sequential chain (transition latency adds
+    # up; artifact passdown must stay metadata-only the whole way)
+    "deep_linear": [
+        ("start", "linear", ["s1"], {}),
+    ] + [
+        ("s%d" % i, "linear", ["s%d" % (i + 1)], {}) for i in range(1, 8)
+    ] + [
+        ("s8", "linear", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
+    # scheduler stress: one split fanning into 6 concurrent branches
+    # under the worker cap, all gated by a single join
+    "wide_branch": [
+        ("start", "split", ["w%d" % i for i in range(6)], {}),
+    ] + [
+        ("w%d" % i, "linear", ["join_w"], {}) for i in range(6)
+    ] + [
+        ("join_w", "join", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
 }
 
 
