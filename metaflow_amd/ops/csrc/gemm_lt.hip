@@ -204,6 +204,61 @@ torch::Tensor gemm_lt_heuristic(torch::Tensor x, torch::Tensor w) {
   return out;
 }
 
+// ---------------------------------------------------------------- fp8
+// OCP E4M3 GEMM (gfx950 native; 2x bf16 peak): D_bf16 = (A_fp8 @ B_fp8^T)
+// with fp32 accumulate. Inputs arrive as uint8 bit-patterns (torch's
+// float8_e4m3fn storage viewed as uint8); scales fold into alpha.
+// Round-2 measurement target — compile-validated in round 1.
+void make_layouts_fp8(Layouts& L, long long M, long long N, long long K) {
+  LT_CHECK(hipblasLtMatmulDescCreate(&L.op, HIPBLAS_COMPUTE_32F,
+                                     HIP_R_32F));
+  hipblasOperation_t t = HIPBLAS_OP_T, n = HIPBLAS_OP_N;
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      L.op, HIPBLASLT_MATMUL_DESC_TRANSA, &t, sizeof(t)));
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      L.op, HIPBLASLT_MATMUL_DESC_TRANSB, &n, sizeof(n)));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&L.a, HIP_R_8F_E4M3, K, N, K));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&L.b, HIP_R_8F_E4M3, K, M, K));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&L.c, HIP_R_16BF, N, M, N));
+}
+
+torch::Tensor gemm_lt_fp8(torch::Tensor x8, torch::Tensor w8,
+                          double alpha_scale) {
+  TORCH_CHECK(x8.is_cuda() && w8.is_cuda());
+  TORCH_CHECK(x8.scalar_type() == torch::kUInt8 &&
+              w8.scalar_type() == torch::kUInt8,
+              "pass float8_e4m3fn storage viewed as uint8");
+  x8 = x8.contiguous();
+  w8 = w8.contiguous();
+  const long long M = x8.size(0), K = x8.size(1), N = w8.size(0);
+  auto out = torch::empty({M, N},
+                          x8.options().dtype(torch::kBFloat16));
+  auto workspace = torch::empty({128 << 20},
+                                x8.options().dtype(torch::kUInt8));
+  Layouts L;
+  make_layouts_fp8(L, M, N, K);
+  hipblasLtMatmulPreference_t pref;
+  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  uint64_t wsmax = (uint64_t)workspace.numel();
+  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsmax,
+      sizeof(wsmax)));
+  hipblasLtMatmulHeuristicResult_t res[1];
+  int found = 0;
+  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(lt_handle(), L.op, L.a, L.b,
+                                           L.c, L.c, pref, 1, res,
+                                           &found));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  TORCH_CHECK(found > 0, "no fp8 solution for this shape");
+  float alpha = (float)alpha_scale, beta = 0.f;
+  LT_CHECK(hipblasLtMatmul(
+      lt_handle(), L.op, &alpha, w8.data_ptr(), L.a, x8.data_ptr(), L.b,
+      &beta, out.data_ptr(), L.c, out.data_ptr(), L.c, &res[0].algo,
+      workspace.data_ptr(), (size_t)workspace.numel(),
+      at::hip::getCurrentHIPStream()));
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -212,4 +267,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "returns (indices, ms) fastest-first");
   m.def("run", &gemm_lt_run, "GEMM with a pinned solution index");
   m.def("heuristic", &gemm_lt_heuristic, "GEMM via hipBLASLt heuristic");
+  m.def("fp8", &gemm_lt_fp8,
+        "OCP E4M3 GEMM -> bf16 out (uint8-viewed fp8 inputs, alpha "
+        "carries the descale)");
 }
