@@ -87,7 +87,8 @@ class DecoderLayer(nn.Module):
         self.gate_up_proj = Linear(h, 2 * cfg.intermediate_size)
         self.down_proj = Linear(cfg.intermediate_size, h)
 
-    def forward(self, res, pending, cos_t, sin_t, cp_group=None):
+    def forward(self, res, pending, cos_t, sin_t, cp_group=None,
+                cache=None, layer_idx=0):
         """Carries (residual_stream, pending_branch): every residual add
         fuses with the next RMSNorm (K.add_rmsnorm). `pending` is the
         previous layer's un-added MLP output (None for layer 0). With
@@ -102,7 +103,18 @@ class DecoderLayer(nn.Module):
         qkv = self.qkv_proj(y)
         # one fused kernel: QKV split + bhsd transpose + RoPE
         q, k, v = K.rope_qkv(qkv, cos_t, sin_t, nq, nkv)
-        if cp_group is not None:
+        if cache is not None:
+            kc, vc, li, pos = cache.k[layer_idx], cache.v[layer_idx], \
+                layer_idx, cache.pos
+            kc[:, :, pos:pos + S] = k
+            vc[:, :, pos:pos + S] = v
+            if pos == 0 and S % 64 == 0:
+                o = K.attention(q, k, v)      # aligned prefill: flash
+            else:
+                o = _attn_with_cache(q, kc[:, :, :pos + S],
+                                     vc[:, :, :pos + S],
+                                     1.0 / math.sqrt(hd), pos)
+        elif cp_group is not None:
             from ..parallel.ring_attention import ring_attention
 
             o = ring_attention(q, k, v, group=cp_group)
@@ -114,6 +126,43 @@ class DecoderLayer(nn.Module):
                                self.post_norm.eps)
         pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
         return res, pending
+
+
+class KVCache(object):
+    """Preallocated per-layer KV cache for autoregressive decode.
+
+    k/v: one [B, Hkv, max_len, 128] bf16 tensor per layer, filled
+    in-place as positions arrive; ``pos`` is the number of cached
+    positions (advanced once per model forward, not per layer).
+    """
+
+    def __init__(self, cfg, batch, max_len, device, dtype=torch.bfloat16):
+        self.k = [torch.zeros(batch, cfg.num_kv_heads, max_len,
+                              cfg.head_dim, device=device, dtype=dtype)
+                  for _ in range(cfg.num_layers)]
+        self.v = [torch.zeros_like(k) for k in self.k]
+        self.pos = 0
+        self.max_len = max_len
+
+
+def _attn_with_cache(q, k_all, v_all, scale, pos):
+    """Attention of the s new queries (global positions pos..pos+s)
+    against all T cached kv positions: full visibility of the prefix,
+    causal within the new block. Plain torch ops (fp32 softmax) — decode
+    blocks are tiny, the flash kernels handle the big prefill."""
+    B, H, s, D = q.shape
+    Hkv = k_all.size(1)
+    T = k_all.size(2)
+    G = H // Hkv
+    kf = k_all.float().repeat_interleave(G, dim=1)
+    vf = v_all.float().repeat_interleave(G, dim=1)
+    att = torch.matmul(q.float(), kf.transpose(-1, -2)) * scale
+    kv_pos = torch.arange(T, device=q.device)
+    q_pos = pos + torch.arange(s, device=q.device)
+    mask = kv_pos[None, :] <= q_pos[:, None]          # [s, T]
+    att = att.masked_fill(~mask, float("-inf"))
+    p = torch.softmax(att, dim=-1)
+    return torch.matmul(p, vf).to(q.dtype)
 
 
 class LlamaForCausalLM(nn.Module):
@@ -155,12 +204,13 @@ class LlamaForCausalLM(nn.Module):
             layer.o_proj.weight.data.mul_(scale)
             layer.down_proj.weight.data.mul_(scale)
 
-    def forward(self, tokens, targets=None):
+    def forward(self, tokens, targets=None, cache=None):
         """tokens [B, S] int64 (the LOCAL shard when cp_group is set);
         returns mean loss over the local tokens if targets given, else
-        logits."""
+        logits. With ``cache`` (a KVCache), S may be a suffix of an
+        ongoing sequence starting at position cache.pos."""
         S = tokens.size(1)
-        off = 0
+        off = cache.pos if cache is not None else 0
         if self.cp_group is not None:
             import torch.distributed as dist
 
@@ -169,9 +219,11 @@ class LlamaForCausalLM(nn.Module):
         sin_t = self.sin_t[off:off + S].contiguous() if off else self.sin_t
         res = self.embed(tokens)
         pending = None
-        for layer in self.layers:
+        for li, layer in enumerate(self.layers):
             res, pending = layer(res, pending, cos_t, sin_t,
-                                 self.cp_group)
+                                 self.cp_group, cache, li)
+        if cache is not None:
+            cache.pos += S
         _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
                              self.final_norm.eps)
         if targets is None:
@@ -181,6 +233,33 @@ class LlamaForCausalLM(nn.Module):
         loss = K.cross_entropy(logits.reshape(B * S, V),
                                targets.reshape(B * S))
         return loss.mean()
+
+    @torch.no_grad()
+    def generate(self, tokens, max_new_tokens, temperature=0.0,
+                 top_k=None):
+        """Autoregressive decode with a KV cache: flash-kernel prefill
+        (when the prompt length is 64-aligned), cached attention for the
+        one-token decode steps. temperature=0 is greedy."""
+        B, S0 = tokens.shape
+        cache = KVCache(self.cfg, B, S0 + max_new_tokens,
+                        tokens.device,
+                        dtype=self.embed.weight.dtype)
+        out = tokens
+        logits = self.forward(tokens, cache=cache)
+        for _ in range(max_new_tokens):
+            last = logits[:, -1].float()
+            if temperature and temperature > 0:
+                last = last / temperature
+                if top_k:
+                    kth = torch.topk(last, top_k, dim=-1).values[:, -1:]
+                    last = last.masked_fill(last < kth, float("-inf"))
+                probs = torch.softmax(last, dim=-1)
+                nxt = torch.multinomial(probs, 1)
+            else:
+                nxt = last.argmax(dim=-1, keepdim=True)
+            out = torch.cat([out, nxt], dim=1)
+            logits = self.forward(nxt, cache=cache)
+        return out
 
     def num_params(self):
         return sum(p.numel() for p in self.parameters())

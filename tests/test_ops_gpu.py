@@ -277,6 +277,28 @@ def test_mixtral_tiny_step(dev):
     assert losses[-1] < losses[0], "loss did not decrease: %s" % losses
 
 
+def test_generate_gpu(dev):
+    """KV-cache decode on hardware: flash-kernel prefill (64-aligned
+    prompt) + cached decode; greedy tokens must match the uncached full
+    forward."""
+    import torch
+
+    from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig.tiny(vocab=256, seq=256)
+    m = LlamaForCausalLM(cfg).to(dev).eval()
+    torch.manual_seed(1)
+    prompt = torch.randint(0, 256, (1, 64), device=dev)
+    out = m.generate(prompt, 4)
+    assert out.shape == (1, 68)
+    with torch.no_grad():
+        for t in range(64, 68):
+            full = m(out[:, :t])
+            nxt = full[:, -1].float().argmax(-1)
+            assert torch.equal(nxt, out[:, t]), t
+
+
 def test_swiglu_fused(dev):
     from metaflow_amd.ops import kernels as K
 
