@@ -244,7 +244,7 @@ class MixtralDecoderLayer(nn.Module):
         self.post_norm = RMSNorm(h, cfg.rms_eps)
         self.moe = MoELayer(cfg, ep_group)
 
-    def forward(self, x, cos_t, sin_t):
+    def forward(self, x, cos_t, sin_t, cache=None, layer_idx=0):
         cfg = self.cfg
         B, S, _ = x.shape
         res = x
@@ -256,8 +256,24 @@ class MixtralDecoderLayer(nn.Module):
                     cos_t, sin_t).transpose(1, 2)
         v = self.v_proj(y).view(B, S, cfg.num_kv_heads,
                                 cfg.head_dim).transpose(1, 2)
-        o = K.attention(q, kk, v).transpose(1, 2).reshape(
-            B, S, cfg.num_heads * cfg.head_dim)
+        if cache is not None:
+            from .llama import _attn_with_cache
+
+            kc, vc, pos = cache.k[layer_idx], cache.v[layer_idx], cache.pos
+            kc[:, :, pos:pos + S] = kk
+            vc[:, :, pos:pos + S] = v
+            if pos == 0 and S % 64 == 0:
+                o = K.attention(q, kk, v)
+            else:
+                import math
+
+                o = _attn_with_cache(q.contiguous(),
+                                     kc[:, :, :pos + S],
+                                     vc[:, :, :pos + S],
+                                     1.0 / math.sqrt(cfg.head_dim), pos)
+        else:
+            o = K.attention(q, kk, v)
+        o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
         x = res + self.o_proj(o)
         x = x + self.moe(self.post_norm(x))
         return x
@@ -285,10 +301,16 @@ class MixtralForCausalLM(nn.Module):
             if p.dim() >= 2:
                 nn.init.normal_(p, mean=0.0, std=0.02)
 
-    def forward(self, tokens, targets=None):
+    def forward(self, tokens, targets=None, cache=None):
+        S = tokens.size(1)
+        off = cache.pos if cache is not None else 0
+        cos_t = self.cos_t[off:off + S].contiguous() if off else self.cos_t
+        sin_t = self.sin_t[off:off + S].contiguous() if off else self.sin_t
         x = self.embed(tokens)
-        for layer in self.layers:
-            x = layer(x, self.cos_t, self.sin_t)
+        for li, layer in enumerate(self.layers):
+            x = layer(x, cos_t, sin_t, cache, li)
+        if cache is not None:
+            cache.pos += S
         x = self.final_norm(x)
         logits = self.lm_head(x)
         if targets is None:
@@ -297,6 +319,32 @@ class MixtralForCausalLM(nn.Module):
         loss = K.cross_entropy(logits.reshape(B * S, V),
                                targets.reshape(B * S))
         return loss.mean()
+
+    @torch.no_grad()
+    def generate(self, tokens, max_new_tokens, temperature=0.0,
+                 top_k=None):
+        """Autoregressive MoE decode with a KV cache (same contract as
+        LlamaForCausalLM.generate; experts route per decoded token)."""
+        from .llama import KVCache
+
+        B, S0 = tokens.shape
+        cache = KVCache(self.cfg, B, S0 + max_new_tokens, tokens.device,
+                        dtype=self.embed.weight.dtype)
+        out = tokens
+        logits = self.forward(tokens, cache=cache)
+        for _ in range(max_new_tokens):
+            last = logits[:, -1].float()
+            if temperature and temperature > 0:
+                last = last / temperature
+                if top_k:
+                    kth = torch.topk(last, top_k, dim=-1).values[:, -1:]
+                    last = last.masked_fill(last < kth, float("-inf"))
+                nxt = torch.multinomial(torch.softmax(last, -1), 1)
+            else:
+                nxt = last.argmax(dim=-1, keepdim=True)
+            out = torch.cat([out, nxt], dim=1)
+            logits = self.forward(nxt, cache=cache)
+        return out
 
     def num_params(self):
         return sum(p.numel() for p in self.parameters())

@@ -31,3 +31,19 @@ def test_generate_sampling_shapes():
                      temperature=0.8, top_k=5)
     assert out.shape == (1, 67)
     assert out[:, :64].equal(out[:, :64])
+
+
+def test_mixtral_generate_matches_full_forward():
+    from metaflow_amd.models.mixtral import MixtralConfig, MixtralForCausalLM
+
+    torch.manual_seed(0)
+    cfg = MixtralConfig.tiny(vocab=256, seq=128)
+    m = MixtralForCausalLM(cfg).eval()
+    torch.manual_seed(1)
+    prompt = torch.randint(0, 256, (1, 11))
+    out = m.generate(prompt, 5)
+    assert out.shape == (1, 16)
+    with torch.no_grad():
+        for t in range(11, 16):
+            nxt = m(out[:, :t])[:, -1].float().argmax(-1)
+            assert torch.equal(nxt, out[:, t]), t
