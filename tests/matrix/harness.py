@@ -192,7 +192,11 @@ def generate_flow(graph_name, test, class_name):
     return "\n".join(lines)
 
 
-def run_matrix_case(graph_name, test, tmp_dir, datastore_root):
+def run_matrix_case(graph_name, test, tmp_dir, datastore_root,
+                    executor="cli"):
+    """executor: "cli" re-invokes the flow file's CLI (the scheduler's
+    own process model); "api" drives the same file through the Runner
+    (reference contexts run both executors per case)."""
     class_name = "MX%s%sFlow" % (
         graph_name.title().replace("_", ""), type(test).__name__)
     src = generate_flow(graph_name, test, class_name)
@@ -202,12 +206,20 @@ def run_matrix_case(graph_name, test, tmp_dir, datastore_root):
     env = dict(os.environ)
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     env["MFX_NUM_GPUS"] = "0"
-    proc = test.execute(flow_file, env, datastore_root)
-    if proc.returncode != 0:
-        raise AssertionError(
-            "matrix flow %s x %s failed:\n%s\n%s"
-            % (graph_name, type(test).__name__, proc.stdout[-3000:],
-               proc.stderr[-3000:]))
+    if executor == "api":
+        os.environ["MFX_NUM_GPUS"] = "0"
+        os.environ["PYTHONPATH"] = env["PYTHONPATH"]
+        from metaflow_amd.runner import Runner
+
+        ex = Runner(flow_file, datastore_root=datastore_root).run()
+        assert ex.status == "successful", ex.status
+    else:
+        proc = test.execute(flow_file, env, datastore_root)
+        if proc.returncode != 0:
+            raise AssertionError(
+                "matrix flow %s x %s failed:\n%s\n%s"
+                % (graph_name, type(test).__name__, proc.stdout[-3000:],
+                   proc.stderr[-3000:]))
 
     # checker 1: client API
     os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = datastore_root

@@ -28,3 +28,17 @@ CASES = [(g, b) for g in GRAPHS for b in BEHAVIORS]
 def test_matrix(graph_name, behavior, tmp_path, tmp_datastore):
     test = BEHAVIORS[behavior]()
     run_matrix_case(graph_name, test, str(tmp_path), tmp_datastore)
+
+
+API_CASES = [("linear", "artifact"), ("foreach", "counter"),
+             ("nested_branch", "current")]
+
+
+@pytest.mark.parametrize("graph_name,behavior", API_CASES,
+                         ids=["api-%s-%s" % c for c in API_CASES])
+def test_matrix_runner_api(graph_name, behavior, tmp_path, tmp_datastore):
+    """Same generated flows driven through the Runner API executor
+    (reference tier 1 runs every case under both executors)."""
+    test = BEHAVIORS[behavior]()
+    run_matrix_case(graph_name, test, str(tmp_path), tmp_datastore,
+                    executor="api")
