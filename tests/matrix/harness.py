@@ -287,23 +287,38 @@ class StepCounterFlow(MatrixTest):
 
 
 class CurrentInfoFlow(MatrixTest):
-    """current.* is coherent in every task."""
+    """current.* is coherent in every task; gang steps see
+    current.parallel (reference basic_parallel)."""
 
     def body(self, name, kind, graph_name):
         if kind == "join":
             return [
                 "assert current.step_name == '%s'" % name,
-                "self.merge_artifacts(inputs, exclude=['seen_step'])",
+                "self.gang_idx = sorted(i.node_idx for i in inputs "
+                "if hasattr(i, 'node_idx'))",
+                "self.merge_artifacts(inputs, "
+                "exclude=['seen_step', 'node_idx'])",
             ]
-        return [
+        lines = [
             "assert current.flow_name == type(self).__name__",
             "assert current.step_name == '%s'" % name,
             "assert current.run_id is not None",
             "self.seen_step = current.step_name",
         ]
+        if graph_name == "parallel" and name == "work":
+            lines += [
+                "assert current.parallel.num_nodes == 2",
+                "assert 0 <= current.parallel.node_index < 2",
+                "assert current.parallel.main_ip",
+                "self.node_idx = current.parallel.node_index",
+            ]
+        return lines
 
     def check(self, run, graph):
         assert run["end"].task.data.seen_step == "end"
+        names = [n for n, _k, _t, _e in graph]
+        if "join_p" in names:
+            assert run["join_p"].task.data.gang_idx == [0, 1]
 
 
 class ForeachContextFlow(MatrixTest):
