@@ -2,7 +2,7 @@
 
 import os
 
-from .test_runtime import latest_run_id, read_artifact, run_flow
+from .test_runtime import REPO, latest_run_id, read_artifact, run_flow
 
 
 def test_retry(tmp_datastore, tmp_path, monkeypatch):
@@ -89,3 +89,40 @@ def test_reentrant_resume(tmp_datastore, tmp_path, monkeypatch):
     # middle ran exactly twice total (once in failed run + once in resume)
     assert int(open(counter_dir / "middle").read()) == 2
     assert int(open(counter_dir / "end").read()) == 1
+
+
+def test_exit_disallow_retry(tmp_datastore, tmp_path):
+    """A task exiting with EXIT_DISALLOW_RETRY (202) must NOT be retried
+    even under @retry (reference METAFLOW_EXIT_DISALLOW_RETRY contract)."""
+    import subprocess
+    import sys
+
+    flow = tmp_path / "disallow_flow.py"
+    flow.write_text(
+        "import os, sys\n"
+        "from metaflow_amd import FlowSpec, step, retry, current\n"
+        "class DisallowFlow(FlowSpec):\n"
+        "    @retry(times=3)\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        with open(os.environ['ATTEMPT_LOG'], 'a') as f:\n"
+        "            f.write('attempt %d\\n' % current.retry_count)\n"
+        "        os._exit(202)\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n"
+        "if __name__ == '__main__':\n"
+        "    DisallowFlow()\n")
+    log = tmp_path / "attempts.log"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    env["ATTEMPT_LOG"] = str(log)
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc.returncode != 0
+    # exactly ONE attempt despite @retry(times=3)
+    assert log.read_text().count("attempt") == 1, log.read_text()
