@@ -70,3 +70,15 @@ def test_tag_cli(tmp_datastore):
     assert "phase:b" in tags and "phase:a" not in tags
     run.remove_tag("phase:b")
     assert "phase:b" not in client.Run("LinearFlow/%s" % run_id).tags
+
+
+def test_spin_foreach_task(tmp_datastore):
+    """spin a single foreach child against origin artifacts via
+    --split-index."""
+    run_flow("foreach_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "ForeachFlow")
+    proc = run_flow("foreach_flow.py", tmp_datastore, "spin", "work",
+                    "--run-id", run_id, "--split-index", "2")
+    assert proc.returncode == 0
+    assert read_artifact(tmp_datastore, "ForeachFlow",
+                         "spin%s" % run_id, "work", "squared") == 9
