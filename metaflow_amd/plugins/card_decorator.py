@@ -46,6 +46,14 @@ pre {{ background: #1e1e1e; color: #d4d4d4; padding: 1rem;
 class CardBuilder(object):
     def __init__(self):
         self.sections = []
+        self._refresh_cb = None
+
+    def refresh(self):
+        """Write the card NOW with the sections appended so far
+        (reference: CardCreator async refresh — long-running tasks can
+        publish progress before task_finished renders the final card)."""
+        if self._refresh_cb is not None:
+            self._refresh_cb()
 
     def append(self, content, title=None):
         """Append a section: a card component (card_components.Markdown/
@@ -112,6 +120,18 @@ class CardDecorator(StepDecorator):
 
         self._builder = CardBuilder()
         self._ds = task_datastore
+        card_id = self.attributes.get("id", "default")
+
+        def _refresh(builder=self._builder, ds=task_datastore,
+                     step=step_name, cid=card_id):
+            try:
+                ds.save_metadata(
+                    "card_%s" % cid,
+                    {"html": render_card(ds, step, True, builder)})
+            except Exception:
+                pass
+
+        self._builder._refresh_cb = _refresh
         current._update_env({"card": self._builder})
 
     def task_finished(self, step_name, flow, graph, is_task_ok, retry_count,

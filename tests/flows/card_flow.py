@@ -20,6 +20,7 @@ class CardFlow(FlowSpec):
         self.metric = 0.125
         current.card.append("training summary", title="Notes")
         current.card.append("<b>custom html</b>")
+        current.card.refresh()   # mid-task publish (progress card)
         current.card.extend([
             Markdown("## Results\n- loss **0.125**\n- `tokens/s` high"),
             Table([[1, 2], [3, 4]], headers=["a", "b"]),
