@@ -126,3 +126,17 @@ def test_exit_disallow_retry(tmp_datastore, tmp_path):
     assert proc.returncode != 0
     # exactly ONE attempt despite @retry(times=3)
     assert log.read_text().count("attempt") == 1, log.read_text()
+
+
+def test_gang_member_failure_and_resume(tmp_datastore):
+    """A dead gang member fails the whole gang; resume reruns the full
+    gang (all-or-nothing clone semantics) and the run completes."""
+    proc = run_flow("gang_fail_flow.py", tmp_datastore, "run",
+                    check=False, env_extra={"GANG_FAIL": "1"})
+    assert proc.returncode != 0
+    proc2 = run_flow("gang_fail_flow.py", tmp_datastore, "resume",
+                     env_extra={"GANG_FAIL": "0"})
+    assert proc2.returncode == 0
+    run_id = latest_run_id(tmp_datastore, "GangFailFlow")
+    assert read_artifact(tmp_datastore, "GangFailFlow", run_id, "join",
+                         "ranks") == [0, 1]
