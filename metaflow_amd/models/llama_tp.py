@@ -1,0 +1,160 @@
+"""Tensor-parallel Llama: Megatron-style head/feature sharding on the
+same HIP kernel library as models/llama.py.
+
+Per layer, rank r holds nq/w query heads + nkv/w kv heads (attention is
+fully local to the rank's heads — GQA ratio preserved), a gate|up shard
+of the MLP, and row shards of o_proj/down_proj; exactly TWO all-reduces
+of [B, S, h] per layer (attention output + MLP output) travel the tp
+group over xGMI. The lm_head is vocab-sharded with an all-gather before
+the loss (a vocab-parallel CE that skips the gather is the round-2
+refinement).
+
+Composes with the other axes: put tp inside the gang's innermost ranks
+and pass the dp group to FlatParamModel(group=...) — sharded params
+must only all-reduce across ranks holding the SAME shard.
+
+`from_full_model` builds a TP model holding exact shards of a reference
+`LlamaForCausalLM` — the gloo equivalence tests
+(tests/test_tensor_parallel.py) check loss and every weight gradient
+against the unsharded model.
+"""
+
+import math
+
+import torch
+import torch.nn as nn
+
+from ..ops import kernels as K
+from ..parallel.tp import (
+    ColumnParallelLinear,
+    RowParallelLinear,
+    gather_from_tp,
+    shard_cols,
+    shard_gate_up_rows,
+    shard_qkv_rows,
+)
+from .llama import LlamaConfig, RMSNorm
+
+
+class TPDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig, group):
+        super().__init__()
+        import torch.distributed as dist
+
+        self.cfg = cfg
+        self.group = group
+        self.world = dist.get_world_size(group)
+        h, hd = cfg.hidden_size, cfg.head_dim
+        assert cfg.num_heads % self.world == 0
+        assert cfg.num_kv_heads % self.world == 0
+        assert cfg.intermediate_size % self.world == 0
+        self.nq_l = cfg.num_heads // self.world
+        self.nkv_l = cfg.num_kv_heads // self.world
+        self.qkv_proj = ColumnParallelLinear(
+            h, (cfg.num_heads + 2 * cfg.num_kv_heads) * hd, group)
+        self.input_norm = RMSNorm(h, cfg.rms_eps)
+        self.o_proj = RowParallelLinear(cfg.num_heads * hd, h, group)
+        self.post_norm = RMSNorm(h, cfg.rms_eps)
+        self.gate_up_proj = ColumnParallelLinear(
+            h, 2 * cfg.intermediate_size, group)
+        self.down_proj = RowParallelLinear(cfg.intermediate_size, h, group)
+
+    def forward(self, res, pending, cos_t, sin_t):
+        cfg = self.cfg
+        B, S, _ = res.shape
+        hd = cfg.head_dim
+        res, y = K.add_rmsnorm(res, pending, self.input_norm.weight,
+                               self.input_norm.eps)
+        qkv = self.qkv_proj(y)          # local heads only
+        q, k, v = K.rope_qkv(qkv, cos_t, sin_t, self.nq_l, self.nkv_l)
+        o = K.attention(q, k, v)        # local GQA group, no comm
+        o = o.transpose(1, 2).reshape(B, S, self.nq_l * hd)
+        res, y = K.add_rmsnorm(res, self.o_proj(o), self.post_norm.weight,
+                               self.post_norm.eps)
+        pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
+        return res, pending
+
+
+class TPLlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp_group):
+        super().__init__()
+        self.cfg = cfg
+        self.group = tp_group
+        import torch.distributed as dist
+
+        self.world = dist.get_world_size(tp_group)
+        assert cfg.vocab_size % self.world == 0
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                  dtype=torch.bfloat16)
+        self.layers = nn.ModuleList(
+            TPDecoderLayer(cfg, tp_group) for _ in range(cfg.num_layers))
+        self.final_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
+        self.lm_head = ColumnParallelLinear(cfg.hidden_size,
+                                            cfg.vocab_size, tp_group)
+        cos_t, sin_t = K.rope_tables(cfg.max_seq_len, cfg.head_dim,
+                                     cfg.rope_theta)
+        self.register_buffer("cos_t", cos_t, persistent=False)
+        self.register_buffer("sin_t", sin_t, persistent=False)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        std = 0.02
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                nn.init.normal_(p, mean=0.0, std=std)
+            elif "norm" in name:
+                nn.init.ones_(p)
+        scale = 1.0 / math.sqrt(2 * self.cfg.num_layers)
+        for layer in self.layers:
+            layer.o_proj.weight.data.mul_(scale)
+            layer.down_proj.weight.data.mul_(scale)
+
+    @classmethod
+    def from_full_model(cls, full, tp_group):
+        """Build the TP model holding exact shards of an unsharded
+        LlamaForCausalLM (equivalence tests / converting checkpoints)."""
+        import torch.distributed as dist
+
+        cfg = full.cfg
+        rank = dist.get_rank(tp_group)
+        world = dist.get_world_size(tp_group)
+        m = cls(cfg, tp_group)
+        with torch.no_grad():
+            m.embed.weight.copy_(full.embed.weight)
+            m.final_norm.weight.copy_(full.final_norm.weight)
+            m.lm_head.weight.copy_(
+                full.lm_head.weight[rank * (cfg.vocab_size // world):
+                                    (rank + 1) * (cfg.vocab_size // world)])
+            for tl, fl in zip(m.layers, full.layers):
+                tl.input_norm.weight.copy_(fl.input_norm.weight)
+                tl.post_norm.weight.copy_(fl.post_norm.weight)
+                tl.qkv_proj.weight.copy_(shard_qkv_rows(
+                    fl.qkv_proj.weight, rank, world, cfg.num_heads,
+                    cfg.num_kv_heads, cfg.head_dim))
+                tl.o_proj.weight.copy_(shard_cols(
+                    fl.o_proj.weight, rank, world))
+                tl.gate_up_proj.weight.copy_(shard_gate_up_rows(
+                    fl.gate_up_proj.weight, rank, world))
+                tl.down_proj.weight.copy_(shard_cols(
+                    fl.down_proj.weight, rank, world))
+        return m
+
+    def forward(self, tokens, targets=None):
+        S = tokens.size(1)
+        cos_t, sin_t = self.cos_t, self.sin_t
+        res = self.embed(tokens)
+        pending = None
+        for layer in self.layers:
+            res, pending = layer(res, pending, cos_t, sin_t)
+        _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
+                             self.final_norm.eps)
+        logits = gather_from_tp(self.lm_head(x), self.group)
+        if targets is None:
+            return logits
+        B, S, V = logits.shape
+        loss = K.cross_entropy(logits.reshape(B * S, V),
+                               targets.reshape(B * S))
+        return loss.mean()
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())

@@ -23,7 +23,12 @@ from ..ops import kernels as K
 class FlatParamModel(object):
     """Flattens a module's parameters into one bf16 buffer + flat grads."""
 
-    def __init__(self, module, bucket_mb=64):
+    def __init__(self, module, bucket_mb=64, group=None):
+        """group: the process group to average gradients over (default
+        WORLD). For dp x tp grids pass the DP group — sharded (tp)
+        params must only all-reduce across ranks holding the same
+        shard."""
+        self.group = group
         self.module = module
         all_params = [p for p in module.parameters() if p.requires_grad]
         # params marked _mfx_no_sync (e.g. expert-parallel weights whose
@@ -79,7 +84,7 @@ class FlatParamModel(object):
         is accumulated (params complete roughly in reverse order, so buckets
         are checked by completion count)."""
         if not (dist.is_available() and dist.is_initialized()
-                and dist.get_world_size() > 1):
+                and dist.get_world_size(self.group) > 1):
             return
         self._done = set()
         bucket_last_param = {}
@@ -112,6 +117,7 @@ class FlatParamModel(object):
                         self._bucket_param_sets[bi]):
                     s, e, _ = self.buckets[bi]
                     work = dist.all_reduce(self.flat_grad[s:e],
+                                           group=self.group,
                                            op=dist.ReduceOp.AVG,
                                            async_op=True)
                     self._pending.append(work)
@@ -138,9 +144,9 @@ class FlatParamModel(object):
             get_system_monitor().gauge(
                 "mfx.ddp.allreduce_wait_ms", (time.time() - t0) * 1000)
         elif (dist.is_available() and dist.is_initialized()
-              and dist.get_world_size() > 1):
+              and dist.get_world_size(self.group) > 1):
             dist.all_reduce(self.flat_grad[:self.sync_end],
-                            op=dist.ReduceOp.AVG)
+                            group=self.group, op=dist.ReduceOp.AVG)
             get_system_monitor().gauge(
                 "mfx.ddp.allreduce_sync_ms", (time.time() - t0) * 1000)
         self.last_comm_wait_ms = (time.time() - t0) * 1000

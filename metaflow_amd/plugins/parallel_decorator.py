@@ -31,7 +31,8 @@ class TorchParallelDecorator(ParallelDecorator):
     are visible, gloo otherwise (CPU tests)."""
 
     name = "torch_parallel"
-    defaults = {"backend": None, "context_parallel": 1}
+    defaults = {"backend": None, "context_parallel": 1,
+                "tensor_parallel": 1}
 
     def task_decorate(self, step_func, flow, graph, retry_count,
                       max_user_code_retries, ubf_context):
@@ -62,7 +63,12 @@ class TorchParallelDecorator(ParallelDecorator):
             cp = (dist.get_world_size() if cp_attr == "all"
                   else int(cp_attr))
             if cp > 1:
-                self._make_grid(dist, cp)
+                self._make_grid(dist, cp, "cp")
+            tp_attr = self.attributes.get("tensor_parallel") or 1
+            tp = (dist.get_world_size() if tp_attr == "all"
+                  else int(tp_attr))
+            if tp > 1:
+                self._make_grid(dist, tp, "tp")
             try:
                 return step_func(*args, **kwargs)
             finally:
@@ -72,41 +78,43 @@ class TorchParallelDecorator(ParallelDecorator):
         wrapped.__name__ = getattr(step_func, "__name__", "step")
         return wrapped
 
-    def _make_grid(self, dist, cp):
-        """Partition the gang into a dp x cp grid (SURVEY §5:
-        `@parallel(context_parallel=k)`): consecutive ranks share a
-        ring-attention group; same-index ranks across rings form the
-        data-parallel gradient group. Exposed as
-        ``current.parallel.cp_group`` / ``dp_group`` / ``cp_rank`` /
-        ``dp_rank`` (pass cp_group to LlamaForCausalLM /
-        ring_attention; FlatParamModel's all-reduce over dp_group x cp
-        averaging stays correct because cp shards see the same batch).
-        """
+    def _make_grid(self, dist, inner, kind):
+        """Partition the gang into a dp x inner grid (SURVEY §5:
+        `@parallel(context_parallel=k)` / `tensor_parallel=k`):
+        consecutive ranks form the inner group (ring-attention peers
+        for cp — pass to LlamaForCausalLM/ring_attention — or
+        Megatron-style shard peers for tp — pass to
+        TPLlamaForCausalLM); same-index ranks across inner groups form
+        the data-parallel gradient group (pass to
+        FlatParamModel(group=...) for tp; for cp the WHOLE gang
+        all-reduce stays correct because cp shards see the same batch).
+        Exposed as ``current.parallel.<kind>_group`` / ``dp_group`` /
+        ``<kind>_rank`` / ``dp_rank``."""
         from ..current import current
         from ..exceptions import MFXException
 
         world = dist.get_world_size()
         rank = dist.get_rank()
-        if world % cp != 0:
+        if world % inner != 0:
             raise MFXException(
-                "context_parallel=%d must divide num_parallel=%d"
-                % (cp, world))
-        cp_group = dp_group = None
+                "%s=%d must divide num_parallel=%d"
+                % (kind, inner, world))
+        in_group = dp_group = None
         # new_group must be called by ALL ranks for EVERY group
-        for start in range(0, world, cp):
-            g = dist.new_group(list(range(start, start + cp)))
-            if start <= rank < start + cp:
-                cp_group = g
-        for idx in range(cp):
-            g = dist.new_group(list(range(idx, world, cp)))
-            if rank % cp == idx:
+        for start in range(0, world, inner):
+            g = dist.new_group(list(range(start, start + inner)))
+            if start <= rank < start + inner:
+                in_group = g
+        for idx in range(inner):
+            g = dist.new_group(list(range(idx, world, inner)))
+            if rank % inner == idx:
                 dp_group = g
         par = getattr(current, "parallel", None)
         info = dict(par._asdict()) if par is not None else {}
         info.update({
-            "cp_group": cp_group, "dp_group": dp_group,
-            "cp_rank": rank % cp, "cp_degree": cp,
-            "dp_rank": rank // cp, "dp_degree": world // cp,
+            "%s_group" % kind: in_group, "dp_group": dp_group,
+            "%s_rank" % kind: rank % inner, "%s_degree" % kind: inner,
+            "dp_rank": rank // inner, "dp_degree": world // inner,
         })
         from collections import namedtuple
 
