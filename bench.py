@@ -35,6 +35,10 @@ def parse_args():
                    help="context parallelism: all ranks form one ring-"
                         "attention group over a single shared batch "
                         "(seq sharded across ranks)")
+    p.add_argument("--tp", action="store_true",
+                   help="tensor parallelism: all ranks form one "
+                        "Megatron-style shard group over a single "
+                        "shared batch")
     p.add_argument("--lr", type=float, default=3e-4)
     p.add_argument("--tunableop", action="store_true",
                    help="enable PyTorch TunableOp (hipBLASLt algorithm "
@@ -95,7 +99,15 @@ def main():
     cfg = cfg_fn()
     seq = min(args.seq, cfg.max_seq_len)
 
-    use_cp = args.cp and distributed
+    use_tp = args.tp and distributed
+    use_cp = args.cp and distributed and not use_tp
+    if use_tp:
+        assert args.model in ("llama3-8b", "llama3-70b", "tiny"), \
+            "--tp supports the Llama family"
+        if cfg.num_kv_heads % world:
+            # tiny config: widen the GQA group so heads shard evenly
+            cfg.num_heads = max(cfg.num_heads, 2 * world)
+            cfg.num_kv_heads = world
     if use_cp:
         assert args.model in ("llama3-8b", "llama3-70b", "tiny"), \
             "--cp supports the Llama family"
@@ -107,11 +119,24 @@ def main():
     torch.manual_seed(1234)  # same init on all ranks (DP)
     t0 = time.time()
     with torch.device(device):  # construct + random-init directly on GPU
-        if use_cp:
+        if use_tp:
+            from metaflow_amd.models.llama_tp import TPLlamaForCausalLM
+
+            model = TPLlamaForCausalLM(cfg, dist.group.WORLD)
+        elif use_cp:
             model = model_cls(cfg, cp_group=dist.group.WORLD)
         else:
             model = model_cls(cfg)
-    flat = FlatParamModel(model, bucket_mb=args.bucket_mb)
+    dp_group = None
+    if use_tp:
+        # tp spans the whole job: each rank is its own dp group (shards
+        # must never be averaged across tp ranks)
+        for r in range(world):
+            g = dist.new_group([r])
+            if r == rank:
+                dp_group = g
+    flat = FlatParamModel(model, bucket_mb=args.bucket_mb,
+                          group=dp_group)
     flat.install_overlap_hooks()
     opt = FusedAdamW(flat, lr=args.lr)
     if rank == 0:
@@ -122,7 +147,7 @@ def main():
     # synthetic batch (fixed per rank but distinct across ranks: loss must
     # fall, proving a real fwd+bwd+optimizer step is in the timed region).
     # Under --cp every rank sees the SAME batch and takes its seq shard.
-    torch.manual_seed(5678 + (0 if use_cp else rank))
+    torch.manual_seed(5678 + (0 if (use_cp or use_tp) else rank))
     tokens = torch.randint(0, cfg.vocab_size, (args.batch, seq + 1),
                            device=device)
     inp, tgt = tokens[:, :-1], tokens[:, 1:].contiguous()
@@ -167,8 +192,9 @@ def main():
     first_loss = float(losses[0].item())
     last_loss = float(losses[-1].item())
 
-    # cp: one shared batch across the group; dp: one batch per rank
-    tokens_per_step = args.batch * seq * (1 if use_cp else n_gpus)
+    # cp/tp: one shared batch across the group; dp: one batch per rank
+    shared = use_cp or use_tp
+    tokens_per_step = args.batch * seq * (1 if shared else n_gpus)
     toks_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -182,15 +208,16 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "strong" if use_cp else "weak",
+            "scaling": "strong" if shared else "weak",
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch * (1 if use_cp else n_gpus),
+                "global_batch": args.batch * (1 if shared else n_gpus),
                 "seq_len": seq,
-                "parallelism": ("cp%d" if use_cp else "dp%d") % n_gpus,
+                "parallelism": ("tp%d" % n_gpus if use_tp else
+                                ("cp%d" if use_cp else "dp%d") % n_gpus),
                 "first_loss": round(first_loss, 4),
                 "last_loss": round(last_loss, 4),
             },

@@ -149,3 +149,38 @@ def _ddp_worker(rank, port, q):
         q.put("ok")
     except Exception as e:  # noqa: BLE001
         q.put("fail: %r" % e)
+
+
+def test_bench_tp_tiny_world2(tmp_path):
+    """bench.py --tp over 2 gloo ranks: tensor-parallel llama through the
+    bench contract; JSON reports tp2 + strong scaling."""
+    port = _free_port()
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({
+            "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": "2",
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(REPO, "bench.py"),
+             "--model", "tiny", "--steps", "2", "--warmup", "1",
+             "--batch", "1", "--seq", "256", "--tp"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, "rank failed:\n%s\n%s" % (out, err)
+        outs.append(out)
+    import json
+
+    json_lines = [l for l in outs[0].splitlines() if l.startswith("{")]
+    rec = json.loads(json_lines[0])
+    assert rec["config"]["parallelism"] == "tp2"
+    assert rec["scaling"] == "strong"
+    assert rec["value"] > 0
