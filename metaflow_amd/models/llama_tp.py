@@ -5,9 +5,9 @@ Per layer, rank r holds nq/w query heads + nkv/w kv heads (attention is
 fully local to the rank's heads — GQA ratio preserved), a gate|up shard
 of the MLP, and row shards of o_proj/down_proj; exactly TWO all-reduces
 of [B, S, h] per layer (attention output + MLP output) travel the tp
-group over xGMI. The lm_head is vocab-sharded with an all-gather before
-the loss (a vocab-parallel CE that skips the gather is the round-2
-refinement).
+group over xGMI. The lm_head is vocab-sharded and the loss is a
+vocab-parallel cross-entropy — three scalar-per-token all-reduces
+instead of gathering [B,S,V] logits.
 
 Composes with the other axes: put tp inside the gang's innermost ranks
 and pass the dp group to FlatParamModel(group=...) — sharded params
@@ -32,6 +32,7 @@ from ..parallel.tp import (
     shard_cols,
     shard_gate_up_rows,
     shard_qkv_rows,
+    vocab_parallel_cross_entropy,
 )
 from .llama import LlamaConfig, RMSNorm
 
@@ -148,12 +149,18 @@ class TPLlamaForCausalLM(nn.Module):
             res, pending = layer(res, pending, cos_t, sin_t)
         _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
                              self.final_norm.eps)
-        logits = gather_from_tp(self.lm_head(x), self.group)
+        logits_shard = self.lm_head(x)
         if targets is None:
-            return logits
-        B, S, V = logits.shape
-        loss = K.cross_entropy(logits.reshape(B * S, V),
-                               targets.reshape(B * S))
+            return gather_from_tp(logits_shard, self.group)
+        import torch.distributed as dist
+
+        B, S, Vl = logits_shard.shape
+        v0 = dist.get_rank(self.group) * Vl
+        # vocab-parallel CE: O(tokens) comm, the full [B,S,V] logits are
+        # never materialized (tp.py: _VocabParallelCE)
+        loss = vocab_parallel_cross_entropy(
+            logits_shard.reshape(B * S, Vl), targets.reshape(B * S),
+            self.group, v0)
         return loss.mean()
 
     def num_params(self):

@@ -149,3 +149,57 @@ def shard_gate_up_rows(full, rank, world):
     half = full.size(0) // 2
     return torch.cat([shard_rows(full[:half], rank, world),
                       shard_rows(full[half:], rank, world)], dim=0)
+
+
+class _VocabParallelCE(torch.autograd.Function):
+    """Cross-entropy over a vocab-sharded logits tensor WITHOUT
+    materializing the full logits: the softmax statistics (max,
+    sum-exp) and the target logit are each one scalar-per-token
+    all-reduce — O(N) communication instead of O(N*V) for an
+    all-gather. Returns per-token loss [N]."""
+
+    @staticmethod
+    def forward(ctx, logits_shard, targets, group, vocab_start):
+        lf = logits_shard.float()
+        N, Vl = lf.shape
+        m = lf.max(dim=-1).values
+        dist.all_reduce(m, op=dist.ReduceOp.MAX, group=group)
+        se = torch.exp(lf - m[:, None]).sum(dim=-1)
+        dist.all_reduce(se, group=group)
+        lse = m + torch.log(se)
+
+        local = targets - vocab_start
+        in_range = (local >= 0) & (local < Vl)
+        idx = local.clamp(0, Vl - 1)
+        tgt_logit = torch.where(
+            in_range, lf.gather(1, idx[:, None]).squeeze(1),
+            torch.zeros_like(m))
+        dist.all_reduce(tgt_logit, group=group)
+        loss = lse - tgt_logit
+        ctx.save_for_backward(logits_shard, targets, lse)
+        ctx.group = group
+        ctx.vocab_start = vocab_start
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits_shard, targets, lse = ctx.saved_tensors
+        lf = logits_shard.float()
+        N, Vl = lf.shape
+        p = torch.exp(lf - lse[:, None])
+        local = targets - ctx.vocab_start
+        in_range = (local >= 0) & (local < Vl)
+        idx = local.clamp(0, Vl - 1)
+        p.scatter_add_(
+            1, idx[:, None],
+            torch.where(in_range, -torch.ones_like(lse),
+                        torch.zeros_like(lse))[:, None])
+        dlogits = (p * dloss.float()[:, None]).to(logits_shard.dtype)
+        return dlogits, None, None, None
+
+
+def vocab_parallel_cross_entropy(logits_shard, targets, group,
+                                 vocab_start):
+    """Per-token loss [N] from vocab-sharded logits [N, V/w]."""
+    return _VocabParallelCE.apply(logits_shard.contiguous(), targets,
+                                  group, vocab_start)
