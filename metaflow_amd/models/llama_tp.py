@@ -28,7 +28,10 @@ from ..ops import kernels as K
 from ..parallel.tp import (
     ColumnParallelLinear,
     RowParallelLinear,
+    all_gather_sp,
     gather_from_tp,
+    mark_sp_partial,
+    reduce_scatter_sp,
     shard_cols,
     shard_gate_up_rows,
     shard_qkv_rows,
@@ -38,12 +41,13 @@ from .llama import LlamaConfig, RMSNorm
 
 
 class TPDecoderLayer(nn.Module):
-    def __init__(self, cfg: LlamaConfig, group):
+    def __init__(self, cfg: LlamaConfig, group, sequence_parallel=False):
         super().__init__()
         import torch.distributed as dist
 
         self.cfg = cfg
         self.group = group
+        self.sp = sequence_parallel
         self.world = dist.get_world_size(group)
         h, hd = cfg.hidden_size, cfg.head_dim
         assert cfg.num_heads % self.world == 0
@@ -61,26 +65,51 @@ class TPDecoderLayer(nn.Module):
         self.down_proj = RowParallelLinear(cfg.intermediate_size, h, group)
 
     def forward(self, res, pending, cos_t, sin_t):
+        """Without SP, res/pending are replicated [B,S,h]; with SP they
+        are sequence shards [B,S/w,h] and every column GEMM is preceded
+        by an all-gather, every row GEMM followed by a reduce-scatter
+        (same comm volume as the plain tp all-reduces, 1/w the
+        norm/residual activation memory)."""
         cfg = self.cfg
-        B, S, _ = res.shape
         hd = cfg.head_dim
         res, y = K.add_rmsnorm(res, pending, self.input_norm.weight,
                                self.input_norm.eps)
-        qkv = self.qkv_proj(y)          # local heads only
+        if self.sp:
+            y = all_gather_sp(y, self.group)
+            qkv = torch.nn.functional.linear(y, self.qkv_proj.weight)
+        else:
+            qkv = self.qkv_proj(y)      # local heads only
+        B, S, _ = qkv.shape
         q, k, v = K.rope_qkv(qkv, cos_t, sin_t, self.nq_l, self.nkv_l)
         o = K.attention(q, k, v)        # local GQA group, no comm
         o = o.transpose(1, 2).reshape(B, S, self.nq_l * hd)
-        res, y = K.add_rmsnorm(res, self.o_proj(o), self.post_norm.weight,
+        if self.sp:
+            att = reduce_scatter_sp(
+                torch.nn.functional.linear(o, self.o_proj.weight),
+                self.group)
+        else:
+            att = self.o_proj(o)
+        res, y = K.add_rmsnorm(res, att, self.post_norm.weight,
                                self.post_norm.eps)
-        pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
+        if self.sp:
+            y = all_gather_sp(y, self.group)
+            gu = torch.nn.functional.linear(y, self.gate_up_proj.weight)
+            pending = reduce_scatter_sp(
+                torch.nn.functional.linear(
+                    K.swiglu_fused(gu), self.down_proj.weight),
+                self.group)
+        else:
+            pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
         return res, pending
 
 
 class TPLlamaForCausalLM(nn.Module):
-    def __init__(self, cfg: LlamaConfig, tp_group):
+    def __init__(self, cfg: LlamaConfig, tp_group,
+                 sequence_parallel=False):
         super().__init__()
         self.cfg = cfg
         self.group = tp_group
+        self.sp = sequence_parallel
         import torch.distributed as dist
 
         self.world = dist.get_world_size(tp_group)
@@ -88,7 +117,8 @@ class TPLlamaForCausalLM(nn.Module):
         self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                   dtype=torch.bfloat16)
         self.layers = nn.ModuleList(
-            TPDecoderLayer(cfg, tp_group) for _ in range(cfg.num_layers))
+            TPDecoderLayer(cfg, tp_group, sequence_parallel)
+            for _ in range(cfg.num_layers))
         self.final_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
         self.lm_head = ColumnParallelLinear(cfg.hidden_size,
                                             cfg.vocab_size, tp_group)
@@ -97,6 +127,12 @@ class TPLlamaForCausalLM(nn.Module):
         self.register_buffer("cos_t", cos_t, persistent=False)
         self.register_buffer("sin_t", sin_t, persistent=False)
         self.reset_parameters()
+        if sequence_parallel:
+            # these params only see this rank's sequence chunk: their
+            # grads need sp_sync_grads(model, tp_group) after backward
+            mark_sp_partial(self.embed.weight, self.final_norm.weight,
+                            *[l.input_norm.weight for l in self.layers],
+                            *[l.post_norm.weight for l in self.layers])
 
     def reset_parameters(self):
         std = 0.02
@@ -111,7 +147,7 @@ class TPLlamaForCausalLM(nn.Module):
             layer.down_proj.weight.data.mul_(scale)
 
     @classmethod
-    def from_full_model(cls, full, tp_group):
+    def from_full_model(cls, full, tp_group, sequence_parallel=False):
         """Build the TP model holding exact shards of an unsharded
         LlamaForCausalLM (equivalence tests / converting checkpoints)."""
         import torch.distributed as dist
@@ -119,7 +155,7 @@ class TPLlamaForCausalLM(nn.Module):
         cfg = full.cfg
         rank = dist.get_rank(tp_group)
         world = dist.get_world_size(tp_group)
-        m = cls(cfg, tp_group)
+        m = cls(cfg, tp_group, sequence_parallel)
         with torch.no_grad():
             m.embed.weight.copy_(full.embed.weight)
             m.final_norm.weight.copy_(full.final_norm.weight)
@@ -141,15 +177,28 @@ class TPLlamaForCausalLM(nn.Module):
         return m
 
     def forward(self, tokens, targets=None):
+        import torch.distributed as dist
+
         S = tokens.size(1)
         cos_t, sin_t = self.cos_t, self.sin_t
         res = self.embed(tokens)
+        if self.sp:  # each rank keeps its sequence chunk
+            sc = S // self.world
+            r = dist.get_rank(self.group)
+            res = res[:, r * sc:(r + 1) * sc].contiguous()
         pending = None
         for layer in self.layers:
             res, pending = layer(res, pending, cos_t, sin_t)
         _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
                              self.final_norm.eps)
-        logits_shard = self.lm_head(x)
+        if self.sp:  # back to the full sequence for the lm_head
+            # NOTE: all_gather_sp's backward already sums input grads
+            # over the group, so bypass copy_to_tp (double reduction)
+            x = all_gather_sp(x, self.group)
+            logits_shard = torch.nn.functional.linear(
+                x, self.lm_head.weight)
+        else:
+            logits_shard = self.lm_head(x)
         if targets is None:
             return gather_from_tp(logits_shard, self.group)
         import torch.distributed as dist

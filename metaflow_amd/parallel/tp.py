@@ -203,3 +203,84 @@ def vocab_parallel_cross_entropy(logits_shard, targets, group,
     """Per-token loss [N] from vocab-sharded logits [N, V/w]."""
     return _VocabParallelCE.apply(logits_shard.contiguous(), targets,
                                   group, vocab_start)
+
+
+# ----------------------------------------------------- sequence parallel
+# Megatron-style SP: the activations BETWEEN sharded GEMMs (norms,
+# residual adds) are sequence-sharded [B, S/w, h] instead of replicated,
+# cutting their memory by w. The tp all-reduces become
+# all-gather (before a column GEMM) / reduce-scatter (after a row GEMM)
+# pairs of identical total volume. reduce-scatter is emulated as
+# all-reduce + take-my-slice so the same code runs gloo (CPU tests) and
+# RCCL (which has the real collective — round-2 swap-in).
+
+def _sp_slice(x, group):
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    sc = x.size(1) // world
+    return x[:, rank * sc:(rank + 1) * sc].contiguous()
+
+
+class _AllGatherSP(torch.autograd.Function):
+    """[B, S/w, h] -> [B, S, h]; backward reduce-scatters the grads."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        world = dist.get_world_size(group)
+        x = x.contiguous()
+        parts = [torch.empty_like(x) for _ in range(world)]
+        dist.all_gather(parts, x, group=group)
+        parts[dist.get_rank(group)] = x
+        return torch.cat(parts, dim=1)
+
+    @staticmethod
+    def backward(ctx, grad):
+        g = grad.contiguous()
+        dist.all_reduce(g, group=ctx.group)
+        return _sp_slice(g, ctx.group), None
+
+
+class _ReduceScatterSP(torch.autograd.Function):
+    """Sum partials over the group and keep this rank's sequence chunk:
+    [B, S, h] -> [B, S/w, h]; backward all-gathers."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        g = x.contiguous()
+        dist.all_reduce(g, group=group)
+        return _sp_slice(g, group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        world = dist.get_world_size(ctx.group)
+        g = grad.contiguous()
+        parts = [torch.empty_like(g) for _ in range(world)]
+        dist.all_gather(parts, g, group=ctx.group)
+        parts[dist.get_rank(ctx.group)] = g
+        return torch.cat(parts, dim=1), None
+
+
+def all_gather_sp(x, group):
+    return _AllGatherSP.apply(x, group)
+
+
+def reduce_scatter_sp(x, group):
+    return _ReduceScatterSP.apply(x, group)
+
+
+def mark_sp_partial(*params):
+    """Tag params whose grads are partial under SP (they only saw this
+    rank's sequence chunk): norms, embeddings."""
+    for p in params:
+        p._mfx_sp_partial = True
+
+
+def sp_sync_grads(model, group):
+    """All-reduce the grads of SP-partial params over the tp group —
+    call after backward, before the optimizer (the analogue of
+    Megatron's sequence-parallel grad sync)."""
+    for p in model.parameters():
+        if getattr(p, "_mfx_sp_partial", False) and p.grad is not None:
+            dist.all_reduce(p.grad, group=group)

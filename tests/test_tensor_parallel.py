@@ -18,13 +18,14 @@ def _free_port():
     return port
 
 
-def test_tp_llama_world2():
+def _run(sp):
     import torch.multiprocessing as mp
 
     port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    ps = [ctx.Process(target=_worker, args=(r, port, q)) for r in range(2)]
+    ps = [ctx.Process(target=_worker, args=(r, port, q, sp))
+          for r in range(2)]
     for p in ps:
         p.start()
     for p in ps:
@@ -33,7 +34,18 @@ def test_tp_llama_world2():
     assert all(r == "ok" for r in results), results
 
 
-def _worker(rank, port, q):
+def test_tp_llama_world2():
+    _run(sp=False)
+
+
+def test_tp_llama_sequence_parallel_world2():
+    """TP + Megatron sequence parallelism: norm/residual activations
+    sequence-sharded; same exactness bar as plain TP (incl. the
+    sp_sync_grads pass for the SP-partial params)."""
+    _run(sp=True)
+
+
+def _worker(rank, port, q, sp=False):
     try:
         import torch
         import torch.distributed as dist
@@ -62,7 +74,8 @@ def _worker(rank, port, q):
         cfg = LlamaConfig.tiny(vocab=256, seq=64)
         cfg.num_heads, cfg.num_kv_heads = 4, 2
         ref = LlamaForCausalLM(cfg)
-        tp = TPLlamaForCausalLM.from_full_model(ref, dist.group.WORLD)
+        tp = TPLlamaForCausalLM.from_full_model(ref, dist.group.WORLD,
+                                                sequence_parallel=sp)
 
         torch.manual_seed(77)
         tok = torch.randint(0, cfg.vocab_size, (2, 65))
@@ -72,6 +85,10 @@ def _worker(rank, port, q):
         loss_ref.backward()
         loss = tp(inp, tgt)
         loss.backward()
+        if sp:
+            from metaflow_amd.parallel.tp import sp_sync_grads
+
+            sp_sync_grads(tp, dist.group.WORLD)
 
         assert abs(float(loss) - float(loss_ref)) < 2e-3, \
             (float(loss), float(loss_ref))
