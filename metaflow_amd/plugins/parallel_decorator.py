@@ -32,7 +32,7 @@ class TorchParallelDecorator(ParallelDecorator):
 
     name = "torch_parallel"
     defaults = {"backend": None, "context_parallel": 1,
-                "tensor_parallel": 1}
+                "tensor_parallel": 1, "pipeline_parallel": 1}
 
     def task_decorate(self, step_func, flow, graph, retry_count,
                       max_user_code_retries, ubf_context):
@@ -69,6 +69,11 @@ class TorchParallelDecorator(ParallelDecorator):
                   else int(tp_attr))
             if tp > 1:
                 self._make_grid(dist, tp, "tp")
+            pp_attr = self.attributes.get("pipeline_parallel") or 1
+            pp = (dist.get_world_size() if pp_attr == "all"
+                  else int(pp_attr))
+            if pp > 1:
+                self._make_grid(dist, pp, "pp")
             try:
                 return step_func(*args, **kwargs)
             finally:

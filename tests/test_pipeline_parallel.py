@@ -107,3 +107,15 @@ def _worker(rank, port, q, microbatches):
         import traceback
 
         q.put("fail rank %d: %r\n%s" % (rank, e, traceback.format_exc()))
+
+
+def test_pp_flow_through_scheduler(tmp_datastore):
+    """@torch_parallel(pipeline_parallel=2) gang through the real
+    scheduler: the 2-stage pipe trains and the loss falls."""
+    from .test_runtime import latest_run_id, read_artifact, run_flow
+
+    run_flow("pp_flow.py", tmp_datastore, "run", timeout=420)
+    run_id = latest_run_id(tmp_datastore, "PPFlow")
+    losses = read_artifact(tmp_datastore, "PPFlow", run_id, "join",
+                           "losses")
+    assert len(losses) == 2
