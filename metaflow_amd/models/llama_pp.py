@@ -10,8 +10,9 @@ consecutive GPUs so each boundary is a dedicated link.
 `pp_train_step` runs the schedule: forward all M micro-batches through
 the pipe (send/recv activations), then backward in reverse order,
 shipping boundary gradients back. Micro-batching bounds the pipeline
-bubble at (w-1)/(M+w-1); the 1F1B interleave that also bounds activation
-memory is the round-2 refinement.
+bubble at (w-1)/(M+w-1); the default 1F1B schedule also bounds live
+activations at pipeline-depth micro-batches (fill-drain available as
+schedule="gpipe").
 
 Works on gloo (CPU equivalence tests vs the unsharded model —
 micro-batch=1 is bit-identical compute) and RCCL alike.
@@ -116,12 +117,19 @@ def _recv(shape, dtype, src, group, device):
     return t
 
 
-def pp_train_step(stage_model, tokens, targets, microbatches=1):
-    """One fill-drain pipeline step; every rank returns the mean loss.
+def pp_train_step(stage_model, tokens, targets, microbatches=1,
+                  schedule="1f1b"):
+    """One pipeline step; every rank returns the mean loss.
 
     tokens/targets are the FULL batch [B, S] on every rank (only the
     first/last stages read them); grads accumulate into the stage's
     params — run your optimizer afterwards.
+
+    schedule="1f1b" (default) runs the one-forward-one-backward steady
+    state: at most (pipeline_depth - stage) micro-batch activations are
+    ever live, so M can grow (shrinking the (w-1)/(M+w-1) bubble)
+    without growing memory. "gpipe" is the plain fill-drain (all M
+    forwards, then all M backwards) — same numerics, simpler trace.
     """
     m = stage_model
     group = m.group
@@ -137,9 +145,18 @@ def pp_train_step(stage_model, tokens, targets, microbatches=1):
     mb = B // microbatches
     shape = (mb, S, cfg.hidden_size)
 
-    saved = []          # per micro-batch: (inputs..., outputs...)
+    from collections import deque
+
+    saved = deque()     # FIFO of (inputs, outputs) awaiting backward
     losses = []
-    for i in range(microbatches):
+    sends = []          # in-flight isends (keep refs until waited)
+    state = {"fwd": 0, "bwd": 0}
+
+    def _isend(t, dst):
+        sends.append(dist.isend(t.contiguous(), dst, group=group))
+
+    def do_fwd():
+        i = state["fwd"]
         sl = slice(i * mb, (i + 1) * mb)
         if m.is_first:
             inp = None
@@ -156,12 +173,13 @@ def pp_train_step(stage_model, tokens, targets, microbatches=1):
             losses.append(loss)
             saved.append((inp, loss))
         else:
-            _send(res.detach(), nxt, group)
-            _send(pending.detach(), nxt, group)
+            _isend(res.detach(), nxt)
+            _isend(pending.detach(), nxt)
             saved.append((inp, (res, pending)))
+        state["fwd"] += 1
 
-    for i in reversed(range(microbatches)):
-        inp, out = saved[i]
+    def do_bwd():
+        inp, out = saved.popleft()
         if m.is_last:
             (out / microbatches).backward()
         else:
@@ -169,8 +187,28 @@ def pp_train_step(stage_model, tokens, targets, microbatches=1):
             g_pending = _recv(shape, torch.bfloat16, nxt, group, dev)
             torch.autograd.backward(list(out), [g_res, g_pending])
         if not m.is_first:
-            _send(inp[0].grad, prv, group)
-            _send(inp[1].grad, prv, group)
+            _isend(inp[0].grad, prv)
+            _isend(inp[1].grad, prv)
+        state["bwd"] += 1
+
+    if schedule == "gpipe":
+        for _ in range(microbatches):
+            do_fwd()
+        # fill-drain backward runs in reverse micro-batch order
+        saved = deque(reversed(saved))
+        for _ in range(microbatches):
+            do_bwd()
+    else:  # 1f1b
+        warmup = min(m.world - 1 - m.stage, microbatches)
+        for _ in range(warmup):
+            do_fwd()
+        while state["fwd"] < microbatches:
+            do_fwd()
+            do_bwd()
+        while state["bwd"] < microbatches:
+            do_bwd()
+    for w in sends:
+        w.wait()
 
     # everyone reports the same mean loss
     loss_val = torch.zeros(1, device=dev, dtype=torch.float32)

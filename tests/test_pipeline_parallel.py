@@ -19,13 +19,14 @@ def _free_port():
     return port
 
 
-def _run(microbatches):
+def _run(microbatches, schedule="1f1b"):
     import torch.multiprocessing as mp
 
     port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    ps = [ctx.Process(target=_worker, args=(r, port, q, microbatches))
+    ps = [ctx.Process(target=_worker,
+                      args=(r, port, q, microbatches, schedule))
           for r in range(2)]
     for p in ps:
         p.start()
@@ -43,7 +44,15 @@ def test_pp_llama_world2_mb2():
     _run(2)
 
 
-def _worker(rank, port, q, microbatches):
+def test_pp_llama_world2_mb2_gpipe():
+    _run(2, schedule="gpipe")
+
+
+def test_pp_llama_world2_mb4_1f1b():
+    _run(4)
+
+
+def _worker(rank, port, q, microbatches, schedule="1f1b"):
     try:
         import torch
         import torch.distributed as dist
@@ -70,13 +79,14 @@ def _worker(rank, port, q, microbatches):
         stage = PPLlamaStage.from_full_model(ref, dist.group.WORLD)
 
         torch.manual_seed(88)
-        tok = torch.randint(0, cfg.vocab_size, (2, 65))
+        tok = torch.randint(0, cfg.vocab_size, (4, 65))
         inp, tgt = tok[:, :-1], tok[:, 1:].contiguous()
 
         loss_ref = ref(inp, tgt)
         loss_ref.backward()
 
-        loss = pp_train_step(stage, inp, tgt, microbatches=microbatches)
+        loss = pp_train_step(stage, inp, tgt, microbatches=microbatches,
+                             schedule=schedule)
         tol = 1e-4 if microbatches == 1 else 5e-3
         assert abs(loss - float(loss_ref)) < tol, (loss, float(loss_ref))
 
