@@ -183,10 +183,15 @@ class FlowGraph(object):
 
     def _traverse_graph(self):
         def traverse(node, seen, split_parents):
-            if node.type in ("split", "split-switch", "foreach",
-                             "split-parallel"):
+            if node.type in ("split", "foreach", "split-parallel"):
                 node.split_parents = split_parents
                 split_parents = split_parents + [node.name]
+            elif node.type == "split-switch":
+                # a switch is not a fan-out: exactly one branch runs, so
+                # it has no matching join and must NOT deepen the split
+                # stack — this is also what makes recursive switches
+                # (back-edges) terminate here
+                node.split_parents = split_parents
             elif node.type == "join":
                 # the matching split is the innermost unjoined one
                 if split_parents:

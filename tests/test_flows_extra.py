@@ -225,3 +225,30 @@ def test_example_train_llama_cp(tmp_datastore):
     losses = read_artifact(tmp_datastore, "TrainLlamaCP", run_id, "join",
                            "losses")
     assert len(losses) == 2 and all(l == l for l in losses)
+
+
+def test_recursive_switch(tmp_datastore):
+    """A switch back-edge loops a step until its condition flips
+    (reference recursive_switch): three iterations, then the exit arm."""
+    run_flow("recursive_switch_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "RecursiveSwitchFlow")
+    assert read_artifact(tmp_datastore, "RecursiveSwitchFlow", run_id,
+                         "finish", "total") == 30
+    # three work iterations really ran as three tasks
+    work_dir = os.path.join(tmp_datastore, "RecursiveSwitchFlow", run_id,
+                            "work")
+    assert len(os.listdir(work_dir)) == 3
+
+
+def test_recursive_switch_resume(tmp_datastore):
+    """Resume of a flow containing a switch loop: the successful loop
+    iterations clone, only the failed tail reruns."""
+    proc = run_flow("recursive_switch_flow.py", tmp_datastore, "run",
+                    check=False, env_extra={"REC_FAIL": "1"})
+    assert proc.returncode != 0
+    proc2 = run_flow("recursive_switch_flow.py", tmp_datastore, "resume",
+                     env_extra={"REC_FAIL": "0"})
+    assert proc2.returncode == 0
+    run_id = latest_run_id(tmp_datastore, "RecursiveSwitchFlow")
+    assert read_artifact(tmp_datastore, "RecursiveSwitchFlow", run_id,
+                         "finish", "total") == 30
