@@ -83,6 +83,27 @@ GRAPHS = {
         ("join_u", "join", ["end"], {}),
         ("end", "end", [], {}),
     ],
+    # composite switch placements (reference switch_in_foreach /
+    # branch_in_switch graphs): only the 'fast' arm executes
+    "switch_in_foreach": [
+        ("start", "foreach", ["mid"], {"var": "fanout", "n": 2}),
+        ("mid", "switch", ["fast", "slow"], {"var": "route"}),
+        ("fast", "linear", ["conv"], {}),
+        ("slow", "linear", ["conv"], {}),
+        ("conv", "linear", ["join_f"], {}),
+        ("join_f", "join", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
+    "branch_in_switch": [
+        ("start", "switch", ["fast", "slow"], {"var": "route"}),
+        ("fast", "split", ["fa", "fb"], {}),
+        ("fa", "linear", ["join_fab"], {}),
+        ("fb", "linear", ["join_fab"], {}),
+        ("join_fab", "join", ["conv"], {}),
+        ("slow", "linear", ["conv"], {}),
+        ("conv", "linear", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
     # scheduler stress: a long sequential chain (transition latency adds
     # up; artifact passdown must stay metadata-only the whole way)
     "deep_linear": [
