@@ -160,6 +160,10 @@ def main(flow):
             top_level_args=state.top_level_args,
             tags=kwargs.get("tag") or (),
         )
+        rid_file = kwargs.get("run_id_file")
+        if rid_file:
+            with open(rid_file, "w") as f:
+                f.write(str(runtime.run_id))
         attr_file = kwargs.get("runner_attribute_file")
         if attr_file:
             with open(attr_file, "w") as f:
@@ -183,6 +187,9 @@ def main(flow):
 
     @cli.command(help="Run the flow locally.")
     @click.option("--run-id", default=None)
+    @click.option("--run-id-file", default=None,
+                  help="Write the run id to this file (reference "
+                       "run_id_file behavior).")
     @click.option("--max-workers", default=MAX_WORKERS, type=int)
     @click.option("--max-num-splits", default=MAX_NUM_SPLITS, type=int)
     @click.option("--tag", multiple=True)
@@ -192,6 +199,8 @@ def main(flow):
         _run_common(kwargs)
 
     @cli.command(help="Resume a failed run from where it left off.")
+    @click.option("--run-id-file", default=None,
+                  help="Write the new run id to this file.")
     @click.option("--origin-run-id", default=None,
                   help="Run to clone from (default: this flow's most "
                        "recent run, reference semantics)")

@@ -102,3 +102,10 @@ def test_dump_command(tmp_datastore):
     proc = run_flow("linear_flow.py", tmp_datastore, "dump",
                     "%s/end/%s" % (run_id, task_id))
     assert "final" in proc.stdout
+
+
+def test_run_id_file(tmp_datastore, tmp_path):
+    rid = tmp_path / "runid.txt"
+    run_flow("linear_flow.py", tmp_datastore, "run",
+             "--run-id-file", str(rid))
+    assert rid.read_text() == latest_run_id(tmp_datastore, "LinearFlow")
