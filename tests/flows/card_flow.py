@@ -29,9 +29,10 @@ class CardFlow(FlowSpec):
         ])
         self.next(self.end)
 
+    @card(id="report")
     @step
     def end(self):
-        pass
+        current.card.append(Markdown("# Final report"))
 
 
 if __name__ == "__main__":

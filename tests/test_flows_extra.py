@@ -125,6 +125,11 @@ def test_card(tmp_datastore):
     assert "data:image/png;base64," in html    # Image data URI
     assert "config" in html and "3e-04" in html.replace(
         "0.0003", "3e-04")                     # Artifact pprint
+    # a second card with an explicit id on another step
+    end_task = fds.list_tasks(run_id, "end")[0]
+    ds_end = fds.get_task_datastore(run_id, "end", end_task)
+    html2 = get_card(ds_end, card_id="report")
+    assert html2 and "Final report" in html2
 
 
 def test_config_and_mutator(tmp_datastore, tmp_path):
