@@ -70,6 +70,11 @@ def _init_state(state, datastore, datastore_root, metadata, quiet, with_,
         raise MFXException("Unknown metadata provider '%s'" % metadata)
     state.metadata = provider(state.flow_cls.__name__, storage)
     state.quiet = quiet
+    # flow-level decorator init (e.g. @project computes its namespace)
+    for deco in getattr(state.flow_cls, "_flow_decorators", []):
+        deco.flow_init(state.flow_cls, state.graph, None,
+                       state.flow_datastore, state.metadata, None, None,
+                       {})
     # args to replay on child processes so they see the same config
     state.top_level_args = [
         "--datastore", datastore,
@@ -158,7 +163,8 @@ def main(flow):
             max_num_splits=kwargs.get("max_num_splits") or MAX_NUM_SPLITS,
             quiet=state.quiet,
             top_level_args=state.top_level_args,
-            tags=kwargs.get("tag") or (),
+            tags=tuple(kwargs.get("tag") or ()) + tuple(
+                getattr(flow_cls, "_project_tags", ())),
         )
         rid_file = kwargs.get("run_id_file")
         if rid_file:

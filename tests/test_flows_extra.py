@@ -257,3 +257,27 @@ def test_recursive_switch_resume(tmp_datastore):
     run_id = latest_run_id(tmp_datastore, "RecursiveSwitchFlow")
     assert read_artifact(tmp_datastore, "RecursiveSwitchFlow", run_id,
                          "finish", "total") == 30
+
+
+def test_project_namespacing(tmp_datastore):
+    """@project: current.project_* populated in tasks and the run tagged
+    with project:/project_branch: (reference project_branch behavior)."""
+    run_flow("project_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "ProjectFlow")
+    assert read_artifact(tmp_datastore, "ProjectFlow", run_id, "start",
+                         "pname") == "mlplat"
+    branch = read_artifact(tmp_datastore, "ProjectFlow", run_id, "start",
+                           "branch")
+    assert branch.startswith("user.")
+    assert read_artifact(tmp_datastore, "ProjectFlow", run_id, "start",
+                         "pflow") == "mlplat.%s.ProjectFlow" % branch
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    import importlib
+
+    import metaflow_amd.client as client
+
+    importlib.reload(client)
+    client.namespace(None)
+    tags = client.Run("ProjectFlow/%s" % run_id).tags
+    assert "project:mlplat" in tags
+    assert any(t.startswith("project_branch:user.") for t in tags)
