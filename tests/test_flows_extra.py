@@ -281,3 +281,47 @@ def test_project_namespacing(tmp_datastore):
     tags = client.Run("ProjectFlow/%s" % run_id).tags
     assert "project:mlplat" in tags
     assert any(t.startswith("project_branch:user.") for t in tags)
+
+
+def test_client_task_logs(tmp_datastore, tmp_path):
+    """Client Task.stdout/loglines return the captured user prints
+    (reference basic_log behavior)."""
+    import subprocess
+    import sys
+
+    flow = tmp_path / "logflow.py"
+    flow.write_text(
+        "import sys\n"
+        "from metaflow_amd import FlowSpec, step\n"
+        "class LogFlow(FlowSpec):\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        print('hello-from-start')\n"
+        "        sys.stderr.write('warn-from-start\\n')\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n"
+        "if __name__ == '__main__':\n"
+        "    LogFlow()\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc.returncode == 0, proc.stderr[-1500:]
+    run_id = latest_run_id(tmp_datastore, "LogFlow")
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    import importlib
+
+    import metaflow_amd.client as client
+
+    importlib.reload(client)
+    client.namespace(None)
+    task = client.Run("LogFlow/%s" % run_id)["start"].task
+    assert "hello-from-start" in task.stdout
+    assert "warn-from-start" in task.stderr
+    lines = list(task.loglines("stdout"))
+    assert any("hello-from-start" in str(l) for l in lines)
