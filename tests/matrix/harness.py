@@ -104,6 +104,15 @@ GRAPHS = {
         ("conv", "linear", ["end"], {}),
         ("end", "end", [], {}),
     ],
+    "foreach_in_switch": [
+        ("start", "switch", ["fast", "slow"], {"var": "route"}),
+        ("fast", "foreach", ["item"], {"var": "fanout", "n": 3}),
+        ("item", "linear", ["join_i"], {}),
+        ("join_i", "join", ["conv"], {}),
+        ("slow", "linear", ["conv"], {}),
+        ("conv", "linear", ["end"], {}),
+        ("end", "end", [], {}),
+    ],
     # scheduler stress: a long sequential chain (transition latency adds
     # up; artifact passdown must stay metadata-only the whole way)
     "deep_linear": [
