@@ -139,3 +139,103 @@ def test_tp_flow_through_scheduler(tmp_datastore):
     losses = read_artifact(tmp_datastore, "TPFlow", run_id, "join",
                            "losses")
     assert len(losses) == 2 and losses[0] == losses[1]
+
+
+def test_tp2_dp2_world4():
+    """Composed grid at world 4: tp=2 inside dp=2. Each dp replica sees a
+    different batch; after the dp-group flat all-reduce every rank's
+    shard grads must equal the batch-averaged reference grads."""
+    import torch.multiprocessing as mp
+
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_grid_worker, args=(r, port, q))
+          for r in range(4)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(420)
+    results = [q.get() for _ in range(4)]
+    assert all(r == "ok" for r in results), results
+
+
+def _grid_worker(rank, port, q):
+    try:
+        import torch
+        import torch.distributed as dist
+
+        if REPO not in sys.path:
+            sys.path.insert(0, REPO)
+        from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
+        from metaflow_amd.parallel.ddp import FlatParamModel
+        from metaflow_amd.models.llama_tp import TPLlamaForCausalLM
+
+        os.environ.update({
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+            "RANK": str(rank),
+            "WORLD_SIZE": "4",
+        })
+        dist.init_process_group("gloo")
+        tp_deg = 2
+        tp_group = dp_group = None
+        for start in (0, 2):  # consecutive ranks share a tp group
+            g = dist.new_group([start, start + 1])
+            if start <= rank < start + 2:
+                tp_group = g
+        for idx in (0, 1):    # same shard index across replicas = dp
+            g = dist.new_group([idx, idx + 2])
+            if rank % tp_deg == idx:
+                dp_group = g
+
+        torch.manual_seed(21)
+        cfg = LlamaConfig.tiny(vocab=256, seq=64)
+        cfg.num_heads, cfg.num_kv_heads = 4, 2
+        ref = LlamaForCausalLM(cfg)
+        tp = TPLlamaForCausalLM.from_full_model(ref, tp_group)
+        flat = FlatParamModel(tp, bucket_mb=1, group=dp_group)
+
+        dp_rank = rank // tp_deg
+        torch.manual_seed(101 + dp_rank)   # one batch per dp replica
+        tok = torch.randint(0, cfg.vocab_size, (2, 65))
+        inp, tgt = tok[:, :-1], tok[:, 1:].contiguous()
+
+        loss = tp(inp, tgt)
+        loss.backward()
+        flat.finish_grad_sync()            # dp-average over dp_group
+
+        # reference: average the two per-batch gradient sets
+        grads = {}
+        for b in (0, 1):
+            refb = LlamaForCausalLM(cfg)
+            refb.load_state_dict(ref.state_dict())
+            torch.manual_seed(101 + b)
+            tb = torch.randint(0, cfg.vocab_size, (2, 65))
+            lb = refb(tb[:, :-1], tb[:, 1:].contiguous())
+            lb.backward()
+            for n, p in refb.named_parameters():
+                grads.setdefault(n, []).append(p.grad.float())
+        avg = {n: (g[0] + g[1]) / 2 for n, g in grads.items()}
+
+        from metaflow_amd.parallel.tp import shard_qkv_rows
+
+        tpr = dist.get_rank(tp_group)
+        l0 = avg["layers.0.qkv_proj.weight"]
+        want = shard_qkv_rows(l0, tpr, tp_deg, cfg.num_heads,
+                              cfg.num_kv_heads, cfg.head_dim)
+        got = tp.layers[0].qkv_proj.weight.grad.float()
+        denom = want.abs().max().item() + 1e-6
+        err = (got - want).abs().max().item() / denom
+        assert err < 8e-2, "qkv dp-averaged grad mismatch %g" % err
+        # replicated param too
+        wn = avg["final_norm.weight"]
+        gn = tp.final_norm.weight.grad.float()
+        errn = (gn - wn).abs().max().item() / (wn.abs().max() + 1e-6)
+        assert errn < 8e-2, "norm dp-averaged grad mismatch %g" % errn
+        dist.destroy_process_group()
+        q.put("ok")
+    except Exception as e:  # noqa: BLE001
+        import traceback
+
+        q.put("fail rank %d: %r\n%s" % (rank, e, traceback.format_exc()))
