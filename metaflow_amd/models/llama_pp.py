@@ -107,10 +107,6 @@ class PPLlamaStage(nn.Module):
         return sum(p.numel() for p in self.parameters())
 
 
-def _send(t, dst, group):
-    dist.send(t.contiguous(), dst, group=group)
-
-
 def _recv(shape, dtype, src, group, device):
     t = torch.empty(shape, dtype=dtype, device=device)
     dist.recv(t, src, group=group)
