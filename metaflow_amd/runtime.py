@@ -476,6 +476,15 @@ class NativeRuntime(object):
                                 for g in range(gpr))
                 env["HIP_VISIBLE_DEVICES"] = devs
                 env["CUDA_VISIBLE_DEVICES"] = devs
+            # CPU affinity: carve the cores evenly so ranks do not fight
+            # over the data-loader/host threads (the NUMA half of GPU
+            # pinning — on MI355X nodes consecutive GPUS map to
+            # consecutive NUMA domains)
+            ncpu = os.cpu_count() or 0
+            if ncpu >= num_parallel * 2:
+                per = ncpu // num_parallel
+                env["MFX_CPU_AFFINITY"] = "%d-%d" % (
+                    rank * per, (rank + 1) * per - 1)
             child_stack = spec.stack + ((spec.step, rank, num_parallel),)
             self._queue_spec(TaskSpec(
                 target, [pathspec], rank, child_stack, task_id=task_id,

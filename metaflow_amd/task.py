@@ -173,6 +173,16 @@ class MFXTask(object):
         self.metadata.register_task(run_id, step_name, task_id, retry_count,
                                     metadata={"attempt_started": True})
 
+        # gang rank CPU pinning (set by the gang scheduler; the NUMA
+        # half of HIP_VISIBLE_DEVICES)
+        aff = os.environ.get("MFX_CPU_AFFINITY")
+        if aff and hasattr(os, "sched_setaffinity"):
+            try:
+                lo, hi = aff.split("-")
+                os.sched_setaffinity(0, range(int(lo), int(hi) + 1))
+            except (ValueError, OSError):
+                pass
+
         # task-level liveness (reference task.py:797: heartbeats for
         # task+run); lossy sidecar, never blocks the task
         hb_sidecar = None

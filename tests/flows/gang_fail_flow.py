@@ -20,6 +20,13 @@ class GangFailFlow(FlowSpec):
         rank = current.parallel.node_index
         if os.environ.get("GANG_FAIL") == "1" and rank == 1:
             os._exit(3)
+        # gang ranks get disjoint CPU carve-outs (when cores allow)
+        aff = os.environ.get("MFX_CPU_AFFINITY")
+        if aff and hasattr(os, "sched_getaffinity"):
+            cur = os.sched_getaffinity(0)
+            lo, hi = (int(x) for x in aff.split("-"))
+            assert cur == set(range(lo, hi + 1)), (aff, cur)
+            self.affinity = aff
         # a collective proving the whole gang is alive
         import torch
 
