@@ -286,14 +286,14 @@ def test_generate_gpu(dev):
     from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
 
     torch.manual_seed(0)
-    cfg = LlamaConfig.tiny(vocab=256, seq=256)
+    cfg = LlamaConfig.tiny(vocab=256, seq=512)
     m = LlamaForCausalLM(cfg).to(dev).eval()
     torch.manual_seed(1)
-    prompt = torch.randint(0, 256, (1, 64), device=dev)
+    prompt = torch.randint(0, 256, (1, 256), device=dev)
     out = m.generate(prompt, 4)
-    assert out.shape == (1, 68)
+    assert out.shape == (1, 260)
     with torch.no_grad():
-        for t in range(64, 68):
+        for t in range(256, 260):
             full = m(out[:, :t])
             nxt = full[:, -1].float().argmax(-1)
             assert torch.equal(nxt, out[:, t]), t
