@@ -325,3 +325,16 @@ def test_client_task_logs(tmp_datastore, tmp_path):
     assert "warn-from-start" in task.stderr
     lines = list(task.loglines("stdout"))
     assert any("hello-from-start" in str(l) for l in lines)
+
+
+def test_nested_unbounded_foreach(tmp_datastore):
+    """UBF control/mapper fan-out nested inside a static foreach: per-
+    branch mappers, two-level join (reference nested_unbounded_foreach)."""
+    run_flow("nested_ubf_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "NestedUBFFlow")
+    assert read_artifact(tmp_datastore, "NestedUBFFlow", run_id, "join_o",
+                         "total") == 96
+    # two control tasks (one per outer branch), each with 3 mappers
+    work_dir = os.path.join(tmp_datastore, "NestedUBFFlow", run_id, "work")
+    tasks = os.listdir(work_dir)
+    assert len([t for t in tasks if "_mapper_" in t]) == 6
