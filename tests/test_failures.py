@@ -140,3 +140,76 @@ def test_gang_member_failure_and_resume(tmp_datastore):
     run_id = latest_run_id(tmp_datastore, "GangFailFlow")
     assert read_artifact(tmp_datastore, "GangFailFlow", run_id, "join",
                          "ranks") == [0, 1]
+
+
+def test_empty_foreach_fails_cleanly(tmp_datastore, tmp_path):
+    """foreach over an empty sequence raises a clear InvalidNext error
+    at the split step (reference behavior: runtime rejects it)."""
+    import subprocess
+    import sys
+
+    flow = tmp_path / "empty_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, step\n"
+        "class EmptyForeachFlow(FlowSpec):\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        self.items = []\n"
+        "        self.next(self.work, foreach='items')\n"
+        "    @step\n"
+        "    def work(self):\n"
+        "        self.next(self.join)\n"
+        "    @step\n"
+        "    def join(self, inputs):\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n"
+        "if __name__ == '__main__':\n"
+        "    EmptyForeachFlow()\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc.returncode != 0
+    assert "foreach over an empty sequence" in (proc.stdout + proc.stderr)
+
+
+def test_bad_switch_value_fails_cleanly(tmp_datastore, tmp_path):
+    """A switch condition naming a non-existent target step fails with a
+    clear error, not a hang or KeyError."""
+    import subprocess
+    import sys
+
+    flow = tmp_path / "badswitch_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, step\n"
+        "class BadSwitchFlow(FlowSpec):\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        self.route = 'nowhere'\n"
+        "        self.next(self.a, self.b, condition='route')\n"
+        "    @step\n"
+        "    def a(self):\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def b(self):\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n"
+        "if __name__ == '__main__':\n"
+        "    BadSwitchFlow()\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc.returncode != 0
+    out = proc.stdout + proc.stderr
+    assert "nowhere" in out or "condition" in out
