@@ -171,7 +171,7 @@ def test_empty_foreach_fails_cleanly(tmp_datastore, tmp_path):
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     env["MFX_NUM_GPUS"] = "0"
     proc = subprocess.run(
-        [sys.executable, str(flow), "--quiet", "--datastore-root",
+        [sys.executable, str(flow), "--datastore-root",
          tmp_datastore, "run"],
         capture_output=True, text=True, env=env, timeout=180)
     assert proc.returncode != 0
