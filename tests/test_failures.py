@@ -213,3 +213,31 @@ def test_bad_switch_value_fails_cleanly(tmp_datastore, tmp_path):
     assert proc.returncode != 0
     out = proc.stdout + proc.stderr
     assert "nowhere" in out or "condition" in out
+
+
+def test_resume_partial_foreach(tmp_datastore):
+    """Resume after ONE foreach child failed: succeeded children clone
+    (metadata 'cloned_from'), only the failed child re-executes
+    (reference resume_foreach_inner behavior)."""
+    import json
+
+    proc = run_flow("foreach_fail_flow.py", tmp_datastore, "run",
+                    check=False, env_extra={"FF_FAIL": "1"})
+    assert proc.returncode != 0
+    proc2 = run_flow("foreach_fail_flow.py", tmp_datastore, "resume",
+                     env_extra={"FF_FAIL": "0"})
+    assert proc2.returncode == 0
+    run_id = latest_run_id(tmp_datastore, "ForeachFailFlow")
+    assert read_artifact(tmp_datastore, "ForeachFailFlow", run_id, "join",
+                         "total") == 60
+    # 3 of the 4 work tasks were cloned, 1 re-executed
+    meta_dir = os.path.join(tmp_datastore, "ForeachFailFlow", "_meta",
+                            run_id)
+    cloned = 0
+    for fn in os.listdir(meta_dir):
+        if fn.startswith("task.work."):
+            info = json.load(open(os.path.join(meta_dir, fn)))
+            blob = json.dumps(info)
+            if "cloned_from" in blob:
+                cloned += 1
+    assert cloned == 3, cloned
