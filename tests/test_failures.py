@@ -241,3 +241,46 @@ def test_resume_partial_foreach(tmp_datastore):
             if "cloned_from" in blob:
                 cloned += 1
     assert cloned == 3, cloned
+
+
+def test_resume_preserves_parameters(tmp_datastore, tmp_path):
+    """resume reuses the ORIGIN run's parameter values (reference
+    semantics) — not CLI defaults."""
+    import subprocess
+    import sys
+
+    flow = tmp_path / "param_resume_flow.py"
+    flow.write_text(
+        "import os\n"
+        "from metaflow_amd import FlowSpec, Parameter, step\n"
+        "class ParamResumeFlow(FlowSpec):\n"
+        "    scale = Parameter('scale', default=1, type=int)\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        self.v = self.scale * 7\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        if os.environ.get('PR_FAIL') == '1':\n"
+        "            raise RuntimeError('planned')\n"
+        "        self.final = self.v + self.scale\n"
+        "if __name__ == '__main__':\n"
+        "    ParamResumeFlow()\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    env["PR_FAIL"] = "1"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run", "--scale", "9"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc.returncode != 0
+    env["PR_FAIL"] = "0"
+    proc2 = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "resume"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc2.returncode == 0, proc2.stderr[-1500:]
+    run_id = latest_run_id(tmp_datastore, "ParamResumeFlow")
+    assert read_artifact(tmp_datastore, "ParamResumeFlow", run_id, "end",
+                         "final") == 9 * 7 + 9
