@@ -284,3 +284,30 @@ def test_resume_preserves_parameters(tmp_datastore, tmp_path):
     run_id = latest_run_id(tmp_datastore, "ParamResumeFlow")
     assert read_artifact(tmp_datastore, "ParamResumeFlow", run_id, "end",
                          "final") == 9 * 7 + 9
+
+
+def test_resume_partial_nested_foreach(tmp_datastore):
+    """One failed leaf inside a 2x3 nested foreach: resume clones the 5
+    succeeded leaves (and the completed inner join of the other branch),
+    reruns only the failed leaf and its dependent joins."""
+    import json
+
+    proc = run_flow("nested_foreach_fail_flow.py", tmp_datastore, "run",
+                    check=False, env_extra={"NF_FAIL": "1"})
+    assert proc.returncode != 0
+    proc2 = run_flow("nested_foreach_fail_flow.py", tmp_datastore,
+                     "resume", env_extra={"NF_FAIL": "0"})
+    assert proc2.returncode == 0
+    run_id = latest_run_id(tmp_datastore, "NestedForeachFailFlow")
+    assert read_artifact(tmp_datastore, "NestedForeachFailFlow", run_id,
+                         "join_o", "total") == 306
+    meta_dir = os.path.join(tmp_datastore, "NestedForeachFailFlow",
+                            "_meta", run_id)
+    cloned = 0
+    for fn in os.listdir(meta_dir):
+        if fn.startswith("task.leaf."):
+            blob = json.dumps(json.load(
+                open(os.path.join(meta_dir, fn))))
+            if "cloned_from" in blob:
+                cloned += 1
+    assert cloned == 5, cloned  # the failing leaf sleeps so siblings finish
