@@ -15,6 +15,9 @@ class ForeachFailFlow(FlowSpec):
     @step
     def work(self):
         if os.environ.get("FF_FAIL") == "1" and self.input == 2:
+            import time
+
+            time.sleep(4)  # let the sibling children finish first
             raise RuntimeError("planned failure on item 2")
         self.val = self.input * 10
         self.next(self.join)
