@@ -104,8 +104,8 @@ class DecoderLayer(nn.Module):
         # one fused kernel: QKV split + bhsd transpose + RoPE
         q, k, v = K.rope_qkv(qkv, cos_t, sin_t, nq, nkv)
         if cache is not None:
-            kc, vc, li, pos = cache.k[layer_idx], cache.v[layer_idx], \
-                layer_idx, cache.pos
+            kc, vc, pos = cache.k[layer_idx], cache.v[layer_idx], \
+                cache.pos
             kc[:, :, pos:pos + S] = k
             vc[:, :, pos:pos + S] = v
             if pos == 0 and S % 64 == 0:
