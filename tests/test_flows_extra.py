@@ -338,3 +338,21 @@ def test_nested_unbounded_foreach(tmp_datastore):
     work_dir = os.path.join(tmp_datastore, "NestedUBFFlow", run_id, "work")
     tasks = os.listdir(work_dir)
     assert len([t for t in tasks if "_mapper_" in t]) == 6
+
+
+def test_example_train_llama_ddp(tmp_datastore):
+    """The DDP example flow end-to-end on CPU (2-rank gloo gang)."""
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, "examples", "train_llama_ddp.py"),
+         "--quiet", "--datastore-root", tmp_datastore, "run",
+         "--num-gpus", "2", "--train-steps", "2", "--batch", "1",
+         "--seq", "256", "--model-size", "tiny"],
+        capture_output=True, text=True, env=env, timeout=420)
+    assert proc.returncode == 0, proc.stderr[-2000:]
