@@ -220,6 +220,18 @@ class Run(MetaflowObject):
             return None
 
     @property
+    def dag(self):
+        """The run's static graph as persisted at execution time
+        ({step: {type, in_funcs, out_funcs, ...}} — reference
+        runtime_dag behavior: the DAG travels with the run)."""
+        try:
+            params = self._fds.get_task_datastore(self.id,
+                                                  "_parameters", "0")
+            return params.get("_graph_info")
+        except Exception:
+            return None
+
+    @property
     def code_package_key(self):
         return (self._meta.get_run(self.id) or {}).get("code_package_key")
 

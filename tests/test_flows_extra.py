@@ -57,6 +57,10 @@ def test_client_api(tmp_datastore):
     mf = client.Metaflow()
     flows = [f.id for f in mf]
     assert "BranchFlow" in flows
+    # the persisted DAG travels with the run (reference runtime_dag)
+    dag = client.Flow("BranchFlow").latest_run.dag
+    assert dag and dag["start"]["out_funcs"] == ["a", "b"]
+    assert dag["join"]["type"] == "join"
     flow = client.Flow("BranchFlow")
     run = flow.latest_run
     assert run.successful
