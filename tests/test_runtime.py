@@ -109,3 +109,46 @@ def test_run_id_file(tmp_datastore, tmp_path):
     run_flow("linear_flow.py", tmp_datastore, "run",
              "--run-id-file", str(rid))
     assert rid.read_text() == latest_run_id(tmp_datastore, "LinearFlow")
+
+
+def test_logs_and_dot_cli(tmp_datastore, tmp_path):
+    """The `logs` and `output-dot` flow CLI commands."""
+    flow = tmp_path / "echo_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, step\n"
+        "class EchoFlow(FlowSpec):\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        print('log-marker-xyz')\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n"
+        "if __name__ == '__main__':\n"
+        "    EchoFlow()\n")
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    run = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert run.returncode == 0
+    run_id = latest_run_id(tmp_datastore, "EchoFlow")
+    task_id = os.listdir(
+        os.path.join(tmp_datastore, "EchoFlow", run_id, "start"))[0]
+    logs = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "logs", "%s/start/%s" % (run_id, task_id)],
+        capture_output=True, text=True, env=env, timeout=120)
+    assert logs.returncode == 0
+    assert "log-marker-xyz" in logs.stdout
+    dot = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "output-dot"],
+        capture_output=True, text=True, env=env, timeout=120)
+    assert dot.returncode == 0
+    assert "digraph" in dot.stdout and "start" in dot.stdout
