@@ -104,3 +104,21 @@ def test_gpu_monitor_sidecar(tmp_path):
         assert out.exists(), "gpu monitor wrote no samples"
     finally:
         sc.terminate()
+
+
+def test_mfx_card_cli(tmp_datastore, tmp_path):
+    """`python -m metaflow_amd card <pathspec>` writes the task's HTML."""
+    run_flow("card_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "CardFlow")
+    task_id = os.listdir(
+        os.path.join(tmp_datastore, "CardFlow", run_id, "start"))[0]
+    out = tmp_path / "card.html"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    proc = subprocess.run(
+        [sys.executable, "-m", "metaflow_amd", "card",
+         "CardFlow/%s/start/%s" % (run_id, task_id), "--out", str(out)],
+        env=env, capture_output=True, text=True, timeout=120)
+    assert proc.returncode == 0, proc.stderr[-1000:]
+    assert "custom html" in out.read_text()
