@@ -311,3 +311,15 @@ def test_resume_partial_nested_foreach(tmp_datastore):
             if "cloned_from" in blob:
                 cloned += 1
     assert cloned == 5, cloned  # the failing leaf sleeps so siblings finish
+
+
+def test_catch_in_foreach(tmp_datastore):
+    """A caught foreach-child failure still reaches the join as a
+    FailureHandledByCatch artifact; the run completes."""
+    run_flow("catch_foreach_flow.py", tmp_datastore, "run")
+    run_id = latest_run_id(tmp_datastore, "CatchForeachFlow")
+    assert read_artifact(tmp_datastore, "CatchForeachFlow", run_id,
+                         "join", "total") == 20
+    fails = read_artifact(tmp_datastore, "CatchForeachFlow", run_id,
+                          "join", "failures")
+    assert len(fails) == 1 and "boom" in fails[0]
