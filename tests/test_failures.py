@@ -323,3 +323,43 @@ def test_catch_in_foreach(tmp_datastore):
     fails = read_artifact(tmp_datastore, "CatchForeachFlow", run_id,
                           "join", "failures")
     assert len(fails) == 1 and "boom" in fails[0]
+
+
+def test_timeout_decorator(tmp_datastore, tmp_path):
+    """@timeout(seconds=2) kills a hung step quickly; with @catch the
+    flow still completes and records the timeout."""
+    import subprocess
+    import sys
+    import time
+
+    flow = tmp_path / "timeout_flow.py"
+    flow.write_text(
+        "import time\n"
+        "from metaflow_amd import FlowSpec, catch, step, timeout\n"
+        "class TimeoutFlow(FlowSpec):\n"
+        "    @catch(var='err')\n"
+        "    @timeout(seconds=2)\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        time.sleep(60)\n"
+        "        self.next(self.end)\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        assert self.err\n"
+        "        self.caught = str(self.err)\n"
+        "if __name__ == '__main__':\n"
+        "    TimeoutFlow()\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    t0 = time.time()
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        capture_output=True, text=True, env=env, timeout=180)
+    assert proc.returncode == 0, proc.stderr[-1500:]
+    assert time.time() - t0 < 40, "timeout did not fire promptly"
+    run_id = latest_run_id(tmp_datastore, "TimeoutFlow")
+    caught = read_artifact(tmp_datastore, "TimeoutFlow", run_id, "end",
+                           "caught")
+    assert "imeout" in caught or "imed out" in caught, caught
