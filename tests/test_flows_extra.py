@@ -360,3 +360,10 @@ def test_example_train_llama_ddp(tmp_datastore):
          "--seq", "256", "--model-size", "tiny"],
         capture_output=True, text=True, env=env, timeout=420)
     assert proc.returncode == 0, proc.stderr[-2000:]
+
+
+def test_gang_gpu_pinning(tmp_datastore):
+    """@resources(gpu=2) on a 2-rank gang with 4 (simulated) GPUs pins
+    disjoint device pairs per rank."""
+    run_flow("gpu_pin_flow.py", tmp_datastore, "run",
+             env_extra={"MFX_NUM_GPUS": "4"})
