@@ -108,8 +108,9 @@ class DecoderLayer(nn.Module):
                 cache.pos
             kc[:, :, pos:pos + S] = k
             vc[:, :, pos:pos + S] = v
-            if pos == 0 and S % 64 == 0:
-                o = K.attention(q, k, v)      # aligned prefill: flash
+            if pos == 0:
+                o = K.attention(q, k, v)      # prefill: flash (padded
+                # to the 256 tile internally for unaligned prompts)
             else:
                 o = _attn_with_cache(q, kc[:, :, :pos + S],
                                      vc[:, :, :pos + S],

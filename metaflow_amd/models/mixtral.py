@@ -262,8 +262,9 @@ class MixtralDecoderLayer(nn.Module):
             kc, vc, pos = cache.k[layer_idx], cache.v[layer_idx], cache.pos
             kc[:, :, pos:pos + S] = kk
             vc[:, :, pos:pos + S] = v
-            if pos == 0 and S % 64 == 0:
-                o = K.attention(q, kk, v)
+            if pos == 0:
+                o = K.attention(q, kk, v)  # flash prefill (any S; the
+                # wrapper pads unaligned seqlens to the 256 tile)
             else:
                 import math
 

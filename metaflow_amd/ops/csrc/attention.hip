@@ -1,10 +1,9 @@
-// Flash attention (fwd + bwd) for gfx950, bf16, D=128, causal, GQA.
-//
-// MI355X-first design (no port): MFMA 16x16x32 bf16 tiles, 64-lane waves,
-// XOR-swizzled LDS tiles (guide §6 G4: row-major D=128 tiles are a 16/32-way
-// bank conflict on ds_read_b128; byte ^= (row&7)<<4 fixes it), online
-// softmax with in-register 16-lane-group row reductions (no serial-lane
-// softmax, guide common-mistake #6).
+// Shared MFMA-16x16 helpers + attention backward delta preprocess for
+// gfx950 (bf16, D=128). The flash fwd/bwd kernels themselves live in
+// attention_v2.hip (8-wave 32x32 swapped-QK^T structure); unaligned
+// seqlens are zero-padded to the 256 boundary by the Python wrapper
+// (exact for causal attention — end-padded keys are causally masked for
+// every real query row).
 //
 // MFMA fragment layout (guide §3, measured m89/m91):
 //   C/D: col = lane&15, row = (lane>>4)*4 + reg
@@ -12,8 +11,6 @@
 //   B (KxN): lane holds B[(lane>>4)*8 + i][lane&15]
 // so every A read wants M-major storage [m][k] and every B read wants
 // N-major storage [n][k]; tiles are staged in the layout each operand needs.
-//
-// Shapes: q [B,H,S,128], kv [B,Hkv,S,128], causal, S % 64 == 0.
 
 #include "common.h"
 
@@ -104,152 +101,6 @@ DEV void wave_lds_fence() {
   __builtin_amdgcn_wave_barrier();
 }
 
-// ============================ FORWARD =====================================
-// BQ=64, BKV=64, 256 threads (4 waves), each wave owns 16 q rows.
-
-template <int BLOCK>
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
-    const short* __restrict__ q, const short* __restrict__ k,
-    const short* __restrict__ v, short* __restrict__ o,
-    float* __restrict__ lse, int B, int H, int Hkv, int S, float scale) {
-  constexpr int BQ = 64, BKV = 64;
-  __shared__ short kt[BKV * ATT_D];     // swizzled [64][128] (N-major for B)
-  __shared__ short vtt[ATT_D * BKV];    // V^T swizzled [128][64]
-  __shared__ short pt[BQ * BKV];        // P swizzled [64][64]
-
-  const int qb = blockIdx.x;            // q tile index
-  const int h = blockIdx.y;
-  const int b = blockIdx.z;
-  const int hk = h / (H / Hkv);
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int lgrp = lane >> 4;           // 16-lane group 0..3
-  const int l16 = lane & 15;
-
-  const long long qoff = (((long long)b * H + h) * S + qb * BQ) * ATT_D;
-  const long long kvoff0 = ((long long)b * Hkv + hk) * S * ATT_D;
-
-  // per-wave per-lane state: 4 rows (reg r), replicated across 16 lanes
-  float m_run[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
-  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
-  f4 acc_o[8];
-#pragma unroll
-  for (int n = 0; n < 8; ++n) acc_o[n] = (f4){0, 0, 0, 0};
-
-  // preload Q A-fragments straight from global (one 16 B read per lane per
-  // k-step; keeping Q out of LDS doubles occupancy)
-  bf16x8 qf[4];
-#pragma unroll
-  for (int kk = 0; kk < 4; ++kk)
-    qf[kk] = *(const bf16x8*)(
-        q + qoff + (long long)(wid * 16 + l16) * ATT_D + kk * 32
-        + lgrp * 8);
-
-  const int kv_tiles = (qb * BQ) / BKV + 1;  // causal bound
-  for (int j = 0; j < kv_tiles; ++j) {
-    __syncthreads();
-    stage_rm<BKV, BLOCK>(kt, k + kvoff0 + (long long)j * BKV * ATT_D,
-                         ATT_D);
-    stage_tr<BKV, BLOCK>(vtt, v + kvoff0 + (long long)j * BKV * ATT_D,
-                         ATT_D);
-    __syncthreads();
-
-    // S strip: 16 q rows x 64 kv cols
-    f4 s[4];
-#pragma unroll
-    for (int n = 0; n < 4; ++n) {
-      s[n] = (f4){0, 0, 0, 0};
-#pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
-        bf16x8 bf = frag8(kt, n * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
-        s[n] = mfma16(qf[kk], bf, s[n]);
-      }
-    }
-
-    // scale + causal mask (diagonal tile only; j < qb is fully unmasked)
-    const bool diag = (j == kv_tiles - 1);
-#pragma unroll
-    for (int n = 0; n < 4; ++n) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float sv = s[n][r] * scale;
-        if (diag) {
-          int qrow = qb * BQ + wid * 16 + lgrp * 4 + r;
-          int kcol = j * BKV + n * 16 + l16;
-          if (kcol > qrow) sv = -INFINITY;
-        }
-        s[n][r] = sv;
-      }
-    }
-
-    // online softmax update (per reg row)
-    float pmax[4], rescale[4], newm[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
-      mx = group_max(mx);
-      newm[r] = fmaxf(m_run[r], mx);
-      rescale[r] = __expf(m_run[r] - newm[r]);
-      pmax[r] = mx;
-      (void)pmax;
-    }
-    float psum[4] = {0, 0, 0, 0};
-#pragma unroll
-    for (int n = 0; n < 4; ++n) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float p = __expf(s[n][r] - newm[r]);
-        s[n][r] = p;
-        psum[r] += p;
-      }
-    }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      l_run[r] = l_run[r] * rescale[r] + group_sum(psum[r]);
-      m_run[r] = newm[r];
-    }
-    // rescale O
-#pragma unroll
-    for (int n = 0; n < 8; ++n)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) acc_o[n][r] *= rescale[r];
-
-    // write P strip to LDS (bf16) and feed PV MFMAs (per-wave region)
-#pragma unroll
-    for (int n = 0; n < 4; ++n)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        *(short*)((char*)pt + swz(wid * 16 + lgrp * 4 + r, n * 16 + l16,
-                                  BKV * 2)) = f2bf(s[n][r]);
-    wave_lds_fence();
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 pa = frag8(pt, wid * 16 + l16, kk * 32 + lgrp * 8, BKV * 2);
-#pragma unroll
-      for (int n = 0; n < 8; ++n) {
-        bf16x8 bv = frag8(vtt, n * 16 + l16, kk * 32 + lgrp * 8, BKV * 2);
-        acc_o[n] = mfma16(pa, bv, acc_o[n]);
-      }
-    }
-  }
-
-  // epilogue: O /= l, write bf16 + lse
-  const long long ooff = qoff;
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qrow = wid * 16 + lgrp * 4 + r;
-    const float inv_l = 1.f / l_run[r];
-#pragma unroll
-    for (int n = 0; n < 8; ++n)
-      o[ooff + (long long)qrow * ATT_D + n * 16 + l16] =
-          f2bf(acc_o[n][r] * inv_l);
-    if (l16 == 0) {
-      lse[(((long long)b * H + h) * S) + qb * BQ + qrow] =
-          m_run[r] + __logf(l_run[r]);
-    }
-  }
-}
-
 // ===================== BACKWARD: delta preprocess ==========================
 // delta[row] = rowsum(dO[row] * O[row]); one wave per row, 2 elems/lane.
 __global__ void attn_bwd_delta_kernel(const short* __restrict__ dout,
@@ -266,285 +117,4 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dout,
             + bf2f(dr[lane + 64]) * bf2f(orow[lane + 64]);
   s = wave_sum(s);
   if (lane == 0) delta[row] = s;
-}
-
-// ========================= BACKWARD: dK/dV ================================
-// Grid (S/BKV, Hkv, B); block 256 = 4 waves, each owns 16 kv rows.
-// Loops over the G q-heads of this kv head and q tiles of BQ2=32.
-
-template <int BLOCK>
-__global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
-    const short* __restrict__ q, const short* __restrict__ k,
-    const short* __restrict__ v, const short* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dk, short* __restrict__ dv,
-    int B, int H, int Hkv, int S, float scale) {
-  constexpr int BKV = 64, BQ2 = 32;
-  __shared__ short qtile[BQ2 * ATT_D];    // Q rm (B for S^T)
-  __shared__ short qtt[ATT_D * BQ2];      // Q^T (B for dK)
-  __shared__ short dot[BQ2 * ATT_D];      // dO rm (B for dP^T)
-  __shared__ short dott[ATT_D * BQ2];     // dO^T (B for dV)
-  __shared__ short ptile[BKV * BQ2];      // P^T / dS^T round-trip
-  __shared__ float lse_s[BQ2];
-  __shared__ float del_s[BQ2];
-
-  const int kvb = blockIdx.x;
-  const int hk = blockIdx.y;
-  const int b = blockIdx.z;
-  const int G = H / Hkv;
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int lgrp = lane >> 4;
-  const int l16 = lane & 15;
-
-  const long long kvoff =
-      (((long long)b * Hkv + hk) * S + kvb * BKV) * ATT_D;
-
-  f4 acc_dk[8], acc_dv[8];
-#pragma unroll
-  for (int n = 0; n < 8; ++n) {
-    acc_dk[n] = (f4){0, 0, 0, 0};
-    acc_dv[n] = (f4){0, 0, 0, 0};
-  }
-
-  // preload K/V A-fragments straight from global (each lane 16 B of its
-  // strip row; one-time cost, saves 32 KiB of LDS)
-  bf16x8 kf[4], vf[4];
-#pragma unroll
-  for (int kk = 0; kk < 4; ++kk) {
-    const long long row_off =
-        kvoff + (long long)(wid * 16 + l16) * ATT_D + kk * 32 + lgrp * 8;
-    kf[kk] = *(const bf16x8*)(k + row_off);
-    vf[kk] = *(const bf16x8*)(v + row_off);
-  }
-
-  const int jq0 = (kvb * BKV) / BQ2;
-  const int nq = S / BQ2;
-  for (int g = 0; g < G; ++g) {
-    const int h = hk * G + g;
-    const long long hoff = ((long long)b * H + h) * S;
-    for (int jq = jq0; jq < nq; ++jq) {
-      __syncthreads();
-      const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
-      const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
-      stage_rm<BQ2, BLOCK>(qtile, qsrc, ATT_D);
-      stage_tr<BQ2, BLOCK>(qtt, qsrc, ATT_D);
-      stage_rm<BQ2, BLOCK>(dot, dsrc, ATT_D);
-      stage_tr<BQ2, BLOCK>(dott, dsrc, ATT_D);
-      if (threadIdx.x < BQ2) {
-        lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
-        del_s[threadIdx.x] = delta[hoff + jq * BQ2 + threadIdx.x];
-      }
-      __syncthreads();
-
-      // S^T strip (16 kv rows x 32 q cols) and dP^T
-      f4 st[2], dpt[2];
-#pragma unroll
-      for (int n = 0; n < 2; ++n) {
-        st[n] = (f4){0, 0, 0, 0};
-        dpt[n] = (f4){0, 0, 0, 0};
-#pragma unroll
-        for (int kk = 0; kk < 4; ++kk) {
-          bf16x8 qb8 = frag8(qtile, n * 16 + l16, kk * 32 + lgrp * 8,
-                             ATT_D * 2);
-          bf16x8 db8 = frag8(dot, n * 16 + l16, kk * 32 + lgrp * 8,
-                             ATT_D * 2);
-          st[n] = mfma16(kf[kk], qb8, st[n]);
-          dpt[n] = mfma16(vf[kk], db8, dpt[n]);
-        }
-      }
-
-      // P^T = exp(S^T*scale - lse[q]), causal mask q < kv
-#pragma unroll
-      for (int n = 0; n < 2; ++n) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kvrow = kvb * BKV + wid * 16 + lgrp * 4 + r;
-          const int qcol = jq * BQ2 + n * 16 + l16;
-          float pv = 0.f;
-          if (qcol >= kvrow)
-            pv = __expf(st[n][r] * scale - lse_s[n * 16 + l16]);
-          st[n][r] = pv;  // now P^T
-        }
-      }
-
-      // dV += P^T @ dO : stage P^T strip, read A frags
-#pragma unroll
-      for (int n = 0; n < 2; ++n)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
-                                       n * 16 + l16, BQ2 * 2)) =
-              f2bf(st[n][r]);
-      wave_lds_fence();
-      {
-        bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BQ2 * 2);
-#pragma unroll
-        for (int n = 0; n < 8; ++n) {
-          bf16x8 db8 = frag8(dott, n * 16 + l16, lgrp * 8, BQ2 * 2);
-          acc_dv[n] = mfma16(pa, db8, acc_dv[n]);
-        }
-      }
-
-      // dS^T = P^T * (dP^T - delta[q]) * scale
-#pragma unroll
-      for (int n = 0; n < 2; ++n)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          st[n][r] = st[n][r] * (dpt[n][r] - del_s[n * 16 + l16]) * scale;
-
-      // dK += dS^T @ Q : round-trip again
-#pragma unroll
-      for (int n = 0; n < 2; ++n)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
-                                       n * 16 + l16, BQ2 * 2)) =
-              f2bf(st[n][r]);
-      wave_lds_fence();
-      {
-        bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BQ2 * 2);
-#pragma unroll
-        for (int n = 0; n < 8; ++n) {
-          bf16x8 qb8 = frag8(qtt, n * 16 + l16, lgrp * 8, BQ2 * 2);
-          acc_dk[n] = mfma16(pa, qb8, acc_dk[n]);
-        }
-      }
-    }
-  }
-
-  // write dK/dV strips
-  const long long dkoff = kvoff;
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int kvrow = wid * 16 + lgrp * 4 + r;
-#pragma unroll
-    for (int n = 0; n < 8; ++n) {
-      dk[dkoff + (long long)kvrow * ATT_D + n * 16 + l16] =
-          f2bf(acc_dk[n][r]);
-      dv[dkoff + (long long)kvrow * ATT_D + n * 16 + l16] =
-          f2bf(acc_dv[n][r]);
-    }
-  }
-}
-
-// =========================== BACKWARD: dQ =================================
-// Grid (S/BQ, H, B); block 256 = 4 waves each owning 16 q rows; kv tiles
-// of 32.
-
-template <int BLOCK>
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
-    const short* __restrict__ q, const short* __restrict__ k,
-    const short* __restrict__ v, const short* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq,
-    int B, int H, int Hkv, int S, float scale) {
-  constexpr int BQ = 64, BKV2 = 32;
-  __shared__ short kt[BKV2 * ATT_D];     // K rm (B for S)
-  __shared__ short ktt[ATT_D * BKV2];    // K^T (B for dQ)
-  __shared__ short vt[BKV2 * ATT_D];     // V rm (B for dP)
-  __shared__ short ptile[BQ * BKV2];     // dS round-trip
-  __shared__ float lse_s[BQ];
-  __shared__ float del_s[BQ];
-
-  const int qb = blockIdx.x;
-  const int h = blockIdx.y;
-  const int b = blockIdx.z;
-  const int hk = h / (H / Hkv);
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int lgrp = lane >> 4;
-  const int l16 = lane & 15;
-
-  const long long hoff = ((long long)b * H + h) * S;
-  const long long qoff = (hoff + qb * BQ) * ATT_D;
-  const long long kvoff0 = ((long long)b * Hkv + hk) * S * ATT_D;
-
-  if (threadIdx.x < BQ) {
-    lse_s[threadIdx.x] = lse[hoff + qb * BQ + threadIdx.x];
-    del_s[threadIdx.x] = delta[hoff + qb * BQ + threadIdx.x];
-  }
-
-  f4 acc_dq[8];
-#pragma unroll
-  for (int n = 0; n < 8; ++n) acc_dq[n] = (f4){0, 0, 0, 0};
-  __syncthreads();
-
-  // preload Q/dO A-fragments straight from global (keeps 32 KiB of LDS
-  // free; one-time cost per block)
-  bf16x8 qf[4], df[4];
-#pragma unroll
-  for (int kk = 0; kk < 4; ++kk) {
-    const long long row_off =
-        qoff + (long long)(wid * 16 + l16) * ATT_D + kk * 32 + lgrp * 8;
-    qf[kk] = *(const bf16x8*)(q + row_off);
-    df[kk] = *(const bf16x8*)(dout + row_off);
-  }
-
-  const int nkv = (qb + 1) * BQ / BKV2;  // causal bound
-  for (int j = 0; j < nkv; ++j) {
-    __syncthreads();
-    const short* ksrc = k + kvoff0 + (long long)j * BKV2 * ATT_D;
-    stage_rm<BKV2, BLOCK>(kt, ksrc, ATT_D);
-    stage_tr<BKV2, BLOCK>(ktt, ksrc, ATT_D);
-    stage_rm<BKV2, BLOCK>(vt, v + kvoff0 + (long long)j * BKV2 * ATT_D,
-                          ATT_D);
-    __syncthreads();
-
-    f4 s[2], dp[2];
-#pragma unroll
-    for (int n = 0; n < 2; ++n) {
-      s[n] = (f4){0, 0, 0, 0};
-      dp[n] = (f4){0, 0, 0, 0};
-#pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
-        bf16x8 kb8 = frag8(kt, n * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
-        bf16x8 vb8 = frag8(vt, n * 16 + l16, kk * 32 + lgrp * 8, ATT_D * 2);
-        s[n] = mfma16(qf[kk], kb8, s[n]);
-        dp[n] = mfma16(df[kk], vb8, dp[n]);
-      }
-    }
-
-    // dS = P * (dP - delta[q]) * scale, P = exp(S*scale - lse[q row])
-#pragma unroll
-    for (int n = 0; n < 2; ++n) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrl = wid * 16 + lgrp * 4 + r;   // local q row
-        const int qrow = qb * BQ + qrl;
-        const int kcol = j * BKV2 + n * 16 + l16;
-        float p = 0.f;
-        if (kcol <= qrow) p = __expf(s[n][r] * scale - lse_s[qrl]);
-        s[n][r] = p * (dp[n][r] - del_s[qrl]) * scale;
-      }
-    }
-
-    // dQ += dS @ K : round-trip dS, B from K^T
-#pragma unroll
-    for (int n = 0; n < 2; ++n)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        *(short*)((char*)ptile + swz(wid * 16 + lgrp * 4 + r,
-                                     n * 16 + l16, BKV2 * 2)) =
-            f2bf(s[n][r]);
-    wave_lds_fence();
-    {
-      bf16x8 pa = frag8(ptile, wid * 16 + l16, lgrp * 8, BKV2 * 2);
-#pragma unroll
-      for (int n = 0; n < 8; ++n) {
-        bf16x8 kb8 = frag8(ktt, n * 16 + l16, lgrp * 8, BKV2 * 2);
-        acc_dq[n] = mfma16(pa, kb8, acc_dq[n]);
-      }
-    }
-  }
-
-  // write dQ
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qrow = wid * 16 + lgrp * 4 + r;
-#pragma unroll
-    for (int n = 0; n < 8; ++n)
-      dq[qoff + (long long)qrow * ATT_D + n * 16 + l16] =
-          f2bf(acc_dq[n][r]);
-  }
 }

@@ -287,24 +287,17 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   const int Hkv = k.size(1);
   TORCH_CHECK(D == 128, "attention requires head dim 128");
-  TORCH_CHECK(S % 64 == 0, "attention requires seqlen % 64 == 0");
+  // the Python wrapper pads unaligned seqlens to the 256 boundary
+  // (zero end-padding is exact for causal attention)
+  TORCH_CHECK(S % 256 == 0, "attention kernel requires seqlen % 256 == 0");
   TORCH_CHECK(H % Hkv == 0, "GQA requires H % Hkv == 0");
-  TORCH_CHECK(causal || S % 256 == 0,
-              "non-causal attention requires seqlen % 256 == 0");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
-  if (S % 256 == 0) {
-    // v2: 8-wave 32x32 swapped-QK^T structure
-    dim3 grid(S / 256, H, B);
-    attn_fwd_v2_kernel<512><<<grid, 512, 0, cur_stream()>>>(
-        bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
-        (float)scale, causal ? 1 : 0);
-  } else {
-    dim3 grid(S / 64, H, B);
-    attn_fwd_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
-        bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
-        (float)scale);
-  }
+  // v2: 8-wave 32x32 swapped-QK^T structure
+  dim3 grid(S / 256, H, B);
+  attn_fwd_v2_kernel<512><<<grid, 512, 0, cur_stream()>>>(
+      bf(q), bf(k), bf(v), bfm(o), lse.data_ptr<float>(), B, H, Hkv, S,
+      (float)scale, causal ? 1 : 0);
   HIP_CHECK_KERNEL();
   return {o, lse};
 }
@@ -328,38 +321,19 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
         bf(dout), bf(o), delta.data_ptr<float>(), rows);
   HIP_CHECK_KERNEL();
   }
-  TORCH_CHECK(causal || S % 256 == 0,
-              "non-causal attention requires seqlen % 256 == 0");
-  if (S % 256 == 0) {
-    dim3 gkv(S / 256, Hkv, B);
-    attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
-        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-        delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
-        (float)scale, causal ? 1 : 0);
-    HIP_CHECK_KERNEL();
-    dim3 gq(S / 256, H, B);
-    attn_bwd_dq_v2_kernel<512><<<gq, 512, 0, cur_stream()>>>(
-        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-        delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale,
-        causal ? 1 : 0);
-    HIP_CHECK_KERNEL();
-    return {dq, dk, dv};
-  }
-  {
-    dim3 grid(S / 64, Hkv, B);
-    attn_bwd_dkdv_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
-        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-        delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
-        (float)scale);
+  TORCH_CHECK(S % 256 == 0, "attention kernel requires seqlen % 256 == 0");
+  dim3 gkv(S / 256, Hkv, B);
+  attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
+      bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+      delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
+      (float)scale, causal ? 1 : 0);
   HIP_CHECK_KERNEL();
-  }
-  {
-    dim3 grid(S / 64, H, B);
-    attn_bwd_dq_kernel<kBlock><<<grid, kBlock, 0, cur_stream()>>>(
-        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-        delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale);
+  dim3 gq(S / 256, H, B);
+  attn_bwd_dq_v2_kernel<512><<<gq, 512, 0, cur_stream()>>>(
+      bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+      delta.data_ptr<float>(), bfm(dq), B, H, Hkv, S, (float)scale,
+      causal ? 1 : 0);
   HIP_CHECK_KERNEL();
-  }
   return {dq, dk, dv};
 }
 

@@ -288,20 +288,65 @@ def rope_qkv(qkv, cos_t, sin_t, nq, nkv):
 
 
 # ============================== attention ==================================
+_ATTN_ALIGN = 256  # v2 kernel tile: S must be a multiple of 256
+
+
+def _pad_seq(t, s_pad):
+    """Zero-pad [B,H,S,D] along dim 2 to s_pad rows."""
+    B, H, S, D = t.shape
+    if S == s_pad:
+        return t.contiguous()
+    out = t.new_zeros((B, H, s_pad, D))
+    out[:, :, :S] = t
+    return out
+
+
+def _attn_pad_len(S, causal):
+    """Padded seqlen for the kernel, or S unchanged if already aligned.
+
+    Zero END-padding is exact for causal attention: every padded key sits
+    at a position >= the real seqlen, so no real query row can attend it;
+    padded query rows produce garbage that the wrappers slice off (their
+    lse stays finite — zero scores — so 0*exp never makes a NaN in bwd).
+    Non-causal attention WOULD attend padded keys, so there we keep the
+    alignment requirement and fail loudly.
+    """
+    if S % _ATTN_ALIGN == 0:
+        return S
+    if not causal:
+        raise ValueError(
+            "non-causal attention requires seqlen %% %d == 0 (got %d); "
+            "pad the inputs (zero-padding keys is only exact for causal)"
+            % (_ATTN_ALIGN, S))
+    return (S + _ATTN_ALIGN - 1) // _ATTN_ALIGN * _ATTN_ALIGN
+
+
 class _Attention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, causal):
-        o, lse = hip_ext().attn_fwd(q, k, v, scale, causal)
-        ctx.save_for_backward(q, k, v, o, lse)
+        S = q.size(2)
+        Sp = _attn_pad_len(S, causal)
+        qp = _pad_seq(q, Sp)
+        kp = _pad_seq(k, Sp)
+        vp = _pad_seq(v, Sp)
+        o, lse = hip_ext().attn_fwd(qp, kp, vp, scale, causal)
+        ctx.save_for_backward(qp, kp, vp, o, lse)
         ctx.scale = scale
         ctx.causal = causal
-        return o
+        ctx.real_s = S
+        return o[:, :, :S].contiguous() if Sp != S else o
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = hip_ext().attn_bwd(q, k, v, o, dout.contiguous(), lse,
+        S, Sp = ctx.real_s, q.size(2)
+        dp = _pad_seq(dout, Sp)
+        dq, dk, dv = hip_ext().attn_bwd(q, k, v, o, dp, lse,
                                         ctx.scale, ctx.causal)
+        if Sp != S:
+            dq = dq[:, :, :S].contiguous()
+            dk = dk[:, :, :S].contiguous()
+            dv = dv[:, :, :S].contiguous()
         return dq, dk, dv, None, None
 
 
@@ -346,8 +391,13 @@ def attn_fwd_raw(q, k, v, scale, causal=True):
     on CPU.
     """
     if q.is_cuda:
-        o, lse = hip_ext().attn_fwd(q.contiguous(), k.contiguous(),
-                                    v.contiguous(), scale, causal)
+        S = q.size(2)
+        Sp = _attn_pad_len(S, causal)
+        o, lse = hip_ext().attn_fwd(_pad_seq(q, Sp), _pad_seq(k, Sp),
+                                    _pad_seq(v, Sp), scale, causal)
+        if Sp != S:
+            o = o[:, :, :S].contiguous()
+            lse = lse[:, :, :S].contiguous()
         return o, lse
     s = _ref_scores(q, k, scale, causal)
     lse = torch.logsumexp(s, dim=-1)
@@ -368,9 +418,23 @@ def attn_bwd_raw(q, k, v, o, dout, lse, scale, causal=True):
     ds = p * (dp - delta) * scale).
     """
     if q.is_cuda:
-        return hip_ext().attn_bwd(q.contiguous(), k.contiguous(),
-                                  v.contiguous(), o.contiguous(),
-                                  dout.contiguous(), lse, scale, causal)
+        S = q.size(2)
+        Sp = _attn_pad_len(S, causal)
+        if Sp == S:
+            return hip_ext().attn_bwd(q.contiguous(), k.contiguous(),
+                                      v.contiguous(), o.contiguous(),
+                                      dout.contiguous(), lse, scale, causal)
+        # padded rows: dout rows are zero so ds == 0 there; padded lse
+        # rows must be finite for exp() — zero q/k gives lse=log(n), but
+        # here lse came from the CALLER (global ring statistic) so pad
+        # with zeros explicitly (p = exp(0-0) is finite, ds still 0).
+        lse_p = lse.new_zeros((lse.size(0), lse.size(1), Sp))
+        lse_p[:, :, :S] = lse
+        dq, dk, dv = hip_ext().attn_bwd(
+            _pad_seq(q, Sp), _pad_seq(k, Sp), _pad_seq(v, Sp),
+            _pad_seq(o, Sp), _pad_seq(dout, Sp), lse_p, scale, causal)
+        return (dq[:, :, :S].contiguous(), dk[:, :, :S].contiguous(),
+                dv[:, :, :S].contiguous())
     G = q.size(1) // k.size(1)
     Hkv = k.size(1)
     s = _ref_scores(q, k, scale, causal)

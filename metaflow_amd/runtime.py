@@ -175,6 +175,19 @@ class NativeRuntime(object):
         self.param_values = param_values or {}
         self.clone_run_id = clone_run_id
         self.steps_to_rerun = set(steps_to_rerun or ())
+        # resume --step-to-rerun reruns the named steps AND everything
+        # downstream (reference runtime.py:415-419): a successor cloned
+        # from the origin run would carry artifacts computed from the
+        # PRE-rerun outputs — stale data. BFS over the static graph; the
+        # seen-check also terminates recursive-switch cycles.
+        if self.steps_to_rerun:
+            frontier = deque(self.steps_to_rerun)
+            while frontier:
+                node = self.graph[frontier.popleft()]
+                for nxt in node.out_funcs:
+                    if nxt not in self.steps_to_rerun:
+                        self.steps_to_rerun.add(nxt)
+                        frontier.append(nxt)
         self.max_workers = max_workers
         self.max_num_splits = max_num_splits
         self.quiet = quiet
@@ -193,6 +206,7 @@ class NativeRuntime(object):
         self._failed = False
         self._failure_msg = None
         self._origin_index = None
+        self._cloned_paths = set()  # new-run pathspecs produced by cloning
         self._selector = selectors.DefaultSelector()
         self._params_pathspec = None
 
@@ -248,6 +262,15 @@ class NativeRuntime(object):
         origin_ds = self._origin_index.get(spec.key)
         if origin_ds is None:
             return None
+        # Only clone when every input task was itself cloned: a freshly
+        # re-executed input means the origin task's artifacts were
+        # computed from data that no longer matches (reference
+        # runtime.py:474-479 requires all input tasks cloned).
+        for p in spec.input_paths or ():
+            if p.split("/")[1] == PARAMETERS_STEP:
+                continue
+            if p not in self._cloned_paths:
+                return None
         task_id = spec.task_id or self._new_task_id()
         spec.task_id = task_id
         new_ds = self.flow_datastore.get_task_datastore(
@@ -337,6 +360,7 @@ class NativeRuntime(object):
         cloned = self._maybe_clone(spec)
         if cloned is not None:
             # treat as finished immediately
+            self._cloned_paths.add(cloned.pathspec.split("/", 1)[1])
             self._task_finished_bookkeeping(spec, cloned)
             return
         self._run_queue.append(spec)

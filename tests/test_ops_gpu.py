@@ -133,6 +133,9 @@ def test_adamw(dev):
     (1, 2, 2, 256),     # MHA
     (2, 4, 1, 256),     # GQA 4:1
     (1, 8, 2, 512),     # GQA 4:2
+    (1, 4, 2, 257),     # unaligned: padded to 512 internally
+    (2, 4, 1, 100),     # unaligned, S < one tile
+    (1, 4, 4, 320),     # 64-aligned but not 256-aligned
 ])
 def test_attention_fwd(dev, B, H, Hkv, S):
     from metaflow_amd.ops import kernels as K
@@ -146,11 +149,12 @@ def test_attention_fwd(dev, B, H, Hkv, S):
     assert rel_err(o, o_ref) < 2e-2, "fwd mismatch"
 
 
-def test_attention_bwd(dev):
+@pytest.mark.parametrize("S", [256, 257, 100])
+def test_attention_bwd(dev, S):
     from metaflow_amd.ops import kernels as K
 
     torch.manual_seed(0)
-    B, H, Hkv, S = 1, 4, 2, 256
+    B, H, Hkv = 1, 4, 2
     q = torch.randn(B, H, S, 128, dtype=torch.bfloat16, device=dev,
                     requires_grad=True)
     k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev,
@@ -277,10 +281,12 @@ def test_mixtral_tiny_step(dev):
     assert losses[-1] < losses[0], "loss did not decrease: %s" % losses
 
 
-def test_generate_gpu(dev):
-    """KV-cache decode on hardware: flash-kernel prefill (64-aligned
-    prompt) + cached decode; greedy tokens must match the uncached full
-    forward."""
+@pytest.mark.parametrize("prompt_len", [256, 200])
+def test_generate_gpu(dev, prompt_len):
+    """KV-cache decode on hardware: flash-kernel prefill (any prompt
+    length — unaligned seqlens are padded to the 256-tile internally) +
+    cached decode; greedy tokens must match the uncached full forward
+    (which itself exercises the padded attention path at every t)."""
     import torch
 
     from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
@@ -289,11 +295,11 @@ def test_generate_gpu(dev):
     cfg = LlamaConfig.tiny(vocab=256, seq=512)
     m = LlamaForCausalLM(cfg).to(dev).eval()
     torch.manual_seed(1)
-    prompt = torch.randint(0, 256, (1, 256), device=dev)
+    prompt = torch.randint(0, 256, (1, prompt_len), device=dev)
     out = m.generate(prompt, 4)
-    assert out.shape == (1, 260)
+    assert out.shape == (1, prompt_len + 4)
     with torch.no_grad():
-        for t in range(256, 260):
+        for t in range(prompt_len, prompt_len + 4):
             full = m(out[:, :t])
             nxt = full[:, -1].float().argmax(-1)
             assert torch.equal(nxt, out[:, t]), t
