@@ -139,10 +139,31 @@ class FlowSpec(object, metaclass=FlowSpecMeta):
         return value
 
     def foreach_stack(self):
-        """[(index, num_splits, input_value_or_None)] per nesting level."""
+        """[(index, num_splits, input_value_or_None)] per nesting level.
+
+        The third element is the resolved INPUT VALUE at that nesting
+        level (reference flowspec.py:654-684) — the element of the
+        foreach sequence this branch processes — not the variable name.
+        None when the value cannot be resolved (e.g. the sequence
+        artifact is unavailable, or a num_parallel gang frame).
+        """
         out = []
         for frame in self._foreach_stack:
-            out.append((frame.index, frame.num_splits, frame.var))
+            value = None
+            if frame.var is not None:
+                try:
+                    seq = getattr(self, frame.var)
+                    from .unbounded_foreach import UnboundedForeachInput
+
+                    if isinstance(seq, UnboundedForeachInput):
+                        value = seq[frame.index]
+                    else:
+                        value = (seq[frame.index]
+                                 if hasattr(seq, "__getitem__")
+                                 else list(seq)[frame.index])
+                except Exception:
+                    value = None
+            out.append((frame.index, frame.num_splits, value))
         return out
 
     # ------------------------------------------------------------ transitions

@@ -52,6 +52,12 @@ def _tensor_to_buffer(t):
     pin_np = pin.numpy()
     if _side_stream is None:
         _side_stream = torch.cuda.Stream()
+    # order the side-stream copies after every kernel the producing
+    # stream still has in flight (e.g. the fused AdamW of the step being
+    # checkpointed) — without this the D2H can capture stale bytes. The
+    # .contiguous() above also ran on the current stream, so one
+    # wait_stream covers both.
+    _side_stream.wait_stream(torch.cuda.current_stream())
     offset = 0
     with torch.cuda.stream(_side_stream):
         while offset < nbytes:

@@ -24,6 +24,26 @@ class CatchDecorator(StepDecorator):
     name = "catch"
     defaults = {"var": None, "print_exception": True}
 
+    def step_init(self, flow, graph, step_name, decorators, datastore_type,
+                  logger):
+        # A swallowed exception means the step never called self.next(),
+        # so the runtime must synthesize the transition from the static
+        # graph. That is only well-defined for linear/static-split/join
+        # shapes: a foreach split has no split size, a switch picks one
+        # branch at runtime, and a parallel split has no gang width.
+        # The reference likewise rejects @catch on foreach splits
+        # (reference catch_decorator.py:45-52).
+        node = graph[step_name]
+        if node.type in ("foreach", "split-switch", "split-parallel"):
+            from ..exceptions import GraphException
+
+            raise GraphException(
+                "@catch is not supported on step '%s': a %s step that "
+                "fails cannot choose its transition (the split size / "
+                "branch / gang width is computed inside the step). Move "
+                "@catch into the child steps instead."
+                % (step_name, node.type))
+
     def task_exception(self, exception, step_name, flow, graph, retry_count,
                        max_user_code_retries):
         # only swallow on the final attempt; earlier attempts should retry

@@ -363,3 +363,65 @@ def test_timeout_decorator(tmp_datastore, tmp_path):
     caught = read_artifact(tmp_datastore, "TimeoutFlow", run_id, "end",
                            "caught")
     assert "imeout" in caught or "imed out" in caught, caught
+
+
+def test_resume_step_to_rerun_transitive(tmp_datastore, tmp_path,
+                                         monkeypatch):
+    """resume --step-to-rerun middle must also rerun downstream steps:
+    cloning `end` from the origin would resurrect artifacts computed
+    from the PRE-rerun middle output (advisor finding r1 #1; reference
+    runtime.py:415-419 expands steps_to_rerun transitively)."""
+    counter_dir = tmp_path / "counters"
+    counter_dir.mkdir()
+    monkeypatch.setenv("RESUME_COUNTER_DIR", str(counter_dir))
+    monkeypatch.setenv("RESUME_FAIL", "0")
+    run_flow("resume_flow.py", tmp_datastore, "run")
+    orig_run = latest_run_id(tmp_datastore, "ResumeFlow")
+
+    run_flow("resume_flow.py", tmp_datastore, "resume",
+             "--origin-run-id", orig_run, "--step-to-rerun", "middle")
+    new_run = latest_run_id(tmp_datastore, "ResumeFlow")
+    assert new_run != orig_run
+    # start cloned (ran once); middle AND end re-ran (twice each)
+    assert int(open(counter_dir / "start").read()) == 1
+    assert int(open(counter_dir / "middle").read()) == 2
+    assert int(open(counter_dir / "end").read()) == 2
+    assert read_artifact(tmp_datastore, "ResumeFlow", new_run, "end",
+                         "z") == 11
+
+
+def test_catch_on_foreach_split_rejected(tmp_datastore, tmp_path):
+    """@catch on a foreach SPLIT step is rejected at graph init: a
+    swallowed failure there cannot synthesize a foreach transition
+    (no split size), so downstream join bookkeeping would corrupt
+    (advisor finding r1 #3; reference catch_decorator.py:45-52)."""
+    flow = tmp_path / "catch_split_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, catch, step\n\n"
+        "class CatchSplitFlow(FlowSpec):\n"
+        "    @catch(var='err')\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        self.items = [1, 2]\n"
+        "        self.next(self.work, foreach='items')\n\n"
+        "    @step\n"
+        "    def work(self):\n"
+        "        self.next(self.join)\n\n"
+        "    @step\n"
+        "    def join(self, inputs):\n"
+        "        self.next(self.end)\n\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n\n"
+        "if __name__ == '__main__':\n"
+        "    CatchSplitFlow()\n")
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--datastore-root", tmp_datastore,
+         "run"], env=env, capture_output=True, text=True, timeout=120)
+    assert proc.returncode != 0
+    assert "not supported" in proc.stderr + proc.stdout

@@ -152,3 +152,50 @@ def test_logs_and_dot_cli(tmp_datastore, tmp_path):
         capture_output=True, text=True, env=env, timeout=120)
     assert dot.returncode == 0
     assert "digraph" in dot.stdout and "start" in dot.stdout
+
+
+def test_foreach_stack_values(tmp_datastore, tmp_path):
+    """foreach_stack() returns the resolved input VALUE per nesting
+    level (reference flowspec.py:654-684), not the variable name
+    (advisor finding r1 #4)."""
+    import subprocess
+    import sys
+
+    flow = tmp_path / "fstack_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, step\n\n"
+        "class FStackFlow(FlowSpec):\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        self.letters = ['a', 'b']\n"
+        "        self.next(self.outer, foreach='letters')\n\n"
+        "    @step\n"
+        "    def outer(self):\n"
+        "        self.digits = [10, 20, 30]\n"
+        "        self.next(self.inner, foreach='digits')\n\n"
+        "    @step\n"
+        "    def inner(self):\n"
+        "        stack = self.foreach_stack()\n"
+        "        assert len(stack) == 2, stack\n"
+        "        (i0, n0, v0), (i1, n1, v1) = stack\n"
+        "        assert n0 == 2 and v0 == ['a', 'b'][i0], stack\n"
+        "        assert n1 == 3 and v1 == [10, 20, 30][i1], stack\n"
+        "        assert v1 == self.input, stack\n"
+        "        self.next(self.join_inner)\n\n"
+        "    @step\n"
+        "    def join_inner(self, inputs):\n"
+        "        self.next(self.join_outer)\n\n"
+        "    @step\n"
+        "    def join_outer(self, inputs):\n"
+        "        self.next(self.end)\n\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        pass\n\n"
+        "if __name__ == '__main__':\n"
+        "    FStackFlow()\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--datastore-root", tmp_datastore,
+         "run"], env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-3000:]
