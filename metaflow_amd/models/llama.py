@@ -47,7 +47,9 @@ class LlamaConfig:
 
 
 class Linear(nn.Module):
-    """Bias-free bf16 linear -> hipBLASLt GEMM."""
+    """Bias-free bf16 linear -> hipBLASLt GEMM (offline-tuned solution
+    indices where the shape is in ops/gemm_table.json; heuristic
+    otherwise)."""
 
     def __init__(self, din, dout, dtype=torch.bfloat16):
         super().__init__()
@@ -55,7 +57,9 @@ class Linear(nn.Module):
             torch.empty(dout, din, dtype=dtype))
 
     def forward(self, x):
-        return torch.nn.functional.linear(x, self.weight)
+        from ..ops.gemm import tuned_linear
+
+        return tuned_linear(x, self.weight)
 
 
 class RMSNorm(nn.Module):

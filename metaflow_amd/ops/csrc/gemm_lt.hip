@@ -1,22 +1,32 @@
-// hipBLASLt algorithm search + pinned-algo GEMM for the Llama hot shapes.
+// hipBLASLt algorithm search + pinned-algo GEMMs for the train-step hot
+// shapes — forward AND both backward GEMMs per projection.
 //
-// The train-step profile (profiles/llama8b_1gpu_r01_final2_kernel_stats.txt)
-// shows hipBLASLt's heuristic picking MI16x16 tiles for ~58% of step time.
-// This extension enumerates EVERY solution hipBLASLt ships for a given
-// (M, N, K) bf16 TN GEMM (hipblaslt_ext::getAllAlgos), times each on the
-// real shape, and replays the winner by solution index — an offline
-// tuning pass without TunableOp's in-process fragility (NOTES_ROUND2.md).
+// The round-1 profile (profiles/llama8b_1gpu_r01_final2_kernel_stats.txt)
+// shows hipBLASLt's heuristic at ~58% of step time and ~49% of its own
+// peak on these shapes. This extension enumerates EVERY solution
+// hipBLASLt ships for a given problem (hipblaslt_ext::getAllAlgos), times
+// each on the real shape, and replays winners by solution index — an
+// offline tuning pass without TunableOp's in-process fragility
+// (NOTES_ROUND2.md). Tuned winners ship in the repo
+// (metaflow_amd/ops/gemm_table.json) and route via ops/gemm.py.
 //
-// Layout convention: torch F.linear semantics out[M,N] = x[M,K] @ w[N,K]^T,
-// all row-major.  hipBLASLt is column-major, so we compute
-// D_cm[N,M] = op_T(W_cm[K,N]) * op_N(X_cm[K,M])  — the TN GEMM the
-// Cijk_Alik_Bljk solutions in the profile implement.
+// Three problem modes (torch F.linear semantics, all tensors row-major):
+//   mode 0 (fwd): out[M,N] = x[M,K] @ w[N,K]^T        (TN)
+//   mode 1 (dx) : dx[M,K]  = dy[M,N] @ w[N,K]         (NN)
+//   mode 2 (dw) : dw[N,K]  = dy[M,N]^T @ x[M,K]       (NT)
+// hipBLASLt is column-major; each mode's layouts are derived below.
+//
+// Per-problem Layouts + algo are cached (the run path is called ~400x
+// per train step; desc creation and getAlgosFromIndex are host-side
+// overhead), and the 128 MiB workspace is a single static allocation.
 
 #include <hipblaslt/hipblaslt.h>
 #include <hipblaslt/hipblaslt-ext.hpp>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <mutex>
+#include <unordered_map>
 #include <vector>
 
 #define LT_CHECK(expr)                                                     \
@@ -34,6 +44,8 @@
 
 namespace {
 
+constexpr size_t kWorkspaceBytes = 128u << 20;
+
 hipblasLtHandle_t lt_handle() {
   static hipblasLtHandle_t handle = [] {
     hipblasLtHandle_t h;
@@ -43,89 +55,226 @@ hipblasLtHandle_t lt_handle() {
   return handle;
 }
 
-struct Layouts {
-  hipblasLtMatmulDesc_t op;
-  hipblasLtMatrixLayout_t a, b, c;
-  ~Layouts() {
-    hipblasLtMatrixLayoutDestroy(a);
-    hipblasLtMatrixLayoutDestroy(b);
-    hipblasLtMatrixLayoutDestroy(c);
-    hipblasLtMatmulDescDestroy(op);
-  }
-};
-
-// D[N,M]cm = W^T[N,K] * X[K,M]cm : A = W (K x N cm, OP_T), B = X (K x M cm)
-void make_layouts(Layouts& L, long long M, long long N, long long K) {
-  LT_CHECK(hipblasLtMatmulDescCreate(&L.op, HIPBLAS_COMPUTE_32F,
-                                     HIP_R_32F));
-  hipblasOperation_t t = HIPBLAS_OP_T, n = HIPBLAS_OP_N;
-  LT_CHECK(hipblasLtMatmulDescSetAttribute(
-      L.op, HIPBLASLT_MATMUL_DESC_TRANSA, &t, sizeof(t)));
-  LT_CHECK(hipblasLtMatmulDescSetAttribute(
-      L.op, HIPBLASLT_MATMUL_DESC_TRANSB, &n, sizeof(n)));
-  LT_CHECK(hipblasLtMatrixLayoutCreate(&L.a, HIP_R_16BF, K, N, K));
-  LT_CHECK(hipblasLtMatrixLayoutCreate(&L.b, HIP_R_16BF, K, M, K));
-  LT_CHECK(hipblasLtMatrixLayoutCreate(&L.c, HIP_R_16BF, N, M, N));
+void* lt_workspace() {
+  static void* ws = [] {
+    void* p = nullptr;
+    HIPRT_CHECK(hipMalloc(&p, kWorkspaceBytes));
+    return p;
+  }();
+  return ws;
 }
 
-void run_algo(const torch::Tensor& x, const torch::Tensor& w,
-              torch::Tensor& out, torch::Tensor& workspace,
+struct Layouts {
+  hipblasLtMatmulDesc_t op = nullptr;
+  hipblasLtMatrixLayout_t a = nullptr, b = nullptr, c = nullptr;
+  hipblasOperation_t opa, opb;
+};
+
+// Column-major layout derivation per mode. A/B below are the hipBLASLt
+// operands; the Python caller always passes (a=weight-side, b=data-side)
+// per run()/search() docstrings.
+//   mode 0: D_cm[N,M] = op_T(W_cm[K,N]) * op_N(X_cm[K,M])
+//   mode 1: D_cm[K,M] = op_N(W_cm[K,N]) * op_N(dY_cm[N,M])
+//   mode 2: D_cm[K,N] = op_N(X_cm[K,M]) * op_T(dY_cm[N,M])
+void make_layouts(Layouts& L, int mode, long long M, long long N,
+                  long long K) {
+  LT_CHECK(hipblasLtMatmulDescCreate(&L.op, HIPBLAS_COMPUTE_32F,
+                                     HIP_R_32F));
+  switch (mode) {
+    case 0:
+      L.opa = HIPBLAS_OP_T;
+      L.opb = HIPBLAS_OP_N;
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.a, HIP_R_16BF, K, N, K));
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.b, HIP_R_16BF, K, M, K));
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.c, HIP_R_16BF, N, M, N));
+      break;
+    case 1:
+      L.opa = HIPBLAS_OP_N;
+      L.opb = HIPBLAS_OP_N;
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.a, HIP_R_16BF, K, N, K));
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.b, HIP_R_16BF, N, M, N));
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.c, HIP_R_16BF, K, M, K));
+      break;
+    case 2:
+      L.opa = HIPBLAS_OP_N;
+      L.opb = HIPBLAS_OP_T;
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.a, HIP_R_16BF, K, M, K));
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.b, HIP_R_16BF, N, M, N));
+      LT_CHECK(hipblasLtMatrixLayoutCreate(&L.c, HIP_R_16BF, K, N, K));
+      break;
+    default:
+      TORCH_CHECK(false, "mode must be 0 (fwd), 1 (dx) or 2 (dw)");
+  }
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      L.op, HIPBLASLT_MATMUL_DESC_TRANSA, &L.opa, sizeof(L.opa)));
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      L.op, HIPBLASLT_MATMUL_DESC_TRANSB, &L.opb, sizeof(L.opb)));
+}
+
+void run_algo(const void* a, const void* b, void* d,
               hipblasLtMatmulAlgo_t* algo, Layouts& L) {
   float alpha = 1.f, beta = 0.f;
   LT_CHECK(hipblasLtMatmul(
-      lt_handle(), L.op, &alpha, w.data_ptr(), L.a, x.data_ptr(), L.b,
-      &beta, out.data_ptr(), L.c, out.data_ptr(), L.c, algo,
-      workspace.data_ptr(), (size_t)workspace.numel(),
+      lt_handle(), L.op, &alpha, a, L.a, b, L.b, &beta, d, L.c, d, L.c,
+      algo, lt_workspace(), kWorkspaceBytes,
       at::hip::getCurrentHIPStream()));
 }
 
-// Enumerate + time every supported solution. Returns (indices, ms) sorted
-// fastest-first.
-std::vector<torch::Tensor> gemm_lt_search(torch::Tensor x, torch::Tensor w,
-                                          long iters, long cap) {
-  TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 2 && w.dim() == 2);
-  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
-              w.scalar_type() == torch::kBFloat16);
-  x = x.contiguous();
-  w = w.contiguous();
-  const long long M = x.size(0), K = x.size(1), N = w.size(0);
-  TORCH_CHECK(w.size(1) == K, "w must be [N, K]");
-  auto out = torch::empty({M, N}, x.options());
-  auto workspace = torch::empty({128 << 20},
-                                x.options().dtype(torch::kUInt8));
+// shapes per mode: returns (out_rows_rm, out_cols_rm) and checks operands
+std::pair<long long, long long> problem_dims(int mode,
+                                             const torch::Tensor& a,
+                                             const torch::Tensor& b,
+                                             long long& M, long long& N,
+                                             long long& K) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2);
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16);
+  switch (mode) {
+    case 0:  // a=w [N,K], b=x [M,K] -> out [M,N]
+      N = a.size(0); K = a.size(1); M = b.size(0);
+      TORCH_CHECK(b.size(1) == K, "x must be [M,K] matching w [N,K]");
+      return {M, N};
+    case 1:  // a=w [N,K], b=dy [M,N] -> dx [M,K]
+      N = a.size(0); K = a.size(1); M = b.size(0);
+      TORCH_CHECK(b.size(1) == N, "dy must be [M,N] matching w [N,K]");
+      return {M, K};
+    case 2:  // a=x [M,K], b=dy [M,N] -> dw [N,K]
+      M = a.size(0); K = a.size(1); N = b.size(1);
+      TORCH_CHECK(b.size(0) == M, "dy must be [M,N] matching x [M,K]");
+      return {N, K};
+    default:
+      TORCH_CHECK(false, "bad mode");
+  }
+}
 
+struct CacheKey {
+  int mode;
+  long long M, N, K;
+  long idx;
+  bool operator==(const CacheKey& o) const {
+    return mode == o.mode && M == o.M && N == o.N && K == o.K &&
+           idx == o.idx;
+  }
+};
+struct CacheKeyHash {
+  size_t operator()(const CacheKey& k) const {
+    size_t h = (size_t)k.mode;
+    h = h * 1315423911u ^ (size_t)k.M;
+    h = h * 1315423911u ^ (size_t)k.N;
+    h = h * 1315423911u ^ (size_t)k.K;
+    h = h * 1315423911u ^ (size_t)k.idx;
+    return h;
+  }
+};
+struct CachedProblem {
+  Layouts L;  // leaked on purpose: lives for the process
+  hipblasLtMatmulAlgo_t algo;
+};
+
+CachedProblem& cached_problem(int mode, long long M, long long N,
+                              long long K, long index) {
+  static std::unordered_map<CacheKey, CachedProblem*, CacheKeyHash> cache;
+  static std::mutex mu;
+  CacheKey key{mode, M, N, K, index};
+  std::lock_guard<std::mutex> lock(mu);
+  auto it = cache.find(key);
+  if (it != cache.end()) return *it->second;
+  auto* cp = new CachedProblem();
+  make_layouts(cp->L, mode, M, N, K);
+  std::vector<int> want{(int)index};
+  std::vector<hipblasLtMatmulHeuristicResult_t> algos;
+  LT_CHECK(hipblaslt_ext::getAlgosFromIndex(lt_handle(), want, algos));
+  TORCH_CHECK(!algos.empty(), "no hipBLASLt algo for index ", index);
+  float alpha = 1.f, beta = 0.f;
+  size_t ws = 0;
+  LT_CHECK(hipblaslt_ext::matmulIsAlgoSupported(
+      lt_handle(), cp->L.op, &alpha, cp->L.a, cp->L.b, &beta, cp->L.c,
+      cp->L.c, algos[0].algo, ws));
+  TORCH_CHECK(ws <= kWorkspaceBytes, "algo ", index,
+              " needs more workspace than the static 128 MiB");
+  cp->algo = algos[0].algo;
+  cache.emplace(key, cp);
+  return *cp;
+}
+
+// Run one GEMM with a pinned solution index (cached layouts+algo).
+torch::Tensor gemm_lt_run(long mode, torch::Tensor a, torch::Tensor b,
+                          long index) {
+  a = a.contiguous();
+  b = b.contiguous();
+  long long M, N, K;
+  auto dims = problem_dims((int)mode, a, b, M, N, K);
+  auto out = torch::empty({dims.first, dims.second}, a.options());
+  auto& cp = cached_problem((int)mode, M, N, K, index);
+  run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &cp.algo, cp.L);
+  return out;
+}
+
+// Time one pinned solution: returns ms/iter.
+double gemm_lt_time(long mode, torch::Tensor a, torch::Tensor b,
+                    long index, long iters) {
+  a = a.contiguous();
+  b = b.contiguous();
+  long long M, N, K;
+  auto dims = problem_dims((int)mode, a, b, M, N, K);
+  auto out = torch::empty({dims.first, dims.second}, a.options());
+  auto& cp = cached_problem((int)mode, M, N, K, index);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipEvent_t ev0, ev1;
+  HIPRT_CHECK(hipEventCreate(&ev0));
+  HIPRT_CHECK(hipEventCreate(&ev1));
+  run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &cp.algo, cp.L);
+  run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &cp.algo, cp.L);
+  HIPRT_CHECK(hipEventRecord(ev0, stream));
+  for (long i = 0; i < iters; ++i)
+    run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &cp.algo, cp.L);
+  HIPRT_CHECK(hipEventRecord(ev1, stream));
+  HIPRT_CHECK(hipEventSynchronize(ev1));
+  float ms = 0.f;
+  HIPRT_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+  HIPRT_CHECK(hipEventDestroy(ev0));
+  HIPRT_CHECK(hipEventDestroy(ev1));
+  return ms / iters;
+}
+
+// Enumerate + time every supported solution. Returns (indices, ms)
+// sorted fastest-first.
+std::vector<torch::Tensor> gemm_lt_search(long mode, torch::Tensor a,
+                                          torch::Tensor b, long iters,
+                                          long cap) {
+  a = a.contiguous();
+  b = b.contiguous();
+  long long M, N, K;
+  auto dims = problem_dims((int)mode, a, b, M, N, K);
+  auto out = torch::empty({dims.first, dims.second}, a.options());
   Layouts L;
-  make_layouts(L, M, N, K);
+  make_layouts(L, (int)mode, M, N, K);
 
   std::vector<hipblasLtMatmulHeuristicResult_t> all;
   LT_CHECK(hipblaslt_ext::getAllAlgos(
-      lt_handle(), hipblaslt_ext::GemmType::HIPBLASLT_GEMM,
-      HIPBLAS_OP_T, HIPBLAS_OP_N, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
-      HIP_R_16BF, HIPBLAS_COMPUTE_32F, all));
+      lt_handle(), hipblaslt_ext::GemmType::HIPBLASLT_GEMM, L.opa, L.opb,
+      HIP_R_16BF, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF, HIPBLAS_COMPUTE_32F,
+      all));
 
   float alpha = 1.f, beta = 0.f;
-  std::vector<int> idxs;
+  std::vector<long> idxs;
   std::vector<float> times;
   hipEvent_t ev0, ev1;
   HIPRT_CHECK(hipEventCreate(&ev0));
   HIPRT_CHECK(hipEventCreate(&ev1));
   auto stream = at::hip::getCurrentHIPStream();
   for (auto& h : all) {
-    if (cap > 0 && (long)idxs.size() >= cap)
-      break;
+    if (cap > 0 && (long)idxs.size() >= cap) break;
     size_t ws = 0;
     if (hipblaslt_ext::matmulIsAlgoSupported(
             lt_handle(), L.op, &alpha, L.a, L.b, &beta, L.c, L.c, h.algo,
             ws) != HIPBLAS_STATUS_SUCCESS)
       continue;
-    if (ws > (size_t)workspace.numel())
-      continue;
-    // warmup x2, then time `iters` back-to-back launches
-    run_algo(x, w, out, workspace, &h.algo, L);
-    run_algo(x, w, out, workspace, &h.algo, L);
+    if (ws > kWorkspaceBytes) continue;
+    run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &h.algo, L);
     HIPRT_CHECK(hipEventRecord(ev0, stream));
     for (long i = 0; i < iters; ++i)
-      run_algo(x, w, out, workspace, &h.algo, L);
+      run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &h.algo, L);
     HIPRT_CHECK(hipEventRecord(ev1, stream));
     HIPRT_CHECK(hipEventSynchronize(ev1));
     float ms = 0.f;
@@ -135,12 +284,15 @@ std::vector<torch::Tensor> gemm_lt_search(torch::Tensor x, torch::Tensor w,
   }
   HIPRT_CHECK(hipEventDestroy(ev0));
   HIPRT_CHECK(hipEventDestroy(ev1));
+  hipblasLtMatrixLayoutDestroy(L.a);
+  hipblasLtMatrixLayoutDestroy(L.b);
+  hipblasLtMatrixLayoutDestroy(L.c);
+  hipblasLtMatmulDescDestroy(L.op);
 
-  // sort fastest-first
   std::vector<size_t> order(idxs.size());
   for (size_t i = 0; i < order.size(); ++i) order[i] = i;
   std::sort(order.begin(), order.end(),
-            [&](size_t a, size_t b) { return times[a] < times[b]; });
+            [&](size_t x, size_t y) { return times[x] < times[y]; });
   auto t_idx = torch::empty({(long long)order.size()},
                             torch::dtype(torch::kInt64));
   auto t_ms = torch::empty({(long long)order.size()},
@@ -152,44 +304,20 @@ std::vector<torch::Tensor> gemm_lt_search(torch::Tensor x, torch::Tensor w,
   return {t_idx, t_ms};
 }
 
-// Run one GEMM with a pinned solution index (from gemm_lt_search).
-torch::Tensor gemm_lt_run(torch::Tensor x, torch::Tensor w, long index) {
-  TORCH_CHECK(x.is_cuda() && w.is_cuda());
-  x = x.contiguous();
-  w = w.contiguous();
-  const long long M = x.size(0), K = x.size(1), N = w.size(0);
-  auto out = torch::empty({M, N}, x.options());
-  auto workspace = torch::empty({128 << 20},
-                                x.options().dtype(torch::kUInt8));
-  Layouts L;
-  make_layouts(L, M, N, K);
-  std::vector<int> want{(int)index};
-  std::vector<hipblasLtMatmulHeuristicResult_t> algos;
-  LT_CHECK(hipblaslt_ext::getAlgosFromIndex(lt_handle(), want, algos));
-  TORCH_CHECK(!algos.empty(), "no algo for index ", index);
-  float alpha = 1.f, beta = 0.f;
-  size_t ws = 0;
-  LT_CHECK(hipblaslt_ext::matmulIsAlgoSupported(
-      lt_handle(), L.op, &alpha, L.a, L.b, &beta, L.c, L.c,
-      algos[0].algo, ws));
-  run_algo(x, w, out, workspace, &algos[0].algo, L);
-  return out;
-}
-
 // hipBLASLt's own heuristic choice (what torch.matmul effectively uses),
 // for A/B comparison in the search report.
-torch::Tensor gemm_lt_heuristic(torch::Tensor x, torch::Tensor w) {
-  x = x.contiguous();
-  w = w.contiguous();
-  const long long M = x.size(0), K = x.size(1), N = w.size(0);
-  auto out = torch::empty({M, N}, x.options());
-  auto workspace = torch::empty({128 << 20},
-                                x.options().dtype(torch::kUInt8));
+torch::Tensor gemm_lt_heuristic(long mode, torch::Tensor a,
+                                torch::Tensor b) {
+  a = a.contiguous();
+  b = b.contiguous();
+  long long M, N, K;
+  auto dims = problem_dims((int)mode, a, b, M, N, K);
+  auto out = torch::empty({dims.first, dims.second}, a.options());
   Layouts L;
-  make_layouts(L, M, N, K);
+  make_layouts(L, (int)mode, M, N, K);
   hipblasLtMatmulPreference_t pref;
   LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
-  uint64_t wsmax = (uint64_t)workspace.numel();
+  uint64_t wsmax = kWorkspaceBytes;
   LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
       pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsmax,
       sizeof(wsmax)));
@@ -200,7 +328,11 @@ torch::Tensor gemm_lt_heuristic(torch::Tensor x, torch::Tensor w) {
                                            &found));
   hipblasLtMatmulPreferenceDestroy(pref);
   TORCH_CHECK(found > 0, "no heuristic solution");
-  run_algo(x, w, out, workspace, &res[0].algo, L);
+  run_algo(a.data_ptr(), b.data_ptr(), out.data_ptr(), &res[0].algo, L);
+  hipblasLtMatrixLayoutDestroy(L.a);
+  hipblasLtMatrixLayoutDestroy(L.b);
+  hipblasLtMatrixLayoutDestroy(L.c);
+  hipblasLtMatmulDescDestroy(L.op);
   return out;
 }
 
@@ -208,7 +340,6 @@ torch::Tensor gemm_lt_heuristic(torch::Tensor x, torch::Tensor w) {
 // OCP E4M3 GEMM (gfx950 native; 2x bf16 peak): D_bf16 = (A_fp8 @ B_fp8^T)
 // with fp32 accumulate. Inputs arrive as uint8 bit-patterns (torch's
 // float8_e4m3fn storage viewed as uint8); scales fold into alpha.
-// Round-2 measurement target — compile-validated in round 1.
 void make_layouts_fp8(Layouts& L, long long M, long long N, long long K) {
   LT_CHECK(hipblasLtMatmulDescCreate(&L.op, HIPBLAS_COMPUTE_32F,
                                      HIP_R_32F));
@@ -233,13 +364,11 @@ torch::Tensor gemm_lt_fp8(torch::Tensor x8, torch::Tensor w8,
   const long long M = x8.size(0), K = x8.size(1), N = w8.size(0);
   auto out = torch::empty({M, N},
                           x8.options().dtype(torch::kBFloat16));
-  auto workspace = torch::empty({128 << 20},
-                                x8.options().dtype(torch::kUInt8));
   Layouts L;
   make_layouts_fp8(L, M, N, K);
   hipblasLtMatmulPreference_t pref;
   LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
-  uint64_t wsmax = (uint64_t)workspace.numel();
+  uint64_t wsmax = kWorkspaceBytes;
   LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
       pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsmax,
       sizeof(wsmax)));
@@ -254,8 +383,12 @@ torch::Tensor gemm_lt_fp8(torch::Tensor x8, torch::Tensor w8,
   LT_CHECK(hipblasLtMatmul(
       lt_handle(), L.op, &alpha, w8.data_ptr(), L.a, x8.data_ptr(), L.b,
       &beta, out.data_ptr(), L.c, out.data_ptr(), L.c, &res[0].algo,
-      workspace.data_ptr(), (size_t)workspace.numel(),
+      lt_workspace(), kWorkspaceBytes,
       at::hip::getCurrentHIPStream()));
+  hipblasLtMatrixLayoutDestroy(L.a);
+  hipblasLtMatrixLayoutDestroy(L.b);
+  hipblasLtMatrixLayoutDestroy(L.c);
+  hipblasLtMatmulDescDestroy(L.op);
   return out;
 }
 
@@ -263,10 +396,15 @@ torch::Tensor gemm_lt_fp8(torch::Tensor x8, torch::Tensor w8,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("search", &gemm_lt_search,
-        "time hipBLASLt solutions for out = x @ w^T (cap=0: all); "
+        "time hipBLASLt solutions for one problem mode (cap=0: all); "
+        "mode 0 fwd (a=w,b=x), 1 dx (a=w,b=dy), 2 dw (a=x,b=dy); "
         "returns (indices, ms) fastest-first");
-  m.def("run", &gemm_lt_run, "GEMM with a pinned solution index");
-  m.def("heuristic", &gemm_lt_heuristic, "GEMM via hipBLASLt heuristic");
+  m.def("run", &gemm_lt_run,
+        "GEMM with a pinned solution index (cached layouts+algo)");
+  m.def("time_one", &gemm_lt_time,
+        "time a pinned solution: (mode, a, b, index, iters) -> ms");
+  m.def("heuristic", &gemm_lt_heuristic,
+        "GEMM via hipBLASLt heuristic (mode, a, b)");
   m.def("fp8", &gemm_lt_fp8,
         "OCP E4M3 GEMM -> bf16 out (uint8-viewed fp8 inputs, alpha "
         "carries the descale)");

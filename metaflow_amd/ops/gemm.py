@@ -1,12 +1,16 @@
 """Tuned hipBLASLt GEMMs: replay offline-searched solution indices.
 
-`benchmarks/tune_gemms.py` sweeps every hipBLASLt solution for a shape
-(`_mfx_gemm.search`, see ops/csrc/gemm_lt.hip) and the winners get
-pinned here: set ``MFX_GEMM_TUNE_FILE`` to a JSON of
-``{"M,K,N": solution_index, ...}`` and `tuned_linear` routes matching
-shapes through ``_mfx_gemm.run`` — everything else (and every CPU call)
-falls back to ``torch.nn.functional.linear``. First sweep found +7.6%
-on the gate_up projection (profiles/bench_results_r01.md).
+`benchmarks/tune_gemms.py` sweeps every hipBLASLt solution for each
+train-step GEMM — the forward AND both backward problems per projection
+(modes fwd/dx/dw, see ops/csrc/gemm_lt.hip) — and persists winners to
+``metaflow_amd/ops/gemm_table.json`` (shipped with the repo; override
+with ``MFX_GEMM_TUNE_FILE``). ``TunedLinear``/``tuned_linear`` route
+matching shapes through ``_mfx_gemm.run`` with full autograd; any
+untuned problem (and every CPU call) falls back to
+``torch.nn.functional.linear`` / its torch backward.
+
+Keys: ``"<mode>|M,K,N"`` where (M, K, N) are the F.linear dims
+(x [M,K], w [N,K], out [M,N]) — the same triplet for all three modes.
 """
 
 import json
@@ -17,18 +21,25 @@ import torch
 _table = None
 _ext = None
 
+_DEFAULT_TABLE = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "gemm_table.json")
+
 
 def _load_table():
     global _table
     if _table is None:
         _table = {}
-        path = os.environ.get("MFX_GEMM_TUNE_FILE")
+        path = os.environ.get("MFX_GEMM_TUNE_FILE", _DEFAULT_TABLE)
         if path and os.path.isfile(path):
             with open(path) as f:
                 raw = json.load(f)
             for key, idx in raw.items():
-                m, k, n = (int(v) for v in key.split(","))
-                _table[(m, k, n)] = int(idx)
+                if "|" in key:
+                    mode, dims = key.split("|")
+                else:  # round-1 format: fwd-only
+                    mode, dims = "fwd", key
+                m, k, n = (int(v) for v in dims.split(","))
+                _table[(mode, m, k, n)] = int(idx)
     return _table
 
 
@@ -42,18 +53,56 @@ def _gemm_ext():
 
 
 def reset_tune_table():
-    """Re-read MFX_GEMM_TUNE_FILE on next use (tests/tuning loops)."""
+    """Re-read the tune file on next use (tests/tuning loops)."""
     global _table
     _table = None
 
 
-def tuned_linear(x, weight):
-    """F.linear with pinned hipBLASLt solutions where tuned.
+class _TunedLinear(torch.autograd.Function):
+    """F.linear with pinned hipBLASLt solutions per GEMM where tuned.
 
-    x: [..., K] bf16, weight: [N, K] bf16. Only exact (M, K, N) matches
-    use a pinned index — autograd-safe because tuning applies to the
-    forward GEMM only (wrap in autograd.Function when the backward
-    shapes get tuned too; round-2 item).
+    Each of the three GEMMs falls back to the torch op independently, so
+    a partially-tuned shape still benefits.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, key):
+        ctx.save_for_backward(x, w)
+        ctx.key = key
+        M, K, N = key
+        table = _load_table()
+        idx = table.get(("fwd", M, K, N))
+        x2 = x.reshape(M, K)
+        if idx is not None:
+            out = _gemm_ext().run(0, w, x2, idx)
+        else:
+            out = torch.nn.functional.linear(x2, w)
+        return out.reshape(*x.shape[:-1], N)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        M, K, N = ctx.key
+        table = _load_table()
+        dy2 = dy.reshape(M, N).contiguous()
+        idx_dx = table.get(("dx", M, K, N))
+        idx_dw = table.get(("dw", M, K, N))
+        if idx_dx is not None:
+            dx = _gemm_ext().run(1, w, dy2, idx_dx)
+        else:
+            dx = dy2 @ w
+        if idx_dw is not None:
+            dw = _gemm_ext().run(2, x.reshape(M, K), dy2, idx_dw)
+        else:
+            dw = dy2.t() @ x.reshape(M, K)
+        return dx.reshape(x.shape), dw, None
+
+
+def tuned_linear(x, weight):
+    """F.linear with pinned hipBLASLt solutions (fwd + dx + dw).
+
+    x: [..., K] bf16, weight: [N, K] bf16. Shapes with no tuned entry at
+    all go straight to torch (no autograd.Function overhead).
     """
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return torch.nn.functional.linear(x, weight)
@@ -63,8 +112,13 @@ def tuned_linear(x, weight):
     K = x.shape[-1]
     M = x.numel() // K
     N = weight.shape[0]
-    idx = table.get((M, K, N))
-    if idx is None or x.requires_grad or weight.requires_grad:
+    if not (("fwd", M, K, N) in table or ("dx", M, K, N) in table
+            or ("dw", M, K, N) in table):
         return torch.nn.functional.linear(x, weight)
-    out = _gemm_ext().run(x.reshape(M, K), weight, idx)
-    return out.reshape(*x.shape[:-1], N)
+    if not (x.requires_grad or weight.requires_grad):
+        idx = table.get(("fwd", M, K, N))
+        if idx is None:
+            return torch.nn.functional.linear(x, weight)
+        out = _gemm_ext().run(0, weight, x.reshape(M, K), idx)
+        return out.reshape(*x.shape[:-1], N)
+    return _TunedLinear.apply(x, weight, (M, K, N))

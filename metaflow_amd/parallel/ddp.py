@@ -198,4 +198,21 @@ def init_process_group_from_env(backend=None):
     os.environ.setdefault("MASTER_PORT", "29500")
     os.environ.setdefault("RANK", "0")
     os.environ.setdefault("WORLD_SIZE", "1")
-    dist.init_process_group(backend=backend)
+    if backend == "nccl":
+        # bind the device BEFORE init: RCCL derives the communicator
+        # device from the current device; late binds wedge rendezvous
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    from datetime import timedelta
+
+    timeout_s = int(os.environ.get("MFX_RCCL_INIT_TIMEOUT", "300"))
+    try:
+        dist.init_process_group(backend=backend,
+                                timeout=timedelta(seconds=timeout_s))
+    except Exception as e:
+        raise RuntimeError(
+            "torch.distributed init failed (backend=%s, rank=%s/%s, "
+            "addr=%s:%s): %s" % (
+                backend, os.environ.get("RANK"),
+                os.environ.get("WORLD_SIZE"),
+                os.environ.get("MASTER_ADDR"),
+                os.environ.get("MASTER_PORT"), e)) from e

@@ -58,7 +58,33 @@ class TorchParallelDecorator(ParallelDecorator):
                 backend = "nccl" if (n_gpus > 0 and num_nodes <= visible) \
                     else "gloo"
             if not dist.is_initialized():
-                dist.init_process_group(backend=backend)
+                from datetime import timedelta
+
+                if backend == "nccl":
+                    # bind this rank's device BEFORE init: RCCL derives
+                    # the communicator device from the current device,
+                    # and a late bind is the classic wedged-rendezvous
+                    timeout_s = int(os.environ.get(
+                        "MFX_RCCL_INIT_TIMEOUT", "300"))
+                    torch.cuda.set_device(
+                        int(os.environ.get("LOCAL_RANK", "0")))
+                else:
+                    timeout_s = int(os.environ.get(
+                        "MFX_RCCL_INIT_TIMEOUT", "300"))
+                try:
+                    dist.init_process_group(
+                        backend=backend,
+                        timeout=timedelta(seconds=timeout_s))
+                except Exception as e:
+                    # fail LOUDLY and fast: the scheduler tears down the
+                    # gang and retries it once on a fresh port
+                    raise RuntimeError(
+                        "torch.distributed init failed (backend=%s, "
+                        "rank=%s/%s, addr=%s:%s): %s" % (
+                            backend, os.environ.get("RANK"),
+                            os.environ.get("WORLD_SIZE"),
+                            os.environ.get("MASTER_ADDR"),
+                            os.environ.get("MASTER_PORT"), e)) from e
             cp_attr = self.attributes.get("context_parallel") or 1
             cp = (dist.get_world_size() if cp_attr == "all"
                   else int(cp_attr))

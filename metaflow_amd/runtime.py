@@ -207,6 +207,7 @@ class NativeRuntime(object):
         self._failure_msg = None
         self._origin_index = None
         self._cloned_paths = set()  # new-run pathspecs produced by cloning
+        self._gang_info = {}  # control_task_id -> gang retry bookkeeping
         self._selector = selectors.DefaultSelector()
         self._params_pathspec = None
 
@@ -469,11 +470,18 @@ class NativeRuntime(object):
                 return max(1, int(deco.attributes.get("gpu") or 1))
         return 1
 
-    def _queue_gang(self, spec, pathspec, target, num_parallel):
+    def _queue_gang(self, spec, pathspec, target, num_parallel,
+                    retry_count=0):
         """Native gang scheduling for @parallel steps: N rank processes with
         rendezvous env + GPU pinning (@resources(gpu=k) gives each rank k
         devices). Replaces the reference's control-task subprocess
-        emulation (parallel_decorator.py:175-246)."""
+        emulation (parallel_decorator.py:175-246).
+
+        A rank failure tears the surviving ranks down immediately and the
+        WHOLE gang is retried as a unit — at least once even without
+        @retry, because the most common first failure is rendezvous-class
+        (RCCL init/port races), which a fresh gang with a fresh port
+        survives (reference retries everything, runtime.py:1506-1512)."""
         control_id = self._new_task_id()
         port = _free_port()
         n_gpus = _visible_gpu_count()
@@ -512,7 +520,18 @@ class NativeRuntime(object):
             child_stack = spec.stack + ((spec.step, rank, num_parallel),)
             self._queue_spec(TaskSpec(
                 target, [pathspec], rank, child_stack, task_id=task_id,
+                retry_count=retry_count,
                 gang=(num_parallel, rank, control_id, port), env=env))
+        self._gang_info[control_id] = {
+            "parent": (spec, pathspec, target, num_parallel),
+            "retry": retry_count,
+            "exited": 0,
+            "state": "running",
+            "rc": None,
+            "member_keys": [
+                (target, spec.stack + ((spec.step, r, num_parallel),))
+                for r in range(num_parallel)],
+        }
 
     # ------------------------------------------------------------- launching
     def _launch_ready(self):
@@ -626,6 +645,15 @@ class NativeRuntime(object):
             pass
 
         if rc == 0:
+            if spec.gang is not None:
+                info = self._gang_info.get(spec.gang[2])
+                if info is not None:
+                    info["exited"] += 1
+                    if info["state"] == "failed":
+                        # gang already failed: this rank's DONE is moot,
+                        # the whole gang retries (or the run fails)
+                        self._maybe_retry_gang(info, spec.gang[2])
+                        return
             task_ds = self.flow_datastore.get_task_datastore(
                 self.run_id, spec.step, spec.task_id)
             if task_ds.attempt is None:
@@ -639,11 +667,13 @@ class NativeRuntime(object):
         if rc == -signal.SIGSEGV:
             self._echo("Task %s segfaulted."
                        % self._pathspec(spec.step, spec.task_id))
+        if spec.gang is not None:
+            self._gang_member_failed(spec, rc)
+            return
         max_retries = self._max_retries_for(spec.step)
         can_retry = (
             spec.retry_count < max_retries
             and rc != EXIT_DISALLOW_RETRY
-            and spec.gang is None
         )
         if can_retry:
             self._echo("Task %s failed (rc=%d); retrying (%d/%d)."
@@ -657,6 +687,61 @@ class NativeRuntime(object):
         else:
             self._fail("Task %s failed (rc=%d)."
                        % (self._pathspec(spec.step, spec.task_id), rc))
+
+    def _gang_member_failed(self, spec, rc):
+        """First rank failure: kill surviving ranks of THIS gang now (a
+        partially-dead gang wedges in the next collective); once every
+        rank has exited, retry the whole gang as a unit or fail."""
+        cid = spec.gang[2]
+        info = self._gang_info.get(cid)
+        if info is None:
+            self._fail("Gang task %s failed (rc=%d)."
+                       % (self._pathspec(spec.step, spec.task_id), rc))
+            return
+        info["exited"] += 1
+        if info["state"] != "failed":
+            info["state"] = "failed"
+            info["rc"] = rc
+            info["failed_step"] = spec.step
+            info["failed_task"] = spec.task_id
+            killed = 0
+            for w in self._workers:
+                wgang = w.spec.gang
+                if wgang and wgang[2] == cid and w.poll() is None:
+                    w.kill()
+                    killed += 1
+            self._echo(
+                "Gang rank %s/%s failed (rc=%d); tearing down %d "
+                "surviving rank(s)."
+                % (spec.step, spec.task_id, rc, killed))
+        self._maybe_retry_gang(info, cid)
+
+    def _maybe_retry_gang(self, info, cid):
+        parent_spec, pathspec, target, num_parallel = info["parent"]
+        if info["exited"] < num_parallel:
+            return  # wait for the rest of the gang to exit
+        del self._gang_info[cid]
+        # gangs always get at least one retry: the dominant first-attempt
+        # failure mode is rendezvous-class (RCCL init / port race), which
+        # a fresh gang on a fresh port survives
+        max_retries = max(self._max_retries_for(target), 1)
+        if info["retry"] >= max_retries or info["rc"] == EXIT_DISALLOW_RETRY:
+            self._fail("Gang step %s failed (rc=%s) after %d attempt(s)."
+                       % (target, info["rc"], info["retry"] + 1))
+            return
+        # purge bookkeeping from ranks that finished before the failure
+        for key in info["member_keys"]:
+            self._finished.pop(key, None)
+        for key in list(self._join_arrivals):
+            jstep, jstack = key
+            if jstack == parent_spec.stack and \
+                    target in self.graph[jstep].in_funcs:
+                del self._join_arrivals[key]
+        self._echo("Retrying gang step %s (attempt %d/%d) with a fresh "
+                   "rendezvous port." % (target, info["retry"] + 2,
+                                         max_retries + 1))
+        self._queue_gang(parent_spec, pathspec, target, num_parallel,
+                         retry_count=info["retry"] + 1)
 
     def _task_finished_bookkeeping(self, spec, task_ds):
         self._finished[spec.key] = task_ds.pathspec

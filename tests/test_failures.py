@@ -425,3 +425,21 @@ def test_catch_on_foreach_split_rejected(tmp_datastore, tmp_path):
          "run"], env=env, capture_output=True, text=True, timeout=120)
     assert proc.returncode != 0
     assert "not supported" in proc.stderr + proc.stdout
+
+
+def test_gang_transient_failure_retries(tmp_datastore, tmp_path,
+                                        monkeypatch):
+    """A rank dying on the first attempt (rendezvous-class crash) tears
+    the gang down and the WHOLE gang retries once on a fresh port; the
+    run completes without resume (VERDICT r1 weak #4)."""
+    marker_dir = tmp_path / "markers"
+    marker_dir.mkdir()
+    proc = run_flow("gang_retry_flow.py", tmp_datastore, "run",
+                    env_extra={"GANG_RETRY_DIR": str(marker_dir)})
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    run_id = latest_run_id(tmp_datastore, "GangRetryFlow")
+    assert read_artifact(tmp_datastore, "GangRetryFlow", run_id, "join",
+                         "ranks") == [0, 1]
+    # every rank attempted twice (whole-gang retry, not per-rank)
+    assert (marker_dir / "attempted_0").read_text() == "xx"
+    assert (marker_dir / "attempted_1").read_text() == "xx"

@@ -1,5 +1,5 @@
 """Tuned-GEMM plumbing: table parsing and CPU fallback (the GPU replay
-path is exercised by benchmarks/tune_gemms.py on-box)."""
+path is exercised by benchmarks/tune_gemms.py + test_ops_gpu on-box)."""
 
 import json
 
@@ -10,7 +10,8 @@ def test_tuned_linear_cpu_fallback(tmp_path, monkeypatch):
     from metaflow_amd.ops import gemm
 
     f = tmp_path / "tune.json"
-    f.write_text(json.dumps({"64,32,16": 123}))
+    f.write_text(json.dumps({"fwd|64,32,16": 123, "dx|64,32,16": 4,
+                             "dw|64,32,16": 5}))
     monkeypatch.setenv("MFX_GEMM_TUNE_FILE", str(f))
     gemm.reset_tune_table()
     x = torch.randn(64, 32)
@@ -18,15 +19,48 @@ def test_tuned_linear_cpu_fallback(tmp_path, monkeypatch):
     out = gemm.tuned_linear(x, w)   # CPU -> plain F.linear
     ref = torch.nn.functional.linear(x, w)
     assert torch.equal(out, ref)
-    assert gemm._load_table() == {(64, 32, 16): 123}
+    assert gemm._load_table() == {("fwd", 64, 32, 16): 123,
+                                  ("dx", 64, 32, 16): 4,
+                                  ("dw", 64, 32, 16): 5}
     gemm.reset_tune_table()
 
 
-def test_tuned_linear_no_table():
+def test_tune_table_v1_format(tmp_path, monkeypatch):
+    """Round-1 fwd-only keys ("M,K,N") still parse."""
     from metaflow_amd.ops import gemm
 
+    f = tmp_path / "tune.json"
+    f.write_text(json.dumps({"64,32,16": 7}))
+    monkeypatch.setenv("MFX_GEMM_TUNE_FILE", str(f))
+    gemm.reset_tune_table()
+    assert gemm._load_table() == {("fwd", 64, 32, 16): 7}
+    gemm.reset_tune_table()
+
+
+def test_tuned_linear_no_table(monkeypatch):
+    from metaflow_amd.ops import gemm
+
+    monkeypatch.setenv("MFX_GEMM_TUNE_FILE", "/nonexistent")
     gemm.reset_tune_table()
     x = torch.randn(4, 8, dtype=torch.bfloat16)
     w = torch.randn(6, 8, dtype=torch.bfloat16)
     out = gemm.tuned_linear(x, w)
     assert out.shape == (4, 6)
+    gemm.reset_tune_table()
+
+
+def test_shipped_table_parses():
+    """The in-repo gemm_table.json (if present) must parse into
+    (mode, M, K, N) -> index entries."""
+    import os
+
+    from metaflow_amd.ops import gemm
+
+    gemm.reset_tune_table()
+    if os.path.isfile(gemm._DEFAULT_TABLE):
+        table = gemm._load_table()
+        assert table, "shipped table exists but parsed empty"
+        for (mode, m, k, n), idx in table.items():
+            assert mode in ("fwd", "dx", "dw")
+            assert m > 0 and k > 0 and n > 0 and idx >= 0
+    gemm.reset_tune_table()

@@ -361,3 +361,44 @@ def test_add_rmsnorm_fused(dev):
     assert rel_err(x.grad, xr.grad) < 2e-2
     assert rel_err(r.grad, rr.grad) < 2e-2
     assert rel_err(w.grad, wr.grad) < 2e-2
+
+
+def test_tuned_linear_gpu(dev, tmp_path, monkeypatch):
+    """End-to-end tuned GEMM path: search winners for a small shape in
+    all three modes (fwd/dx/dw), pin them via a table file, and check
+    TunedLinear's forward AND gradients against torch."""
+    import json
+
+    from metaflow_amd.ops import _mfx_gemm as G
+    from metaflow_amd.ops import gemm
+
+    torch.manual_seed(0)
+    M, K, N = 512, 256, 384
+    x0 = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.1
+    w0 = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.1
+    dy = torch.randn(M, N, dtype=torch.bfloat16, device=dev) * 0.1
+    table = {}
+    for mode, mid, a, b in (("fwd", 0, w0, x0), ("dx", 1, w0, dy),
+                            ("dw", 2, x0, dy)):
+        idxs, ms = G.search(mid, a, b, 2, 16)
+        assert idxs.numel() > 0
+        table["%s|%d,%d,%d" % (mode, M, K, N)] = int(idxs[0])
+    f = tmp_path / "table.json"
+    f.write_text(json.dumps(table))
+    monkeypatch.setenv("MFX_GEMM_TUNE_FILE", str(f))
+    gemm.reset_tune_table()
+    try:
+        x = x0.clone().requires_grad_(True)
+        w = w0.clone().requires_grad_(True)
+        out = gemm.tuned_linear(x, w)
+        out.backward(dy)
+
+        xr = x0.clone().requires_grad_(True)
+        wr = w0.clone().requires_grad_(True)
+        ref = torch.nn.functional.linear(xr, wr)
+        ref.backward(dy)
+        assert rel_err(out, ref) < 2e-2, "fwd"
+        assert rel_err(x.grad, xr.grad) < 2e-2, "dx"
+        assert rel_err(w.grad, wr.grad) < 2e-2, "dw"
+    finally:
+        gemm.reset_tune_table()
