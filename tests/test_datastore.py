@@ -197,3 +197,146 @@ def test_checkpoint_save_load_roundtrip(tmp_path):
         assert out[k].dtype == state[k].dtype
         assert torch.equal(out[k], state[k]), k
     out["w"] += 1  # writable (trained on after resume)
+
+
+def test_serializer_registry_order_and_override():
+    """Priority-ordered registry: lower priority wins; registering the
+    same encoding again replaces the instance (extension override)."""
+    from metaflow_amd.datastore import serializers as S
+
+    order = S.get_ordered_serializers()
+    prios = [s.priority for s in order]
+    assert prios == sorted(prios)
+    assert order[-1].encoding == S.ENC_PICKLE  # universal fallback last
+
+    class Grabby(S.ArtifactSerializer):
+        encoding = "test-grabby-v1"
+        priority = 1  # runs before tensor + pickle
+
+        def can_serialize(self, obj):
+            return isinstance(obj, dict) and obj.get("grab") is True
+
+        def serialize(self, obj):
+            return b"GRABBED"
+
+        def deserialize(self, data):
+            assert data == b"GRABBED"
+            return {"grab": True, "restored": True}
+
+    try:
+        blob, enc = S.serialize({"grab": True})
+        assert enc == "test-grabby-v1" and blob == b"GRABBED"
+        assert S.deserialize(blob, enc) == {"grab": True,
+                                            "restored": True}
+        # non-matching objects still fall through to pickle
+        blob2, enc2 = S.serialize({"grab": False})
+        assert enc2 == S.ENC_PICKLE
+        assert S.deserialize(blob2, enc2) == {"grab": False}
+    finally:
+        S._REGISTRY.pop("test-grabby-v1", None)
+        S._ORDERED = None
+
+
+def test_serializer_failure_falls_through():
+    """A serializer whose serialize() raises falls through to pickle
+    instead of failing the artifact save."""
+    from metaflow_amd.datastore import serializers as S
+
+    class Broken(S.ArtifactSerializer):
+        encoding = "test-broken-v1"
+        priority = 1
+
+        def can_serialize(self, obj):
+            return isinstance(obj, (list, dict, int, str))
+
+        def serialize(self, obj):
+            raise RuntimeError("boom")
+
+        def deserialize(self, data):
+            raise RuntimeError("boom")
+
+    try:
+        blob, enc = S.serialize([1, 2, 3])
+        assert enc == S.ENC_PICKLE
+        assert S.deserialize(blob, enc) == [1, 2, 3]
+    finally:
+        S._REGISTRY.pop("test-broken-v1", None)
+        S._ORDERED = None
+
+
+def test_serializer_extension_flow(tmp_path, tmp_datastore):
+    """An extension package contributes an artifact serializer via
+    ARTIFACT_SERIALIZERS; a flow artifact round-trips through it with
+    its encoding recorded (reference serializer.py:252 bootstrap)."""
+    import os
+    import subprocess
+    import sys
+    import textwrap
+
+    ext = tmp_path / "metaflow_amd_extensions" / "serext"
+    ext.mkdir(parents=True)
+    (ext / "__init__.py").write_text(textwrap.dedent("""
+        from metaflow_amd.datastore.serializers import ArtifactSerializer
+
+        class Payload(object):
+            def __init__(self, text):
+                self.text = text
+
+        class PayloadSerializer(ArtifactSerializer):
+            encoding = "payload-v1"
+            priority = 10
+
+            def can_serialize(self, obj):
+                return isinstance(obj, Payload)
+
+            def serialize(self, obj):
+                return obj.text.encode() + b"|EXT"
+
+            def deserialize(self, data):
+                assert data.endswith(b"|EXT")
+                return Payload(data[:-4].decode())
+
+        ARTIFACT_SERIALIZERS = [PayloadSerializer]
+    """))
+    flow = tmp_path / "ser_flow.py"
+    flow.write_text(textwrap.dedent("""
+        from metaflow_amd import FlowSpec, step
+        from metaflow_amd_extensions.serext import Payload
+
+        class SerFlow(FlowSpec):
+            @step
+            def start(self):
+                self.art = Payload("hello")
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.art.text == "hello", self.art
+                self.ok = True
+
+        if __name__ == "__main__":
+            SerFlow()
+    """))
+    from .test_runtime import REPO
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.pathsep.join(
+        [str(tmp_path), REPO, env.get("PYTHONPATH", "")])
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet",
+         "--datastore-root", tmp_datastore, "run"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    # the artifact's recorded encoding is the extension codec
+    import json as _json
+
+    found = []
+    for root, _dirs, files in os.walk(tmp_datastore):
+        for fn in files:
+            if fn.endswith(".data.json"):
+                info = _json.load(open(os.path.join(root, fn)))
+                arts = info.get("artifacts", {})
+                if "art" in arts:
+                    found.append(arts["art"].get("encoding"))
+    assert "payload-v1" in found, found
