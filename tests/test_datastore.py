@@ -334,7 +334,7 @@ def test_serializer_extension_flow(tmp_path, tmp_datastore):
     found = []
     for root, _dirs, files in os.walk(tmp_datastore):
         for fn in files:
-            if fn.endswith(".data.json"):
+            if fn.endswith(".data"):
                 info = _json.load(open(os.path.join(root, fn)))
                 arts = info.get("artifacts", {})
                 if "art" in arts:
