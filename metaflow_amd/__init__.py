@@ -15,6 +15,11 @@ from .includefile import IncludeFile
 from .current import current
 from .unbounded_foreach import UnboundedForeachInput
 from .user_config import Config, ConfigValue, FlowMutator, MutableFlow
+from .user_decorators import (
+    USER_SKIP_STEP,
+    UserStepDecorator,
+    user_step_decorator,
+)
 from .plugins.retry_decorator import retry
 from .plugins.catch_decorator import catch
 from .plugins.timeout_decorator import timeout

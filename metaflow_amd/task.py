@@ -303,6 +303,14 @@ class MFXTask(object):
                 func = deco.task_decorate(
                     func, flow, self.graph, retry_count,
                     max_user_code_retries, ubf_context)
+            # user step wrappers (pre/post/skip semantics) nest OUTSIDE
+            # the plugin-decorator chain (reference task.py:67)
+            user_wrappers = getattr(step_func, "user_wrappers", None)
+            if user_wrappers:
+                from .user_decorators import apply_user_wrappers
+
+                func = apply_user_wrappers(func, user_wrappers,
+                                           step_name, self.graph)
 
             with monitor.measure("mfx.task.user_code"), \
                     monitor.count("mfx.task.runs"):

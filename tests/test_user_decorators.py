@@ -1,0 +1,282 @@
+"""UserStepDecorator subsystem: wrap, replace, skip, ordering, exception
+swallowing, generator form — run through real subprocess-driven flows
+(reference user_decorators/user_step_decorator.py semantics)."""
+
+import os
+import subprocess
+import sys
+import textwrap
+
+from .test_runtime import REPO, latest_run_id, read_artifact
+
+
+def _run_inline_flow(tmp_path, tmp_datastore, body, name="udeco_flow.py",
+                     check=True, env_extra=None):
+    flow = tmp_path / name
+    flow.write_text(textwrap.dedent(body))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    env.update(env_extra or {})
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet",
+         "--datastore-root", tmp_datastore, "run"],
+        env=env, capture_output=True, text=True, timeout=300)
+    if check:
+        assert proc.returncode == 0, proc.stderr[-3000:]
+    return proc
+
+
+def test_wrap_pre_post_and_ordering(tmp_path, tmp_datastore):
+    """pre/post hooks fire around the step; stacked wrappers nest with
+    the decorator closest to @step innermost."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, UserStepDecorator, step
+
+        class Tag(UserStepDecorator):
+            def init(self, **kw):
+                self.tag = kw["tag"]
+            def pre_step(self, step_name, flow, inputs=None):
+                flow.trace = getattr(flow, "trace", []) + \\
+                    ["pre-" + self.tag]
+            def post_step(self, step_name, flow, exception=None):
+                flow.trace = flow.trace + ["post-" + self.tag]
+                return exception
+
+        class OuterTag(Tag):
+            pass
+
+        class FlowA(FlowSpec):
+            @OuterTag(tag="outer")
+            @Tag(tag="inner")
+            @step
+            def start(self):
+                self.trace = self.trace + ["body"]
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.trace == [
+                    "pre-outer", "pre-inner", "body",
+                    "post-inner", "post-outer"], self.trace
+                self.ok = True
+
+        if __name__ == "__main__":
+            FlowA()
+    """)
+    run_id = latest_run_id(tmp_datastore, "FlowA")
+    assert read_artifact(tmp_datastore, "FlowA", run_id, "end", "ok")
+
+
+def test_replace_step(tmp_path, tmp_datastore):
+    """pre_step returning a callable replaces the step body."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, UserStepDecorator, step
+
+        class Replace(UserStepDecorator):
+            def pre_step(self, step_name, flow, inputs=None):
+                def other(flow_obj):
+                    flow_obj.who = "replacement"
+                    flow_obj.next(flow_obj.end)
+                return other
+
+        class FlowB(FlowSpec):
+            @Replace
+            @step
+            def start(self):
+                self.who = "original"
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.who == "replacement", self.who
+
+        if __name__ == "__main__":
+            FlowB()
+    """)
+
+
+def test_skip_step(tmp_path, tmp_datastore):
+    """skip_step=True skips the body and synthesizes the default
+    self.next() from the static graph."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, UserStepDecorator, step
+
+        class Skip(UserStepDecorator):
+            def pre_step(self, step_name, flow, inputs=None):
+                self.skip_step = True
+
+        class FlowC(FlowSpec):
+            @step
+            def start(self):
+                self.ran = ["start"]
+                self.next(self.middle)
+
+            @Skip
+            @step
+            def middle(self):
+                self.ran = self.ran + ["middle"]
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.ran == ["start"], self.ran
+                self.ok = True
+
+        if __name__ == "__main__":
+            FlowC()
+    """)
+    run_id = latest_run_id(tmp_datastore, "FlowC")
+    assert read_artifact(tmp_datastore, "FlowC", run_id, "end", "ok")
+
+
+def test_swallow_exception(tmp_path, tmp_datastore):
+    """post_step returning None swallows the step's exception; the
+    wrapper records it as an artifact and the run completes."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, UserStepDecorator, step
+
+        class Guard(UserStepDecorator):
+            def post_step(self, step_name, flow, exception=None):
+                flow.guarded = repr(exception) if exception else None
+                return None  # swallow
+
+        class FlowD(FlowSpec):
+            @Guard
+            @step
+            def start(self):
+                raise ValueError("boom")
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert "boom" in self.guarded, self.guarded
+                self.ok = True
+
+        if __name__ == "__main__":
+            FlowD()
+    """)
+    run_id = latest_run_id(tmp_datastore, "FlowD")
+    assert read_artifact(tmp_datastore, "FlowD", run_id, "end", "ok")
+
+
+def test_generator_form(tmp_path, tmp_datastore):
+    """@user_step_decorator generator: code before/after the yield runs
+    around the step; attributes arrive as the 4th argument."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, step, user_step_decorator
+
+        @user_step_decorator
+        def stamp(step_name, flow, inputs, attributes):
+            flow.pre_mark = attributes.get("mark", "?")
+            yield
+            flow.post_mark = flow.pre_mark + "-done"
+
+        class FlowE(FlowSpec):
+            @stamp(mark="m1")
+            @step
+            def start(self):
+                assert self.pre_mark == "m1"
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.post_mark == "m1-done", self.post_mark
+                self.ok = True
+
+        if __name__ == "__main__":
+            FlowE()
+    """)
+    run_id = latest_run_id(tmp_datastore, "FlowE")
+    assert read_artifact(tmp_datastore, "FlowE", run_id, "end", "ok")
+
+
+def test_generator_skip_and_catch(tmp_path, tmp_datastore):
+    """Generator protocol: yielding a dict skips the step; catching the
+    exception around the yield makes a failing step successful."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, step, user_step_decorator
+
+        @user_step_decorator
+        def skipper(step_name, flow, inputs):
+            yield {}
+
+        @user_step_decorator
+        def catcher(step_name, flow, inputs):
+            try:
+                yield
+            except RuntimeError as e:
+                flow.caught = str(e)
+
+        class FlowF(FlowSpec):
+            @step
+            def start(self):
+                self.ran = []
+                self.next(self.skipped)
+
+            @skipper
+            @step
+            def skipped(self):
+                self.ran = self.ran + ["skipped"]
+                self.next(self.fails)
+
+            @catcher
+            @step
+            def fails(self):
+                raise RuntimeError("caught-me")
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.ran == [], self.ran
+                assert self.caught == "caught-me", self.caught
+                self.ok = True
+
+        if __name__ == "__main__":
+            FlowF()
+    """)
+    run_id = latest_run_id(tmp_datastore, "FlowF")
+    assert read_artifact(tmp_datastore, "FlowF", run_id, "end", "ok")
+
+
+def test_wrapper_composes_with_catch_and_foreach(tmp_path, tmp_datastore):
+    """A user wrapper on a foreach CHILD composes with @catch: the
+    wrapper runs per-iteration and @catch still swallows a failure."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        from metaflow_amd import FlowSpec, UserStepDecorator, catch, step
+
+        class Count(UserStepDecorator):
+            def pre_step(self, step_name, flow, inputs=None):
+                flow.counted = True
+
+        class FlowG(FlowSpec):
+            @step
+            def start(self):
+                self.items = [0, 1, 2]
+                self.next(self.work, foreach="items")
+
+            @catch(var="err")
+            @Count
+            @step
+            def work(self):
+                assert self.counted
+                if self.input == 1:
+                    raise ValueError("boom")
+                self.val = self.input * 2
+                self.next(self.join)
+
+            @step
+            def join(self, inputs):
+                self.total = sum(getattr(i, "val", 0) for i in inputs)
+                self.n_failed = sum(
+                    1 for i in inputs if getattr(i, "err", None))
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.total == 4, self.total
+                assert self.n_failed == 1
+
+        if __name__ == "__main__":
+            FlowG()
+    """)
