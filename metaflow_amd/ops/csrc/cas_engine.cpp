@@ -10,6 +10,7 @@
 
 #include <torch/extension.h>
 
+#include <fcntl.h>
 #include <sys/stat.h>
 #include <unistd.h>
 
@@ -228,6 +229,69 @@ class Engine {
     return py::bytes(out);
   }
 
+  // Parallel pread straight into the result bytes object: one
+  // allocation, zero intermediate copies (the Python path pays
+  // f.read() + a header-slice copy, both single-threaded — the round-1
+  // profile had load at 2.45 GB/s vs save at 8). Chunked pread from the
+  // page cache scales with cores.
+  py::bytes load_blob_parallel(const std::string& path,
+                               size_t header_skip) {
+    struct stat st;
+    if (::stat(path.c_str(), &st) != 0)
+      throw std::runtime_error("cas_engine: missing " + path);
+    size_t total = (size_t)st.st_size;
+    if (total < header_skip)
+      throw std::runtime_error("cas_engine: short blob " + path);
+    size_t n = total - header_skip;
+    PyObject* obj = PyBytes_FromStringAndSize(nullptr, (Py_ssize_t)n);
+    if (!obj) throw std::runtime_error("cas_engine: alloc failed");
+    char* dst = PyBytes_AS_STRING(obj);
+    bool ok = true;
+    {
+      py::gil_scoped_release rel;
+      int fd = ::open(path.c_str(), O_RDONLY);
+      if (fd < 0) {
+        ok = false;
+      } else {
+        constexpr size_t kChunk = 16u << 20;
+        size_t nchunks = (n + kChunk - 1) / kChunk;
+        int nt = std::min<int>(threads_, (int)std::max<size_t>(nchunks,
+                                                               1));
+        std::atomic<size_t> next{0};
+        std::atomic<bool> failed{false};
+        auto worker = [&]() {
+          size_t i;
+          while (!failed.load(std::memory_order_relaxed) &&
+                 (i = next.fetch_add(1)) < nchunks) {
+            size_t off = i * kChunk;
+            size_t len = std::min(kChunk, n - off);
+            size_t done = 0;
+            while (done < len) {
+              ssize_t r = ::pread(fd, dst + off + done, len - done,
+                                  (off_t)(header_skip + off + done));
+              if (r <= 0) {
+                failed.store(true);
+                return;
+              }
+              done += (size_t)r;
+            }
+          }
+        };
+        std::vector<std::thread> pool;
+        for (int t = 0; t < nt; ++t) pool.emplace_back(worker);
+        for (auto& t : pool) t.join();
+        ::close(fd);
+        ok = !failed.load();
+      }
+    }
+    if (!ok) {
+      Py_DECREF(obj);
+      throw std::runtime_error("cas_engine: parallel read failed: " +
+                               path);
+    }
+    return py::reinterpret_steal<py::bytes>(obj);
+  }
+
   int threads() const { return threads_; }
 
  private:
@@ -245,5 +309,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("compute_key", &Engine::compute_key)
       .def("save_blob", &Engine::save_blob)
       .def("load_blob", &Engine::load_blob)
+      .def("load_blob_parallel", &Engine::load_blob_parallel)
       .def("threads", &Engine::threads);
 }
