@@ -471,11 +471,16 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
   constexpr int BKVB = 256, BQ2 = 64;
   // panel layout [8][BQ2][16] serves BOTH the A-fragment reads (contiguous
   // within a panel row) and the B-fragments via ds_read_b64_tr_b16 — no
-  // transposed copies, staging is pure vec8 (see panel helpers above)
-  __shared__ short qp[BQ2 * ATT_D];    // Q panels
-  __shared__ short dop[BQ2 * ATT_D];   // dO panels
-  __shared__ float lse_s[BQ2];
-  __shared__ float del_s[BQ2];
+  // transposed copies, staging is pure vec8 (see panel helpers above).
+  // DOUBLE-BUFFERED (T14 async-stage split, same structure as the fwd
+  // kernel): tile i+1's global loads issue before tile i's compute, the
+  // LDS writes land after, so staging latency hides under the MFMAs —
+  // the single-buffered version serialized stage→sync→compute and was
+  // the most expensive kernel of the step (20.5%).
+  __shared__ short qp[2][BQ2 * ATT_D];    // Q panels
+  __shared__ short dop[2][BQ2 * ATT_D];   // dO panels
+  __shared__ float lse_s[2][BQ2];
+  __shared__ float del_s[2][BQ2];
 
   const int kvb = blockIdx.x;
   const int hk = blockIdx.y;
@@ -504,32 +509,74 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
 
   const int jq0 = causal ? (kvb * BKVB) / BQ2 : 0;
   const int nq = S / BQ2;
-  for (int g = 0; g < G; ++g) {
-    const int h = hk * G + g;
-    const long long hoff = ((long long)b * H + h) * S;
-    for (int jq = jq0; jq < nq; ++jq) {
-      __syncthreads();
-      const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
-      const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
-      stage_panel<BQ2, BLOCK>(qp, qsrc, ATT_D);
-      stage_panel<BQ2, BLOCK>(dop, dsrc, ATT_D);
-      if (threadIdx.x < BQ2) {
-        lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
-        del_s[threadIdx.x] = delta[hoff + jq * BQ2 + threadIdx.x];
-      }
-      __syncthreads();
-      // wave-uniform skip: all of this wave's kv rows above every q row
-      if (causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
+  const int per_g = nq - jq0;      // q tiles per head
+  const int total = G * per_g;     // flattened (g, jq) iterations
 
+  // per-thread staging slices: BQ2*16 = 1024 vec8 slots, 2 per thread
+  const int tid = threadIdx.x;
+  bf16x8 stg_q[2], stg_d[2];
+  float stg_lse = 0.f, stg_del = 0.f;
+  int stg_r[2], stg_c8[2], stg_pel[2];
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = u * BLOCK + tid;
+    stg_r[u] = idx % BQ2;
+    stg_c8[u] = idx / BQ2;
+    stg_pel[u] = (stg_c8[u] >> 1) * (BQ2 * 16) + stg_r[u] * 16
+                 + (stg_c8[u] & 1) * 8;
+  }
+
+#define MFX_DKDV_LOAD(it)                                                  \
+  {                                                                        \
+    const int g_ = (it) / per_g;                                           \
+    const int jq_ = jq0 + (it) % per_g;                                    \
+    const long long hoff_ = ((long long)b * H + hk * G + g_) * S;          \
+    const short* qsrc_ = q + (hoff_ + (long long)jq_ * BQ2) * ATT_D;       \
+    const short* dsrc_ = dout + (hoff_ + (long long)jq_ * BQ2) * ATT_D;    \
+    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                        \
+      stg_q[u] = *(const bf16x8*)(qsrc_ + stg_r[u] * ATT_D                 \
+                                  + stg_c8[u] * 8);                        \
+      stg_d[u] = *(const bf16x8*)(dsrc_ + stg_r[u] * ATT_D                 \
+                                  + stg_c8[u] * 8);                        \
+    }                                                                      \
+    if (tid < BQ2) {                                                       \
+      stg_lse = lse[hoff_ + jq_ * BQ2 + tid];                              \
+      stg_del = delta[hoff_ + jq_ * BQ2 + tid];                            \
+    }                                                                      \
+  }
+
+#define MFX_DKDV_WRITE(buf)                                                \
+  {                                                                        \
+    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                        \
+      *(bf16x8*)(qp[buf] + stg_pel[u]) = stg_q[u];                         \
+      *(bf16x8*)(dop[buf] + stg_pel[u]) = stg_d[u];                        \
+    }                                                                      \
+    if (tid < BQ2) {                                                       \
+      lse_s[buf][tid] = stg_lse;                                           \
+      del_s[buf][tid] = stg_del;                                           \
+    }                                                                      \
+  }
+
+  MFX_DKDV_LOAD(0);
+  MFX_DKDV_WRITE(0);
+  __syncthreads();
+  for (int it = 0; it < total; ++it) {
+    const int cur = it & 1;
+    const int jq = jq0 + it % per_g;
+    if (it + 1 < total) MFX_DKDV_LOAD(it + 1);  // issue loads early
+
+    // wave-uniform skip: all of this wave's kv rows above every q row
+    // (staging + barrier below still run)
+    if (!(causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32)) {
       // per 32-q sub-tile: S, dP, P/dS, dV, dK — keeps live regs low
 #pragma unroll
       for (int t = 0; t < 2; ++t) {
         f16f st = (f16f){}, dpt = (f16f){};
 #pragma unroll
         for (int s = 0; s < 8; ++s) {
-          bf16x8 qf = frag8_panel<BQ2>(qp, t * 32 + l32,
+          bf16x8 qf = frag8_panel<BQ2>(qp[cur], t * 32 + l32,
                                        s * 16 + hi * 8);
-          bf16x8 df = frag8_panel<BQ2>(dop, t * 32 + l32,
+          bf16x8 df = frag8_panel<BQ2>(dop[cur], t * 32 + l32,
                                        s * 16 + hi * 8);
           bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
           bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
@@ -543,9 +590,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           const int qrg = jq * BQ2 + qrl;
           float p = 0.f;
           if (!causal || qrg >= my_kvrow)
-            p = __expf(st[r] * scale - lse_s[qrl]);
+            p = __expf(st[r] * scale - lse_s[cur][qrl]);
           st[r] = p;
-          dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
+          dpt[r] = p * (dpt[r] - del_s[cur][qrl]) * scale;
         }
         // ds_read_b64_tr_b16 semantics: each lane loads 64b at its OWN
         // address; the HW transposes 16-bit elements within each 16-lane
@@ -565,8 +612,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           for (int n = 0; n < 4; ++n) {
             const int base = n * 2 * (BQ2 * 16) + tr_panel
                              + (q0 + hi * 8) * 16 + tr_lane_off;
-            bfr[n].u[0] = tr_read(dop + base);
-            bfr[n].u[1] = tr_read(dop + base + 4 * 16);
+            bfr[n].u[0] = tr_read(dop[cur] + base);
+            bfr[n].u[1] = tr_read(dop[cur] + base + 4 * 16);
           }
           asm volatile("s_waitcnt lgkmcnt(0)");
           __builtin_amdgcn_sched_barrier(0);
@@ -584,8 +631,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
           for (int n = 0; n < 4; ++n) {
             const int base = n * 2 * (BQ2 * 16) + tr_panel
                              + (q0 + hi * 8) * 16 + tr_lane_off;
-            bfr[n].u[0] = tr_read(qp + base);
-            bfr[n].u[1] = tr_read(qp + base + 4 * 16);
+            bfr[n].u[0] = tr_read(qp[cur] + base);
+            bfr[n].u[1] = tr_read(qp[cur] + base + 4 * 16);
           }
           asm volatile("s_waitcnt lgkmcnt(0)");
           __builtin_amdgcn_sched_barrier(0);
@@ -595,7 +642,12 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
         }
       }
     }
+
+    if (it + 1 < total) MFX_DKDV_WRITE(cur ^ 1);
+    __syncthreads();
   }
+#undef MFX_DKDV_LOAD
+#undef MFX_DKDV_WRITE
 
   // epilogue: C row = kv local, col = d
 #pragma unroll
