@@ -46,6 +46,14 @@ MAX_NUM_SPLITS = from_conf("MAX_NUM_SPLITS", 100, int)
 MAX_LOG_SIZE = from_conf("MAX_LOG_SIZE", 1024 * 1024, int)
 POLL_TIMEOUT_MS = from_conf("POLL_TIMEOUT_MS", 1000, int)
 MAX_ATTEMPTS = from_conf("MAX_ATTEMPTS", 4, int)
+# scheduler liveness: kill a task whose wall-clock exceeds
+# TASK_STALL_TIMEOUT seconds (0 = disabled; @timeout is the opt-in
+# per-step variant INSIDE the task — this one catches tasks that can't
+# even run their own signal handler, e.g. a wedged RCCL rendezvous), or
+# whose heartbeat sidecar has gone silent for HEARTBEAT_TIMEOUT seconds
+# (only enforced once a first heartbeat was seen; 0 = disabled).
+TASK_STALL_TIMEOUT = from_conf("TASK_STALL_TIMEOUT", 0, float)
+HEARTBEAT_TIMEOUT = from_conf("HEARTBEAT_TIMEOUT", 600, float)
 
 # --- datastore ---------------------------------------------------------------
 DATASTORE_LOCAL_DIR = from_conf("DATASTORE_LOCAL_DIR", ".mfx")
