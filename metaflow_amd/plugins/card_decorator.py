@@ -108,10 +108,116 @@ def render_card(task_datastore, step_name, ok, builder=None):
     )
 
 
+def _telemetry_section(jsonl_path):
+    """Summarize the gpu_monitor sidecar's samples into an HTML table."""
+    import json as _json
+    import os as _os
+
+    if not _os.path.isfile(jsonl_path):
+        return None
+    mem, busy, n = [], [], 0
+    try:
+        with open(jsonl_path) as f:
+            for line in f:
+                try:
+                    rec = _json.loads(line)
+                except ValueError:
+                    continue
+                n += 1
+                if "mem_used_gb" in rec:
+                    mem.append(rec["mem_used_gb"])
+                smi = rec.get("rocm_smi") or {}
+                for dev in smi.values():
+                    if isinstance(dev, dict):
+                        use = dev.get("GPU use (%)")
+                        if use is not None:
+                            try:
+                                busy.append(float(use))
+                            except (TypeError, ValueError):
+                                pass
+    except OSError:
+        return None
+    if not n:
+        return None
+    rows = ["<tr><th>samples</th><td>%d</td></tr>" % n]
+    if mem:
+        rows.append("<tr><th>HBM used (GB)</th><td>max %.1f / mean %.1f"
+                    "</td></tr>" % (max(mem), sum(mem) / len(mem)))
+    if busy:
+        rows.append("<tr><th>GPU busy (%%)</th><td>max %.0f / mean %.0f"
+                    "</td></tr>" % (max(busy), sum(busy) / len(busy)))
+    return "<table>%s</table>" % "".join(rows)
+
+
+def rocprof_stats_section(stats_dir, top=15):
+    """Kernel-time breakdown table from a rocprofv3 --stats output dir
+    (any *kernel_stats.csv under it). Used by the runtime to splice a
+    profile section into the task card when @card(profile=True)."""
+    import csv
+    import glob
+    import os as _os
+
+    files = sorted(glob.glob(_os.path.join(stats_dir, "**",
+                                           "*kernel_stats.csv"),
+                             recursive=True))
+    if not files:
+        return None
+    rows = []
+    with open(files[-1]) as f:
+        reader = csv.DictReader(f)
+        for rec in reader:
+            keys = {k.lower().replace("_", "").replace(" ", ""): k
+                    for k in rec}
+            name = rec.get(keys.get("name", ""), "")
+            pct = rec.get(keys.get("percentage", ""), "")
+            total = rec.get(keys.get("totaldurationns", ""), "")
+            calls = rec.get(keys.get("calls", ""), "")
+            try:
+                total_ms = float(total) / 1e6
+            except (TypeError, ValueError):
+                total_ms = 0.0
+            rows.append((total_ms, pct, calls, name))
+    rows.sort(reverse=True)
+    out = ["<table><tr><th>total ms</th><th>%</th><th>calls</th>"
+           "<th>kernel</th></tr>"]
+    for total_ms, pct, calls, name in rows[:top]:
+        try:
+            pct_s = "%.1f" % float(pct)
+        except (TypeError, ValueError):
+            pct_s = str(pct)
+        out.append("<tr><td>%.1f</td><td>%s</td><td>%s</td>"
+                   "<td><code>%s</code></td></tr>"
+                   % (total_ms, pct_s, html.escape(str(calls)),
+                      html.escape(str(name)[:120])))
+    out.append("</table>")
+    return "\n".join(out)
+
+
 class CardDecorator(StepDecorator):
     name = "card"
-    defaults = {"id": "default"}
+    # profile=True reruns the task under `rocprofv3 --kernel-trace
+    # --stats` (when available on the box) and the scheduler splices the
+    # kernel-time breakdown into the card after the task exits;
+    # gpu_telemetry=True samples rocm-smi/HBM via the gpu_monitor
+    # sidecar for the card's telemetry section.
+    defaults = {"id": "default", "profile": False, "gpu_telemetry": True}
     allow_multiple = True
+
+    def runtime_step_cli(self, args, retry_count, max_user_code_retries,
+                         ubf_context):
+        if not self.attributes.get("profile"):
+            return
+        import shutil
+        import tempfile
+
+        rocprof = shutil.which("rocprofv3")
+        if rocprof is None:
+            return
+        out_dir = tempfile.mkdtemp(prefix="mfx_rocprof_")
+        args["cmd"] = [rocprof, "--kernel-trace", "--stats", "-d",
+                       out_dir, "--"] + args["cmd"]
+        args["env"]["MFX_ROCPROF_OUT"] = out_dir
+        args["env"].setdefault("TMPDIR", "/tmp")
 
     def task_pre_step(self, step_name, task_datastore, metadata, run_id,
                       task_id, flow, graph, retry_count,
@@ -120,7 +226,28 @@ class CardDecorator(StepDecorator):
 
         self._builder = CardBuilder()
         self._ds = task_datastore
+        self._telemetry_path = None
+        self._gpu_sidecar = None
         card_id = self.attributes.get("id", "default")
+        if self.attributes.get("gpu_telemetry", True):
+            try:
+                import torch
+
+                if torch.cuda.is_available():
+                    import tempfile
+
+                    from ..sidecar import SidecarSubProcess
+
+                    fd, path = tempfile.mkstemp(prefix="mfx_gpu_card_",
+                                                suffix=".jsonl")
+                    import os as _os
+
+                    _os.close(fd)
+                    self._telemetry_path = path
+                    self._gpu_sidecar = SidecarSubProcess(
+                        "gpu_monitor", {"out_path": path})
+            except Exception:
+                self._gpu_sidecar = None
 
         def _refresh(builder=self._builder, ds=task_datastore,
                      step=step_name, cid=card_id):
@@ -137,6 +264,15 @@ class CardDecorator(StepDecorator):
     def task_finished(self, step_name, flow, graph, is_task_ok, retry_count,
                       max_user_code_retries):
         try:
+            if self._gpu_sidecar is not None:
+                try:
+                    self._gpu_sidecar.terminate()
+                except Exception:
+                    pass
+            if self._telemetry_path:
+                section = _telemetry_section(self._telemetry_path)
+                if section:
+                    self._builder.append(section, title="GPU telemetry")
             html_doc = render_card(self._ds, step_name, is_task_ok,
                                    self._builder)
             self._ds.save_metadata(
@@ -151,4 +287,16 @@ card = make_step_decorator(CardDecorator)
 
 def get_card(task_datastore, card_id="default"):
     meta = task_datastore.load_metadata("card_%s" % card_id)
-    return meta["html"] if meta else None
+    if meta is None:
+        return None
+    doc = meta["html"]
+    # splice in the post-exit rocprof section when the scheduler saved
+    # one (@card(profile=True): the stats CSV only exists after the task
+    # process — and rocprofv3 around it — exited)
+    prof = task_datastore.load_metadata("card_profile")
+    if prof and prof.get("html") and "</body>" in doc:
+        doc = doc.replace(
+            "</body>",
+            "<h2>Kernel-time breakdown (rocprofv3)</h2>\n%s\n</body>"
+            % prof["html"])
+    return doc

@@ -594,6 +594,9 @@ class NativeRuntime(object):
             deco.runtime_step_cli(args_holder, spec.retry_count, max_retries,
                                   spec.ubf_context)
 
+        # keep decorator env mutations visible post-exit (e.g. the
+        # @card(profile=True) rocprof output dir)
+        spec.env = args_holder["env"]
         prefix = "[%s/%s]" % (spec.step, spec.task_id)
         if os.environ.get("MFX_DEBUG_SUBCOMMAND"):
             self._echo("exec: %s" % " ".join(shlex.quote(c)
@@ -695,6 +698,24 @@ class NativeRuntime(object):
                 source, worker.stderr_buf.get_bytes()))
         except Exception:
             pass
+
+        # @card(profile=True): the rocprofv3 stats CSV exists only now
+        # (written when the wrapped process exited); splice the kernel
+        # breakdown into the card via a side metadata record
+        prof_dir = (spec.env or {}).get("MFX_ROCPROF_OUT")
+        if prof_dir:
+            try:
+                from .plugins.card_decorator import rocprof_stats_section
+
+                section = rocprof_stats_section(prof_dir)
+                if section:
+                    prof_ds = self.flow_datastore.get_task_datastore(
+                        self.run_id, spec.step, spec.task_id,
+                        attempt=spec.retry_count, mode="w")
+                    prof_ds.save_metadata("card_profile",
+                                          {"html": section})
+            except Exception:
+                pass
 
         if rc == 0:
             if spec.gang is not None:

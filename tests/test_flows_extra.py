@@ -367,3 +367,43 @@ def test_gang_gpu_pinning(tmp_datastore):
     disjoint device pairs per rank."""
     run_flow("gpu_pin_flow.py", tmp_datastore, "run",
              env_extra={"MFX_NUM_GPUS": "4"})
+
+
+def test_rocprof_stats_section(tmp_path):
+    """Kernel-stats CSV -> HTML breakdown table (the @card(profile=True)
+    post-exit splice)."""
+    from metaflow_amd.plugins.card_decorator import rocprof_stats_section
+
+    d = tmp_path / "prof"
+    d.mkdir()
+    (d / "run_kernel_stats.csv").write_text(
+        '"Name","Calls","TotalDurationNs","AverageNs","Percentage",'
+        '"MinNs","MaxNs","StdDev"\n'
+        '"attn_fwd_v2_kernel",96,500000000,5208333,41.7,1,2,0.1\n'
+        '"Cijk_gemm_big",291,400000000,1374570,33.3,1,2,0.1\n'
+        '"tiny_kernel",10,1000,100,0.0,1,2,0.0\n')
+    html = rocprof_stats_section(str(d))
+    assert html is not None
+    assert "attn_fwd_v2_kernel" in html
+    assert html.index("attn_fwd_v2_kernel") < html.index("Cijk_gemm_big")
+    assert "500.0" in html  # total ms
+
+
+def test_card_profile_splice(tmp_path):
+    """get_card splices the scheduler-saved rocprof section into the
+    rendered card."""
+    from metaflow_amd.datastore import FlowDataStore
+    from metaflow_amd.datastore.storage import LocalStorage
+    from metaflow_amd.plugins.card_decorator import get_card
+
+    fds = FlowDataStore("CardFlow", LocalStorage(str(tmp_path)))
+    ds = fds.get_task_datastore("1", "s", "1", attempt=0, mode="w")
+    ds.init_task()
+    ds.save_metadata("card_default",
+                     {"html": "<html><body><h1>t</h1></body></html>"})
+    ds.save_metadata("card_profile", {"html": "<table>KERNELS</table>"})
+    ds.done()
+    rd = fds.get_task_datastore("1", "s", "1")
+    doc = get_card(rd)
+    assert "KERNELS" in doc
+    assert "Kernel-time breakdown" in doc
