@@ -456,36 +456,36 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v2_kernel(
 }
 
 // ===================== BACKWARD v2: dK/dV =================================
-// Grid (S/256, Hkv, B), 8 waves x 32 kv rows; K/V rows live in registers
-// (B-operands); loops over the G q-heads and q tiles of 64. C col = own
-// kv row, so P/dS columns feed col_to_afrags for the dV/dK MFMAs.
-// LDS: Q rm + Q^T + dO rm + dO^T (64 KiB) + lse/delta rows.
+// Grid (S/256, H, B) — one Q-HEAD per block, mirroring the dq kernel's
+// structure (which measures 2.2x more TF/s than the round-1 dkdv that
+// looped all G q-heads inside a (S/256, Hkv, B) grid: 4x fewer blocks
+// with a skewed causal trapezoid starved/imbalanced the chip). Each
+// block accumulates its head's (dK, dV) contribution in registers and
+// writes FP32 PARTIALS laid out [B, H, S, D]; dkdv_reduce_kernel sums
+// the G partials per kv head and converts to bf16 — numerically
+// identical to the old in-register fp32 accumulation across g. 8 waves
+// x 32 kv rows; K/V rows stay register-resident (the compiler hoists
+// them at the 256-VGPR budget, 0 spill — probed locally with
+// -Rpass-analysis=kernel-resource-usage). LDS: Q + dO panels (32 KiB).
 
 template <int BLOCK>
 __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dk, short* __restrict__ dv,
+    float* __restrict__ dk_part, float* __restrict__ dv_part,
     int B, int H, int Hkv, int S, float scale, int causal) {
   constexpr int BKVB = 256, BQ2 = 64;
-  // panel layout [8][BQ2][16] serves BOTH the A-fragment reads (contiguous
-  // within a panel row) and the B-fragments via ds_read_b64_tr_b16 — no
-  // transposed copies, staging is pure vec8 (see panel helpers above).
-  // DOUBLE-BUFFERED (T14 async-stage split, same structure as the fwd
-  // kernel): tile i+1's global loads issue before tile i's compute, the
-  // LDS writes land after, so staging latency hides under the MFMAs —
-  // the single-buffered version serialized stage→sync→compute and was
-  // the most expensive kernel of the step (20.5%).
-  __shared__ short qp[2][BQ2 * ATT_D];    // Q panels
-  __shared__ short dop[2][BQ2 * ATT_D];   // dO panels
-  __shared__ float lse_s[2][BQ2];
-  __shared__ float del_s[2][BQ2];
+  __shared__ short qp[BQ2 * ATT_D];    // Q panels
+  __shared__ short dop[BQ2 * ATT_D];   // dO panels
+  __shared__ float lse_s[BQ2];
+  __shared__ float del_s[BQ2];
 
   const int kvb = blockIdx.x;
-  const int hk = blockIdx.y;
+  const int h = blockIdx.y;           // q head
   const int b = blockIdx.z;
   const int G = H / Hkv;
+  const int hk = h / G;               // kv head this block feeds
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int l32 = lane & 31;
@@ -495,8 +495,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
   const long long kvoff =
       (((long long)b * Hkv + hk) * S + kvb * BKVB) * ATT_D;
 
-  // K rows live in registers (hot in every S^T mfma); V rows are
-  // re-read from L2 per q-tile — both in registers spills ~500 B/lane
+  // K/V rows for this wave's 32 kv rows; register-resident across the
+  // whole q loop at the 256-VGPR budget
   const short* krow = k + kvoff + (long long)(wid * 32 + l32) * ATT_D;
   const short* vrow = v + kvoff + (long long)(wid * 32 + l32) * ATT_D;
 
@@ -507,158 +507,148 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
     acc_dv[n] = (f16f){};
   }
 
+  const long long hoff = ((long long)b * H + h) * S;
   const int jq0 = causal ? (kvb * BKVB) / BQ2 : 0;
   const int nq = S / BQ2;
-  const int per_g = nq - jq0;      // q tiles per head
-  const int total = G * per_g;     // flattened (g, jq) iterations
-
-  // per-thread staging slices: BQ2*16 = 1024 vec8 slots, 2 per thread
-  const int tid = threadIdx.x;
-  bf16x8 stg_q[2], stg_d[2];
-  float stg_lse = 0.f, stg_del = 0.f;
-  int stg_r[2], stg_c8[2], stg_pel[2];
-#pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    const int idx = u * BLOCK + tid;
-    stg_r[u] = idx % BQ2;
-    stg_c8[u] = idx / BQ2;
-    stg_pel[u] = (stg_c8[u] >> 1) * (BQ2 * 16) + stg_r[u] * 16
-                 + (stg_c8[u] & 1) * 8;
-  }
-
-#define MFX_DKDV_LOAD(it)                                                  \
-  {                                                                        \
-    const int g_ = (it) / per_g;                                           \
-    const int jq_ = jq0 + (it) % per_g;                                    \
-    const long long hoff_ = ((long long)b * H + hk * G + g_) * S;          \
-    const short* qsrc_ = q + (hoff_ + (long long)jq_ * BQ2) * ATT_D;       \
-    const short* dsrc_ = dout + (hoff_ + (long long)jq_ * BQ2) * ATT_D;    \
-    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                        \
-      stg_q[u] = *(const bf16x8*)(qsrc_ + stg_r[u] * ATT_D                 \
-                                  + stg_c8[u] * 8);                        \
-      stg_d[u] = *(const bf16x8*)(dsrc_ + stg_r[u] * ATT_D                 \
-                                  + stg_c8[u] * 8);                        \
-    }                                                                      \
-    if (tid < BQ2) {                                                       \
-      stg_lse = lse[hoff_ + jq_ * BQ2 + tid];                              \
-      stg_del = delta[hoff_ + jq_ * BQ2 + tid];                            \
-    }                                                                      \
-  }
-
-#define MFX_DKDV_WRITE(buf)                                                \
-  {                                                                        \
-    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                        \
-      *(bf16x8*)(qp[buf] + stg_pel[u]) = stg_q[u];                         \
-      *(bf16x8*)(dop[buf] + stg_pel[u]) = stg_d[u];                        \
-    }                                                                      \
-    if (tid < BQ2) {                                                       \
-      lse_s[buf][tid] = stg_lse;                                           \
-      del_s[buf][tid] = stg_del;                                           \
-    }                                                                      \
-  }
-
-  MFX_DKDV_LOAD(0);
-  MFX_DKDV_WRITE(0);
-  __syncthreads();
-  for (int it = 0; it < total; ++it) {
-    const int cur = it & 1;
-    const int jq = jq0 + it % per_g;
-    if (it + 1 < total) MFX_DKDV_LOAD(it + 1);  // issue loads early
-
+  for (int jq = jq0; jq < nq; ++jq) {
+    __syncthreads();
+    const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
+    const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
+    stage_panel<BQ2, BLOCK>(qp, qsrc, ATT_D);
+    stage_panel<BQ2, BLOCK>(dop, dsrc, ATT_D);
+    if (threadIdx.x < BQ2) {
+      lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
+      del_s[threadIdx.x] = delta[hoff + jq * BQ2 + threadIdx.x];
+    }
+    __syncthreads();
     // wave-uniform skip: all of this wave's kv rows above every q row
-    // (staging + barrier below still run)
-    if (!(causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32)) {
-      // per 32-q sub-tile: S, dP, P/dS, dV, dK — keeps live regs low
+    if (causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
+
+    // per 32-q sub-tile: S, dP, P/dS, dV, dK — keeps live regs low
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
-        f16f st = (f16f){}, dpt = (f16f){};
+    for (int t = 0; t < 2; ++t) {
+      f16f st = (f16f){}, dpt = (f16f){};
 #pragma unroll
-        for (int s = 0; s < 8; ++s) {
-          bf16x8 qf = frag8_panel<BQ2>(qp[cur], t * 32 + l32,
-                                       s * 16 + hi * 8);
-          bf16x8 df = frag8_panel<BQ2>(dop[cur], t * 32 + l32,
-                                       s * 16 + hi * 8);
-          bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
-          bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
-          st = mfma32(qf, kr, st);
-          dpt = mfma32(df, vr, dpt);
+      for (int s = 0; s < 8; ++s) {
+        bf16x8 qf = frag8_panel<BQ2>(qp, t * 32 + l32, s * 16 + hi * 8);
+        bf16x8 df = frag8_panel<BQ2>(dop, t * 32 + l32, s * 16 + hi * 8);
+        bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
+        bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
+        st = mfma32(qf, kr, st);
+        dpt = mfma32(df, vr, dpt);
+      }
+      // P (into st) with causal mask q >= kv; dS (into dpt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qrg = jq * BQ2 + qrl;
+        float p = 0.f;
+        if (!causal || qrg >= my_kvrow)
+          p = __expf(st[r] * scale - lse_s[qrl]);
+        st[r] = p;
+        dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
+      }
+      // ds_read_b64_tr_b16 semantics: each lane loads 64b at its OWN
+      // address; the HW transposes 16-bit elements within each 16-lane
+      // group (a [4][16] tile). To make lane l=32*hi+l32 RECEIVE rows
+      // q0+hi*8+j at panel column n*32+l32, lane l must LOAD the 4
+      // elements of row q0+hi*8+((l>>2)&3) at col offset 4*(l&3):
+      const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
+      const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
+      // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d)
+      bf16x8 pa[2];
+      col_to_afrags1(st, pa, hi);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int q0 = t * 32 + ks * 16;
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int base = n * 2 * (BQ2 * 16) + tr_panel
+                           + (q0 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(dop + base);
+          bfr[n].u[1] = tr_read(dop + base + 4 * 16);
         }
-        // P (into st) with causal mask q >= kv; dS (into dpt)
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const int qrg = jq * BQ2 + qrl;
-          float p = 0.f;
-          if (!causal || qrg >= my_kvrow)
-            p = __expf(st[r] * scale - lse_s[cur][qrl]);
-          st[r] = p;
-          dpt[r] = p * (dpt[r] - del_s[cur][qrl]) * scale;
+        for (int n = 0; n < 4; ++n)
+          acc_dv[n] = mfma32(pa[ks], bfr[n].v, acc_dv[n]);
+      }
+      // dK(32kv x 128d) += dS^T(32kv x 32q) @ Q(32q x 128d)
+      col_to_afrags1(dpt, pa, hi);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int q0 = t * 32 + ks * 16;
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int base = n * 2 * (BQ2 * 16) + tr_panel
+                           + (q0 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(qp + base);
+          bfr[n].u[1] = tr_read(qp + base + 4 * 16);
         }
-        // ds_read_b64_tr_b16 semantics: each lane loads 64b at its OWN
-        // address; the HW transposes 16-bit elements within each 16-lane
-        // group (a [4][16] tile). To make lane l=32*hi+l32 RECEIVE rows
-        // q0+hi*8+j at panel column n*32+l32, lane l must LOAD the 4
-        // elements of row q0+hi*8+((l>>2)&3) at col offset 4*(l&3):
-        const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
-        const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
-        // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d)
-        bf16x8 pa[2];
-        col_to_afrags1(st, pa, hi);
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          const int q0 = t * 32 + ks * 16;
-          union { unsigned long long u[2]; bf16x8 v; } bfr[4];
-#pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int base = n * 2 * (BQ2 * 16) + tr_panel
-                             + (q0 + hi * 8) * 16 + tr_lane_off;
-            bfr[n].u[0] = tr_read(dop[cur] + base);
-            bfr[n].u[1] = tr_read(dop[cur] + base + 4 * 16);
-          }
-          asm volatile("s_waitcnt lgkmcnt(0)");
-          __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-          for (int n = 0; n < 4; ++n)
-            acc_dv[n] = mfma32(pa[ks], bfr[n].v, acc_dv[n]);
-        }
-        // dK(32kv x 128d) += dS^T(32kv x 32q) @ Q(32q x 128d)
-        col_to_afrags1(dpt, pa, hi);
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          const int q0 = t * 32 + ks * 16;
-          union { unsigned long long u[2]; bf16x8 v; } bfr[4];
-#pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int base = n * 2 * (BQ2 * 16) + tr_panel
-                             + (q0 + hi * 8) * 16 + tr_lane_off;
-            bfr[n].u[0] = tr_read(qp[cur] + base);
-            bfr[n].u[1] = tr_read(qp[cur] + base + 4 * 16);
-          }
-          asm volatile("s_waitcnt lgkmcnt(0)");
-          __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-          for (int n = 0; n < 4; ++n)
-            acc_dk[n] = mfma32(pa[ks], bfr[n].v, acc_dk[n]);
-        }
+        for (int n = 0; n < 4; ++n)
+          acc_dk[n] = mfma32(pa[ks], bfr[n].v, acc_dk[n]);
       }
     }
-
-    if (it + 1 < total) MFX_DKDV_WRITE(cur ^ 1);
-    __syncthreads();
   }
-#undef MFX_DKDV_LOAD
-#undef MFX_DKDV_WRITE
 
-  // epilogue: C row = kv local, col = d
+  // epilogue: fp32 partials at [b, h, kv row, d]
+  const long long poff = (hoff + (long long)kvb * BKVB) * ATT_D;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
     const long long obase =
-        kvoff + (long long)(wid * 32 + row_local) * ATT_D;
+        poff + (long long)(wid * 32 + row_local) * ATT_D;
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
-      dk[obase + n * 32 + l32] = f2bf(acc_dk[n][r]);
-      dv[obase + n * 32 + l32] = f2bf(acc_dv[n][r]);
+      dk_part[obase + n * 32 + l32] = acc_dk[n][r];
+      dv_part[obase + n * 32 + l32] = acc_dv[n][r];
     }
+  }
+}
+
+// Sum the G per-q-head fp32 partials into the kv head's bf16 (dK, dV).
+// Memory-bound; vectorized f32x4 loads, grid-stride.
+__global__ void dkdv_reduce_kernel(const float* __restrict__ dk_part,
+                                   const float* __restrict__ dv_part,
+                                   short* __restrict__ dk,
+                                   short* __restrict__ dv,
+                                   long long n4_kv, int G,
+                                   long long head_elems) {
+  // n4_kv = B*Hkv*S*D/4. Flat kv vec4-index i4 -> kv slot (b*Hkv + hk);
+  // its G partials live at heads slot*G + g of the [B, H, S, D] buffer
+  // (h = hk*G + g).
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i4 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i4 < n4_kv; i4 += stride) {
+    const long long i = i4 * 4;
+    const long long slot = i / head_elems;
+    const long long rem = i - slot * head_elems;
+    const float* kp = dk_part + slot * G * head_elems + rem;
+    const float* vp = dv_part + slot * G * head_elems + rem;
+    f32x4 ks = *(const f32x4*)kp;
+    f32x4 vs = *(const f32x4*)vp;
+    for (int g = 1; g < G; ++g) {
+      const f32x4 k2 = *(const f32x4*)(kp + (long long)g * head_elems);
+      const f32x4 v2 = *(const f32x4*)(vp + (long long)g * head_elems);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        ks[j] += k2[j];
+        vs[j] += v2[j];
+      }
+    }
+    bf16x4 ko, vo;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      ko[j] = f2bf(ks[j]);
+      vo[j] = f2bf(vs[j]);
+    }
+    *(bf16x4*)(dk + i) = ko;
+    *(bf16x4*)(dv + i) = vo;
   }
 }

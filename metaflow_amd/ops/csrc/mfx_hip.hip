@@ -322,12 +322,30 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   HIP_CHECK_KERNEL();
   }
   TORCH_CHECK(S % 256 == 0, "attention kernel requires seqlen % 256 == 0");
-  dim3 gkv(S / 256, Hkv, B);
+  // dkdv: one q-head per block into fp32 partials [B,H,S,D], then a
+  // memory-bound reduction over the G partials per kv head (see
+  // attention_v2.hip rationale — 4x the workgroups of the per-kv-head
+  // grid and balanced causal trapezoids)
+  auto f32opts = q.options().dtype(torch::kFloat32);
+  auto dk_part = torch::empty({B, H, S, 128}, f32opts);
+  auto dv_part = torch::empty({B, H, S, 128}, f32opts);
+  dim3 gkv(S / 256, H, B);
   attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
       bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-      delta.data_ptr<float>(), bfm(dk), bfm(dv), B, H, Hkv, S,
-      (float)scale, causal ? 1 : 0);
+      delta.data_ptr<float>(), dk_part.data_ptr<float>(),
+      dv_part.data_ptr<float>(), B, H, Hkv, S, (float)scale,
+      causal ? 1 : 0);
   HIP_CHECK_KERNEL();
+  {
+    const long long n4_kv = (long long)B * Hkv * S * 128 / 4;
+    const long long head_elems = (long long)S * 128;
+    int blocks = (int)std::min<long long>(
+        (n4_kv + kBlock - 1) / kBlock, kGridCap);
+    dkdv_reduce_kernel<<<blocks, kBlock, 0, cur_stream()>>>(
+        dk_part.data_ptr<float>(), dv_part.data_ptr<float>(), bfm(dk),
+        bfm(dv), n4_kv, H / Hkv, head_elems);
+    HIP_CHECK_KERNEL();
+  }
   dim3 gq(S / 256, H, B);
   attn_bwd_dq_v2_kernel<512><<<gq, 512, 0, cur_stream()>>>(
       bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
