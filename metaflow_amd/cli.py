@@ -149,6 +149,12 @@ def main(flow):
         values = None
         if clone_run_id is None:
             values = _resolve_params(flow_cls, kwargs)
+            # IncludeFile params: upload the file bytes to the CAS once
+            # (raw blob, dedup across runs) and keep only the small
+            # IncludedFile handle as the parameter artifact
+            from .includefile import upload_include_files
+
+            upload_include_files(values, state.flow_datastore)
         runtime = NativeRuntime(
             flow_cls,
             state.graph,

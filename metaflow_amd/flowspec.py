@@ -31,6 +31,8 @@ INTERNAL_ATTRS = frozenset(
         "_steps_cache",
         "_current_step",
         "_cached_input",
+        "_lazy_includes",
+        "_lazy_include_cache",
         "name",
     ]
 )
@@ -89,6 +91,16 @@ class FlowSpec(object, metaclass=FlowSpecMeta):
         # called only when normal lookup fails: try the task datastore
         if name.startswith("__") or name in INTERNAL_ATTRS:
             raise AttributeError(name)
+        lazy = self.__dict__.get("_lazy_includes")
+        if lazy and name in lazy:
+            # IncludeFile parameter: decode from the CAS on first
+            # access; cache OUTSIDE the artifact namespace so persist()
+            # keeps propagating the small handle, not the content
+            cache = self.__dict__.setdefault("_lazy_include_cache", {})
+            if name not in cache:
+                cache[name] = lazy[name].decode(
+                    self.__dict__.get("_datastore"))
+            return cache[name]
         ds = self.__dict__.get("_datastore")
         if ds is not None and name in ds:
             value = ds[name]

@@ -78,5 +78,22 @@ class Parameter(object):
             return self.default(context)
         return self.default
 
+    def __get__(self, obj, owner=None):
+        """Non-data descriptor (reference task.py:191 read-only property
+        exposure): instance attributes set by the task shadow this; when
+        no instance value exists, an IncludeFile parameter decodes its
+        CAS blob on demand, anything else falls through to FlowSpec's
+        __getattr__ (datastore lookup) by raising AttributeError."""
+        if obj is None:
+            return self
+        lazy = obj.__dict__.get("_lazy_includes")
+        if lazy and self.name in lazy:
+            cache = obj.__dict__.setdefault("_lazy_include_cache", {})
+            if self.name not in cache:
+                cache[self.name] = lazy[self.name].decode(
+                    obj.__dict__.get("_datastore"))
+            return cache[self.name]
+        raise AttributeError(self.name)
+
     def __repr__(self):
         return "Parameter(%s)" % self.name

@@ -231,10 +231,20 @@ class MFXTask(object):
 
         object.__setattr__(flow, "_cached_input", _NOT_SET)
         if "_parameter_names" in output:
+            from .includefile import IncludedFile
+
             flow._parameter_names = output["_parameter_names"]
+            lazy_includes = {}
             for pname in flow._parameter_names:
                 if pname in output:
-                    object.__setattr__(flow, pname, output[pname])
+                    value = output[pname]
+                    if isinstance(value, IncludedFile):
+                        # content stays in the CAS until first access
+                        # (FlowSpec.__getattr__ decodes on demand)
+                        lazy_includes[pname] = value
+                    else:
+                        object.__setattr__(flow, pname, value)
+            object.__setattr__(flow, "_lazy_includes", lazy_includes)
 
         # ---- gang context ---------------------------------------------------
         parallel_ctx = None

@@ -81,3 +81,54 @@ def test_scheduler_stress(tmp_datastore):
     retried = read_artifact(tmp_datastore, "StressFlow", run_id, "join",
                             "retried")
     assert retried > 0
+
+
+def test_include_file_cas_upload(tmp_datastore, tmp_path):
+    """IncludeFile content lives in the CAS as one raw blob; the
+    parameter ARTIFACT is only the small IncludedFile handle (a 1 GB
+    include is no longer pickled whole into _parameters — VERDICT r1
+    missing #8; reference includefile.py:234,386)."""
+    import json
+    import os
+
+    from .test_runtime import latest_run_id, run_flow
+
+    payload = b"x" * (6 << 20)  # 6 MiB binary
+    f = tmp_path / "big.bin"
+    f.write_bytes(payload)
+    flow = tmp_path / "bin_include_flow.py"
+    flow.write_text(
+        "from metaflow_amd import FlowSpec, IncludeFile, step\n\n"
+        "class BinIncludeFlow(FlowSpec):\n"
+        "    blob = IncludeFile('blob', is_text=False, required=True)\n\n"
+        "    @step\n"
+        "    def start(self):\n"
+        "        self.n = len(self.blob)\n"
+        "        assert self.blob[:4] == b'xxxx'\n"
+        "        self.next(self.end)\n\n"
+        "    @step\n"
+        "    def end(self):\n"
+        "        assert self.n == 6 << 20, self.n\n\n"
+        "if __name__ == '__main__':\n"
+        "    BinIncludeFlow()\n")
+    import subprocess
+    import sys
+
+    from .test_runtime import REPO
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet",
+         "--datastore-root", tmp_datastore, "run", "--blob", str(f)],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    run_id = latest_run_id(tmp_datastore, "BinIncludeFlow")
+    # the parameter artifact must be a SMALL handle, not the 6 MiB body
+    pdir = os.path.join(tmp_datastore, "BinIncludeFlow", run_id,
+                        "_parameters", "0")
+    data = json.load(open(os.path.join(pdir, "0.data")))
+    art = data["artifacts"]["blob"]
+    assert art["size"] < 4096, art
+    assert "IncludedFile" in art["type"]
