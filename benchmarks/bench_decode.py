@@ -5,10 +5,8 @@
         [--prompt 512] [--new 128]
 
 Measures prefill time and per-token decode rate separately. The decode
-attention is currently the torch cached-attention path
-(models/llama.py: _attn_with_cache) — a fused decode kernel
-(paged/flash-decode over the KV cache) is the round-2 serving lever;
-this bench is its baseline.
+attention runs the split-K flash-decode kernel (ops/csrc/decode.hip)
+straight over the cache allocation; prefill runs the flash fwd kernel.
 """
 
 import argparse

@@ -115,6 +115,11 @@ class DecoderLayer(nn.Module):
             if pos == 0:
                 o = K.attention(q, k, v)      # prefill: flash (padded
                 # to the 256 tile internally for unaligned prompts)
+            elif S == 1:
+                # one-token decode: split-K flash-decode straight over
+                # the cache allocation (decode.hip)
+                o = K.attn_decode(q, kc, vc, pos + 1,
+                                  1.0 / math.sqrt(hd))
             else:
                 o = _attn_with_cache(q, kc[:, :, :pos + S],
                                      vc[:, :, :pos + S],

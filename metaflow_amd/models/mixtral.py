@@ -265,6 +265,11 @@ class MixtralDecoderLayer(nn.Module):
             if pos == 0:
                 o = K.attention(q, kk, v)  # flash prefill (any S; the
                 # wrapper pads unaligned seqlens to the 256 tile)
+            elif S == 1:
+                import math
+
+                o = K.attn_decode(q.contiguous(), kc, vc, pos + 1,
+                                  1.0 / math.sqrt(cfg.head_dim))
             else:
                 import math
 

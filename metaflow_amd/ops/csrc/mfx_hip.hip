@@ -12,6 +12,7 @@
 #include "cross_entropy.hip"
 #include "attention.hip"
 #include "attention_v2.hip"
+#include "decode.hip"
 #include "debug_kernels.hip"
 
 namespace {
@@ -355,6 +356,39 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   return {dq, dk, dv};
 }
 
+// ---------------------------------------------------------- flash-decode
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc,
+                          torch::Tensor vc, long L, double scale) {
+  check_bf16(q, "q");
+  check_bf16(kc, "kc");
+  check_bf16(vc, "vc");
+  const int B = q.size(0), H = q.size(1);
+  const int Hkv = kc.size(1), Lmax = kc.size(2);
+  TORCH_CHECK(q.size(2) == 1 && q.size(3) == 128,
+              "decode expects q [B,H,1,128]");
+  TORCH_CHECK(kc.size(3) == 128 && L >= 1 && L <= Lmax);
+  TORCH_CHECK(H % Hkv == 0);
+  // enough splits to fill 256 CUs at small B*H, chunks >= ~256 rows
+  int splits = (int)std::max<long>(1, 1024 / ((long)B * H));
+  splits = (int)std::min<long>(splits, (L + 255) / 256);
+  splits = std::max(splits, 1);
+  auto f32 = q.options().dtype(torch::kFloat32);
+  auto o_part = torch::empty({splits, B, H, 128}, f32);
+  auto ml_part = torch::empty({splits, B, H, 2}, f32);
+  auto o = torch::empty_like(q);
+  dim3 grid(splits, H, B);
+  attn_decode_partial_kernel<<<grid, 256, 0, cur_stream()>>>(
+      bf(q), bf(kc), bf(vc), o_part.data_ptr<float>(),
+      ml_part.data_ptr<float>(), B, H, Hkv, Lmax, (int)L, splits,
+      (float)scale);
+  HIP_CHECK_KERNEL();
+  attn_decode_merge_kernel<<<B * H, 64, 0, cur_stream()>>>(
+      o_part.data_ptr<float>(), ml_part.data_ptr<float>(), bfm(o), B, H,
+      splits);
+  HIP_CHECK_KERNEL();
+  return o;
+}
+
 }  // namespace
 
 
@@ -406,6 +440,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused QKV rope backward -> GEMM-grad layout");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (causal, GQA)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward");
+  m.def("attn_decode", &attn_decode,
+        "split-K flash-decode over the KV cache (q [B,H,1,128])");
   m.def("dbg_st", &dbg_st, "debug S^T path");
   m.def("dbg_dv", &dbg_dv, "debug dV path");
   m.def("dbg_mfma32", &dbg_mfma32, "mfma 32x32x16 layout probe");

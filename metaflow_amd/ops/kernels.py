@@ -457,6 +457,22 @@ def attn_bwd_raw(q, k, v, o, dout, lse, scale, causal=True):
     return dq, dk, dv
 
 
+def attn_decode(q, k_cache, v_cache, length, scale):
+    """Split-K flash-decode: one query row per (b, h) against the first
+    ``length`` positions of the preallocated KV cache (no slicing
+    copies — the kernel reads at the cache's full stride).
+
+    q: [B, H, 1, 128] bf16; k_cache/v_cache: [B, Hkv, Lmax, 128] bf16.
+    GPU -> decode.hip (grid (splits, H, B), memory-bound streaming);
+    CPU -> fp32 reference.
+    """
+    if q.is_cuda:
+        return hip_ext().attn_decode(q.contiguous(), k_cache, v_cache,
+                                     length, scale)
+    return attention_ref(q, k_cache[:, :, :length],
+                         v_cache[:, :, :length], scale, causal=False)
+
+
 # ================================ adam =====================================
 def adamw_step(p, g, m, v, step, lr, beta1=0.9, beta2=0.95, eps=1e-8,
                weight_decay=0.1, master=None, grad_scale=1.0):

@@ -402,3 +402,30 @@ def test_tuned_linear_gpu(dev, tmp_path, monkeypatch):
         assert rel_err(w.grad, wr.grad) < 2e-2, "dw"
     finally:
         gemm.reset_tune_table()
+
+
+@pytest.mark.parametrize("B,H,Hkv,L", [
+    (1, 4, 2, 257),     # unaligned cache length, GQA
+    (2, 8, 8, 1000),    # MHA, bigger cache
+    (1, 4, 1, 64),      # small L (single split)
+])
+def test_attn_decode(dev, B, H, Hkv, L):
+    """Split-K flash-decode vs the fp32 reference: one query per head
+    against the first L cache positions (full cache stride, GQA)."""
+    import math
+
+    from metaflow_amd.ops import kernels as K
+
+    torch.manual_seed(0)
+    Lmax = L + 37  # cache longer than the valid prefix
+    q = torch.randn(B, H, 1, 128, dtype=torch.bfloat16, device=dev)
+    kc = torch.randn(B, Hkv, Lmax, 128, dtype=torch.bfloat16,
+                     device=dev)
+    vc = torch.randn(B, Hkv, Lmax, 128, dtype=torch.bfloat16,
+                     device=dev)
+    scale = 1.0 / math.sqrt(128)
+    o = K.attn_decode(q, kc, vc, L, scale)
+    ref = K.attention_ref(q, kc[:, :, :L], vc[:, :, :L], scale,
+                          causal=False)
+    assert o.shape == (B, H, 1, 128)
+    assert rel_err(o, ref) < 2e-2
