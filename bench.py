@@ -29,7 +29,8 @@ def parse_args():
     p.add_argument("--seq", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3-8b",
                    choices=["llama3-8b", "llama3-70b", "tiny",
-                            "mixtral-8x7b", "mixtral-tiny"])
+                            "mixtral-8x7b", "mixtral-2x7b",
+                            "mixtral-tiny"])
     p.add_argument("--bucket-mb", type=int, default=64)
     p.add_argument("--cp", action="store_true",
                    help="context parallelism: all ranks form one ring-"
@@ -93,6 +94,7 @@ def main():
         "llama3-70b": (LlamaConfig.llama3_70b, LlamaForCausalLM),
         "tiny": (LlamaConfig.tiny, LlamaForCausalLM),
         "mixtral-8x7b": (MixtralConfig.mixtral_8x7b, MixtralForCausalLM),
+        "mixtral-2x7b": (MixtralConfig.mixtral_2x7b, MixtralForCausalLM),
         "mixtral-tiny": (MixtralConfig.tiny, MixtralForCausalLM),
     }
     cfg_fn, model_cls = model_factories[args.model]

@@ -39,6 +39,17 @@ class MixtralConfig:
         return cls()
 
     @classmethod
+    def mixtral_2x7b(cls):
+        """Same per-layer/per-expert geometry as 8x7b but 2 experts
+        (top-1): ~13.5B params — the largest Mixtral shape whose
+        12 B/param training state (bf16 p+g, fp32 m+v) fits ONE
+        MI355X's 288 GB, so the MoE hot path (router, grouped expert
+        GEMMs, token scatter/gather) is MEASURABLE on a 1-GPU box.
+        Honest label: this is NOT config 5's 8x7b — the full model needs
+        the 8-GPU expert-parallel run (ep_group all-to-all path)."""
+        return cls(num_experts=2, top_k=1)
+
+    @classmethod
     def tiny(cls, vocab=1024, seq=256):
         return cls(vocab_size=vocab, hidden_size=256, intermediate_size=512,
                    num_layers=2, num_heads=2, num_kv_heads=1, head_dim=128,
