@@ -1,0 +1,94 @@
+"""FP8 (OCP E4M3) training linear with delayed scaling — gfx950-native.
+
+gfx950's matrix cores run E4M3 at 2x the bf16 rate; hipBLASLt exposes it
+via HIP_R_8F_E4M3 (ops/csrc/gemm_lt.hip: gemm_lt_fp8, fp32 accumulate,
+descale folded into alpha). This module implements the standard
+delayed-scaling recipe (per-tensor scale from an amax HISTORY so the
+cast is a single fused pass without a pre-scan of the current tensor):
+
+  * forward GEMM in E4M3 (x and w quantized with their delayed scales);
+  * backward GEMMs in bf16 (dgrad/wgrad keep bf16 accuracy — the
+    standard first rung of fp8 training);
+  * amax history window (default 16 steps) updated per forward.
+
+``Fp8Linear`` is a drop-in for models.llama.Linear; CPU (and missing
+extension) falls back to bf16 F.linear so the module stays testable
+everywhere. E4M3 max normal = 448.
+"""
+
+import torch
+
+E4M3_MAX = 448.0
+
+
+def _gemm_ext():
+    from . import _mfx_gemm
+
+    return _mfx_gemm
+
+
+def quantize_e4m3(t, scale):
+    """bf16/fp32 -> float8_e4m3fn storage viewed as uint8 (saturating)."""
+    x = (t.float() * scale).clamp(-E4M3_MAX, E4M3_MAX)
+    return x.to(torch.float8_e4m3fn).view(torch.uint8)
+
+
+class _Fp8LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, scale_x, scale_w):
+        ctx.save_for_backward(x, w)
+        K = x.shape[-1]
+        M = x.numel() // K
+        x8 = quantize_e4m3(x.reshape(M, K), scale_x)
+        w8 = quantize_e4m3(w, scale_w)
+        out = _gemm_ext().fp8(x8, w8, 1.0 / (scale_x * scale_w))
+        return out.reshape(*x.shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        K = x.shape[-1]
+        M = x.numel() // K
+        dy2 = dy.reshape(M, -1)
+        dx = (dy2 @ w).reshape(x.shape)
+        dw = dy2.t() @ x.reshape(M, K)
+        return dx, dw, None, None
+
+
+class Fp8Linear(torch.nn.Module):
+    """Bias-free linear with an E4M3 forward GEMM and delayed scaling."""
+
+    def __init__(self, din, dout, dtype=torch.bfloat16, history=16):
+        super().__init__()
+        self.weight = torch.nn.Parameter(
+            torch.empty(dout, din, dtype=dtype))
+        self.history = history
+        self.register_buffer(
+            "amax_x", torch.zeros(history), persistent=False)
+        self.register_buffer(
+            "amax_w", torch.zeros(history), persistent=False)
+        self._step = 0
+
+    def _scales(self):
+        ax = float(self.amax_x.max())
+        aw = float(self.amax_w.max())
+        sx = E4M3_MAX / ax if ax > 0 else 1.0
+        sw = E4M3_MAX / aw if aw > 0 else 1.0
+        return sx, sw
+
+    def _update_amax(self, x):
+        i = self._step % self.history
+        with torch.no_grad():
+            self.amax_x[i] = x.detach().abs().max().float()
+            self.amax_w[i] = self.weight.detach().abs().max().float()
+        self._step += 1
+
+    def forward(self, x):
+        if not (x.is_cuda and hasattr(torch, "float8_e4m3fn")):
+            return torch.nn.functional.linear(x, self.weight)
+        sx, sw = self._scales()  # delayed: previous steps' amax window
+        self._update_amax(x)
+        if self._step == 1:
+            # no history yet: first step runs bf16 (standard warmup)
+            return torch.nn.functional.linear(x, self.weight)
+        return _Fp8LinearFn.apply(x, self.weight, sx, sw)
