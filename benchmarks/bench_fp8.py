@@ -69,18 +69,19 @@ def main():
     K, N = SHAPES["gate_up"]
     lin = Fp8Linear(K, N).to(dev)
     torch.nn.init.normal_(lin.weight, std=0.02)
-    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev,
-                    requires_grad=True) * 0.1
-    x.requires_grad_(True)
+    x = (torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+         * 0.1).detach().requires_grad_(True)
     lin(x)  # warmup step fills amax history
 
     def fp8_step():
+        x.grad = None
         out = lin(x)
         out.backward(torch.ones_like(out) * 1e-3)
 
     wb = torch.nn.Parameter(lin.weight.detach().clone())
 
     def bf16_step():
+        x.grad = None
         out = torch.nn.functional.linear(x, wb)
         out.backward(torch.ones_like(out) * 1e-3)
 

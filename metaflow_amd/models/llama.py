@@ -137,6 +137,49 @@ class DecoderLayer(nn.Module):
         pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
         return res, pending
 
+    def decode_step(self, res, pending, c_rows, s_rows, cache, layer_idx,
+                    pos_t, lengths):
+        """Continuous-batching decode: one new token per slot, each at
+        its OWN cache position (pos_t [B]); the attention is the varlen
+        flash-decode kernel over the full cache allocation."""
+        cfg = self.cfg
+        B = res.size(0)
+        hd = cfg.head_dim
+        nq, nkv = cfg.num_heads, cfg.num_kv_heads
+        res, y = K.add_rmsnorm(res, pending, self.input_norm.weight,
+                               self.input_norm.eps)
+        qkv = self.qkv_proj(y)
+        q, k, v = qkv.split([nq * hd, nkv * hd, nkv * hd], dim=-1)
+        q = q.view(B, 1, nq, hd).transpose(1, 2)     # [B, nq, 1, hd]
+        k = k.view(B, 1, nkv, hd).transpose(1, 2)
+        v = v.view(B, 1, nkv, hd).transpose(1, 2)
+        q = _rope_rows(q, c_rows, s_rows)
+        k = _rope_rows(k, c_rows, s_rows)
+        kc, vc = cache.k[layer_idx], cache.v[layer_idx]
+        bidx = torch.arange(B, device=res.device)
+        kc[bidx, :, pos_t] = k[:, :, 0]
+        vc[bidx, :, pos_t] = v[:, :, 0]
+        o = K.attn_decode_varlen(q.contiguous(), kc, vc, lengths,
+                                 1.0 / math.sqrt(hd))
+        o = o.transpose(1, 2).reshape(B, 1, nq * hd)
+        res, y = K.add_rmsnorm(res, self.o_proj(o), self.post_norm.weight,
+                               self.post_norm.eps)
+        pending = self.down_proj(K.swiglu_fused(self.gate_up_proj(y)))
+        return res, pending
+
+
+def _rope_rows(x, c, s):
+    """Half-rotation RoPE with a per-BATCH position row: x [B,Hh,1,D],
+    c/s [B, D/2] (each slot's own position — the batched-decode case the
+    positional kernel's single-offset layout can't express)."""
+    B, _hh, _one, D = x.shape
+    half = D // 2
+    c = c.view(B, 1, 1, half).float()
+    s = s.view(B, 1, 1, half).float()
+    xf = x.float()
+    x1, x2 = xf[..., :half], xf[..., half:]
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1).to(x.dtype)
+
 
 class KVCache(object):
     """Preallocated per-layer KV cache for autoregressive decode.
@@ -243,6 +286,29 @@ class LlamaForCausalLM(nn.Module):
         loss = K.cross_entropy(logits.reshape(B * S, V),
                                targets.reshape(B * S))
         return loss.mean()
+
+    @torch.no_grad()
+    def decode_step(self, tokens, cache, positions):
+        """One batched decode step for continuous batching: tokens
+        [B, 1], positions[b] = this token's cache position for slot b
+        (slots may sit at DIFFERENT positions; inactive slots pass 0 and
+        their output row is garbage the batcher discards). Returns
+        logits [B, 1, V]."""
+        B = tokens.size(0)
+        pos_t = torch.as_tensor(positions, device=tokens.device,
+                                dtype=torch.long)
+        lengths = (pos_t + 1).to(torch.int32)
+        c_rows = self.cos_t[pos_t]
+        s_rows = self.sin_t[pos_t]
+        res = self.embed(tokens)
+        pending = None
+        for li, layer in enumerate(self.layers):
+            res, pending = layer.decode_step(res, pending, c_rows,
+                                             s_rows, cache, li, pos_t,
+                                             lengths)
+        _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
+                             self.final_norm.eps)
+        return self.lm_head(x)
 
     @torch.no_grad()
     def generate(self, tokens, max_new_tokens, temperature=0.0,

@@ -27,12 +27,14 @@ __global__ __launch_bounds__(256) void attn_decode_partial_kernel(
     const short* __restrict__ q,    // [B, H, 1, 128]
     const short* __restrict__ kc,   // [B, Hkv, Lmax, 128]
     const short* __restrict__ vc,
+    const int* __restrict__ lengths,  // [B] valid prefix per sequence
     float* __restrict__ o_part, float* __restrict__ ml_part,
-    int B, int H, int Hkv, int Lmax, int L, int splits, float scale) {
+    int B, int H, int Hkv, int Lmax, int splits, float scale) {
   const int sp = blockIdx.x;
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hk = h / (H / Hkv);
+  const int L = lengths[b];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
 
@@ -145,7 +147,8 @@ __global__ __launch_bounds__(64) void attn_decode_merge_kernel(
     o0 += f * o_part[po + 2 * lane];
     o1 += f * o_part[po + 2 * lane + 1];
   }
-  const float inv = 1.f / l_all;
+  // inactive slots (length 0) produce l_all == 0: write 0, not NaN
+  const float inv = l_all > 0.f ? 1.f / l_all : 0.f;
   const long long oo = ((long long)b * H + h) * DEC_D;
   o[oo + 2 * lane] = f2bf(o0 * inv);
   o[oo + 2 * lane + 1] = f2bf(o1 * inv);

@@ -357,8 +357,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
 }
 
 // ---------------------------------------------------------- flash-decode
-torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc,
-                          torch::Tensor vc, long L, double scale) {
+torch::Tensor attn_decode_varlen(torch::Tensor q, torch::Tensor kc,
+                                 torch::Tensor vc, torch::Tensor lengths,
+                                 long max_len, double scale) {
   check_bf16(q, "q");
   check_bf16(kc, "kc");
   check_bf16(vc, "vc");
@@ -366,11 +367,14 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc,
   const int Hkv = kc.size(1), Lmax = kc.size(2);
   TORCH_CHECK(q.size(2) == 1 && q.size(3) == 128,
               "decode expects q [B,H,1,128]");
-  TORCH_CHECK(kc.size(3) == 128 && L >= 1 && L <= Lmax);
+  TORCH_CHECK(kc.size(3) == 128 && max_len >= 1 && max_len <= Lmax);
   TORCH_CHECK(H % Hkv == 0);
+  TORCH_CHECK(lengths.is_cuda() &&
+              lengths.scalar_type() == torch::kInt32 &&
+              lengths.numel() == B, "lengths must be int32 [B] on GPU");
   // enough splits to fill 256 CUs at small B*H, chunks >= ~256 rows
   int splits = (int)std::max<long>(1, 1024 / ((long)B * H));
-  splits = (int)std::min<long>(splits, (L + 255) / 256);
+  splits = (int)std::min<long>(splits, (max_len + 255) / 256);
   splits = std::max(splits, 1);
   auto f32 = q.options().dtype(torch::kFloat32);
   auto o_part = torch::empty({splits, B, H, 128}, f32);
@@ -378,15 +382,22 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc,
   auto o = torch::empty_like(q);
   dim3 grid(splits, H, B);
   attn_decode_partial_kernel<<<grid, 256, 0, cur_stream()>>>(
-      bf(q), bf(kc), bf(vc), o_part.data_ptr<float>(),
-      ml_part.data_ptr<float>(), B, H, Hkv, Lmax, (int)L, splits,
-      (float)scale);
+      bf(q), bf(kc), bf(vc), lengths.data_ptr<int>(),
+      o_part.data_ptr<float>(), ml_part.data_ptr<float>(), B, H, Hkv,
+      Lmax, splits, (float)scale);
   HIP_CHECK_KERNEL();
   attn_decode_merge_kernel<<<B * H, 64, 0, cur_stream()>>>(
       o_part.data_ptr<float>(), ml_part.data_ptr<float>(), bfm(o), B, H,
       splits);
   HIP_CHECK_KERNEL();
   return o;
+}
+
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc,
+                          torch::Tensor vc, long L, double scale) {
+  auto lengths = torch::full({q.size(0)}, (long)L,
+                             q.options().dtype(torch::kInt32));
+  return attn_decode_varlen(q, kc, vc, lengths, L, scale);
 }
 
 }  // namespace
@@ -442,6 +453,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, "flash attention backward");
   m.def("attn_decode", &attn_decode,
         "split-K flash-decode over the KV cache (q [B,H,1,128])");
+  m.def("attn_decode_varlen", &attn_decode_varlen,
+        "flash-decode with per-sequence cache lengths (int32 [B]; "
+        "length 0 = inactive slot, output 0)");
   m.def("dbg_st", &dbg_st, "debug S^T path");
   m.def("dbg_dv", &dbg_dv, "debug dV path");
   m.def("dbg_mfma32", &dbg_mfma32, "mfma 32x32x16 layout probe");

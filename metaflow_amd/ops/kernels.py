@@ -473,6 +473,35 @@ def attn_decode(q, k_cache, v_cache, length, scale):
                          v_cache[:, :, :length], scale, causal=False)
 
 
+def attn_decode_varlen(q, k_cache, v_cache, lengths, scale):
+    """Flash-decode with a per-sequence valid prefix (continuous
+    batching: slots decode at different positions; length 0 marks an
+    inactive slot, whose output row is zero).
+
+    q: [B, H, 1, 128]; lengths: int32 [B] (GPU) or list/IntTensor.
+    """
+    import torch as _torch
+
+    if q.is_cuda:
+        if not _torch.is_tensor(lengths):
+            lengths = _torch.tensor(lengths, dtype=_torch.int32,
+                                    device=q.device)
+        lengths = lengths.to(device=q.device, dtype=_torch.int32)
+        max_len = int(lengths.max().item())
+        return hip_ext().attn_decode_varlen(
+            q.contiguous(), k_cache, v_cache, lengths, max_len, scale)
+    outs = []
+    for b in range(q.size(0)):
+        n = int(lengths[b])
+        if n == 0:
+            outs.append(_torch.zeros_like(q[b:b + 1]))
+            continue
+        outs.append(attention_ref(q[b:b + 1], k_cache[b:b + 1, :, :n],
+                                  v_cache[b:b + 1, :, :n], scale,
+                                  causal=False))
+    return _torch.cat(outs, 0)
+
+
 # ================================ adam =====================================
 def adamw_step(p, g, m, v, step, lr, beta1=0.9, beta2=0.95, eps=1e-8,
                weight_decay=0.1, master=None, grad_scale=1.0):

@@ -1,0 +1,130 @@
+"""Continuous-batching serving engine for the in-house model family.
+
+Iteration-level scheduling (the vLLM/Orca idea, built MI355X-first on
+our own kernels — no reference counterpart, SURVEY §2.4 note):
+
+* a fixed pool of KV-cache SLOTS (one preallocated batched cache, full
+  stride — the varlen flash-decode kernel reads each slot at its own
+  valid length, decode.hip);
+* new requests are admitted BETWEEN decode steps: the prompt prefills
+  into a free slot through the flash fwd kernel (a [slot:slot+1] view
+  of the batched cache — zero copies);
+* every step decodes ONE token for all active slots in a single batched
+  kernel pass (slots at different positions — per-slot RoPE rows +
+  per-slot cache scatter, models/llama.py decode_step);
+* finished slots free immediately and the queue refills them, so
+  throughput tracks the arrival rate instead of the slowest member of a
+  static batch.
+
+Greedy decoding is token-exact with ``model.generate`` per request.
+"""
+
+from collections import deque
+
+
+class Request(object):
+    _next_id = 0
+
+    def __init__(self, prompt_tokens, max_new_tokens):
+        self.id = Request._next_id
+        Request._next_id += 1
+        self.prompt = list(prompt_tokens)
+        self.max_new = int(max_new_tokens)
+        self.generated = []
+        self.done = False
+
+
+class _SlotView(object):
+    """Single-slot view of the batched KV cache: the model's prefill
+    path writes through it in place (views, no copies)."""
+
+    def __init__(self, cache, slot):
+        self.k = [t[slot:slot + 1] for t in cache.k]
+        self.v = [t[slot:slot + 1] for t in cache.v]
+        self.pos = 0
+
+
+class ContinuousBatcher(object):
+    def __init__(self, model, max_batch=8, max_len=2048):
+        import torch
+
+        from .models.llama import KVCache
+
+        self.model = model
+        self.max_batch = max_batch
+        self.max_len = max_len
+        device = next(model.parameters()).device
+        self.device = device
+        self.cache = KVCache(model.cfg, max_batch, max_len, device,
+                             dtype=model.embed.weight.dtype)
+        self.positions = [0] * max_batch    # cached tokens per slot
+        self.slots = [None] * max_batch     # Request or None
+        self.next_token = [0] * max_batch   # token to feed next step
+        self.queue = deque()
+        self._torch = torch
+
+    def submit(self, prompt_tokens, max_new_tokens):
+        req = Request(prompt_tokens, max_new_tokens)
+        self.queue.append(req)
+        return req
+
+    # ------------------------------------------------------------- internals
+    def _admit(self):
+        torch = self._torch
+        for slot in range(self.max_batch):
+            if self.slots[slot] is not None or not self.queue:
+                continue
+            req = self.queue.popleft()
+            assert len(req.prompt) + req.max_new <= self.max_len, \
+                "request longer than the slot capacity"
+            prompt = torch.tensor([req.prompt], device=self.device)
+            view = _SlotView(self.cache, slot)
+            with torch.no_grad():
+                logits = self.model(prompt, cache=view)
+            nxt = int(logits[0, -1].float().argmax())
+            req.generated.append(nxt)
+            self.slots[slot] = req
+            self.positions[slot] = len(req.prompt)
+            self.next_token[slot] = nxt
+            if req.max_new <= 1:
+                self._finish(slot)
+
+    def _finish(self, slot):
+        self.slots[slot].done = True
+        self.slots[slot] = None
+        self.positions[slot] = 0
+
+    def step(self):
+        """Admit waiting requests, then decode one token for every
+        active slot in a single batched pass. Returns the number of
+        active slots decoded."""
+        torch = self._torch
+        self._admit()
+        active = [i for i, r in enumerate(self.slots) if r is not None]
+        if not active:
+            return 0
+        tokens = torch.zeros(self.max_batch, 1, dtype=torch.long,
+                             device=self.device)
+        for i in active:
+            tokens[i, 0] = self.next_token[i]
+        logits = self.model.decode_step(tokens, self.cache,
+                                        list(self.positions))
+        for i in active:
+            req = self.slots[i]
+            nxt = int(logits[i, -1].float().argmax())
+            req.generated.append(nxt)
+            self.positions[i] += 1
+            self.next_token[i] = nxt
+            if len(req.generated) >= req.max_new:
+                self._finish(i)
+        return len(active)
+
+    def run(self):
+        """Drain the queue; returns {request_id: generated tokens}."""
+        results = {}
+        pending = list(self.queue)
+        while self.queue or any(r is not None for r in self.slots):
+            self.step()
+        for req in pending:
+            results[req.id] = req.generated
+        return results
