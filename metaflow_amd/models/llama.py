@@ -138,7 +138,7 @@ class DecoderLayer(nn.Module):
         return res, pending
 
     def decode_step(self, res, pending, c_rows, s_rows, cache, layer_idx,
-                    pos_t, lengths):
+                    pos_t, lengths, max_len=None):
         """Continuous-batching decode: one new token per slot, each at
         its OWN cache position (pos_t [B]); the attention is the varlen
         flash-decode kernel over the full cache allocation."""
@@ -160,7 +160,7 @@ class DecoderLayer(nn.Module):
         kc[bidx, :, pos_t] = k[:, :, 0]
         vc[bidx, :, pos_t] = v[:, :, 0]
         o = K.attn_decode_varlen(q.contiguous(), kc, vc, lengths,
-                                 1.0 / math.sqrt(hd))
+                                 1.0 / math.sqrt(hd), max_len=max_len)
         o = o.transpose(1, 2).reshape(B, 1, nq * hd)
         res, y = K.add_rmsnorm(res, self.o_proj(o), self.post_norm.weight,
                                self.post_norm.eps)
@@ -288,13 +288,17 @@ class LlamaForCausalLM(nn.Module):
         return loss.mean()
 
     @torch.no_grad()
-    def decode_step(self, tokens, cache, positions):
+    def decode_step(self, tokens, cache, positions, max_len=None):
         """One batched decode step for continuous batching: tokens
         [B, 1], positions[b] = this token's cache position for slot b
         (slots may sit at DIFFERENT positions; inactive slots pass 0 and
         their output row is garbage the batcher discards). Returns
-        logits [B, 1, V]."""
-        B = tokens.size(0)
+        logits [B, 1, V].
+
+        With ``max_len`` (the cache capacity) the whole step is free of
+        host reads and is hipGraph-capturable: the serving engine
+        captures it once and replays per token (decode is launch-bound;
+        see serving.ContinuousBatcher(graph=True))."""
         pos_t = torch.as_tensor(positions, device=tokens.device,
                                 dtype=torch.long)
         lengths = (pos_t + 1).to(torch.int32)
@@ -305,7 +309,7 @@ class LlamaForCausalLM(nn.Module):
         for li, layer in enumerate(self.layers):
             res, pending = layer.decode_step(res, pending, c_rows,
                                              s_rows, cache, li, pos_t,
-                                             lengths)
+                                             lengths, max_len=max_len)
         _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,
                              self.final_norm.eps)
         return self.lm_head(x)

@@ -473,12 +473,16 @@ def attn_decode(q, k_cache, v_cache, length, scale):
                          v_cache[:, :, :length], scale, causal=False)
 
 
-def attn_decode_varlen(q, k_cache, v_cache, lengths, scale):
+def attn_decode_varlen(q, k_cache, v_cache, lengths, scale,
+                       max_len=None):
     """Flash-decode with a per-sequence valid prefix (continuous
     batching: slots decode at different positions; length 0 marks an
     inactive slot, whose output row is zero).
 
     q: [B, H, 1, 128]; lengths: int32 [B] (GPU) or list/IntTensor.
+    ``max_len`` bounds the split count; passing the cache capacity makes
+    the call hipGraph-capturable (no host read of lengths — the kernel
+    chunks from the device lengths, empty splits cost nothing).
     """
     import torch as _torch
 
@@ -487,7 +491,8 @@ def attn_decode_varlen(q, k_cache, v_cache, lengths, scale):
             lengths = _torch.tensor(lengths, dtype=_torch.int32,
                                     device=q.device)
         lengths = lengths.to(device=q.device, dtype=_torch.int32)
-        max_len = int(lengths.max().item())
+        if max_len is None:
+            max_len = int(lengths.max().item())
         return hip_ext().attn_decode_varlen(
             q.contiguous(), k_cache, v_cache, lengths, max_len, scale)
     outs = []

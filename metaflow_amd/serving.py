@@ -45,7 +45,13 @@ class _SlotView(object):
 
 
 class ContinuousBatcher(object):
-    def __init__(self, model, max_batch=8, max_len=2048):
+    def __init__(self, model, max_batch=8, max_len=2048, graph="auto"):
+        """graph: capture the whole decode step in a hipGraph and replay
+        it per token (decode is LAUNCH-bound — ~200 small kernels per
+        step; replay collapses them to one submit). "auto" captures on
+        GPU and falls back silently; the step stays host-read-free
+        because the varlen kernel chunks from device lengths
+        (decode.hip) and splits come from the cache capacity."""
         import torch
 
         from .models.llama import KVCache
@@ -62,6 +68,46 @@ class ContinuousBatcher(object):
         self.next_token = [0] * max_batch   # token to feed next step
         self.queue = deque()
         self._torch = torch
+        self._graph = None
+        if graph and device.type == "cuda":
+            try:
+                self._capture_graph()
+            except Exception:
+                if graph is True:
+                    raise
+                self._graph = None  # auto: eager decode
+
+    def _capture_graph(self):
+        torch = self._torch
+        self._tok_buf = torch.zeros(self.max_batch, 1, dtype=torch.long,
+                                    device=self.device)
+        self._pos_buf = torch.zeros(self.max_batch, dtype=torch.long,
+                                    device=self.device)
+        # warmup on a side stream (allocator primes its graph pool)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.model.decode_step(self._tok_buf, self.cache,
+                                       self._pos_buf,
+                                       max_len=self.max_len)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._logits_buf = self.model.decode_step(
+                self._tok_buf, self.cache, self._pos_buf,
+                max_len=self.max_len)
+        self._graph = g
+
+    def _decode(self, tokens, positions):
+        if self._graph is not None:
+            torch = self._torch
+            self._tok_buf.copy_(tokens)
+            self._pos_buf.copy_(torch.as_tensor(positions,
+                                                device=self.device))
+            self._graph.replay()
+            return self._logits_buf
+        return self.model.decode_step(tokens, self.cache, positions)
 
     def submit(self, prompt_tokens, max_new_tokens):
         req = Request(prompt_tokens, max_new_tokens)
@@ -107,8 +153,7 @@ class ContinuousBatcher(object):
                              device=self.device)
         for i in active:
             tokens[i, 0] = self.next_token[i]
-        logits = self.model.decode_step(tokens, self.cache,
-                                        list(self.positions))
+        logits = self._decode(tokens, list(self.positions))
         for i in active:
             req = self.slots[i]
             nxt = int(logits[i, -1].float().argmax())
