@@ -159,11 +159,14 @@ def main():
         inp = inp[:, sl].contiguous()
         tgt = tgt[:, sl].contiguous()
 
+    comm_ms = []
+
     def one_step():
         flat.zero_grad()
         loss = model(inp, tgt)
         loss.backward()
         flat.finish_grad_sync()
+        comm_ms.append(getattr(flat, "last_comm_wait_ms", 0.0))
         opt.step()
         return loss
 
@@ -222,6 +225,11 @@ def main():
                                 ("cp%d" if use_cp else "dp%d") % n_gpus),
                 "first_loss": round(first_loss, 4),
                 "last_loss": round(last_loss, 4),
+                # NON-hidden gradient-comm wait per timed step (rank 0):
+                # the number that explains the 1/2/4/8 scaling curve
+                "comm_wait_ms": round(
+                    sum(comm_ms[args.warmup:])
+                    / max(1, len(comm_ms) - args.warmup), 2),
             },
         }
         print(json.dumps(out), flush=True)

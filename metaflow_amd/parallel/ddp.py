@@ -130,10 +130,21 @@ class FlatParamModel(object):
         part of gradient communication — the number that explains the
         1/2/4/8-GPU scaling curve (collective-time breakdown
         through the monitor; enable with MFX_MONITOR=debug|sidecar)."""
+        import os
         import time
 
         from ..monitor import get_system_monitor
 
+        # work.wait() on NCCL/RCCL is stream-ordered (host barely
+        # blocks), so wall-clocking it reads ~0 even when comm
+        # dominates. MFX_COMM_TIMING=1 brackets the wait with device
+        # synchronizes: the measured delta is the comm still running
+        # AFTER compute drained — the exposed (non-overlapped) tail.
+        # Costs a sync per step, so it is opt-in for diagnosis runs.
+        comm_timing = os.environ.get("MFX_COMM_TIMING") == "1" and \
+            torch.cuda.is_available() and self.flat_grad.is_cuda
+        if comm_timing:
+            torch.cuda.synchronize()
         t0 = time.time()
         if self._pending:
             for work in self._pending:
@@ -141,6 +152,8 @@ class FlatParamModel(object):
             self._pending.clear()
             for bi in self._bucket_done_count:
                 self._bucket_done_count[bi] = 0
+            if comm_timing:
+                torch.cuda.synchronize()
             get_system_monitor().gauge(
                 "mfx.ddp.allreduce_wait_ms", (time.time() - t0) * 1000)
         elif (dist.is_available() and dist.is_initialized()
