@@ -69,15 +69,62 @@ def _tensor_to_buffer(t):
     return memoryview(host), dtype, shape
 
 
+def _gpu_chunks(t, torch):
+    """Yield the tensor's bytes as pinned-host chunks: D2H of half i+1
+    overlaps the caller's consumption of half i (hash+write in
+    cas.save_stream). Chunks are 8 MiB-aligned (Merkle leaf contract)."""
+    global _side_stream
+    nbytes = t.element_size() * t.numel()
+    flat = t.reshape(-1).view(torch.uint8)
+    pin = _get_pin_buf(torch)
+    half = _PIN_BUF_BYTES // 2  # 256 MiB: a multiple of the 8 MiB leaf
+    views = [pin[:half], pin[half:]]
+    np_views = [v.numpy() for v in views]
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
+    _side_stream.wait_stream(torch.cuda.current_stream())
+    events = [torch.cuda.Event(), torch.cuda.Event()]
+    with torch.cuda.stream(_side_stream):
+        n0 = min(half, nbytes)
+        views[0][:n0].copy_(flat[:n0], non_blocking=True)
+        events[0].record(_side_stream)
+        offset, i = 0, 0
+        while offset < nbytes:
+            n = min(half, nbytes - offset)
+            nxt = offset + n
+            if nxt < nbytes:
+                n2 = min(half, nbytes - nxt)
+                views[i ^ 1][:n2].copy_(flat[nxt:nxt + n2],
+                                        non_blocking=True)
+                events[i ^ 1].record(_side_stream)
+            events[i].synchronize()
+            # NOTE: save_stream copies each 8 MiB leaf out of this view
+            # before hashing, so reusing the half next iteration is safe
+            yield memoryview(np_views[i][:n])
+            offset = nxt
+            i ^= 1
+
+
 def save_state_dict(task_datastore, state_dict, name="checkpoint"):
     """Persist a state dict; returns {tensor_name: {sha, dtype, shape}}."""
     import torch
 
     cas = task_datastore._ca_store
+    can_stream = hasattr(cas._storage, "_abs")
     index = {}
     other = {}
     for key, value in state_dict.items():
         if isinstance(value, torch.Tensor):
+            if value.is_cuda and can_stream:
+                # pipelined D2H -> leaf hash -> write (config 4 GB/s)
+                t = value.detach().contiguous()
+                nbytes = t.element_size() * t.numel()
+                _uri, sha = cas.save_stream(_gpu_chunks(t, torch),
+                                            nbytes)
+                index[key] = {"sha": sha,
+                              "dtype": str(t.dtype).replace("torch.", ""),
+                              "shape": list(t.shape), "nbytes": nbytes}
+                continue
             buf, dtype, shape = _tensor_to_buffer(value)
             (_uri, sha), = cas.save_blobs([buf], raw=True)
             index[key] = {"sha": sha, "dtype": dtype, "shape": list(shape),

@@ -370,3 +370,39 @@ def test_native_parallel_load_matches_python(tmp_path):
         # the public API (which routes through the native path) agrees
         [(k2, got_api)] = list(store.load_blobs([key]))
         assert k2 == key and bytes(got_api) == want
+
+
+def test_cas_save_stream_matches_save_blobs(tmp_path):
+    """Streaming save (pipelined hash+write) must produce the SAME keys
+    as save_blobs for identical content (dedup across paths), for both
+    the plain-sha (<16 MiB) and Merkle (>=16 MiB) regimes."""
+    import numpy as np
+
+    from metaflow_amd.datastore.cas import ContentAddressedStore
+    from metaflow_amd.datastore.storage import LocalStorage
+
+    store = ContentAddressedStore("data", LocalStorage(str(tmp_path)))
+    rng = np.random.default_rng(3)
+    for size in (3 << 20, 40 << 20):
+        blob = rng.bytes(size)
+        [(_u, key_ref)] = store.save_blobs([blob], raw=True)
+        # chunks of 8 MiB-multiples (leaf contract), ragged tail
+        chunks = [memoryview(blob)[o:o + (16 << 20)]
+                  for o in range(0, size, 16 << 20)]
+        _uri, key_stream = store.save_stream(iter(chunks), size)
+        assert key_stream == key_ref, size
+        [(_k, back)] = list(store.load_blobs([key_stream]))
+        assert bytes(back) == blob
+    # dedup: re-streaming identical content leaves one physical file
+    blob = rng.bytes(20 << 20)
+    _u1, k1 = store.save_stream(
+        iter([memoryview(blob)]), len(blob))
+    _u2, k2 = store.save_stream(
+        iter([memoryview(blob)]), len(blob))
+    assert k1 == k2
+    import os as _os
+
+    files = []
+    for dirpath, _d, names in _os.walk(str(tmp_path)):
+        files += [n for n in names if k1[:8] in n]
+    assert len(files) == 1
