@@ -95,13 +95,21 @@ class Flow(MetaflowObject):
                 return run
         return None
 
-    def runs(self, *tags):
+    def runs(self, *tags, limit=None):
+        """Newest-first runs with all the given tags; ``limit`` bounds
+        how many matching runs are yielded (the metadata provider pages
+        lazily, so a deep history costs what you read)."""
+        n = 0
         for run in self:
             if all(t in run.tags for t in tags):
                 yield run
+                n += 1
+                if limit is not None and n >= limit:
+                    return
 
     def __iter__(self):
-        for info in self._meta.list_runs():
+        iter_fn = getattr(self._meta, "iter_runs", self._meta.list_runs)
+        for info in iter_fn():
             tags = _full_tags(info)
             if _check_namespace(tags):
                 yield Run("%s/%s" % (self.id, info["run_id"]), _info=info)

@@ -123,15 +123,40 @@ class LocalMetadataProvider(object):
         return hb.get("ts")
 
     # ----------------------------------------------------------------- query
-    def list_runs(self):
-        runs = []
-        for path, is_file in self._storage.list_content([self._meta_root()]):
-            if not is_file:
-                info = self._load(self._storage.path_join(path, "run.json"))
-                if info:
-                    runs.append(info)
-        runs.sort(key=lambda r: r.get("ts_epoch", 0), reverse=True)
-        return runs
+    @staticmethod
+    def _run_sort_key(name):
+        # run ids are creation-time microseconds (resume runs prefix the
+        # origin id with "resume"), so DIRECTORY NAMES order runs
+        # chronologically without opening any run.json
+        digits = name[6:] if name.startswith("resume") else name
+        try:
+            return int(digits)
+        except ValueError:
+            return 0
+
+    def iter_runs(self, limit=None):
+        """Newest-first run records, loading at most ``limit`` run.json
+        files: ordering comes from the directory names alone (µs
+        timestamps), so a flow with a million runs pays for the page it
+        reads, not the history (reference metadata.py:352 query
+        routing / paging)."""
+        names = [self._storage.basename(path)
+                 for path, is_file in
+                 self._storage.list_content([self._meta_root()])
+                 if not is_file]
+        names.sort(key=self._run_sort_key, reverse=True)
+        count = 0
+        for name in names:
+            if limit is not None and count >= limit:
+                return
+            info = self._load(self._storage.path_join(
+                self._meta_root(), name, "run.json"))
+            if info:
+                count += 1
+                yield info
+
+    def list_runs(self, limit=None):
+        return list(self.iter_runs(limit=limit))
 
     def get_run(self, run_id):
         return self._load(self._run_path(run_id))

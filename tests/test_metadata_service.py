@@ -83,3 +83,25 @@ def test_flow_run_with_service_metadata(service, tmp_datastore,
     assert len(p.list_tasks(run_id)) >= 3
     # heartbeat sidecar went through the service provider too
     assert p._request("GET", "/version")["api_version"] == 1
+
+
+def test_local_metadata_paging(tmp_path):
+    """iter_runs(limit=N) loads only N run records, newest first, with
+    ordering derived from directory names (no full-history scan)."""
+    from metaflow_amd.datastore.storage import LocalStorage
+    from metaflow_amd.metadata.local import LocalMetadataProvider
+
+    meta = LocalMetadataProvider("PagedFlow", LocalStorage(str(tmp_path)))
+    ids = []
+    for _ in range(25):
+        rid = meta.new_run_id()
+        meta.register_run(rid)
+        ids.append(rid)
+    page = list(meta.iter_runs(limit=5))
+    assert len(page) == 5
+    assert [r["run_id"] for r in page] == sorted(ids, reverse=True)[:5]
+    # resume-prefixed ids sort with their origin's timestamp
+    meta.register_run("resume%s" % ids[0])
+    newest = next(iter(meta.iter_runs(limit=1)))
+    assert newest["run_id"] in ("resume%s" % ids[0], ids[-1])
+    assert len(meta.list_runs()) == 26
