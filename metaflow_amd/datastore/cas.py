@@ -157,6 +157,9 @@ class ContentAddressedStore(object):
             _hash_pool = ThreadPoolExecutor(
                 max_workers=_os.cpu_count() or 8)
 
+        import queue
+        import threading
+
         root_dir = self._storage._abs(self._prefix)
         _os.makedirs(root_dir, exist_ok=True)
         tmp = _os.path.join(root_dir, ".stream.%d.%d.tmp"
@@ -165,22 +168,46 @@ class ContentAddressedStore(object):
         small = nbytes < PARALLEL_KEY_MIN
         plain = hashlib.sha256() if small else None
         raw_header = MAGIC + bytes([1, CODEC_RAW, 0, 0])
+        # dedicated writer thread: the leaf bytes objects are hashed by
+        # the pool AND written by the writer, so the main thread's only
+        # serial work is the leaf copy out of the caller's (pinned)
+        # buffer — write, hash, copy and the caller's D2H all overlap
+        wq = queue.Queue(maxsize=128)
+        werr = []
+
+        def _writer():
+            try:
+                with open(tmp, "wb") as f:
+                    f.write(raw_header)
+                    while True:
+                        item = wq.get()
+                        if item is None:
+                            return
+                        f.write(item)
+            except Exception as ex:  # surfaced after join
+                werr.append(ex)
+                while wq.get() is not None:
+                    pass
+
+        wt = threading.Thread(target=_writer, daemon=True)
+        wt.start()
         try:
-            with open(tmp, "wb") as f:
-                f.write(raw_header)
-                for chunk in chunk_iter:
-                    mv = memoryview(chunk)
+            for chunk in chunk_iter:
+                mv = memoryview(chunk)
+                for off in range(0, len(mv), _LEAF):
+                    # bytes() copy: the caller reuses its pinned halves
+                    # once we return from this iteration
+                    leaf = bytes(mv[off:off + _LEAF])
                     if small:
-                        plain.update(mv)
+                        plain.update(leaf)
                     else:
-                        for off in range(0, len(mv), _LEAF):
-                            # bytes() copy: the caller reuses its pinned
-                            # halves once we return from this iteration
-                            leaf = bytes(mv[off:off + _LEAF])
-                            futures.append(_hash_pool.submit(
-                                lambda b: hashlib.sha256(b).digest(),
-                                leaf))
-                    f.write(mv)
+                        futures.append(_hash_pool.submit(
+                            lambda b: hashlib.sha256(b).digest(), leaf))
+                    wq.put(leaf)
+            wq.put(None)
+            wt.join()
+            if werr:
+                raise werr[0]
             if small:
                 key = plain.hexdigest()
             else:
@@ -195,6 +222,10 @@ class ContentAddressedStore(object):
                 _os.replace(tmp, ap)
             return self._storage.full_uri(path), key
         except Exception:
+            try:
+                wq.put(None)
+            except Exception:
+                pass
             try:
                 _os.unlink(tmp)
             except OSError:
