@@ -208,6 +208,9 @@ def main(flow):
     @click.option("--runner-attribute-file", default=None)
     @(lambda f: _param_options(f, flow_cls))
     def run(**kwargs):
+        from .system_context import SCHEDULER, set_phase
+
+        set_phase(SCHEDULER)
         _run_common(kwargs)
 
     @cli.command(help="Resume a failed run from where it left off.")
@@ -227,6 +230,9 @@ def main(flow):
                   help="Safe under concurrent invocation: one caller wins "
                        "leader election and resumes; the rest wait.")
     def resume(origin_run_id, step_to_rerun, reentrant, **kwargs):
+        from .system_context import SCHEDULER, set_phase
+
+        set_phase(SCHEDULER)
         if origin_run_id is None:
             runs = state.metadata.list_runs()
             if not runs:
@@ -278,6 +284,9 @@ def main(flow):
     @click.option("--ubf-context", default=None)
     def step(step_name, run_id, task_id, input_paths, split_index,
              retry_count, max_user_code_retries, origin_run_id, ubf_context):
+        from .system_context import TASK, set_phase
+
+        set_phase(TASK)
         task = MFXTask(state.flow, state.graph, state.flow_datastore,
                        state.metadata)
         paths = [p for p in input_paths.split(",") if p]

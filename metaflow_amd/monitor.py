@@ -122,6 +122,10 @@ class DebugEventLogger(NullEventLogger):
     TYPE = "debug"
 
     def log(self, payload):
+        from .system_context import current_phase
+
+        payload = dict(payload)
+        payload.setdefault("phase", current_phase())
         sys.stderr.write("[mfx-event] %s\n" % json.dumps(payload))
 
 

@@ -101,3 +101,17 @@ def test_extension_package_discovery(tmp_path, tmp_datastore):
     run_id = latest_run_id(tmp_datastore, "LinearFlow")
     assert read_artifact(tmp_datastore, "LinearFlow", run_id, "start",
                          "stamped") == "by-extension"
+
+
+def test_system_context_phase(tmp_datastore):
+    """Telemetry records carry the process's execution phase (reference
+    system_context.py): task subprocesses log phase=task."""
+    proc = run_flow("linear_flow.py", tmp_datastore, "run",
+                    env_extra={"MFX_EVENT_LOGGER": "debug"})
+    assert '"phase": "task"' in proc.stderr
+
+    from metaflow_amd import system_context as sc
+
+    assert sc.phase_from_subcommand("run") == sc.SCHEDULER
+    assert sc.phase_from_subcommand("step") == sc.TASK
+    assert sc.phase_from_subcommand("dump") == sc.CLIENT
