@@ -97,5 +97,45 @@ def main():
         shutil.rmtree(d)
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "--probe-pinned" not in sys.argv:
     main()
+
+
+def probe_pinned():
+    """Is reading TORCH-PINNED host memory slow (uncached mapping)?"""
+    import numpy as np
+    import torch
+
+    n = 1 << 30
+    pin = torch.empty(n, dtype=torch.uint8, pin_memory=True)
+    pin_np = pin.numpy()
+    pin_np[::4096] = 1  # fault in
+    page = np.empty(n, dtype=np.uint8)
+
+    t = time.time()
+    for o in range(0, n, 8 << 20):
+        bytes(memoryview(pin_np)[o:o + (8 << 20)])
+    print("leaf copies FROM PINNED: %.2f GB/s"
+          % (n / (time.time() - t) / GB), flush=True)
+
+    t = time.time()
+    np.copyto(page, pin_np)
+    print("np.copyto pinned->pageable: %.2f GB/s"
+          % (n / (time.time() - t) / GB), flush=True)
+
+    cpu_t = torch.from_numpy(page)
+    t = time.time()
+    cpu_t.copy_(pin)
+    print("torch copy pinned->pageable: %.2f GB/s"
+          % (n / (time.time() - t) / GB), flush=True)
+
+    t = time.time()
+    h = hashlib.sha256()
+    for o in range(0, 256 << 20, 8 << 20):
+        h.update(memoryview(pin_np)[o:o + (8 << 20)])
+    print("sha256 straight from pinned: %.2f GB/s"
+          % ((256 << 20) / (time.time() - t) / GB), flush=True)
+
+
+if __name__ == "__main__" and "--probe-pinned" in sys.argv:
+    probe_pinned()
