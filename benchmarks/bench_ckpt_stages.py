@@ -139,3 +139,64 @@ def probe_pinned():
 
 if __name__ == "__main__" and "--probe-pinned" in sys.argv:
     probe_pinned()
+
+
+def probe_compose():
+    """Decompose save_stream overhead: host-source streaming vs the old
+    buffer-then-save path, with and without the GPU in the loop."""
+    import shutil
+
+    import numpy as np
+    import torch
+
+    from metaflow_amd.datastore.cas import ContentAddressedStore
+    from metaflow_amd.datastore.storage import LocalStorage
+
+    n = 2 * GB
+    big = np.random.bytes(n)
+    chunks = [memoryview(big)[o:o + (256 << 20)]
+              for o in range(0, n, 256 << 20)]
+
+    d = tempfile.mkdtemp(prefix="ckpt_cmp_")
+    store = ContentAddressedStore("data", LocalStorage(d))
+    t = time.time()
+    store.save_stream(iter(list(chunks)), n)
+    print("save_stream(host chunks): %.2f GB/s"
+          % (n / (time.time() - t) / GB), flush=True)
+    shutil.rmtree(d)
+
+    d = tempfile.mkdtemp(prefix="ckpt_cmp2_")
+    store = ContentAddressedStore("data", LocalStorage(d))
+    t = time.time()
+    store.save_blobs([big], raw=True)
+    print("save_blobs(host, old path): %.2f GB/s"
+          % (n / (time.time() - t) / GB), flush=True)
+    shutil.rmtree(d)
+
+    if torch.cuda.is_available():
+        from metaflow_amd.parallel import checkpoint as CK
+
+        t_gpu = torch.randn(n // 2, dtype=torch.bfloat16, device="cuda")
+        torch.cuda.synchronize()
+        t = time.time()
+        buf, _d, _s = CK._tensor_to_buffer(t_gpu)
+        print("_tensor_to_buffer(D2H serial): %.2f GB/s"
+              % (n / (time.time() - t) / GB), flush=True)
+        d = tempfile.mkdtemp(prefix="ckpt_cmp3_")
+        store = ContentAddressedStore("data", LocalStorage(d))
+        t = time.time()
+        store.save_blobs([buf], raw=True)
+        print("save_blobs(of that buffer): %.2f GB/s"
+              % (n / (time.time() - t) / GB), flush=True)
+        shutil.rmtree(d)
+        # chunk generator alone (no hash/write): drain _gpu_chunks
+        t = time.time()
+        total = 0
+        for mv in CK._gpu_chunks(t_gpu, torch):
+            total += len(mv)
+        print("_gpu_chunks drain only: %.2f GB/s"
+              % (total / (time.time() - t) / GB), flush=True)
+
+
+if __name__ == "__main__" and "--probe-compose" in sys.argv:
+    probe_compose()
