@@ -41,6 +41,11 @@ def parse_args():
                         "Megatron-style shard group over a single "
                         "shared batch")
     p.add_argument("--lr", type=float, default=3e-4)
+    p.add_argument("--fp8", action="store_true",
+                   help="decoder projections' forward GEMMs in OCP E4M3 "
+                        "(delayed scaling, bf16 backward) — reported "
+                        "honestly as dtype fp8-e4m3-fwd, NOT the bf16 "
+                        "headline")
     p.add_argument("--tunableop", action="store_true",
                    help="enable PyTorch TunableOp (hipBLASLt algorithm "
                         "autotuning) during warmup")
@@ -99,6 +104,10 @@ def main():
     }
     cfg_fn, model_cls = model_factories[args.model]
     cfg = cfg_fn()
+    if args.fp8:
+        assert args.model in ("llama3-8b", "llama3-70b", "tiny"), \
+            "--fp8 supports the Llama family"
+        cfg.fp8 = True
     seq = min(args.seq, cfg.max_seq_len)
 
     use_tp = args.tp and distributed
@@ -215,7 +224,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong" if shared else "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "fp8-e4m3-fwd" if args.fp8 else "bf16",
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -28,6 +28,9 @@ class LlamaConfig:
     rope_theta: float = 500000.0
     rms_eps: float = 1e-5
     tie_embeddings: bool = False
+    # fp8=True runs the decoder projections' FORWARD GEMMs in OCP E4M3
+    # (delayed scaling, bf16 backward — ops/fp8.py); lm_head stays bf16
+    fp8: bool = False
 
     @classmethod
     def llama3_8b(cls):
@@ -83,13 +86,17 @@ class DecoderLayer(nn.Module):
         super().__init__()
         self.cfg = cfg
         h, hd = cfg.hidden_size, cfg.head_dim
-        self.qkv_proj = Linear(
+        if cfg.fp8:
+            from ..ops.fp8 import Fp8Linear as Lin
+        else:
+            Lin = Linear
+        self.qkv_proj = Lin(
             h, (cfg.num_heads + 2 * cfg.num_kv_heads) * hd)
         self.input_norm = RMSNorm(h, cfg.rms_eps)
-        self.o_proj = Linear(cfg.num_heads * hd, h)
+        self.o_proj = Lin(cfg.num_heads * hd, h)
         self.post_norm = RMSNorm(h, cfg.rms_eps)
-        self.gate_up_proj = Linear(h, 2 * cfg.intermediate_size)
-        self.down_proj = Linear(cfg.intermediate_size, h)
+        self.gate_up_proj = Lin(h, 2 * cfg.intermediate_size)
+        self.down_proj = Lin(cfg.intermediate_size, h)
 
     def forward(self, res, pending, cos_t, sin_t, cp_group=None,
                 cache=None, layer_idx=0):
