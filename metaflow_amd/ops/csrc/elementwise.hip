@@ -415,3 +415,34 @@ __global__ void swiglu_gu_bwd_kernel(const short* __restrict__ gu,
     *(bf16x8*)(dgu + base + I) = ou;
   }
 }
+
+// ------------------------------------------------------ fp8 quantize
+// One-pass bf16 -> OCP E4M3 with scale + saturation, via the gfx950
+// native packed convert (v_cvt_pk_fp8_f32). The torch composition
+// (float() * scale, clamp, .to(fp8)) materializes fp32 intermediates —
+// 4+ memory passes that made fp8 GEMMs a net LOSS end-to-end.
+__global__ void quant_e4m3_kernel(const short* __restrict__ x,
+                                  unsigned char* __restrict__ y,
+                                  float scale, long long n8) {
+  typedef __attribute__((ext_vector_type(2))) int int2v;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n8; i += stride) {
+    bf16x8 v = *(const bf16x8*)(x + i * 8);
+    float f[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float t = bf2f(v[j]) * scale;
+      f[j] = fminf(fmaxf(t, -448.f), 448.f);
+    }
+    int2v out;
+    int w0 = 0, w1 = 0;
+    w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], w0, false);
+    w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], w0, true);
+    w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], w1, false);
+    w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], w1, true);
+    out[0] = w0;
+    out[1] = w1;
+    *(int2v*)(y + i * 8) = out;
+  }
+}

@@ -207,6 +207,17 @@ torch::Tensor swiglu_gu_bwd(torch::Tensor gu, torch::Tensor dy) {
   return dgu;
 }
 
+torch::Tensor quant_e4m3(torch::Tensor x, double scale) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0, "quant_e4m3 needs numel % 8 == 0");
+  auto y = torch::empty_like(x, x.options().dtype(torch::kUInt8));
+  long long n8 = x.numel() / 8;
+  quant_e4m3_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
+      bf(x), y.data_ptr<unsigned char>(), (float)scale, n8);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
 torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a");
   TORCH_CHECK(a.numel() % 8 == 0);
@@ -442,6 +453,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_gu_fwd", [](torch::Tensor gu){ return swiglu_gu_fwd(gu)[0]; }, "fused-layout SwiGLU fwd");
   m.def("swiglu_gu_bwd", &swiglu_gu_bwd, "fused-layout SwiGLU bwd");
   m.def("add_bf16", &add_bf16, "fused bf16 add");
+  m.def("quant_e4m3", &quant_e4m3,
+        "one-pass bf16 -> OCP E4M3 (uint8 storage) with scale");
   m.def("adamw", &adamw, "fused AdamW (bf16 p/g, fp32 m/v[, master])");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward");

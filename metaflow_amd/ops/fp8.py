@@ -28,7 +28,14 @@ def _gemm_ext():
 
 
 def quantize_e4m3(t, scale):
-    """bf16/fp32 -> float8_e4m3fn storage viewed as uint8 (saturating)."""
+    """bf16/fp32 -> float8_e4m3fn storage viewed as uint8 (saturating).
+
+    GPU bf16 goes through the one-pass quant_e4m3 HIP kernel (native
+    v_cvt_pk_fp8_f32); everything else uses the torch composition."""
+    if t.is_cuda and t.dtype == torch.bfloat16 and t.numel() % 8 == 0:
+        from .kernels import hip_ext
+
+        return hip_ext().quant_e4m3(t.contiguous(), float(scale))
     x = (t.float() * scale).clamp(-E4M3_MAX, E4M3_MAX)
     return x.to(torch.float8_e4m3fn).view(torch.uint8)
 

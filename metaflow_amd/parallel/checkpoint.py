@@ -105,6 +105,77 @@ def _gpu_chunks(t, torch):
             i ^= 1
 
 
+def _gpu_chunks_torch(t, torch):
+    """Like _gpu_chunks but yields torch PINNED slices (for the native
+    StreamSaver, which reads them GIL-free in C++)."""
+    global _side_stream
+    nbytes = t.element_size() * t.numel()
+    flat = t.reshape(-1).view(torch.uint8)
+    pin = _get_pin_buf(torch)
+    half = _PIN_BUF_BYTES // 2
+    views = [pin[:half], pin[half:]]
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
+    _side_stream.wait_stream(torch.cuda.current_stream())
+    events = [torch.cuda.Event(), torch.cuda.Event()]
+    with torch.cuda.stream(_side_stream):
+        n0 = min(half, nbytes)
+        views[0][:n0].copy_(flat[:n0], non_blocking=True)
+        events[0].record(_side_stream)
+        offset, i = 0, 0
+        while offset < nbytes:
+            n = min(half, nbytes - offset)
+            nxt = offset + n
+            if nxt < nbytes:
+                n2 = min(half, nbytes - nxt)
+                views[i ^ 1][:n2].copy_(flat[nxt:nxt + n2],
+                                        non_blocking=True)
+                events[i ^ 1].record(_side_stream)
+            events[i].synchronize()
+            # feed() completes (hash + write) before the generator
+            # resumes, so reusing this half next round is safe
+            yield views[i][:n]
+            offset = nxt
+            i ^= 1
+
+
+def _save_gpu_tensor_native(t, cas, torch):
+    """GPU shard -> CAS with zero Python-side data work: D2H into
+    pinned halves on the side stream while the C++ StreamSaver hashes
+    Merkle leaves across threads and writes the previous half
+    (cas_engine.cpp). Returns (sha, nbytes)."""
+    import os as _os
+
+    from ..datastore.cas import CODEC_RAW, MAGIC
+    from ..ops import _mfx_cas
+
+    t = t.detach().contiguous()
+    nbytes = t.element_size() * t.numel()
+    root_dir = cas._storage._abs(cas._prefix)
+    _os.makedirs(root_dir, exist_ok=True)
+    tmp = _os.path.join(root_dir, ".ckpt.%d.%x.tmp"
+                        % (_os.getpid(), id(t)))
+    header = MAGIC + bytes([1, CODEC_RAW, 0, 0])
+    try:
+        saver = _mfx_cas.StreamSaver(tmp, header, 0)
+        for chunk in _gpu_chunks_torch(t, torch):
+            saver.feed(chunk)
+        key = saver.finish()
+        ap = cas._storage._abs(cas._key_path(key))
+        if _os.path.isfile(ap):
+            _os.unlink(tmp)  # dedup
+        else:
+            _os.makedirs(_os.path.dirname(ap), exist_ok=True)
+            _os.replace(tmp, ap)
+        return key, nbytes
+    except Exception:
+        try:
+            _os.unlink(tmp)
+        except OSError:
+            pass
+        raise
+
+
 def save_state_dict(task_datastore, state_dict, name="checkpoint"):
     """Persist a state dict; returns {tensor_name: {sha, dtype, shape}}.
 
@@ -129,9 +200,32 @@ def save_state_dict(task_datastore, state_dict, name="checkpoint"):
         index[k] = {"sha": sha, "dtype": dtype, "shape": list(shape),
                     "nbytes": nbytes}
 
+    can_native = hasattr(cas._storage, "_abs")
+    if can_native:
+        try:
+            from ..ops import _mfx_cas  # noqa: F401
+        except Exception:
+            can_native = False
+
     with ThreadPoolExecutor(max_workers=1) as io_pool:
         for key, value in state_dict.items():
             if isinstance(value, torch.Tensor):
+                if value.is_cuda and can_native:
+                    # native path: C++ hashes+writes straight from the
+                    # pinned halves (no host-side numpy copies at all);
+                    # the pinned buffer is shared, so drain any pending
+                    # io_pool shard first
+                    if pending is not None:
+                        _resolve(pending)
+                        pending = None
+                    sha, nbytes = _save_gpu_tensor_native(value, cas,
+                                                          torch)
+                    t = value
+                    index[key] = {
+                        "sha": sha,
+                        "dtype": str(t.dtype).replace("torch.", ""),
+                        "shape": list(t.shape), "nbytes": nbytes}
+                    continue
                 buf, dtype, shape = _tensor_to_buffer(value)
                 if pending is not None:
                     _resolve(pending)

@@ -406,3 +406,44 @@ def test_cas_save_stream_matches_save_blobs(tmp_path):
     for dirpath, _d, names in _os.walk(str(tmp_path)):
         files += [n for n in names if k1[:8] in n]
     assert len(files) == 1
+
+
+def test_stream_saver_native_key_parity(tmp_path):
+    """C++ StreamSaver produces the same content keys as parallel_key
+    for both regimes, and the file round-trips through load_blobs."""
+    import numpy as np
+    import torch
+
+    from metaflow_amd.datastore.cas import (
+        CODEC_RAW,
+        MAGIC,
+        ContentAddressedStore,
+        parallel_key,
+    )
+    from metaflow_amd.datastore.storage import LocalStorage
+
+    try:
+        from metaflow_amd.ops import _mfx_cas
+    except ImportError:
+        import pytest
+
+        pytest.skip("_mfx_cas not built")
+    store = ContentAddressedStore("data", LocalStorage(str(tmp_path)))
+    header = MAGIC + bytes([1, CODEC_RAW, 0, 0])
+    rng = np.random.default_rng(11)
+    for size in (2 << 20, 40 << 20):
+        blob = rng.bytes(size)
+        t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+        import os as _os
+
+        tmp = str(tmp_path / ("s%d.tmp" % size))
+        sv = _mfx_cas.StreamSaver(tmp, header, 0)
+        for o in range(0, size, 16 << 20):
+            sv.feed(t[o:o + (16 << 20)])
+        key = sv.finish()
+        assert key == parallel_key(blob), size
+        ap = store._storage._abs(store._key_path(key))
+        _os.makedirs(_os.path.dirname(ap), exist_ok=True)
+        _os.replace(tmp, ap)
+        [(_k, back)] = list(store.load_blobs([key]))
+        assert bytes(back) == blob

@@ -535,3 +535,21 @@ def test_checkpoint_stream_roundtrip_gpu(dev, tmp_path):
     cpu_bytes = state["big"].cpu().reshape(-1).view(
         torch.uint8).numpy().tobytes()
     assert parallel_key(cpu_bytes) == index["big"]["sha"]
+
+
+def test_quant_e4m3_kernel(dev):
+    """One-pass quantize kernel vs the torch composition (bitwise —
+    both saturate to +-448 and round-to-nearest-even)."""
+    from metaflow_amd.ops.fp8 import E4M3_MAX
+    from metaflow_amd.ops.kernels import hip_ext
+
+    torch.manual_seed(0)
+    x = torch.randn(4096, 512, dtype=torch.bfloat16, device=dev) * 3
+    x[0, 0] = 10000.0   # saturation
+    x[0, 1] = -10000.0
+    scale = 7.5
+    got = hip_ext().quant_e4m3(x, scale)
+    ref = ((x.float() * scale).clamp(-E4M3_MAX, E4M3_MAX)
+           .to(torch.float8_e4m3fn).view(torch.uint8))
+    mismatch = (got != ref).float().mean().item()
+    assert mismatch < 1e-3, mismatch  # allow boundary rounding ties
