@@ -100,7 +100,19 @@ class ObjectStore(object):
     # ------------------------------------------------------------------ gets
     def get(self, key):
         def do(k):
-            with open(self._path(k), "rb") as f:
+            path = self._path(k)
+            _maybe_inject_failure()
+            # large objects go through the native parallel-pread engine
+            # (cas_engine.cpp: 16 MiB chunks across a thread pool,
+            # single allocation) — Python's f.read() is single-threaded
+            try:
+                if os.path.getsize(path) >= (32 << 20):
+                    from .ops import cas_native
+
+                    return cas_native.engine().load_blob_parallel(path, 0)
+            except ImportError:
+                pass
+            with open(path, "rb") as f:
                 return f.read()
 
         return self._with_retries(do, key)
