@@ -101,7 +101,6 @@ class ObjectStore(object):
     def get(self, key):
         def do(k):
             path = self._path(k)
-            _maybe_inject_failure()
             # large objects go through the native parallel-pread engine
             # (cas_engine.cpp: 16 MiB chunks across a thread pool,
             # single allocation) — Python's f.read() is single-threaded
