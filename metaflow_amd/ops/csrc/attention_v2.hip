@@ -555,12 +555,31 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
       // elements of row q0+hi*8+((l>>2)&3) at col offset 4*(l&3):
       const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
       const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
-      // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d)
-      bf16x8 pa[2];
+      // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d).
+      // dK's A-fragments (col_to_afrags1 on dS) are assembled BETWEEN
+      // the first dV tr_reads and their waitcnt: ~35 VALU ops of pack/
+      // shfl hide the DS-transpose latency instead of stalling on it.
+      bf16x8 pa[2], pak[2];
       col_to_afrags1(st, pa, hi);
+      {
+        const int q0 = t * 32;
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const int q0 = t * 32 + ks * 16;
+        for (int n = 0; n < 4; ++n) {
+          const int base = n * 2 * (BQ2 * 16) + tr_panel
+                           + (q0 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(dop + base);
+          bfr[n].u[1] = tr_read(dop + base + 4 * 16);
+        }
+        col_to_afrags1(dpt, pak, hi);   // VALU under the DS reads
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc_dv[n] = mfma32(pa[0], bfr[n].v, acc_dv[n]);
+      }
+      {
+        const int q0 = t * 32 + 16;
         union { unsigned long long u[2]; bf16x8 v; } bfr[4];
 #pragma unroll
         for (int n = 0; n < 4; ++n) {
@@ -573,10 +592,11 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
         __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
         for (int n = 0; n < 4; ++n)
-          acc_dv[n] = mfma32(pa[ks], bfr[n].v, acc_dv[n]);
+          acc_dv[n] = mfma32(pa[1], bfr[n].v, acc_dv[n]);
       }
       // dK(32kv x 128d) += dS^T(32kv x 32q) @ Q(32q x 128d)
-      col_to_afrags1(dpt, pa, hi);
+      pa[0] = pak[0];
+      pa[1] = pak[1];
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const int q0 = t * 32 + ks * 16;
