@@ -41,6 +41,10 @@ def parse_args():
                         "Megatron-style shard group over a single "
                         "shared batch")
     p.add_argument("--lr", type=float, default=3e-4)
+    p.add_argument("--zero", action="store_true",
+                   help="ZeRO-1: shard optimizer state (fp32 m/v) and "
+                        "grad ownership across DP ranks — the memory "
+                        "mode that fits llama3-70b on 8x288 GB")
     p.add_argument("--fp8", action="store_true",
                    help="decoder projections' forward GEMMs in OCP E4M3 "
                         "(delayed scaling, bf16 backward) — reported "
@@ -147,7 +151,7 @@ def main():
             if r == rank:
                 dp_group = g
     flat = FlatParamModel(model, bucket_mb=args.bucket_mb,
-                          group=dp_group)
+                          group=dp_group, zero=args.zero)
     flat.install_overlap_hooks()
     opt = FusedAdamW(flat, lr=args.lr)
     if rank == 0:
@@ -231,7 +235,9 @@ def main():
                 "global_batch": args.batch * (1 if shared else n_gpus),
                 "seq_len": seq,
                 "parallelism": ("tp%d" % n_gpus if use_tp else
-                                ("cp%d" if use_cp else "dp%d") % n_gpus),
+                                ("cp%d" if use_cp else
+                                 ("dp%d-zero1" if args.zero else "dp%d"))
+                                % n_gpus),
                 "first_loss": round(first_loss, 4),
                 "last_loss": round(last_loss, 4),
                 # NON-hidden gradient-comm wait per timed step (rank 0):

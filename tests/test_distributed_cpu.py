@@ -222,3 +222,85 @@ def test_bench_tiny_world8(tmp_path):
     assert rec["n_gpus"] == 8
     assert rec["config"]["parallelism"] == "dp8"
     assert rec["value"] > 0
+
+
+def _zero1_worker(rank, world, port, out_q):
+    import os
+
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+    })
+    import torch
+    import torch.distributed as dist
+
+    from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from metaflow_amd.parallel.ddp import FlatParamModel, FusedAdamW
+
+    dist.init_process_group("gloo")
+    torch.manual_seed(7)
+    model = LlamaForCausalLM(LlamaConfig.tiny(vocab=128, seq=64))
+    flat = FlatParamModel(model, zero=True)
+    flat.install_overlap_hooks()
+    opt = FusedAdamW(flat, lr=1e-3)
+    assert flat.zero_world == world
+    assert opt.m.numel() == flat.flat_param.numel() // world
+    torch.manual_seed(100 + rank)  # distinct per-rank batches (DP)
+    toks = torch.randint(0, 128, (2, 33))
+    for _ in range(3):
+        flat.zero_grad()
+        loss = model(toks[:, :-1], toks[:, 1:].contiguous())
+        loss.backward()
+        flat.finish_grad_sync()
+        opt.step()
+    out_q.put((rank, flat.flat_param.clone()))
+    dist.destroy_process_group()
+
+
+def test_zero1_matches_ddp_world2():
+    """ZeRO-1 (sharded optimizer state + grad ownership) produces the
+    SAME parameters on every rank, and they match a single-process run
+    fed the averaged gradients (the plain-DDP semantics)."""
+    import multiprocessing as mp
+
+    import torch
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_zero1_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, fp = q.get(timeout=300)
+        results[rank] = fp
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    # all-gathered params identical across ranks
+    assert torch.equal(results[0], results[1])
+
+    # single-process reference consuming the rank-averaged grads
+    from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from metaflow_amd.parallel.ddp import FlatParamModel, FusedAdamW
+
+    torch.manual_seed(7)
+    model = LlamaForCausalLM(LlamaConfig.tiny(vocab=128, seq=64))
+    flat = FlatParamModel(model)
+    opt = FusedAdamW(flat, lr=1e-3)
+    batches = []
+    for r in range(2):
+        torch.manual_seed(100 + r)
+        batches.append(torch.randint(0, 128, (2, 33)))
+    for _ in range(3):
+        flat.zero_grad()
+        for toks in batches:
+            loss = model(toks[:, :-1], toks[:, 1:].contiguous()) / 2
+            loss.backward()
+        opt.step()
+    n = flat.flat_param.numel()
+    diff = (results[0][:n].float()
+            - flat.flat_param.float()).abs().max().item()
+    assert diff < 1e-2, diff
