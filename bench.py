@@ -41,6 +41,10 @@ def parse_args():
                         "Megatron-style shard group over a single "
                         "shared batch")
     p.add_argument("--lr", type=float, default=3e-4)
+    p.add_argument("--recompute", action="store_true",
+                   help="checkpoint decoder-layer activations "
+                        "(recompute in backward): O(1)-layer activation "
+                        "memory for 70B/long-seq configs")
     p.add_argument("--zero", action="store_true",
                    help="ZeRO-1: shard optimizer state (fp32 m/v) and "
                         "grad ownership across DP ranks — the memory "
@@ -112,6 +116,8 @@ def main():
         assert args.model in ("llama3-8b", "llama3-70b", "tiny"), \
             "--fp8 supports the Llama family"
         cfg.fp8 = True
+    if args.recompute:
+        cfg.recompute = True
     seq = min(args.seq, cfg.max_seq_len)
 
     use_tp = args.tp and distributed

@@ -31,6 +31,10 @@ class LlamaConfig:
     # fp8=True runs the decoder projections' FORWARD GEMMs in OCP E4M3
     # (delayed scaling, bf16 backward — ops/fp8.py); lm_head stays bf16
     fp8: bool = False
+    # recompute=True checkpoints each decoder layer's activations
+    # (recomputed in backward): activation memory drops from O(layers)
+    # to O(1) layers — the knob that fits 70B/long-seq training
+    recompute: bool = False
 
     @classmethod
     def llama3_8b(cls):
@@ -279,9 +283,18 @@ class LlamaForCausalLM(nn.Module):
         sin_t = self.sin_t[off:off + S].contiguous() if off else self.sin_t
         res = self.embed(tokens)
         pending = None
+        recompute = (self.cfg.recompute and torch.is_grad_enabled()
+                     and cache is None)
         for li, layer in enumerate(self.layers):
-            res, pending = layer(res, pending, cos_t, sin_t,
-                                 self.cp_group, cache, li)
+            if recompute:
+                from torch.utils.checkpoint import checkpoint
+
+                res, pending = checkpoint(
+                    layer, res, pending, cos_t, sin_t, self.cp_group,
+                    cache, li, use_reentrant=False)
+            else:
+                res, pending = layer(res, pending, cos_t, sin_t,
+                                     self.cp_group, cache, li)
         if cache is not None:
             cache.pos += S
         _, x = K.add_rmsnorm(res, pending, self.final_norm.weight,

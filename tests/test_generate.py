@@ -47,3 +47,30 @@ def test_mixtral_generate_matches_full_forward():
         for t in range(11, 16):
             nxt = m(out[:, :t])[:, -1].float().argmax(-1)
             assert torch.equal(nxt, out[:, t]), t
+
+
+def test_activation_recompute_grads_match():
+    """cfg.recompute=True (per-layer activation checkpointing through
+    the custom-kernel autograd functions) reproduces the exact
+    gradients of the plain forward."""
+    import torch
+
+    from metaflow_amd.models.llama import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig.tiny(vocab=64, seq=64)
+    m1 = LlamaForCausalLM(cfg)
+    cfg2 = LlamaConfig.tiny(vocab=64, seq=64)
+    cfg2.recompute = True
+    torch.manual_seed(0)
+    m2 = LlamaForCausalLM(cfg2)
+    toks = torch.randint(0, 64, (2, 33))
+    l1 = m1(toks[:, :-1], toks[:, 1:].contiguous())
+    l1.backward()
+    l2 = m2(toks[:, :-1], toks[:, 1:].contiguous())
+    l2.backward()
+    assert torch.equal(l1.detach(), l2.detach())
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(),
+                                  m2.named_parameters()):
+        assert n1 == n2
+        assert torch.equal(p1.grad, p2.grad), n1
