@@ -69,3 +69,27 @@ def test_varlen_decode_cpu_matches_per_slot():
         err = ((o[b:b + 1] - ref).float().norm()
                / (ref.float().norm() + 1e-8)).item()
         assert err < 2e-2
+
+
+def test_serving_flow_example(tmp_path):
+    """examples/serving_flow.py end to end: a 2-rank gang drains the
+    request stream through ContinuousBatcher (gloo/CPU here; the same
+    flow runs GPU-resident on a box)."""
+    import os
+    import subprocess
+    import sys
+
+    from .test_runtime import REPO
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    ds = tmp_path / "ds"
+    proc = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples",
+                                      "serving_flow.py"),
+         "--datastore-root", str(ds), "run",
+         "--ranks", "2", "--num-requests", "6"],
+        env=env, capture_output=True, text=True, timeout=600)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    assert "served" in proc.stdout
