@@ -446,3 +446,22 @@ __global__ void quant_e4m3_kernel(const short* __restrict__ x,
     *(int2v*)(y + i * 8) = out;
   }
 }
+
+// -------------------------------------------------- token decode
+// Pretraining shards store tokens as packed uint16 (2 B/token); models
+// consume int64. Decoding ON the GPU means the PCIe/H2D copy moves 2
+// bytes per token instead of 8 — the foreach data-pipe decode kernel
+// (SURVEY §2.4): one memory-bound pass, vectorized 8 tokens/lane.
+__global__ void decode_tokens_u16_kernel(
+    const unsigned short* __restrict__ x, long long* __restrict__ y,
+    long long n8) {
+  typedef __attribute__((ext_vector_type(8))) short u16x8;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n8; i += stride) {
+    u16x8 v = *(const u16x8*)(x + i * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      y[i * 8 + j] = (long long)(unsigned short)v[j];
+  }
+}

@@ -207,6 +207,20 @@ torch::Tensor swiglu_gu_bwd(torch::Tensor gu, torch::Tensor dy) {
   return dgu;
 }
 
+torch::Tensor decode_tokens_u16(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kUInt8, "pass raw bytes (uint8)");
+  TORCH_CHECK(x.numel() % 16 == 0,
+              "token buffer must hold a multiple of 8 uint16 tokens");
+  const long long n = x.numel() / 2;
+  auto y = torch::empty({n}, x.options().dtype(torch::kInt64));
+  decode_tokens_u16_kernel<<<grid_for(n / 8), kBlock, 0, cur_stream()>>>(
+      reinterpret_cast<const unsigned short*>(x.data_ptr()),
+      reinterpret_cast<long long*>(y.data_ptr<int64_t>()), n / 8);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
 torch::Tensor quant_e4m3(torch::Tensor x, double scale) {
   check_bf16(x, "x");
   TORCH_CHECK(x.numel() % 8 == 0, "quant_e4m3 needs numel % 8 == 0");
@@ -453,6 +467,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_gu_fwd", [](torch::Tensor gu){ return swiglu_gu_fwd(gu)[0]; }, "fused-layout SwiGLU fwd");
   m.def("swiglu_gu_bwd", &swiglu_gu_bwd, "fused-layout SwiGLU bwd");
   m.def("add_bf16", &add_bf16, "fused bf16 add");
+  m.def("decode_tokens_u16", &decode_tokens_u16,
+        "packed uint16 token bytes -> int64 tokens (one pass on-GPU)");
   m.def("quant_e4m3", &quant_e4m3,
         "one-pass bf16 -> OCP E4M3 (uint8 storage) with scale");
   m.def("adamw", &adamw, "fused AdamW (bf16 p/g, fp32 m/v[, master])");

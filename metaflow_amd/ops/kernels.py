@@ -529,6 +529,20 @@ def adamw_step(p, g, m, v, step, lr, beta1=0.9, beta2=0.95, eps=1e-8,
     p.copy_(upd.to(p.dtype))
 
 
+def decode_tokens_u16(raw):
+    """Packed little-endian uint16 token bytes -> int64 token ids.
+
+    raw: uint8 tensor (GPU: one-pass decode kernel, so the host->device
+    copy moves 2 B/token instead of 8; CPU: numpy view). The data-pipe
+    decode for pretraining shards (SURVEY §2.4 foreach data path).
+    """
+    import torch as _torch
+
+    if raw.is_cuda:
+        return hip_ext().decode_tokens_u16(raw.contiguous())
+    return raw.view(_torch.int16).to(_torch.int64) & 0xFFFF
+
+
 def add_bf16(a, b):
     if a.is_cuda:
         return hip_ext().add_bf16(a.contiguous(), b.contiguous())

@@ -57,7 +57,7 @@ ext_modules = [
         sources=[os.path.join(CSRC, "mfx_hip.hip")],
         extra_compile_args={
             "cxx": ["-O3", "-std=c++17"],
-            "nvcc": ["-O3", "-std=c++17", "-Rpass-analysis=kernel-resource-usage"],
+            "nvcc": ["-O3", "-std=c++17"],
         },
     ),
     CUDAExtension(

@@ -553,3 +553,18 @@ def test_quant_e4m3_kernel(dev):
            .to(torch.float8_e4m3fn).view(torch.uint8))
     mismatch = (got != ref).float().mean().item()
     assert mismatch < 1e-3, mismatch  # allow boundary rounding ties
+
+
+def test_decode_tokens_u16(dev):
+    """On-GPU uint16 token decode matches the numpy view (2 B/token over
+    the bus instead of 8)."""
+    from metaflow_amd.ops import kernels as K
+
+    torch.manual_seed(0)
+    toks = torch.randint(0, 65536, (1 << 16,), dtype=torch.int32)
+    raw = toks.to(torch.int16).view(torch.uint8)
+    got = K.decode_tokens_u16(raw.to(dev))
+    want = K.decode_tokens_u16(raw)
+    assert got.dtype == torch.int64
+    assert torch.equal(got.cpu(), want)
+    assert torch.equal(want, toks.to(torch.int64))
