@@ -555,64 +555,53 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
       // elements of row q0+hi*8+((l>>2)&3) at col offset 4*(l&3):
       const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
       const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
-      // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d).
-      // dK's A-fragments (col_to_afrags1 on dS) are assembled BETWEEN
-      // the first dV tr_reads and their waitcnt: ~35 VALU ops of pack/
-      // shfl hide the DS-transpose latency instead of stalling on it.
+      // dV(32kv x 128d) += P^T(32kv x 32q) @ dO(32q x 128d), then
+      // dK += dS^T @ Q — FOUR ds_read_b64_tr groups, software-pipelined
+      // with COUNTED waits: group g+1's 8 DS reads issue before waiting
+      // on group g (DS returns in order, so lgkmcnt(8) = "the earlier 8
+      // are done"), so the transpose-read latency of every group but
+      // the last hides under the previous group's MFMAs. The PMC run
+      // (gpurun_out/pmc) showed both backward kernels stalled on these
+      // full-stop groups, not on LDS bandwidth (LdsUtil 12%, conflicts
+      // 4%).
       bf16x8 pa[2], pak[2];
       col_to_afrags1(st, pa, hi);
-      {
-        const int q0 = t * 32;
-        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
+      union { unsigned long long u[2]; bf16x8 v; } bfr[2][4];
+      const short* g_tile[4] = {dop, dop, qp, qp};
+      int g_q0[4];
+      g_q0[0] = t * 32;
+      g_q0[1] = t * 32 + 16;
+      g_q0[2] = t * 32;
+      g_q0[3] = t * 32 + 16;
 #pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int base = n * 2 * (BQ2 * 16) + tr_panel
-                           + (q0 + hi * 8) * 16 + tr_lane_off;
-          bfr[n].u[0] = tr_read(dop + base);
-          bfr[n].u[1] = tr_read(dop + base + 4 * 16);
-        }
-        col_to_afrags1(dpt, pak, hi);   // VALU under the DS reads
-        asm volatile("s_waitcnt lgkmcnt(0)");
-        __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-        for (int n = 0; n < 4; ++n)
-          acc_dv[n] = mfma32(pa[0], bfr[n].v, acc_dv[n]);
+      for (int n = 0; n < 4; ++n) {
+        const int base = n * 2 * (BQ2 * 16) + tr_panel
+                         + (g_q0[0] + hi * 8) * 16 + tr_lane_off;
+        bfr[0][n].u[0] = tr_read(g_tile[0] + base);
+        bfr[0][n].u[1] = tr_read(g_tile[0] + base + 4 * 16);
       }
-      {
-        const int q0 = t * 32 + 16;
-        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
+      col_to_afrags1(dpt, pak, hi);     // VALU under the first DS group
 #pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int base = n * 2 * (BQ2 * 16) + tr_panel
-                           + (q0 + hi * 8) * 16 + tr_lane_off;
-          bfr[n].u[0] = tr_read(dop + base);
-          bfr[n].u[1] = tr_read(dop + base + 4 * 16);
+      for (int g = 0; g < 4; ++g) {
+        const int cur = g & 1;
+        if (g < 3) {
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int base = n * 2 * (BQ2 * 16) + tr_panel
+                             + (g_q0[g + 1] + hi * 8) * 16 + tr_lane_off;
+            bfr[cur ^ 1][n].u[0] = tr_read(g_tile[g + 1] + base);
+            bfr[cur ^ 1][n].u[1] = tr_read(g_tile[g + 1] + base + 4 * 16);
+          }
+          asm volatile("s_waitcnt lgkmcnt(8)");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(0)");
         }
-        asm volatile("s_waitcnt lgkmcnt(0)");
         __builtin_amdgcn_sched_barrier(0);
+        const bf16x8 a = (g < 2) ? pa[g] : pak[g - 2];
+        f16f* acc = (g < 2) ? acc_dv : acc_dk;
 #pragma unroll
         for (int n = 0; n < 4; ++n)
-          acc_dv[n] = mfma32(pa[1], bfr[n].v, acc_dv[n]);
-      }
-      // dK(32kv x 128d) += dS^T(32kv x 32q) @ Q(32q x 128d)
-      pa[0] = pak[0];
-      pa[1] = pak[1];
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const int q0 = t * 32 + ks * 16;
-        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int base = n * 2 * (BQ2 * 16) + tr_panel
-                           + (q0 + hi * 8) * 16 + tr_lane_off;
-          bfr[n].u[0] = tr_read(qp + base);
-          bfr[n].u[1] = tr_read(qp + base + 4 * 16);
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)");
-        __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-        for (int n = 0; n < 4; ++n)
-          acc_dk[n] = mfma32(pa[ks], bfr[n].v, acc_dk[n]);
+          acc[n] = mfma32(a, bfr[cur][n].v, acc[n]);
       }
     }
   }
