@@ -26,10 +26,14 @@ DEV f16f mfma32(bf16x8 a, bf16x8 b, f16f c) {
   return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
 }
 
-// pack two fp32 into one u32 of two bf16 (lo, hi)
+// pack two fp32 into one u32 of two bf16 (lo, hi): single
+// v_cvt_pk_bf16_f32 on gfx950 (RNE) — the manual shift/round sequence
+// was ~12 VALU ops and the P->A-fragment repack runs twice per subtile
+// in every attention kernel
 DEV unsigned int pack_bf2(float lo, float hi) {
-  return ((unsigned int)(unsigned short)f2bf(lo))
-         | ((unsigned int)(unsigned short)f2bf(hi) << 16);
+  union { __hip_bfloat162 h; unsigned int u; } c;
+  c.h = __float22bfloat162_rn(make_float2(lo, hi));
+  return c.u;
 }
 
 // Turn a per-lane C-tile column (st[2][16]: lane holds values X[row][mycol]
