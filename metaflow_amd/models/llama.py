@@ -239,6 +239,10 @@ class LlamaForCausalLM(nn.Module):
     cp ranks join the flat-buffer DDP all-reduce exactly like extra data-
     parallel ranks (parallel/ddp.py needs no changes)."""
 
+    #: decode_step is free of data-dependent shapes -> the serving
+    #: engine may capture it in a hipGraph and replay per token
+    graph_safe_decode = True
+
     def __init__(self, cfg: LlamaConfig, cp_group=None):
         super().__init__()
         self.cfg = cfg

@@ -296,6 +296,37 @@ class MixtralDecoderLayer(nn.Module):
         return x
 
 
+    def decode_step(self, x, c_rows, s_rows, cache, layer_idx, pos_t,
+                    lengths, max_len=None):
+        """Continuous-batching decode for the MoE layer: per-slot cache
+        scatter + varlen flash-decode; experts route per decoded
+        token."""
+        import math
+
+        from .llama import _rope_rows
+
+        cfg = self.cfg
+        B = x.size(0)
+        hd = cfg.head_dim
+        res = x
+        y = self.input_norm(x)
+        q = self.q_proj(y).view(B, 1, cfg.num_heads, hd).transpose(1, 2)
+        k = self.k_proj(y).view(B, 1, cfg.num_kv_heads, hd).transpose(1, 2)
+        v = self.v_proj(y).view(B, 1, cfg.num_kv_heads, hd).transpose(1, 2)
+        q = _rope_rows(q, c_rows, s_rows)
+        k = _rope_rows(k, c_rows, s_rows)
+        kc, vc = cache.k[layer_idx], cache.v[layer_idx]
+        bidx = torch.arange(B, device=x.device)
+        kc[bidx, :, pos_t] = k[:, :, 0]
+        vc[bidx, :, pos_t] = v[:, :, 0]
+        o = K.attn_decode_varlen(q.contiguous(), kc, vc, lengths,
+                                 1.0 / math.sqrt(hd), max_len=max_len)
+        o = o.transpose(1, 2).reshape(B, 1, cfg.num_heads * hd)
+        x = res + self.o_proj(o)
+        x = x + self.moe(self.post_norm(x))
+        return x
+
+
 class MixtralForCausalLM(nn.Module):
     def __init__(self, cfg: MixtralConfig, ep_group=None):
         super().__init__()
@@ -336,6 +367,27 @@ class MixtralForCausalLM(nn.Module):
         loss = K.cross_entropy(logits.reshape(B * S, V),
                                targets.reshape(B * S))
         return loss.mean()
+
+    @torch.no_grad()
+    def decode_step(self, tokens, cache, positions, max_len=None):
+        """One batched decode step for continuous batching (same
+        contract as LlamaForCausalLM.decode_step: per-slot cache
+        positions, varlen flash-decode). NOT hipGraph-capturable — the
+        MoE router's token dispatch is shape-dynamic — so the serving
+        engine keeps Mixtral on the eager decode path
+        (graph_safe_decode stays False)."""
+        from .llama import _rope_rows
+
+        pos_t = torch.as_tensor(positions, device=tokens.device,
+                                dtype=torch.long)
+        lengths = (pos_t + 1).to(torch.int32)
+        c_rows = self.cos_t[pos_t]
+        s_rows = self.sin_t[pos_t]
+        x = self.embed(tokens)
+        for li, layer in enumerate(self.layers):
+            x = layer.decode_step(x, c_rows, s_rows, cache, li, pos_t,
+                                  lengths, max_len=max_len)
+        return self.lm_head(self.final_norm(x))
 
     @torch.no_grad()
     def generate(self, tokens, max_new_tokens, temperature=0.0,

@@ -69,7 +69,13 @@ class ContinuousBatcher(object):
         self.queue = deque()
         self._torch = torch
         self._graph = None
-        if graph and device.type == "cuda":
+        if graph is True and not getattr(model, "graph_safe_decode",
+                                         False):
+            raise ValueError(
+                "model's decode_step is not hipGraph-safe (data-"
+                "dependent shapes, e.g. MoE routing)")
+        if graph and device.type == "cuda" and \
+                getattr(model, "graph_safe_decode", False):
             try:
                 self._capture_graph()
             except Exception:

@@ -93,3 +93,25 @@ def test_serving_flow_example(tmp_path):
         env=env, capture_output=True, text=True, timeout=600)
     assert proc.returncode == 0, proc.stderr[-3000:]
     assert "served" in proc.stdout
+
+
+def test_continuous_batching_mixtral():
+    """The serving engine drives the MoE family too: greedy tokens from
+    the batcher match per-request Mixtral generate() (eager decode —
+    MoE routing is shape-dynamic, so no graph capture)."""
+    from metaflow_amd.models.mixtral import (
+        MixtralConfig,
+        MixtralForCausalLM,
+    )
+
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(MixtralConfig.tiny(vocab=128, seq=256)).eval()
+    prompts = [([5, 9, 17, 4], 5), (list(range(2, 20)), 4),
+               ([100, 101], 6)]
+    batcher = ContinuousBatcher(m, max_batch=2, max_len=64)
+    reqs = [batcher.submit(p, n) for p, n in prompts]
+    out = batcher.run()
+    for req, (prompt, n) in zip(reqs, prompts):
+        ref = m.generate(torch.tensor([prompt]),
+                         n)[0, len(prompt):].tolist()
+        assert out[req.id] == ref, (req.id, out[req.id], ref)
