@@ -214,8 +214,11 @@ class CardDecorator(StepDecorator):
         if rocprof is None:
             return
         out_dir = tempfile.mkdtemp(prefix="mfx_rocprof_")
-        args["cmd"] = [rocprof, "--kernel-trace", "--stats", "-d",
-                       out_dir, "--"] + args["cmd"]
+        # --output-format csv: rocprofv3's default output is a SQLite
+        # results.db; the stats CSV the splicer parses needs asking for
+        args["cmd"] = [rocprof, "--kernel-trace", "--stats",
+                       "--output-format", "csv", "-d", out_dir,
+                       "--"] + args["cmd"]
         args["env"]["MFX_ROCPROF_OUT"] = out_dir
         args["env"].setdefault("TMPDIR", "/tmp")
 

@@ -142,6 +142,7 @@ def _gpu_monitor_worker(context):
         with open(out_path, "a") as f:
             f.write(json.dumps(rec) + "\n")
 
+    sample()  # immediate first sample: short tasks get telemetry too
     while True:
         ready, _w, _x = select.select([sys.stdin], [], [], 5.0)
         if ready:
