@@ -390,6 +390,15 @@ class Task(MetaflowObject):
         return list(merge_logs([raw]))
 
     @property
+    def card_html(self):
+        """The rendered @card HTML for this task (with the scheduler's
+        rocprof kernel-breakdown spliced in when @card(profile=True)),
+        or None."""
+        from ..plugins.card_decorator import get_card
+
+        return get_card(self._ds)
+
+    @property
     def parent(self):
         return Step("%s/%s/%s" % (self.flow_name, self.run_id,
                                   self.step_name))

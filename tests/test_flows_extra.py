@@ -407,3 +407,24 @@ def test_card_profile_splice(tmp_path):
     doc = get_card(rd)
     assert "KERNELS" in doc
     assert "Kernel-time breakdown" in doc
+
+
+def test_client_card_html(tmp_datastore):
+    """Client Task.card_html returns the rendered card."""
+    import os
+
+    from .test_runtime import latest_run_id, run_flow
+
+    run_flow("card_flow.py", tmp_datastore, "run")
+    os.environ["MFX_DATASTORE_SYSROOT_LOCAL"] = tmp_datastore
+    try:
+        from metaflow_amd.client import Flow, namespace
+
+        namespace(None)
+        run_id = latest_run_id(tmp_datastore, "CardFlow")
+        flow = Flow("CardFlow")
+        task = flow[run_id]["start"].task
+        doc = task.card_html
+        assert doc and "<html" in doc
+    finally:
+        del os.environ["MFX_DATASTORE_SYSROOT_LOCAL"]
