@@ -259,7 +259,10 @@ class FusedAdamW(object):
                     fp[r * ss:(r + 1) * ss].copy_(sh)
 
     def state_dict_tensors(self):
-        """Shard tensors for @checkpoint (per-rank)."""
+        """Shard tensors for @checkpoint (per-rank). Under ZeRO-1, m/v
+        (and master) are this rank's 1/world shard — restore requires
+        the same world size and rank mapping (the gang scheduler's
+        stable rank->GPU pinning guarantees it on resume)."""
         out = {"flat_param": self.flat.flat_param, "adam_m": self.m,
                "adam_v": self.v}
         if self.master is not None:
