@@ -11,6 +11,18 @@ _tried = False
 
 def engine():
     global _engine, _tried
+    if _engine is None:
+        import sys
+
+        if "torch" not in sys.modules:
+            # the extension links torch's libs, so first use would pull
+            # in the FULL torch import (~1.5 s) — that cost made every
+            # small-artifact task subprocess 10x slower (bench_flow
+            # 0.47 s -> 5.3 s). Processes that already use torch
+            # (training, checkpoint, datatools on big tensors) get the
+            # native engine; everyone else keeps the plain-python path.
+            raise ImportError(
+                "native CAS engine deferred: torch not imported")
     if not _tried:
         _tried = True
         try:
