@@ -23,6 +23,9 @@ sys.path.insert(0, ROOT)
 
 def bench_cas_io(size_gb):
     import numpy as np
+    import torch  # noqa: F401 — unlock the native CAS engine (it is a
+    # torch extension; torch-less processes deliberately keep the plain
+    # python IO path to avoid the 1.5 s import in small tasks)
 
     from metaflow_amd.datastore.cas import ContentAddressedStore
     from metaflow_amd.datastore.storage import LocalStorage
