@@ -62,6 +62,26 @@ class _Fp8LinearFn(torch.autograd.Function):
         return dx, dw, None, None
 
 
+def _in_recompute():
+    """True inside torch.utils.checkpoint's backward recompute — state
+    updates (amax history, step counter) must NOT re-run there, and the
+    recompute must reuse the ORIGINAL forward's scales or the
+    recomputed activations diverge from what autograd saved."""
+    try:
+        from torch.utils.checkpoint import is_recomputing
+
+        return bool(is_recomputing())
+    except Exception:
+        pass
+    try:
+        # this torch predates is_recomputing(): checkpoint recompute is
+        # the only way this module's forward runs INSIDE a backward
+        # graph task
+        return torch._C._current_graph_task_id() != -1
+    except Exception:
+        return False
+
+
 class Fp8Linear(torch.nn.Module):
     """Bias-free linear with an E4M3 forward GEMM and delayed scaling."""
 
@@ -93,8 +113,14 @@ class Fp8Linear(torch.nn.Module):
     def forward(self, x):
         if not (x.is_cuda and hasattr(torch, "float8_e4m3fn")):
             return torch.nn.functional.linear(x, self.weight)
-        sx, sw = self._scales()  # delayed: previous steps' amax window
-        self._update_amax(x)
+        if _in_recompute():
+            # activation-checkpoint recompute: reuse the original
+            # forward's scales, touch no state
+            sx, sw = self._last_scales
+        else:
+            sx, sw = self._scales()  # delayed: prior steps' amax window
+            self._last_scales = (sx, sw)
+            self._update_amax(x)
         if self._step == 1:
             # no history yet: first step runs bf16 (standard warmup)
             return torch.nn.functional.linear(x, self.weight)
