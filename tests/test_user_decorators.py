@@ -280,3 +280,69 @@ def test_wrapper_composes_with_catch_and_foreach(tmp_path, tmp_datastore):
         if __name__ == "__main__":
             FlowG()
     """)
+
+
+def test_user_wrapper_with_resume(tmp_path, tmp_datastore):
+    """User wrappers re-run on resumed (non-cloned) steps and stay
+    silent on cloned ones; composes with IncludeFile lazy decode."""
+    data = tmp_path / "inc.txt"
+    data.write_text("alpha\nbeta\n")
+    marker = tmp_path / "wrapped_runs"
+    flow = tmp_path / "wr_flow.py"
+    flow.write_text('''
+import os
+
+from metaflow_amd import FlowSpec, IncludeFile, UserStepDecorator, step
+
+
+class Count(UserStepDecorator):
+    def pre_step(self, step_name, flow, inputs=None):
+        with open(os.environ["WRAP_MARKER"], "a") as f:
+            f.write(step_name + "\\n")
+
+
+class WRFlow(FlowSpec):
+    notes = IncludeFile("notes", required=True)
+
+    @step
+    def start(self):
+        self.n = len(self.notes.splitlines())
+        self.next(self.mid)
+
+    @Count
+    @step
+    def mid(self):
+        if os.environ.get("WR_FAIL") == "1":
+            raise RuntimeError("fail first")
+        self.m = self.n * 2
+        self.next(self.end)
+
+    @step
+    def end(self):
+        assert self.m == 4, self.m
+
+
+if __name__ == "__main__":
+    WRFlow()
+''')
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    env["WRAP_MARKER"] = str(marker)
+    env["WR_FAIL"] = "1"
+    p1 = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run", "--notes", str(data)],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert p1.returncode != 0
+    env["WR_FAIL"] = "0"
+    p2 = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "resume"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert p2.returncode == 0, p2.stderr[-3000:]
+    # wrapper ran on the failed attempt AND the resumed re-run of mid
+    assert marker.read_text().count("mid") == 2
