@@ -665,3 +665,219 @@ __global__ void dkdv_reduce_kernel(const float* __restrict__ dk_part,
     *(bf16x4*)(dv + i) = vo;
   }
 }
+
+// ============== BACKWARD v2 (split variant): dV-only / dK-only ============
+// Experiment (MFX_ATTN_DKDV_SPLIT=1): the fused dkdv kernel holds 128
+// live accumulator VGPRs (dk+dv) vs the dq kernel's 64, and measures
+// 187-260 TF/s vs dq's 357 at identical occupancy with every
+// latency-hiding experiment flat — pointing at scheduler ILP starved
+// by register pressure. Splitting halves the live accumulators per
+// kernel at the cost of recomputing S^T (5 GEMM-equivalents total vs
+// 4): dV needs only S^T -> P -> P^T·dO (K resident, no V, no delta);
+// dK needs S^T, dP^T, dS -> dS^T·Q (K+V resident). waves_per_eu(3) on
+// the dV kernel caps it at 170 VGPRs for 3 waves/SIMD.
+
+template <int BLOCK>
+__global__ __launch_bounds__(512)
+__attribute__((amdgpu_waves_per_eu(3))) void attn_bwd_dv_v2_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ dout, const float* __restrict__ lse,
+    float* __restrict__ dv_part,
+    int B, int H, int Hkv, int S, float scale, int causal) {
+  constexpr int BKVB = 256, BQ2 = 64;
+  __shared__ short qp[BQ2 * ATT_D];
+  __shared__ short dop[BQ2 * ATT_D];
+  __shared__ float lse_s[BQ2];
+
+  const int kvb = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int G = H / Hkv;
+  const int hk = h / G;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int my_kvrow = kvb * BKVB + wid * 32 + l32;
+  const long long kvoff =
+      (((long long)b * Hkv + hk) * S + kvb * BKVB) * ATT_D;
+  const short* krow = k + kvoff + (long long)(wid * 32 + l32) * ATT_D;
+
+  f16f acc_dv[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) acc_dv[n] = (f16f){};
+
+  const long long hoff = ((long long)b * H + h) * S;
+  const int jq0 = causal ? (kvb * BKVB) / BQ2 : 0;
+  const int nq = S / BQ2;
+  for (int jq = jq0; jq < nq; ++jq) {
+    __syncthreads();
+    const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
+    const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
+    stage_panel<BQ2, BLOCK>(qp, qsrc, ATT_D);
+    stage_panel<BQ2, BLOCK>(dop, dsrc, ATT_D);
+    if (threadIdx.x < BQ2)
+      lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
+    __syncthreads();
+    if (causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
+
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f16f st = (f16f){};
+#pragma unroll
+      for (int s = 0; s < 8; ++s) {
+        bf16x8 qf = frag8_panel<BQ2>(qp, t * 32 + l32, s * 16 + hi * 8);
+        bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
+        st = mfma32(qf, kr, st);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qrg = jq * BQ2 + qrl;
+        float p = 0.f;
+        if (!causal || qrg >= my_kvrow)
+          p = __expf(st[r] * scale - lse_s[qrl]);
+        st[r] = p;
+      }
+      const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
+      const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
+      bf16x8 pa[2];
+      col_to_afrags1(st, pa, hi);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int q0 = t * 32 + ks * 16;
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int base = n * 2 * (BQ2 * 16) + tr_panel
+                           + (q0 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(dop + base);
+          bfr[n].u[1] = tr_read(dop + base + 4 * 16);
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc_dv[n] = mfma32(pa[ks], bfr[n].v, acc_dv[n]);
+      }
+    }
+  }
+
+  const long long poff = (hoff + (long long)kvb * BKVB) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const long long obase =
+        poff + (long long)(wid * 32 + row_local) * ATT_D;
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      dv_part[obase + n * 32 + l32] = acc_dv[n][r];
+  }
+}
+
+template <int BLOCK>
+__global__ __launch_bounds__(512) void attn_bwd_dk_v2_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    float* __restrict__ dk_part,
+    int B, int H, int Hkv, int S, float scale, int causal) {
+  constexpr int BKVB = 256, BQ2 = 64;
+  __shared__ short qp[BQ2 * ATT_D];
+  __shared__ short dop[BQ2 * ATT_D];
+  __shared__ float lse_s[BQ2];
+  __shared__ float del_s[BQ2];
+
+  const int kvb = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int G = H / Hkv;
+  const int hk = h / G;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int my_kvrow = kvb * BKVB + wid * 32 + l32;
+  const long long kvoff =
+      (((long long)b * Hkv + hk) * S + kvb * BKVB) * ATT_D;
+  const short* krow = k + kvoff + (long long)(wid * 32 + l32) * ATT_D;
+  const short* vrow = v + kvoff + (long long)(wid * 32 + l32) * ATT_D;
+
+  f16f acc_dk[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) acc_dk[n] = (f16f){};
+
+  const long long hoff = ((long long)b * H + h) * S;
+  const int jq0 = causal ? (kvb * BKVB) / BQ2 : 0;
+  const int nq = S / BQ2;
+  for (int jq = jq0; jq < nq; ++jq) {
+    __syncthreads();
+    const short* qsrc = q + (hoff + (long long)jq * BQ2) * ATT_D;
+    const short* dsrc = dout + (hoff + (long long)jq * BQ2) * ATT_D;
+    stage_panel<BQ2, BLOCK>(qp, qsrc, ATT_D);
+    stage_panel<BQ2, BLOCK>(dop, dsrc, ATT_D);
+    if (threadIdx.x < BQ2) {
+      lse_s[threadIdx.x] = lse[hoff + jq * BQ2 + threadIdx.x];
+      del_s[threadIdx.x] = delta[hoff + jq * BQ2 + threadIdx.x];
+    }
+    __syncthreads();
+    if (causal && jq * BQ2 + BQ2 - 1 < kvb * BKVB + wid * 32) continue;
+
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f16f st = (f16f){}, dpt = (f16f){};
+#pragma unroll
+      for (int s = 0; s < 8; ++s) {
+        bf16x8 qf = frag8_panel<BQ2>(qp, t * 32 + l32, s * 16 + hi * 8);
+        bf16x8 df = frag8_panel<BQ2>(dop, t * 32 + l32, s * 16 + hi * 8);
+        bf16x8 kr = *(const bf16x8*)(krow + s * 16 + hi * 8);
+        bf16x8 vr = *(const bf16x8*)(vrow + s * 16 + hi * 8);
+        st = mfma32(qf, kr, st);
+        dpt = mfma32(df, vr, dpt);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qrg = jq * BQ2 + qrl;
+        float p = 0.f;
+        if (!causal || qrg >= my_kvrow)
+          p = __expf(st[r] * scale - lse_s[qrl]);
+        dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
+      }
+      const int tr_lane_off = (((lane >> 2) & 3) * 16) + (lane & 3) * 4;
+      const int tr_panel = ((lane >> 4) & 1) * (BQ2 * 16);
+      bf16x8 pa[2];
+      col_to_afrags1(dpt, pa, hi);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int q0 = t * 32 + ks * 16;
+        union { unsigned long long u[2]; bf16x8 v; } bfr[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int base = n * 2 * (BQ2 * 16) + tr_panel
+                           + (q0 + hi * 8) * 16 + tr_lane_off;
+          bfr[n].u[0] = tr_read(qp + base);
+          bfr[n].u[1] = tr_read(qp + base + 4 * 16);
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc_dk[n] = mfma32(pa[ks], bfr[n].v, acc_dk[n]);
+      }
+    }
+  }
+
+  const long long poff = (hoff + (long long)kvb * BKVB) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const long long obase =
+        poff + (long long)(wid * 32 + row_local) * ATT_D;
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      dk_part[obase + n * 32 + l32] = acc_dk[n][r];
+  }
+}

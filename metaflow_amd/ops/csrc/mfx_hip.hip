@@ -356,12 +356,32 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto dk_part = torch::empty({B, H, S, 128}, f32opts);
   auto dv_part = torch::empty({B, H, S, 128}, f32opts);
   dim3 gkv(S / 256, H, B);
-  attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
-      bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
-      delta.data_ptr<float>(), dk_part.data_ptr<float>(),
-      dv_part.data_ptr<float>(), B, H, Hkv, S, (float)scale,
-      causal ? 1 : 0);
-  HIP_CHECK_KERNEL();
+  // MFX_ATTN_DKDV_SPLIT=1: dV-only + dK-only kernels (64 live
+  // accumulator VGPRs each instead of 128; S^T recomputed) — the
+  // register-pressure/ILP experiment documented in attention_v2.hip
+  static const bool split_dkdv = [] {
+    const char* e = getenv("MFX_ATTN_DKDV_SPLIT");
+    return e && e[0] == '1';
+  }();
+  if (split_dkdv) {
+    attn_bwd_dv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(dout), lse.data_ptr<float>(),
+        dv_part.data_ptr<float>(), B, H, Hkv, S, (float)scale,
+        causal ? 1 : 0);
+    HIP_CHECK_KERNEL();
+    attn_bwd_dk_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+        delta.data_ptr<float>(), dk_part.data_ptr<float>(), B, H, Hkv,
+        S, (float)scale, causal ? 1 : 0);
+    HIP_CHECK_KERNEL();
+  } else {
+    attn_bwd_dkdv_v2_kernel<512><<<gkv, 512, 0, cur_stream()>>>(
+        bf(q), bf(k), bf(v), bf(dout), lse.data_ptr<float>(),
+        delta.data_ptr<float>(), dk_part.data_ptr<float>(),
+        dv_part.data_ptr<float>(), B, H, Hkv, S, (float)scale,
+        causal ? 1 : 0);
+    HIP_CHECK_KERNEL();
+  }
   {
     const long long n4_kv = (long long)B * Hkv * S * 128 / 4;
     const long long head_elems = (long long)S * 128;
