@@ -541,16 +541,28 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v2_kernel(
         st = mfma32(qf, kr, st);
         dpt = mfma32(df, vr, dpt);
       }
-      // P (into st) with causal mask q >= kv; dS (into dpt)
+      // P (into st) with causal mask q >= kv; dS (into dpt). lse/delta
+      // come as f32x4 GROUP loads: the per-element lse_s[qrl] reads were
+      // 64 scalar ds_read_b32 per tile (the dq kernel holds its lse in
+      // a register) — the disassembly showed them, plus their waitcnts,
+      // matching the mfma count, which is where dkdv's efficiency gap
+      // vs dq was hiding. qrl = t*32 + (r&3) + 8*(r>>2) + 4*hi, so
+      // r = g*4+j walks 4 consecutive floats at t*32 + 8g + 4hi.
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int qrl = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const int qrg = jq * BQ2 + qrl;
-        float p = 0.f;
-        if (!causal || qrg >= my_kvrow)
-          p = __expf(st[r] * scale - lse_s[qrl]);
-        st[r] = p;
-        dpt[r] = p * (dpt[r] - del_s[qrl]) * scale;
+      for (int g = 0; g < 4; ++g) {
+        const int q0l = t * 32 + g * 8 + 4 * hi;
+        const f32x4 lv = *(const f32x4*)&lse_s[q0l];
+        const f32x4 dl = *(const f32x4*)&del_s[q0l];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int r = g * 4 + j;
+          const int qrg = jq * BQ2 + q0l + j;
+          float p = 0.f;
+          if (!causal || qrg >= my_kvrow)
+            p = __expf(st[r] * scale - lv[j]);
+          st[r] = p;
+          dpt[r] = p * (dpt[r] - dl[j]) * scale;
+        }
       }
       // ds_read_b64_tr_b16 semantics: each lane loads 64b at its OWN
       // address; the HW transposes 16-bit elements within each 16-lane
