@@ -22,7 +22,8 @@ def run_one(model, graph, args, device):
 
     batcher = ContinuousBatcher(model, max_batch=args.slots,
                                 max_len=args.prompt + args.new + 8,
-                                graph=graph)
+                                graph=graph,
+                                prefill_chunk=args.prefill_chunk)
     random.seed(0)
     reqs = [batcher.submit(
         [random.randrange(2, 1000) for _ in
@@ -47,6 +48,8 @@ def main():
     p.add_argument("--requests", type=int, default=24)
     p.add_argument("--prompt", type=int, default=256)
     p.add_argument("--new", type=int, default=64)
+    p.add_argument("--prefill-chunk", type=int, default=None,
+                   help="bound prompt tokens prefetched per step")
     args = p.parse_args()
 
     import torch
@@ -68,7 +71,8 @@ def main():
             "graph_replay": used_graph,
             "config": {"model": args.model, "slots": args.slots,
                        "requests": args.requests,
-                       "prompt_max": args.prompt, "new": args.new},
+                       "prompt_max": args.prompt, "new": args.new,
+                       "prefill_chunk": args.prefill_chunk},
         }), flush=True)
 
 

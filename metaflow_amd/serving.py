@@ -45,13 +45,23 @@ class _SlotView(object):
 
 
 class ContinuousBatcher(object):
-    def __init__(self, model, max_batch=8, max_len=2048, graph="auto"):
+    def __init__(self, model, max_batch=8, max_len=2048, graph="auto",
+                 prefill_chunk=None):
         """graph: capture the whole decode step in a hipGraph and replay
         it per token (decode is LAUNCH-bound — ~200 small kernels per
         step; replay collapses them to one submit). "auto" captures on
         GPU and falls back silently; the step stays host-read-free
         because the varlen kernel chunks from device lengths
-        (decode.hip) and splits come from the cache capacity."""
+        (decode.hip) and splits come from the cache capacity.
+
+        prefill_chunk: bound per-step latency under long prompts — a
+        newly admitted request prefills at most this many prompt tokens
+        per step() instead of all at once, so active slots keep
+        decoding every step instead of stalling behind a monolithic
+        prefill (a slot mid-prefill sits at cache length
+        positions[slot] and its decode row is masked by the varlen
+        kernel's per-slot lengths). None = whole-prompt prefill at
+        admission (lowest total latency when prompts are short)."""
         import torch
 
         from .models.llama import KVCache
@@ -66,6 +76,9 @@ class ContinuousBatcher(object):
         self.positions = [0] * max_batch    # cached tokens per slot
         self.slots = [None] * max_batch     # Request or None
         self.next_token = [0] * max_batch   # token to feed next step
+        self.prefill_chunk = prefill_chunk
+        self._views = [None] * max_batch    # _SlotView while prefilling
+        self._prefill_done = [0] * max_batch  # prompt tokens cached
         self.queue = deque()
         self._torch = torch
         self._graph = None
@@ -129,13 +142,45 @@ class ContinuousBatcher(object):
             req = self.queue.popleft()
             assert len(req.prompt) + req.max_new <= self.max_len, \
                 "request longer than the slot capacity"
+            self.slots[slot] = req
+            if self.prefill_chunk is not None:
+                # chunked: cache nothing yet; step() feeds chunks
+                self._views[slot] = _SlotView(self.cache, slot)
+                self._prefill_done[slot] = 0
+                self.positions[slot] = 0
+                continue
             prompt = torch.tensor([req.prompt], device=self.device)
             view = _SlotView(self.cache, slot)
             with torch.no_grad():
                 logits = self.model(prompt, cache=view)
             nxt = int(logits[0, -1].float().argmax())
             req.generated.append(nxt)
-            self.slots[slot] = req
+            self.positions[slot] = len(req.prompt)
+            self.next_token[slot] = nxt
+            if req.max_new <= 1:
+                self._finish(slot)
+
+    def _prefill_step(self):
+        """Advance every mid-prefill slot by one prompt chunk; a slot
+        whose final chunk just ran emits its first token and joins the
+        decode batch next step."""
+        torch = self._torch
+        for slot in range(self.max_batch):
+            view = self._views[slot]
+            if view is None:
+                continue
+            req = self.slots[slot]
+            a = self._prefill_done[slot]
+            b = min(a + self.prefill_chunk, len(req.prompt))
+            chunk = torch.tensor([req.prompt[a:b]], device=self.device)
+            with torch.no_grad():
+                logits = self.model(chunk, cache=view)
+            self._prefill_done[slot] = b
+            if b < len(req.prompt):
+                continue
+            self._views[slot] = None
+            nxt = int(logits[0, -1].float().argmax())
+            req.generated.append(nxt)
             self.positions[slot] = len(req.prompt)
             self.next_token[slot] = nxt
             if req.max_new <= 1:
@@ -145,21 +190,34 @@ class ContinuousBatcher(object):
         self.slots[slot].done = True
         self.slots[slot] = None
         self.positions[slot] = 0
+        self._views[slot] = None
+        self._prefill_done[slot] = 0
 
     def step(self):
-        """Admit waiting requests, then decode one token for every
-        active slot in a single batched pass. Returns the number of
-        active slots decoded."""
+        """Admit waiting requests, advance mid-prefill slots by one
+        chunk, then decode one token for every active slot in a single
+        batched pass. Returns the number of active slots decoded."""
         torch = self._torch
         self._admit()
-        active = [i for i, r in enumerate(self.slots) if r is not None]
+        if self.prefill_chunk is not None:
+            self._prefill_step()
+        active = [i for i, r in enumerate(self.slots) if r is not None
+                  and self._views[i] is None]
+        # the batched decode scatters a (garbage) K/V row for EVERY
+        # slot at its position; a mid-prefill slot must aim that write
+        # at its next-unwritten row — the following prefill chunk
+        # overwrites it, so the cached prefix stays intact
+        pos_list = list(self.positions)
+        for i in range(self.max_batch):
+            if self._views[i] is not None:
+                pos_list[i] = self._prefill_done[i]
         if not active:
             return 0
         tokens = torch.zeros(self.max_batch, 1, dtype=torch.long,
                              device=self.device)
         for i in active:
             tokens[i, 0] = self.next_token[i]
-        logits = self._decode(tokens, list(self.positions))
+        logits = self._decode(tokens, pos_list)
         for i in active:
             req = self.slots[i]
             nxt = int(logits[i, -1].float().argmax())

@@ -115,3 +115,49 @@ def test_continuous_batching_mixtral():
         ref = m.generate(torch.tensor([prompt]),
                          n)[0, len(prompt):].tolist()
         assert out[req.id] == ref, (req.id, out[req.id], ref)
+
+
+def test_chunked_prefill_token_exact(tiny_model):
+    """prefill_chunk: chunked admission produces exactly the tokens the
+    whole-prompt batcher produces (chunk boundary mid-prompt, chunk
+    bigger than a short prompt, several requests racing slots)."""
+    prompts = [
+        (list(range(2, 25)), 5),      # 23 tokens -> 3 chunks of 8
+        ([5, 9, 17], 6),              # shorter than one chunk
+        ([7] * 16, 4),                # exact chunk multiple
+        ([100, 101, 102, 103, 104], 7),
+    ]
+    ref_b = ContinuousBatcher(tiny_model, max_batch=2, max_len=128)
+    ref_reqs = [ref_b.submit(p, n) for p, n in prompts]
+    ref_out = ref_b.run()
+    chk_b = ContinuousBatcher(tiny_model, max_batch=2, max_len=128,
+                              prefill_chunk=8)
+    chk_reqs = [chk_b.submit(p, n) for p, n in prompts]
+    chk_out = chk_b.run()
+    for rr, cr in zip(ref_reqs, chk_reqs):
+        assert chk_out[cr.id] == ref_out[rr.id], (chk_out[cr.id],
+                                                  ref_out[rr.id])
+        assert cr.done
+
+
+def test_chunked_prefill_bounds_step_latency(tiny_model):
+    """A long prompt never runs >prefill_chunk prompt tokens in one
+    step(): active decoders are not stalled behind a monolithic
+    prefill."""
+    seen = []
+    orig_forward = tiny_model.forward
+
+    def spy(tokens, *a, **kw):
+        seen.append(tokens.shape[1])
+        return orig_forward(tokens, *a, **kw)
+
+    tiny_model.forward = spy
+    try:
+        b = ContinuousBatcher(tiny_model, max_batch=2, max_len=256,
+                              prefill_chunk=16)
+        b.submit(list(range(2, 2 + 100)), 3)   # 100-token prompt
+        b.submit([5, 6, 7], 3)
+        b.run()
+    finally:
+        tiny_model.forward = orig_forward
+    assert max(seen) <= 16, seen
