@@ -45,7 +45,7 @@ from .client import (
 )
 from .multicore_utils import parallel_map, parallel_imap_unordered
 from .datatools import ObjectStore
-from .runner import Runner
+from .runner import DeployedFlow, Deployer, Runner, TriggeredRun
 
 __version__ = "0.1.0"
 
@@ -88,6 +88,9 @@ __all__ = [
     "get_namespace",
     "ObjectStore",
     "Runner",
+    "Deployer",
+    "DeployedFlow",
+    "TriggeredRun",
     "parallel_map",
     "parallel_imap_unordered",
 ]

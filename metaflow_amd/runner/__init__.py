@@ -1,3 +1,5 @@
-from .metaflow_runner import Runner, ExecutingRun
+from .deployer import DeployedFlow, Deployer, TriggeredRun
+from .metaflow_runner import ExecutingRun, Runner
 
-__all__ = ["Runner", "ExecutingRun"]
+__all__ = ["Runner", "ExecutingRun", "Deployer", "DeployedFlow",
+           "TriggeredRun"]
