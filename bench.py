@@ -54,6 +54,9 @@ def parse_args():
                         "(delayed scaling, bf16 backward) — reported "
                         "honestly as dtype fp8-e4m3-fwd, NOT the bf16 "
                         "headline")
+    p.add_argument("--fp8-bwd", action="store_true",
+                   help="with --fp8: dgrad/wgrad GEMMs in E4M3 too "
+                        "(delayed-scaled dy) — dtype fp8-e4m3-fwd-bwd")
     p.add_argument("--tunableop", action="store_true",
                    help="enable PyTorch TunableOp (hipBLASLt algorithm "
                         "autotuning) during warmup")
@@ -112,10 +115,13 @@ def main():
     }
     cfg_fn, model_cls = model_factories[args.model]
     cfg = cfg_fn()
+    if args.fp8_bwd:
+        args.fp8 = True
     if args.fp8:
         assert args.model in ("llama3-8b", "llama3-70b", "tiny"), \
             "--fp8 supports the Llama family"
         cfg.fp8 = True
+        cfg.fp8_bwd = args.fp8_bwd
     if args.recompute:
         cfg.recompute = True
     seq = min(args.seq, cfg.max_seq_len)
@@ -234,7 +240,8 @@ def main():
             "higher_is_better": True,
             "scaling": "strong" if shared else "weak",
             "vs_baseline": None,
-            "dtype": "fp8-e4m3-fwd" if args.fp8 else "bf16",
+            "dtype": ("fp8-e4m3-fwd-bwd" if args.fp8_bwd else
+                      "fp8-e4m3-fwd") if args.fp8 else "bf16",
             "data": "synthetic",
             "config": {
                 "model": args.model,

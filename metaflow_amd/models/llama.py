@@ -29,8 +29,11 @@ class LlamaConfig:
     rms_eps: float = 1e-5
     tie_embeddings: bool = False
     # fp8=True runs the decoder projections' FORWARD GEMMs in OCP E4M3
-    # (delayed scaling, bf16 backward — ops/fp8.py); lm_head stays bf16
+    # (delayed scaling, bf16 backward — ops/fp8.py); lm_head stays bf16.
+    # fp8_bwd=True additionally runs dgrad/wgrad in E4M3 (delayed-scaled
+    # dy) — the full-fp8 rung; watch loss quality
     fp8: bool = False
+    fp8_bwd: bool = False
     # recompute=True checkpoints each decoder layer's activations
     # (recomputed in backward): activation memory drops from O(layers)
     # to O(1) layers — the knob that fits 70B/long-seq training
@@ -91,7 +94,11 @@ class DecoderLayer(nn.Module):
         self.cfg = cfg
         h, hd = cfg.hidden_size, cfg.head_dim
         if cfg.fp8:
-            from ..ops.fp8 import Fp8Linear as Lin
+            import functools
+
+            from ..ops.fp8 import Fp8Linear
+
+            Lin = functools.partial(Fp8Linear, fp8_bwd=cfg.fp8_bwd)
         else:
             Lin = Linear
         self.qkv_proj = Lin(

@@ -42,24 +42,54 @@ def quantize_e4m3(t, scale):
 
 class _Fp8LinearFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, scale_x, scale_w):
-        ctx.save_for_backward(x, w)
+    def forward(ctx, x, w, scale_x, scale_w, mod):
         K = x.shape[-1]
         M = x.numel() // K
         x8 = quantize_e4m3(x.reshape(M, K), scale_x)
         w8 = quantize_e4m3(w, scale_w)
         out = _gemm_ext().fp8(x8, w8, 1.0 / (scale_x * scale_w))
+        ctx.mod = mod
+        if mod is not None and mod.fp8_bwd:
+            # fp8 backward: keep the QUANTIZED operands (half the
+            # activation memory of saving bf16 x) — dgrad/wgrad rerun
+            # the same TN-form fp8 GEMM on byte-transposed views
+            ctx.save_for_backward(x8, w8)
+            ctx.meta = (scale_x, scale_w, x.shape, K, M)
+        else:
+            ctx.save_for_backward(x, w)
         return out.reshape(*x.shape[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        x, w = ctx.saved_tensors
-        K = x.shape[-1]
-        M = x.numel() // K
-        dy2 = dy.reshape(M, -1)
-        dx = (dy2 @ w).reshape(x.shape)
-        dw = dy2.t() @ x.reshape(M, K)
-        return dx, dw, None, None
+        mod = ctx.mod
+        if mod is None or not mod.fp8_bwd:
+            x, w = ctx.saved_tensors
+            K = x.shape[-1]
+            M = x.numel() // K
+            dy2 = dy.reshape(M, -1)
+            dx = (dy2 @ w).reshape(x.shape)
+            dw = dy2.t() @ x.reshape(M, K)
+            return dx, dw, None, None, None
+        x8, w8 = ctx.saved_tensors
+        sx, sw, xshape, K, M = ctx.meta
+        dy2 = dy.reshape(M, -1).contiguous()
+        # dy scale is DELAYED from the previous backwards' amax history
+        # (first fp8 backward seeds it from this dy: one host sync once)
+        ady = float(mod.amax_dy.max())
+        if ady <= 0:
+            ady = float(dy2.detach().abs().max())
+        sdy = E4M3_MAX / ady if ady > 0 else 1.0
+        dy8 = quantize_e4m3(dy2, sdy)
+        # dgrad: dx[M,K] = dy[M,N] @ W[N,K]   = fp8(dy8, w8^T)
+        dx = _gemm_ext().fp8(dy8, w8.t().contiguous(), 1.0 / (sdy * sw))
+        # wgrad: dw[N,K] = dy^T[N,M] @ x[M,K] = fp8(dy8^T, x8^T)
+        dw = _gemm_ext().fp8(dy8.t().contiguous(), x8.t().contiguous(),
+                             1.0 / (sdy * sx))
+        with torch.no_grad():
+            mod.amax_dy[mod._bstep % mod.history] = \
+                dy2.detach().abs().max().float()
+        mod._bstep += 1
+        return dx.reshape(xshape), dw, None, None, None
 
 
 def _in_recompute():
@@ -85,16 +115,24 @@ def _in_recompute():
 class Fp8Linear(torch.nn.Module):
     """Bias-free linear with an E4M3 forward GEMM and delayed scaling."""
 
-    def __init__(self, din, dout, dtype=torch.bfloat16, history=16):
+    def __init__(self, din, dout, dtype=torch.bfloat16, history=16,
+                 fp8_bwd=False):
         super().__init__()
         self.weight = torch.nn.Parameter(
             torch.empty(dout, din, dtype=dtype))
         self.history = history
+        #: fp8_bwd=True additionally runs dgrad AND wgrad in E4M3
+        #: (delayed-scaled dy) — the full-fp8 rung; measure loss
+        #: quality before production use
+        self.fp8_bwd = fp8_bwd
         self.register_buffer(
             "amax_x", torch.zeros(history), persistent=False)
         self.register_buffer(
             "amax_w", torch.zeros(history), persistent=False)
+        self.register_buffer(
+            "amax_dy", torch.zeros(history), persistent=False)
         self._step = 0
+        self._bstep = 0
 
     def _scales(self):
         ax = float(self.amax_x.max())
@@ -124,4 +162,4 @@ class Fp8Linear(torch.nn.Module):
         if self._step == 1:
             # no history yet: first step runs bf16 (standard warmup)
             return torch.nn.functional.linear(x, self.weight)
-        return _Fp8LinearFn.apply(x, self.weight, sx, sw)
+        return _Fp8LinearFn.apply(x, self.weight, sx, sw, self)

@@ -459,6 +459,38 @@ def test_fp8_linear_gpu(dev):
     assert xg.grad is not None and lin.weight.grad is not None
 
 
+def test_fp8_bwd_linear_gpu(dev):
+    """fp8_bwd=True: dgrad AND wgrad through the E4M3 GEMM (delayed-
+    scaled dy, byte-transposed operands) match the bf16 autograd
+    gradients to fp8 tolerance."""
+    from metaflow_amd.ops.fp8 import Fp8Linear
+
+    torch.manual_seed(1)
+    M, K, N = 512, 256, 384
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.1
+    dy = torch.randn(M, N, dtype=torch.bfloat16, device=dev) * 1e-3
+
+    xr = x.clone().requires_grad_(True)
+    wr = w.clone().requires_grad_(True)
+    torch.nn.functional.linear(xr, wr).backward(dy)
+
+    lin = Fp8Linear(K, N, fp8_bwd=True).to(dev)
+    with torch.no_grad():
+        lin.weight.copy_(w)
+    xg = x.clone().requires_grad_(True)
+    lin(xg)                         # bf16 warmup step
+    out = lin(xg)                   # fp8 fwd + fp8 bwd
+    out.backward(dy)
+    assert rel_err(xg.grad, xr.grad) < 8e-2
+    assert rel_err(lin.weight.grad, wr.grad) < 8e-2
+    # second backward uses the RECORDED dy amax (delayed scaling)
+    assert float(lin.amax_dy.max()) > 0
+    xg2 = x.clone().requires_grad_(True)
+    lin(xg2).backward(dy)
+    assert rel_err(xg2.grad, xr.grad) < 8e-2
+
+
 def test_attn_decode_varlen_gpu(dev):
     """Varlen flash-decode: per-slot lengths (incl. an inactive slot)
     vs the per-slot fp32 reference."""
