@@ -102,7 +102,11 @@ if __name__ == "__main__":
 '''
 
 
-def bench_foreach(splits, mb):
+def bench_foreach(splits, mb, max_workers=None):
+    if max_workers is None:
+        # config-2 tasks are small (load + checksum): size the worker
+        # pool to the host, not the reference's 16-subprocess default
+        max_workers = min(64, max(16, os.cpu_count() or 16))
     tmpdir = tempfile.mkdtemp(prefix="mfx_fanout_")
     flow_file = os.path.join(tmpdir, "fanout_flow.py")
     with open(flow_file, "w") as f:
@@ -113,7 +117,8 @@ def bench_foreach(splits, mb):
     proc = subprocess.run(
         [sys.executable, flow_file, "--quiet", "--datastore-root",
          os.path.join(tmpdir, "ds"), "run", "--splits", str(splits),
-         "--mb", str(mb), "--max-num-splits", str(splits)],
+         "--mb", str(mb), "--max-num-splits", str(splits),
+         "--max-workers", str(max_workers)],
         env=env, capture_output=True, text=True)
     wall = time.time() - t
     ok = proc.returncode == 0
@@ -123,6 +128,7 @@ def bench_foreach(splits, mb):
         print(proc.stderr[-3000:], file=sys.stderr)
         raise SystemExit("foreach flow failed")
     return {"foreach_wall_s": wall, "tasks": splits + 4,
+            "max_workers": max_workers,
             "tasks_per_sec": (splits + 4) / wall}
 
 
@@ -132,6 +138,7 @@ def main():
     p.add_argument("--splits", type=int, default=1024)
     p.add_argument("--artifact-mb", type=int, default=1024)
     p.add_argument("--skip-foreach", action="store_true")
+    p.add_argument("--max-workers", type=int, default=None)
     args = p.parse_args()
 
     out = {"metric": "artifact save GB/s", "higher_is_better": True,
@@ -140,7 +147,8 @@ def main():
     out["value"] = out["save_gbps"]
     out["unit"] = "GB/s"
     if not args.skip_foreach:
-        out.update(bench_foreach(args.splits, args.artifact_mb))
+        out.update(bench_foreach(args.splits, args.artifact_mb,
+                                args.max_workers))
     print(json.dumps(out), flush=True)
 
 
