@@ -53,6 +53,21 @@ def parallel_key(blob):
     return hashlib.sha256(b"MFXP1" + digests).hexdigest()
 
 
+class _IoEngine(object):
+    """load-only engine backed by the STANDALONE _mfx_io extension (no
+    torch link, ~1 ms import): torch-less task subprocesses get the
+    parallel-pread read path without paying the 1.5 s torch import the
+    full _mfx_cas engine would drag in (ops/cas_native.py)."""
+
+    __slots__ = ("_mod",)
+
+    def __init__(self, mod):
+        self._mod = mod
+
+    def load_blob_parallel(self, path, header_skip):
+        return self._mod.load_file(path, header_skip)
+
+
 def _native_engine():
     if not CAS_NATIVE:
         return None
@@ -60,6 +75,12 @@ def _native_engine():
         from ..ops import cas_native
 
         return cas_native.engine()
+    except Exception:
+        pass
+    try:
+        from ..ops import _mfx_io
+
+        return _IoEngine(_mfx_io)
     except Exception:
         return None
 

@@ -4,6 +4,7 @@
 
 * metaflow_amd/ops/_mfx_hip.so  — gfx950 HIP kernel library (torch ext)
 * metaflow_amd/ops/_mfx_cas.so  — C++ CAS engine (threaded SHA-256 + IO)
+* metaflow_amd/ops/_mfx_io.so   — standalone pread pool (NO torch link)
 
 gfx950-only by design (PYTORCH_ROCM_ARCH=gfx950); the .so files travel with
 the repo snapshot to the GPU box.
@@ -14,7 +15,7 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 os.environ.setdefault("MAX_JOBS", str(os.cpu_count() or 8))
 
-from setuptools import setup  # noqa: E402
+from setuptools import Extension, setup  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension, CppExtension  # noqa: E402
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
@@ -73,6 +74,15 @@ ext_modules = [
         name="metaflow_amd.ops._mfx_cas",
         sources=[os.path.join(CSRC, "cas_engine.cpp")],
         extra_compile_args={"cxx": ["-O3", "-std=c++17", "-pthread"]},
+    ),
+    # plain setuptools Extension ON PURPOSE: _mfx_io must not link
+    # torch so torch-less task subprocesses can import it in ~10 ms
+    # (ops/cas_native.py explains the 10x regression that rule avoids)
+    Extension(
+        name="metaflow_amd.ops._mfx_io",
+        sources=[os.path.join(CSRC, "io_engine.cpp")],
+        extra_compile_args=["-O3", "-std=c++17", "-pthread"],
+        extra_link_args=["-pthread"],
     ),
 ]
 

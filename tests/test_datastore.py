@@ -447,3 +447,38 @@ def test_stream_saver_native_key_parity(tmp_path):
         _os.replace(tmp, ap)
         [(_k, back)] = list(store.load_blobs([key]))
         assert bytes(back) == blob
+
+
+def test_torchless_native_load_engine(tmp_path):
+    """A torch-less subprocess gets the STANDALONE _mfx_io pread engine
+    for raw blob loads (the torch-linked engine is refused there —
+    ops/cas_native.py's 10x-regression guard)."""
+    import subprocess
+    import sys
+    import textwrap
+
+    from .test_runtime import REPO
+
+    code = textwrap.dedent("""
+        import sys
+        sys.path.insert(0, %r)
+        from metaflow_amd.datastore import cas as C
+        from metaflow_amd.datastore.cas import ContentAddressedStore
+        from metaflow_amd.datastore.storage import LocalStorage
+        import os
+        store = ContentAddressedStore("f", LocalStorage(%r))
+        data = os.urandom(6 << 20)
+        [(_, key)] = store.save_blobs([data])
+        [(_, got)] = list(store.load_blobs([key]))
+        assert got == data
+        eng = C._native_engine()
+        assert "torch" not in sys.modules, "torch leaked into the task"
+        print(type(eng).__name__)
+    """) % (REPO, str(tmp_path))
+    out = subprocess.run([sys.executable, "-c", code],
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-2000:]
+    # _IoEngine when the extension is built; None only if it is not
+    assert out.stdout.strip() in ("_IoEngine", "None")
+    assert out.stdout.strip() == "_IoEngine", \
+        "_mfx_io extension missing — build_ext --inplace"
