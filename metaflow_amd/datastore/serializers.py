@@ -172,6 +172,51 @@ class TensorSerializer(ArtifactSerializer):
         return t.view(dtype).reshape(shape)
 
 
+class NumpySerializer(ArtifactSerializer):
+    """Zero-pickle numpy codec: dtype/shape header + raw buffer bytes.
+
+    pickle round-trips a 1 GiB uint8 array in ~2.6 s (dumps) + ~1.0 s
+    (loads) of pure buffer shuffling — the dominant per-task cost of
+    BASELINE config 2 (profiles/bench_results_r02.md). This codec
+    serializes with a single memcpy and deserializes ZERO-copy:
+    ``np.frombuffer`` over the CAS bytes, so the returned array is
+    READ-ONLY (content-addressed artifacts are immutable by design;
+    call ``.copy()`` for a mutable scratch array). Object dtypes fall
+    through to pickle."""
+
+    encoding = "numpy-v1"
+    priority = 60
+    _MAGIC = b"MFXN1"
+
+    def can_serialize(self, obj):
+        np = sys.modules.get("numpy")
+        return (np is not None and isinstance(obj, np.ndarray)
+                and obj.dtype.hasobject is False)
+
+    def serialize(self, obj):
+        import numpy as np
+
+        # (ascontiguousarray promotes 0-d to 1-d; 0-d is contiguous)
+        arr = obj if obj.flags.c_contiguous \
+            else np.ascontiguousarray(obj)
+        dt = arr.dtype.str.encode()          # e.g. b"<u1", b"<f4"
+        header = self._MAGIC + struct.pack("<BB", len(dt), arr.ndim)
+        header += dt
+        header += struct.pack("<%dq" % arr.ndim, *arr.shape)
+        return header + arr.tobytes()
+
+    def deserialize(self, data):
+        import numpy as np
+
+        assert data[:5] == self._MAGIC, "bad numpy blob"
+        dtlen, ndim = struct.unpack_from("<BB", data, 5)
+        dt = data[7:7 + dtlen].decode()
+        shape = struct.unpack_from("<%dq" % ndim, data, 7 + dtlen)
+        offset = 7 + dtlen + 8 * ndim
+        return np.frombuffer(data, dtype=np.dtype(dt),
+                             offset=offset).reshape(shape)
+
+
 class PickleSerializer(ArtifactSerializer):
     """Universal fallback (reference registers it last,
     plugins/__init__.py:192)."""

@@ -33,6 +33,7 @@ INTERNAL_ATTRS = frozenset(
         "_cached_input",
         "_lazy_includes",
         "_lazy_include_cache",
+        "_artifact_provenance",
         "name",
     ]
 )
@@ -106,6 +107,19 @@ class FlowSpec(object, metaclass=FlowSpecMeta):
             value = ds[name]
             # cache so repeated access doesn't re-deserialize
             object.__setattr__(self, name, value)
+            # provenance: if this exact object is still the attribute at
+            # persist time AND is immutable (bytes/str/scalars, read-only
+            # numpy views), persist() reuses the original index entry
+            # instead of re-serializing + re-hashing the payload — the
+            # dominant cost of big read-only artifacts in fan-outs
+            try:
+                info = ds.artifact_info(name)
+                if info:
+                    prov = self.__dict__.setdefault(
+                        "_artifact_provenance", {})
+                    prov[name] = (id(value), dict(info))
+            except Exception:
+                pass
             return value
         raise AttributeError(
             "Flow %s has no artifact or attribute '%s'"

@@ -215,8 +215,24 @@ class NativeRuntime(object):
         self._gang_info = {}  # control_task_id -> gang retry bookkeeping
         self._selector = selectors.DefaultSelector()
         self._params_pathspec = None
+        # MFX_SCHED_TIMING=1: accumulate per-phase scheduler time and
+        # print a breakdown at run end (the scheduler is single-
+        # threaded, so serial per-task cost here caps fan-out
+        # throughput — profiles/bench_results_r02.md config-2)
+        self._timing = {} if os.environ.get("MFX_SCHED_TIMING") else None
 
     # ------------------------------------------------------------- utilities
+    def _timed(self, phase, fn, *args, **kwargs):
+        if self._timing is None:
+            return fn(*args, **kwargs)
+        t0 = time.perf_counter()
+        try:
+            return fn(*args, **kwargs)
+        finally:
+            d = time.perf_counter() - t0
+            tot, n = self._timing.get(phase, (0.0, 0))
+            self._timing[phase] = (tot + d, n + 1)
+
     def _new_task_id(self):
         self._task_seq += 1
         return str(self._task_seq)
@@ -342,6 +358,13 @@ class NativeRuntime(object):
                 if time.time() - last_hb > 10:
                     self.metadata.heartbeat(self.run_id)
                     last_hb = time.time()
+            if self._timing is not None:
+                for phase, (tot, n) in sorted(self._timing.items()):
+                    sys.stdout.write(
+                        "[mfx-sched-timing] %-12s total %7.2fs  n=%-5d "
+                        "avg %6.1f ms\n" % (phase, tot, n,
+                                             1e3 * tot / max(n, 1)))
+                sys.stdout.flush()
         except KeyboardInterrupt:
             self._failed = True
             self._failure_msg = "interrupted"
@@ -550,7 +573,7 @@ class NativeRuntime(object):
             # gang members bypass the cap: a partially-launched gang
             # deadlocks in rendezvous
             self._run_queue.popleft()
-            self._launch(spec)
+            self._timed("launch", self._launch, spec)
 
     def _max_retries_for(self, step):
         func = getattr(self.flow_cls, step)
@@ -679,25 +702,28 @@ class NativeRuntime(object):
                 except KeyError:
                     pass
             worker.flush_echo()
-            self._worker_exited(worker, rc)
+            self._timed("exit", self._worker_exited, worker, rc)
         self._workers = still
 
     def _worker_exited(self, worker, rc):
         spec = worker.spec
         # persist captured logs (mflog-structured) into the task's attempt
-        try:
-            from . import mflog
+        def _persist_logs():
+            try:
+                from . import mflog
 
-            source = "%s/%s" % (spec.step, spec.task_id)
-            log_ds = self.flow_datastore.get_task_datastore(
-                self.run_id, spec.step, spec.task_id,
-                attempt=spec.retry_count, mode="w")
-            log_ds.save_logs("stdout", mflog.decorate_stream(
-                source, worker.stdout_buf.get_bytes()))
-            log_ds.save_logs("stderr", mflog.decorate_stream(
-                source, worker.stderr_buf.get_bytes()))
-        except Exception:
-            pass
+                source = "%s/%s" % (spec.step, spec.task_id)
+                log_ds = self.flow_datastore.get_task_datastore(
+                    self.run_id, spec.step, spec.task_id,
+                    attempt=spec.retry_count, mode="w")
+                log_ds.save_logs("stdout", mflog.decorate_stream(
+                    source, worker.stdout_buf.get_bytes()))
+                log_ds.save_logs("stderr", mflog.decorate_stream(
+                    source, worker.stderr_buf.get_bytes()))
+            except Exception:
+                pass
+
+        self._timed("exit.logs", _persist_logs)
 
         # @card(profile=True): the rocprofv3 stats CSV exists only now
         # (written when the wrapped process exited); splice the kernel

@@ -19,6 +19,38 @@ import sys
 import time
 
 
+# Self-contained heartbeat loop for the LOCAL metadata provider: spawned
+# as `python -S -c ...` so the child never imports metaflow_amd (the full
+# package import costs ~0.3-0.5 s of CPU, which DOUBLED the cost of every
+# small task — the config-2 fan-out is task-CPU bound, see
+# profiles/bench_results_r02.md). Paths must mirror
+# metadata/local.py:_heartbeat_path/_task_heartbeat_path.
+_LOCAL_HB_SCRIPT = r"""
+import json, os, select, sys, time
+ctx = json.loads(os.environ["MFX_SIDECAR_CONTEXT"])
+base = os.path.join(ctx["datastore_root"], ctx["flow_name"], "_meta",
+                    str(ctx["run_id"]))
+paths = [os.path.join(base, "heartbeat.json")]
+if ctx.get("step_name") and ctx.get("task_id"):
+    paths.append(os.path.join(base, "task_heartbeat_%s_%s.json"
+                              % (ctx["step_name"], ctx["task_id"])))
+os.makedirs(base, exist_ok=True)
+while True:
+    payload = json.dumps({"ts": time.time()}).encode()
+    for p in paths:
+        try:
+            tmp = p + ".tmp%d" % os.getpid()
+            with open(tmp, "wb") as f:
+                f.write(payload)
+            os.replace(tmp, p)
+        except OSError:
+            pass
+    r, _w, _x = select.select([sys.stdin], [], [], 5.0)
+    if r and not sys.stdin.buffer.read(1):
+        break  # stdin EOF: parent gone / terminate()
+"""
+
+
 class SidecarSubProcess(object):
     def __init__(self, worker_name, context=None):
         self.worker_name = worker_name
@@ -28,10 +60,15 @@ class SidecarSubProcess(object):
         self._start()
 
     def _start(self):
+        if self.worker_name == "heartbeat" and \
+                self.context.get("provider") != "service":
+            argv = [sys.executable, "-S", "-u", "-c", _LOCAL_HB_SCRIPT]
+        else:
+            argv = [sys.executable, "-m", "metaflow_amd.sidecar",
+                    self.worker_name]
         try:
             self._proc = subprocess.Popen(
-                [sys.executable, "-m", "metaflow_amd.sidecar",
-                 self.worker_name],
+                argv,
                 stdin=subprocess.PIPE,
                 stdout=subprocess.DEVNULL,
                 stderr=subprocess.DEVNULL,

@@ -164,14 +164,18 @@ class MFXTask(object):
     def run_step(self, step_name, run_id, task_id, input_paths, split_index,
                  retry_count, max_user_code_retries, origin_run_id=None,
                  ubf_context=None, namespace=None):
+        from .profile_util import from_start
+
         flow = self.flow
         node = self.graph[step_name]
         # unbound class function: decorators wrap it, then we call f(flow)
         step_func = getattr(type(flow), step_name)
         decorators = getattr(step_func, "decorators", [])
 
+        from_start("task: enter run_step")
         self.metadata.register_task(run_id, step_name, task_id, retry_count,
                                     metadata={"attempt_started": True})
+        from_start("task: attempt registered")
 
         # gang rank CPU pinning (set by the gang scheduler; the NUMA
         # half of HIP_VISIBLE_DEVICES)
@@ -200,7 +204,9 @@ class MFXTask(object):
         except Exception:
             pass
 
+        from_start("task: sidecar up")
         input_dss = self._input_datastores(input_paths)
+        from_start("task: input datastores built")
         output = self.flow_datastore.get_task_datastore(
             run_id, step_name, task_id, attempt=retry_count, mode="w")
         output.init_task()
@@ -368,6 +374,7 @@ class MFXTask(object):
                         % (step_name, bad, node.out_funcs))
 
         # ---- persist ---------------------------------------------------------
+        from_start("task: step function done")
         try:
             flow._task_ok = task_ok
             if error is not None:
@@ -376,6 +383,7 @@ class MFXTask(object):
         except Exception:
             task_ok = False
             traceback.print_exc()
+        from_start("task: artifacts persisted")
 
         if transition is not None:
             output.save_metadata("transition", transition)
@@ -396,8 +404,10 @@ class MFXTask(object):
 
         if hb_sidecar is not None:
             hb_sidecar.terminate()
+        from_start("task: sidecar terminated")
         if task_ok:
             output.done()
+            from_start("task: DONE committed")
         else:
             raise TaskFailed(error)
 

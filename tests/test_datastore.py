@@ -482,3 +482,185 @@ def test_torchless_native_load_engine(tmp_path):
     assert out.stdout.strip() in ("_IoEngine", "None")
     assert out.stdout.strip() == "_IoEngine", \
         "_mfx_io extension missing — build_ext --inplace"
+
+
+def test_numpy_serializer_roundtrip_and_readonly():
+    """numpy-v1 codec: header+raw bytes round-trip for assorted
+    dtypes/shapes; deserialized arrays are zero-copy READ-ONLY views
+    over the CAS bytes; object dtype falls through to pickle."""
+    import numpy as np
+
+    from metaflow_amd.datastore import serializers
+
+    cases = [np.arange(12, dtype=np.float32).reshape(3, 4),
+             np.array(3.5),
+             np.zeros((0, 5), dtype=np.int64),
+             np.asfortranarray(np.arange(6).reshape(2, 3)),
+             np.arange(6).reshape(2, 3)[:, ::2],      # non-contiguous
+             np.array([True, False])]
+    for a in cases:
+        payload, enc = serializers.serialize(a)
+        assert enc == "numpy-v1"
+        b = serializers.deserialize(payload, enc)
+        assert np.array_equal(b, a) and b.dtype == a.dtype
+        assert not b.flags.writeable     # content-addressed = immutable
+    payload, enc = serializers.serialize(np.array(["a", "b"],
+                                                  dtype=object))
+    assert enc == "pickle-v4"            # object dtype: pickle fallback
+
+
+def test_numpy_artifact_through_flow(tmp_path, tmp_datastore):
+    """A numpy artifact crosses steps via the numpy-v1 codec (the
+    datastore records the encoding) and reads back equal."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import textwrap
+
+    from .test_runtime import REPO, latest_run_id
+
+    flow = tmp_path / "np_flow.py"
+    flow.write_text(textwrap.dedent("""
+        import numpy as np
+
+        from metaflow_amd import FlowSpec, step
+
+        class NpFlow(FlowSpec):
+            @step
+            def start(self):
+                self.arr = np.arange(1024, dtype=np.float32)
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.arr.sum() == 1023 * 1024 / 2
+                assert not self.arr.flags.writeable
+                self.ok = True
+
+        if __name__ == "__main__":
+            NpFlow()
+    """))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    run_id = latest_run_id(tmp_datastore, "NpFlow")
+    data = json.load(open(os.path.join(
+        tmp_datastore, "NpFlow", run_id, "start", "1", "0.data")))
+    assert data["artifacts"]["arr"]["encoding"] == "numpy-v1"
+
+
+def test_persist_provenance_skip(tmp_datastore):
+    """persist() reuses the loaded artifact's index entry (no
+    re-serialize/re-hash) ONLY when the attribute is the identical
+    object AND immutable (read-only numpy / bytes); replaced or
+    mutable values are re-serialized."""
+    import numpy as np
+
+    from metaflow_amd.datastore.flow_datastore import FlowDataStore
+    from metaflow_amd.datastore.storage import LocalStorage
+    from metaflow_amd.datastore.task_datastore import TaskDataStore
+
+    fds = FlowDataStore("ProvFlow", LocalStorage(tmp_datastore))
+    up = TaskDataStore(fds, "1", "start", "1", attempt=0, mode="w")
+    up.init_task()
+    arr = np.arange(1 << 16, dtype=np.uint8)
+    up.save_artifacts([("big", arr), ("note", "hello")])
+    up.done()
+
+    down = TaskDataStore(fds, "1", "work", "2", attempt=0, mode="w")
+    rd = TaskDataStore(fds, "1", "start", "1", mode="r")
+    loaded = dict(rd.load_artifacts(["big", "note"]))
+    assert not loaded["big"].flags.writeable
+
+    class FakeFlow(object):
+        pass
+
+    flow = FakeFlow()
+    flow.big = loaded["big"]
+    flow.note = loaded["note"]
+    flow.fresh = [1, 2, 3]
+    flow.__dict__["_artifact_provenance"] = {
+        "big": (id(loaded["big"]), rd.artifact_info("big")),
+        "note": (id(loaded["note"]), rd.artifact_info("note")),
+    }
+    flow._artifacts_to_persist = lambda: [
+        ("big", flow.big), ("note", flow.note), ("fresh", flow.fresh)]
+
+    calls = []
+    orig = down.save_artifacts
+    down.save_artifacts = lambda pairs: (
+        calls.extend(n for n, _ in pairs), orig(pairs))
+    down.persist(flow)
+    # big + note skipped via provenance; only the new artifact saved
+    assert calls == ["fresh"]
+    assert down.artifact_info("big")["sha"] == \
+        rd.artifact_info("big")["sha"]
+
+    # a REPLACED object (same content, different identity) re-saves
+    flow.big = loaded["big"].copy()          # writable copy
+    calls.clear()
+    down.persist(flow)
+    assert "big" in calls
+
+
+def test_provenance_skip_through_real_flow(tmp_path, tmp_datastore):
+    """End-to-end: a read-only numpy artifact crosses a pass-through
+    step keeping the SAME sha (no re-serialization drift), and a step
+    that replaces it produces a new sha."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import textwrap
+
+    from .test_runtime import REPO, latest_run_id
+
+    flow = tmp_path / "prov_flow.py"
+    flow.write_text(textwrap.dedent("""
+        import numpy as np
+
+        from metaflow_amd import FlowSpec, step
+
+        class PvFlow(FlowSpec):
+            @step
+            def start(self):
+                self.big = np.arange(1 << 16, dtype=np.uint8)
+                self.next(self.mid)
+
+            @step
+            def mid(self):
+                self.peek = int(self.big[:16].sum())   # loads big
+                self.next(self.end)
+
+            @step
+            def end(self):
+                self.big = self.big.copy() * 0 + 1     # replace
+                self.total = int(self.big[:4].sum())
+
+        if __name__ == "__main__":
+            PvFlow()
+    """))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    run_id = latest_run_id(tmp_datastore, "PvFlow")
+
+    def sha(step):
+        data = json.load(open(os.path.join(
+            tmp_datastore, "PvFlow", run_id, step, {"start": "1",
+            "mid": "2", "end": "3"}[step], "0.data")))
+        return data["artifacts"]["big"]["sha"]
+
+    assert sha("mid") == sha("start")       # provenance-skip kept sha
+    assert sha("end") != sha("start")       # replacement re-saved
