@@ -18,6 +18,15 @@ from ..exceptions import DataArtifactMissingError, DataException
 from . import serializers
 
 
+# id(obj) -> (obj, index_info) for every artifact THIS process
+# deserialized. The strong ref keeps the id stable; retention cost is
+# nil because TaskDataStore._cache holds loaded artifacts anyway (one
+# task per process). persist() uses it to skip re-serializing loaded
+# immutable artifacts wherever they came from — self.<attr> lazy loads
+# AND join inputs (self.x = inputs[0].x).
+_PROVENANCE = {}
+
+
 def _immutable_artifact(obj):
     """True only for objects that CANNOT have been mutated in place
     since load — the soundness condition for provenance-based persist
@@ -179,11 +188,10 @@ class TaskDataStore(object):
         is reused — no re-serialize, no re-hash. For a 1 GiB read-only
         input that is ~2 s of per-task CPU saved (config 2,
         profiles/bench_results_r02.md)."""
-        prov = flow.__dict__.get("_artifact_provenance") or {}
         pairs = []
         for name, obj in flow._artifacts_to_persist():
-            p = prov.get(name)
-            if p is not None and p[0] == id(obj) and p[1] and \
+            p = _PROVENANCE.get(id(obj))
+            if p is not None and p[0] is obj and \
                     _immutable_artifact(obj):
                 self._objects[name] = dict(p[1])
             else:
@@ -210,6 +218,8 @@ class TaskDataStore(object):
                 obj = serializers.deserialize(
                     data, self._objects[name]["encoding"])
                 self._cache[name] = obj
+                if _immutable_artifact(obj):
+                    _PROVENANCE[id(obj)] = (obj, self._objects[name])
                 yield name, obj
 
     def __contains__(self, name):

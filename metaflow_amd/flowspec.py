@@ -107,19 +107,8 @@ class FlowSpec(object, metaclass=FlowSpecMeta):
             value = ds[name]
             # cache so repeated access doesn't re-deserialize
             object.__setattr__(self, name, value)
-            # provenance: if this exact object is still the attribute at
-            # persist time AND is immutable (bytes/str/scalars, read-only
-            # numpy views), persist() reuses the original index entry
-            # instead of re-serializing + re-hashing the payload — the
-            # dominant cost of big read-only artifacts in fan-outs
-            try:
-                info = ds.artifact_info(name)
-                if info:
-                    prov = self.__dict__.setdefault(
-                        "_artifact_provenance", {})
-                    prov[name] = (id(value), dict(info))
-            except Exception:
-                pass
+            # (provenance for persist()'s no-reserialize fast path is
+            # recorded at deserialization time — task_datastore.py)
             return value
         raise AttributeError(
             "Flow %s has no artifact or attribute '%s'"

@@ -585,10 +585,7 @@ def test_persist_provenance_skip(tmp_datastore):
     flow.big = loaded["big"]
     flow.note = loaded["note"]
     flow.fresh = [1, 2, 3]
-    flow.__dict__["_artifact_provenance"] = {
-        "big": (id(loaded["big"]), rd.artifact_info("big")),
-        "note": (id(loaded["note"]), rd.artifact_info("note")),
-    }
+    # provenance is registered automatically at deserialization time
     flow._artifacts_to_persist = lambda: [
         ("big", flow.big), ("note", flow.note), ("fresh", flow.fresh)]
 
@@ -664,3 +661,63 @@ def test_provenance_skip_through_real_flow(tmp_path, tmp_datastore):
 
     assert sha("mid") == sha("start")       # provenance-skip kept sha
     assert sha("end") != sha("start")       # replacement re-saved
+
+
+def test_provenance_skip_join_inputs(tmp_path, tmp_datastore):
+    """The provenance fast path also covers JOIN inputs: a read-only
+    array carried through `self.x = inputs[0].x` keeps its sha."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import textwrap
+
+    from .test_runtime import REPO, latest_run_id
+
+    flow = tmp_path / "provj_flow.py"
+    flow.write_text(textwrap.dedent("""
+        import numpy as np
+
+        from metaflow_amd import FlowSpec, step
+
+        class PvJFlow(FlowSpec):
+            @step
+            def start(self):
+                self.big = np.arange(1 << 16, dtype=np.uint8)
+                self.items = [0, 1]
+                self.next(self.work, foreach="items")
+
+            @step
+            def work(self):
+                self.part = int(self.input)
+                self.next(self.join)
+
+            @step
+            def join(self, inputs):
+                self.big = inputs[0].big       # join-input carry
+                self.total = sum(i.part for i in inputs)
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert int(self.big[:4].sum()) == 6
+
+        if __name__ == "__main__":
+            PvJFlow()
+    """))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    run_id = latest_run_id(tmp_datastore, "PvJFlow")
+
+    def sha(step, tid):
+        data = json.load(open(os.path.join(
+            tmp_datastore, "PvJFlow", run_id, step, tid, "0.data")))
+        return data["artifacts"]["big"]["sha"]
+
+    assert sha("join", "4") == sha("start", "1")
