@@ -129,6 +129,51 @@ def card_server(port):
     server.serve_forever()
 
 
+@cli.command(help="Register a flow file as a named local deployment "
+                  "(content-addressed code snapshot).")
+@click.argument("flow_file")
+@click.option("--name", default="prod")
+def deploy(flow_file, name):
+    from .runner import Deployer
+
+    df = Deployer(flow_file).local().create(name=name)
+    click.echo("deployed %s/%s  code=%s" % (df.flow_name, df.name,
+                                            df.code_package_key[:16]))
+
+
+@cli.command(help="List deployments of a flow.")
+@click.argument("flow_name")
+def deployments(flow_name):
+    from .runner import DeployedFlow
+
+    rows = DeployedFlow.list_deployed(flow_name)
+    if not rows:
+        click.echo("no deployments for %s" % flow_name)
+        return
+    for df in rows:
+        click.echo("%-16s created %s  code=%s"
+                   % (df.name, df.created_at, df.code_package_key[:16]))
+
+
+@cli.command(help="Run a deployed flow's code snapshot; blocks until "
+                  "done and prints the run id.")
+@click.argument("flow_name")
+@click.option("--name", default="prod")
+@click.option("--param", "params", multiple=True,
+              help="NAME=VALUE flow parameter (repeatable)")
+def trigger(flow_name, name, params):
+    from .runner import DeployedFlow
+
+    df = DeployedFlow.get(flow_name, name)
+    kwargs = {}
+    for p in params:
+        k, _sep, v = p.partition("=")
+        kwargs[k] = v
+    tr = df.trigger(**kwargs)
+    tr.wait()
+    click.echo("run %s: %s" % (tr.run_id, tr.status))
+
+
 @cli.command(help="Show visible GPUs (rocm-smi summary).")
 def gpus():
     try:

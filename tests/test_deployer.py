@@ -91,3 +91,27 @@ def test_get_missing_deployment_raises(tmp_datastore, monkeypatch):
     with pytest.raises(MFXException):
         DeployedFlow.get("NoSuchFlow", "prod",
                          datastore_root=tmp_datastore)
+
+
+def test_mfx_deploy_cli(flow_dir, tmp_datastore, monkeypatch):
+    """mfx deploy / deployments / trigger round-trip."""
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    env["MFX_DATASTORE_ROOT"] = tmp_datastore
+
+    def mfx(*args):
+        return subprocess.run(
+            [sys.executable, "-m", "metaflow_amd.cmd"] + list(args),
+            env=env, capture_output=True, text=True, timeout=300)
+
+    r = mfx("deploy", str(flow_dir / "dep_flow.py"), "--name", "prod")
+    assert r.returncode == 0 and "deployed DepFlow/prod" in r.stdout, \
+        r.stderr[-2000:]
+    r = mfx("deployments", "DepFlow")
+    assert r.returncode == 0 and "prod" in r.stdout
+    r = mfx("trigger", "DepFlow", "--param", "alpha=3")
+    assert r.returncode == 0 and "successful" in r.stdout, r.stderr[-2000:]
