@@ -46,6 +46,7 @@ from .client import (
 from .multicore_utils import parallel_map, parallel_imap_unordered
 from .datatools import ObjectStore
 from .runner import DeployedFlow, Deployer, Runner, TriggeredRun
+from .speculative import speculative_generate
 
 __version__ = "0.1.0"
 
@@ -91,6 +92,7 @@ __all__ = [
     "Deployer",
     "DeployedFlow",
     "TriggeredRun",
+    "speculative_generate",
     "parallel_map",
     "parallel_imap_unordered",
 ]
