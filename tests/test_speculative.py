@@ -42,6 +42,20 @@ def test_speculative_selfdraft_accepts_everything(models):
     assert stats["target_steps"] <= 6
 
 
+def test_speculative_smaller_draft(models):
+    """Draft with a different architecture (fewer layers/heads) — the
+    realistic deployment shape — still token-exact."""
+    target, _ = models
+    torch.manual_seed(7)
+    cfg = LlamaConfig.tiny(vocab=128, seq=256)
+    cfg.num_layers = max(1, cfg.num_layers // 2)
+    draft = LlamaForCausalLM(cfg).eval()
+    prompt = torch.tensor([[9, 1, 2, 3, 4]])
+    ref = target.generate(prompt, 16)
+    out, _ = speculative_generate(target, draft, prompt, 16, k=3)
+    assert torch.equal(out, ref)
+
+
 def test_speculative_short_budget(models):
     target, draft = models
     prompt = torch.tensor([[42, 43]])
