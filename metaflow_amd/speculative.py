@@ -15,15 +15,11 @@ Cache bookkeeping leans on KVCache.pos being the single read boundary
 but are never read once ``pos`` is rolled back, so rollback is free.
 """
 
-import torch
-
-
 def _greedy_row(logits):
     # logits [1, S, V] -> [S] int64 greedy tokens
     return logits[0].float().argmax(dim=-1)
 
 
-@torch.no_grad()
 def speculative_generate(target, draft, tokens, max_new_tokens,
                          k=4):
     """Greedy decode ``max_new_tokens`` continuation tokens for ONE
@@ -34,10 +30,23 @@ def speculative_generate(target, draft, tokens, max_new_tokens,
     acceptance rate. Output is identical to ``target.generate(tokens,
     max_new_tokens)`` (greedy) — asserted by tests, guaranteed by the
     acceptance rule."""
+    # torch imported lazily: this module is exported from the package
+    # root, and a module-level torch import would load torch into every
+    # small task subprocess (the bisected 10x startup regression)
+    import torch
+
     from .models.llama import KVCache
 
     assert tokens.dim() == 2 and tokens.size(0) == 1, \
         "speculative_generate is per-sequence (B=1)"
+    with torch.no_grad():
+        return _speculative_generate(torch, target, draft, tokens,
+                                     max_new_tokens, k)
+
+
+def _speculative_generate(torch, target, draft, tokens,
+                          max_new_tokens, k):
+    from .models.llama import KVCache
     device = tokens.device
     S0 = tokens.size(1)
     cap = S0 + max_new_tokens + k + 2
