@@ -132,3 +132,23 @@ def test_include_file_cas_upload(tmp_datastore, tmp_path):
     art = data["artifacts"]["blob"]
     assert art["size"] < 4096, art
     assert "IncludedFile" in art["type"]
+
+
+def test_package_import_is_torch_free():
+    """Importing metaflow_amd must NOT import torch: every task
+    subprocess pays this import, and torch costs ~1.5 s (the bisected
+    10x task-startup regression). Anything exported from the package
+    root must defer its torch import to call time."""
+    import subprocess
+    import sys
+
+    from .test_runtime import REPO
+
+    code = ("import sys; sys.path.insert(0, %r); import metaflow_amd; "
+            "import metaflow_amd.task, metaflow_amd.runtime, "
+            "metaflow_amd.cli; "
+            "assert 'torch' not in sys.modules, 'torch leaked'; "
+            "print('clean')" % REPO)
+    r = subprocess.run([sys.executable, "-c", code],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0 and "clean" in r.stdout, r.stderr[-2000:]
