@@ -18,13 +18,42 @@ from ..exceptions import DataArtifactMissingError, DataException
 from . import serializers
 
 
-# id(obj) -> (obj, index_info) for every artifact THIS process
-# deserialized. The strong ref keeps the id stable; retention cost is
-# nil because TaskDataStore._cache holds loaded artifacts anyway (one
-# task per process). persist() uses it to skip re-serializing loaded
+# id(obj) -> (ref, index_info) for artifacts THIS process
+# deserialized; persist() uses it to skip re-serializing loaded
 # immutable artifacts wherever they came from — self.<attr> lazy loads
-# AND join inputs (self.x = inputs[0].x).
+# AND join inputs (self.x = inputs[0].x). A task subprocess lives for
+# one task, but the Client API runs in long-lived notebooks, so the
+# registry must not leak: numpy arrays (the big case) are held by
+# WEAKREF (the finalize callback removes the entry before the id can
+# be reused), and non-weakrefable immutables (bytes/str/scalars) go
+# into a small FIFO — evicting one only forfeits the optimization.
 _PROVENANCE = {}
+_PROVENANCE_FIFO = []
+_PROVENANCE_FIFO_MAX = 64
+
+
+def _provenance_register(obj, info):
+    import weakref
+
+    oid = id(obj)
+    try:
+        ref = weakref.ref(
+            obj, lambda _r, oid=oid: _PROVENANCE.pop(oid, None))
+        _PROVENANCE[oid] = (ref, info)
+        return
+    except TypeError:
+        pass
+    _PROVENANCE[oid] = (lambda obj=obj: obj, info)
+    _PROVENANCE_FIFO.append(oid)
+    if len(_PROVENANCE_FIFO) > _PROVENANCE_FIFO_MAX:
+        _PROVENANCE.pop(_PROVENANCE_FIFO.pop(0), None)
+
+
+def _provenance_lookup(obj):
+    entry = _PROVENANCE.get(id(obj))
+    if entry is not None and entry[0]() is obj:
+        return entry[1]
+    return None
 
 
 def _immutable_artifact(obj):
@@ -190,10 +219,10 @@ class TaskDataStore(object):
         profiles/bench_results_r02.md)."""
         pairs = []
         for name, obj in flow._artifacts_to_persist():
-            p = _PROVENANCE.get(id(obj))
-            if p is not None and p[0] is obj and \
-                    _immutable_artifact(obj):
-                self._objects[name] = dict(p[1])
+            info = _provenance_lookup(obj) if \
+                _immutable_artifact(obj) else None
+            if info is not None:
+                self._objects[name] = dict(info)
             else:
                 pairs.append((name, obj))
         self.save_artifacts(pairs)
@@ -219,7 +248,7 @@ class TaskDataStore(object):
                     data, self._objects[name]["encoding"])
                 self._cache[name] = obj
                 if _immutable_artifact(obj):
-                    _PROVENANCE[id(obj)] = (obj, self._objects[name])
+                    _provenance_register(obj, self._objects[name])
                 yield name, obj
 
     def __contains__(self, name):
