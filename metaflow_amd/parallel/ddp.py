@@ -51,19 +51,25 @@ class FlatParamModel(object):
         params = sync + nosync
         self.n_sync_params = len(sync)
         self.params = params
-        total = sum(self._padded(p.numel()) for p in params)
         self.zero_world = 1
         if self.zero and dist.is_available() and dist.is_initialized():
             self.zero_world = dist.get_world_size(self.group)
-        if self.zero_world > 1 and nosync:
-            raise NotImplementedError(
-                "ZeRO-1 + expert-parallel (_mfx_no_sync) params: the "
-                "whole-buffer reduce-scatter would average expert grads")
+        # ZeRO-1 + expert parallelism: only the SYNC region is sharded
+        # (reduce-scattered / all-gathered); _mfx_no_sync (expert)
+        # params differ per rank — their grads are already complete
+        # after the token all-to-all and their optimizer state stays
+        # owner-local and full. The buffer is laid out
+        # [sync | align-gap | nosync] so the sync region splits into
+        # world equal shards (each a multiple of 4 for fused adam).
+        sync_sum = sum(self._padded(p.numel()) for p in sync)
         if self.zero_world > 1:
-            # tail-pad so the buffer splits into world equal shards,
-            # each a multiple of 4 (fused-adam vector width)
             align = 4 * self.zero_world
-            total = (total + align - 1) // align * align
+            self.sync_total = (sync_sum + align - 1) // align * align
+        else:
+            self.sync_total = sync_sum
+        total = self.sync_total + sum(self._padded(p.numel())
+                                      for p in nosync)
+        self.total = total
         device = params[0].device
         dtype = params[0].dtype
         self.flat_param = torch.empty(total, dtype=dtype, device=device)
@@ -71,7 +77,9 @@ class FlatParamModel(object):
 
         offset = 0
         self.offsets = []
-        for p in params:
+        for i, p in enumerate(params):
+            if i == self.n_sync_params:
+                offset = self.sync_total  # skip the alignment gap
             n = p.numel()
             self.flat_param[offset:offset + n].copy_(p.detach().reshape(-1))
             # re-point the parameter at the flat storage
@@ -93,11 +101,16 @@ class FlatParamModel(object):
         self._pending = []
         self._hooks = []
         if self.zero_world > 1:
-            ss = total // self.zero_world
+            ss = self.sync_total // self.zero_world
             r = dist.get_rank(self.group)
             self.zero_shard = (r * ss, (r + 1) * ss)
         else:
             self.zero_shard = (0, total)
+        # optimizer segment OUTSIDE the sharded region (owner-local
+        # expert params under ZeRO; empty otherwise — the plain path's
+        # zero_shard already spans the whole buffer)
+        self.local_seg = (self.sync_total, total) \
+            if self.zero_world > 1 and nosync else None
 
     @staticmethod
     def _padded(n):
@@ -180,14 +193,16 @@ class FlatParamModel(object):
             s_, e_ = self.zero_shard
             if dist.get_backend(self.group or dist.group.WORLD) == "nccl":
                 # each rank receives its averaged shard; (1-1/w) of the
-                # grad volume moves vs 2(1-1/w) for all-reduce
+                # grad volume moves vs 2(1-1/w) for all-reduce. Only
+                # the sync region — expert grads stay owner-local.
                 dist.reduce_scatter_tensor(
-                    self.flat_grad[s_:e_], self.flat_grad,
+                    self.flat_grad[s_:e_],
+                    self.flat_grad[:self.sync_total],
                     op=dist.ReduceOp.AVG, group=self.group)
             else:
                 # gloo has no reduce_scatter_tensor: CPU-test emulation
-                dist.all_reduce(self.flat_grad, group=self.group,
-                                op=dist.ReduceOp.AVG)
+                dist.all_reduce(self.flat_grad[:self.sync_total],
+                                group=self.group, op=dist.ReduceOp.AVG)
             if comm_timing:
                 torch.cuda.synchronize()
             get_system_monitor().gauge(
@@ -226,30 +241,61 @@ class FusedAdamW(object):
         self.weight_decay = weight_decay
         self.step_count = 0
         s_, e_ = flat_model.zero_shard
-        # ZeRO-1: fp32 state only for this rank's shard (1/world of it)
-        self.m = torch.zeros(e_ - s_, dtype=torch.float32,
+        # ZeRO-1: fp32 state for this rank's shard (1/world of the
+        # sync region) PLUS, under expert parallelism, the full
+        # owner-local nosync segment
+        n1 = e_ - s_
+        seg = flat_model.local_seg if hasattr(flat_model, "local_seg") \
+            else None
+        n2 = (seg[1] - seg[0]) if seg else 0
+        self._n1 = n1
+        self.m = torch.zeros(n1 + n2, dtype=torch.float32,
                              device=fp.device)
         self.v = torch.zeros_like(self.m)
-        self.master = fp[s_:e_].float() if master_weights else None
+        if master_weights:
+            self.master = torch.empty(n1 + n2, dtype=torch.float32,
+                                      device=fp.device)
+            self.master[:n1] = fp[s_:e_].float()
+            if seg:
+                self.master[n1:] = fp[seg[0]:seg[1]].float()
+        else:
+            self.master = None
 
     def step(self, grad_scale=1.0):
         self.step_count += 1
         s_, e_ = self.flat.zero_shard
+        n1 = self._n1
         K.adamw_step(self.flat.flat_param[s_:e_],
-                     self.flat.flat_grad[s_:e_], self.m,
-                     self.v, self.step_count, self.lr, self.beta1,
+                     self.flat.flat_grad[s_:e_], self.m[:n1],
+                     self.v[:n1], self.step_count, self.lr, self.beta1,
                      self.beta2, self.eps, self.weight_decay,
-                     master=self.master, grad_scale=grad_scale)
+                     master=self.master[:n1] if self.master is not None
+                     else None, grad_scale=grad_scale)
+        seg = getattr(self.flat, "local_seg", None)
+        if seg:
+            # owner-local expert segment: full state, local grads
+            a, b = seg
+            K.adamw_step(self.flat.flat_param[a:b],
+                         self.flat.flat_grad[a:b], self.m[n1:],
+                         self.v[n1:], self.step_count, self.lr,
+                         self.beta1, self.beta2, self.eps,
+                         self.weight_decay,
+                         master=self.master[n1:]
+                         if self.master is not None else None,
+                         grad_scale=grad_scale)
         if self.flat.zero_world > 1:
-            # publish the updated bf16 shard to every rank
+            # publish the updated bf16 SYNC shard to every rank
+            # (expert params are per-rank; nothing to gather there)
             fp = self.flat.flat_param
+            st = self.flat.sync_total
             group = self.flat.group
             if dist.get_backend(group or dist.group.WORLD) == "nccl":
-                dist.all_gather_into_tensor(fp, fp[s_:e_].contiguous(),
+                dist.all_gather_into_tensor(fp[:st],
+                                            fp[s_:e_].contiguous(),
                                             group=group)
             else:
                 world = self.flat.zero_world
-                ss = fp.numel() // world
+                ss = st // world
                 shards = [torch.empty(ss, dtype=fp.dtype,
                                       device=fp.device)
                           for _ in range(world)]

@@ -304,3 +304,75 @@ def test_zero1_matches_ddp_world2():
     diff = (results[0][:n].float()
             - flat.flat_param.float()).abs().max().item()
     assert diff < 1e-2, diff
+
+
+def _zero_ep_worker(rank, world, port, out_q):
+    import os
+
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+    })
+    import torch
+    import torch.distributed as dist
+    import torch.nn as nn
+
+    from metaflow_amd.parallel.ddp import FlatParamModel, FusedAdamW
+
+    dist.init_process_group("gloo")
+
+    class Hybrid(nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(3)               # shared weights: same
+            self.shared = nn.Linear(16, 16, bias=False,
+                                    dtype=torch.bfloat16)
+            torch.manual_seed(50 + rank)       # expert: per-rank
+            self.expert = nn.Linear(16, 16, bias=False,
+                                    dtype=torch.bfloat16)
+            self.expert.weight._mfx_no_sync = True
+
+    m = Hybrid()
+    flat = FlatParamModel(m, zero=True)
+    assert flat.zero_world == world and flat.local_seg is not None
+    opt = FusedAdamW(flat, lr=1e-2)
+    torch.manual_seed(200 + rank)
+    x = torch.randn(8, 16, dtype=torch.bfloat16)
+    for _ in range(2):
+        flat.zero_grad()
+        y = m.expert(m.shared(x))
+        y.float().square().mean().backward()
+        flat.finish_grad_sync()
+        opt.step()
+    # plain lists: tensor fd-sharing races worker exit (the zero1
+    # test's single big tensor survives it; be explicit here)
+    out_q.put((rank, m.shared.weight.detach().float().tolist(),
+               m.expert.weight.detach().float().tolist()))
+    dist.destroy_process_group()
+
+
+def test_zero1_with_expert_parallel_world2():
+    """ZeRO-1 + _mfx_no_sync: the sync region shards/averages across
+    ranks (identical shared params) while expert params keep their
+    OWN rank-local gradients and optimizer state (they stay
+    different)."""
+    import multiprocessing as mp
+
+    import torch
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_zero_ep_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(2):
+        rank, shared, expert = q.get(timeout=300)
+        res[rank] = (shared, expert)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert res[0][0] == res[1][0]        # sync: identical
+    assert res[0][1] != res[1][1]        # experts: stay rank-local
