@@ -190,6 +190,9 @@ class FlatParamModel(object):
             torch.cuda.synchronize()
         t0 = time.time()
         if self.zero and self.zero_world > 1:
+            if self.sync_total == 0:    # pure-expert module: no comm
+                self.last_comm_wait_ms = 0.0
+                return
             s_, e_ = self.zero_shard
             if dist.get_backend(self.group or dist.group.WORLD) == "nccl":
                 # each rank receives its averaged shard; (1-1/w) of the
@@ -265,12 +268,15 @@ class FusedAdamW(object):
         self.step_count += 1
         s_, e_ = self.flat.zero_shard
         n1 = self._n1
-        K.adamw_step(self.flat.flat_param[s_:e_],
-                     self.flat.flat_grad[s_:e_], self.m[:n1],
-                     self.v[:n1], self.step_count, self.lr, self.beta1,
-                     self.beta2, self.eps, self.weight_decay,
-                     master=self.master[:n1] if self.master is not None
-                     else None, grad_scale=grad_scale)
+        if n1:
+            K.adamw_step(self.flat.flat_param[s_:e_],
+                         self.flat.flat_grad[s_:e_], self.m[:n1],
+                         self.v[:n1], self.step_count, self.lr,
+                         self.beta1, self.beta2, self.eps,
+                         self.weight_decay,
+                         master=self.master[:n1]
+                         if self.master is not None else None,
+                         grad_scale=grad_scale)
         seg = getattr(self.flat, "local_seg", None)
         if seg:
             # owner-local expert segment: full state, local grads
@@ -283,7 +289,7 @@ class FusedAdamW(object):
                          master=self.master[n1:]
                          if self.master is not None else None,
                          grad_scale=grad_scale)
-        if self.flat.zero_world > 1:
+        if self.flat.zero_world > 1 and self.flat.sync_total > 0:
             # publish the updated bf16 SYNC shard to every rank
             # (expert params are per-rank; nothing to gather there)
             fp = self.flat.flat_param
