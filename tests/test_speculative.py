@@ -63,3 +63,14 @@ def test_speculative_short_budget(models):
         ref = target.generate(prompt, n)
         out, _ = speculative_generate(target, draft, prompt, n, k=4)
         assert torch.equal(out, ref), n
+
+
+def test_speculative_edge_shapes(models):
+    """Prompt of length 1, and k far exceeding the budget."""
+    target, draft = models
+    for prompt, n, k in [([7], 8, 4), ([3, 5], 4, 16), ([2] * 30, 6, 5)]:
+        pt = torch.tensor([prompt])
+        ref = target.generate(pt, n)
+        out, stats = speculative_generate(target, draft, pt, n, k=k)
+        assert torch.equal(out, ref), (prompt, n, k)
+        assert len(out[0]) == len(prompt) + n
