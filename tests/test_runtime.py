@@ -199,3 +199,51 @@ def test_foreach_stack_values(tmp_datastore, tmp_path):
         [sys.executable, str(flow), "--datastore-root", tmp_datastore,
          "run"], env=env, capture_output=True, text=True, timeout=300)
     assert proc.returncode == 0, proc.stderr[-3000:]
+
+
+def test_sched_timing_breakdown(tmp_path, tmp_datastore):
+    """MFX_SCHED_TIMING=1 prints the scheduler's per-phase serial-time
+    breakdown at run end (launch / exit / exit.logs)."""
+    import os
+    import subprocess
+    import sys
+    import textwrap
+
+    flow = tmp_path / "timing_flow.py"
+    flow.write_text(textwrap.dedent("""
+        from metaflow_amd import FlowSpec, step
+
+        class TFlow(FlowSpec):
+            @step
+            def start(self):
+                self.items = [0, 1, 2]
+                self.next(self.work, foreach="items")
+
+            @step
+            def work(self):
+                self.v = self.input
+                self.next(self.join)
+
+            @step
+            def join(self, inputs):
+                self.next(self.end)
+
+            @step
+            def end(self):
+                pass
+
+        if __name__ == "__main__":
+            TFlow()
+    """))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["MFX_NUM_GPUS"] = "0"
+    env["MFX_SCHED_TIMING"] = "1"
+    proc = subprocess.run(
+        [sys.executable, str(flow), "--quiet", "--datastore-root",
+         tmp_datastore, "run"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    out = proc.stdout
+    for phase in ("launch", "exit", "exit.logs"):
+        assert "[mfx-sched-timing] %s" % phase in out, out[-1500:]
