@@ -447,6 +447,36 @@ __global__ void quant_e4m3_kernel(const short* __restrict__ x,
   }
 }
 
+// E5M2 variant (v_cvt_pk_bf8_f32): wider exponent range for GRADIENT
+// tensors in full-fp8 training (dy spans more orders of magnitude than
+// activations; e5m2 max normal = 57344). Written at round-2 end for the
+// round-3 e5m2-dy experiment — compile-checked, unmeasured.
+__global__ void quant_e5m2_kernel(const short* __restrict__ x,
+                                  unsigned char* __restrict__ y,
+                                  float scale, long long n8) {
+  typedef __attribute__((ext_vector_type(2))) int int2v;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n8; i += stride) {
+    bf16x8 v = *(const bf16x8*)(x + i * 8);
+    float f[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float t = bf2f(v[j]) * scale;
+      f[j] = fminf(fmaxf(t, -57344.f), 57344.f);
+    }
+    int2v out;
+    int w0 = 0, w1 = 0;
+    w0 = __builtin_amdgcn_cvt_pk_bf8_f32(f[0], f[1], w0, false);
+    w0 = __builtin_amdgcn_cvt_pk_bf8_f32(f[2], f[3], w0, true);
+    w1 = __builtin_amdgcn_cvt_pk_bf8_f32(f[4], f[5], w1, false);
+    w1 = __builtin_amdgcn_cvt_pk_bf8_f32(f[6], f[7], w1, true);
+    out[0] = w0;
+    out[1] = w1;
+    *(int2v*)(y + i * 8) = out;
+  }
+}
+
 // -------------------------------------------------- token decode
 // Pretraining shards store tokens as packed uint16 (2 B/token); models
 // consume int64. Decoding ON the GPU means the PCIe/H2D copy moves 2

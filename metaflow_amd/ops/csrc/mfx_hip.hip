@@ -232,6 +232,17 @@ torch::Tensor quant_e4m3(torch::Tensor x, double scale) {
   return y;
 }
 
+torch::Tensor quant_e5m2(torch::Tensor x, double scale) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0, "quant_e5m2 needs numel % 8 == 0");
+  auto y = torch::empty_like(x, x.options().dtype(torch::kUInt8));
+  long long n8 = x.numel() / 8;
+  quant_e5m2_kernel<<<grid_for(n8), kBlock, 0, cur_stream()>>>(
+      bf(x), y.data_ptr<unsigned char>(), (float)scale, n8);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
 torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a");
   TORCH_CHECK(a.numel() % 8 == 0);
@@ -489,6 +500,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_bf16", &add_bf16, "fused bf16 add");
   m.def("decode_tokens_u16", &decode_tokens_u16,
         "packed uint16 token bytes -> int64 tokens (one pass on-GPU)");
+  m.def("quant_e5m2", &quant_e5m2,
+        "bf16 -> OCP E5M2 bytes, one fused scale+saturate+convert pass "
+        "(v_cvt_pk_bf8_f32; for delayed-scaled GRADIENT quantization)");
   m.def("quant_e4m3", &quant_e4m3,
         "one-pass bf16 -> OCP E4M3 (uint8 storage) with scale");
   m.def("adamw", &adamw, "fused AdamW (bf16 p/g, fp32 m/v[, master])");

@@ -19,6 +19,7 @@ everywhere. E4M3 max normal = 448.
 import torch
 
 E4M3_MAX = 448.0
+E5M2_MAX = 57344.0
 
 
 def _gemm_ext():
@@ -38,6 +39,22 @@ def quantize_e4m3(t, scale):
         return hip_ext().quant_e4m3(t.contiguous(), float(scale))
     x = (t.float() * scale).clamp(-E4M3_MAX, E4M3_MAX)
     return x.to(torch.float8_e4m3fn).view(torch.uint8)
+
+
+def quantize_e5m2(t, scale):
+    """bf16/fp32 -> float8_e5m2 storage viewed as uint8 (saturating).
+
+    E5M2's wider exponent range suits GRADIENT tensors (dy spans more
+    orders of magnitude than activations). GPU bf16 goes through the
+    one-pass quant_e5m2 HIP kernel (v_cvt_pk_bf8_f32) — written at
+    round-2 end, compile-checked; the kernel's first on-HW numerics
+    run is a round-3 task before switching the dy path to e5m2."""
+    if t.is_cuda and t.dtype == torch.bfloat16 and t.numel() % 8 == 0:
+        from .kernels import hip_ext
+
+        return hip_ext().quant_e5m2(t.contiguous(), float(scale))
+    x = (t.float() * scale).clamp(-E5M2_MAX, E5M2_MAX)
+    return x.to(torch.float8_e5m2).view(torch.uint8)
 
 
 class _Fp8LinearFn(torch.autograd.Function):

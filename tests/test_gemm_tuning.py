@@ -64,3 +64,25 @@ def test_shipped_table_parses():
             assert mode in ("fwd", "dx", "dw")
             assert m > 0 and k > 0 and n > 0 and idx >= 0
     gemm.reset_tune_table()
+
+
+def test_quantize_e5m2_cpu_reference():
+    """quantize_e5m2's CPU composition: saturating scale->e5m2 cast
+    matches a manual reference; values round-trip within e5m2's 2-bit
+    mantissa; saturation clamps at 57344."""
+    import torch
+
+    from metaflow_amd.ops.fp8 import E5M2_MAX, quantize_e5m2
+
+    torch.manual_seed(0)
+    x = torch.randn(64, 8, dtype=torch.bfloat16) * 1e-3
+    scale = E5M2_MAX / x.abs().max().float().item()
+    q = quantize_e5m2(x, scale)
+    assert q.dtype == torch.uint8 and q.shape == x.shape
+    back = q.view(torch.float8_e5m2).float() / scale
+    rel = (back - x.float()).abs() / x.float().abs().clamp_min(1e-12)
+    assert rel.median() < 0.15          # 2 mantissa bits ~= 12.5% ulp
+    # saturation
+    big = torch.full((8,), 1e6, dtype=torch.bfloat16)
+    qb = quantize_e5m2(big, 1.0).view(torch.float8_e5m2).float()
+    assert (qb == E5M2_MAX).all()
