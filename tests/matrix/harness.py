@@ -414,3 +414,32 @@ class ResumeFlow(MatrixTest):
             if name in skipped:
                 continue
             assert len(list(run[name])) >= 1, "no tasks for %s" % name
+
+
+class NumpyArtifactFlow(MatrixTest):
+    """A numpy artifact created at start is LOADED in every non-join
+    step (exercising the numpy-v1 zero-copy codec + provenance-skip
+    persist across every DAG shape): the deserialized view is
+    read-only, and the SAME blob sha propagates to the end
+    unchanged (no re-serialization drift through splits, foreach,
+    gangs or joins)."""
+
+    def body(self, name, kind, graph_name):
+        if name == "start":
+            return ["import numpy as np",
+                    "self.arr = np.arange(4096, dtype=np.int32)"]
+        if kind == "join":
+            return ["self.merge_artifacts(inputs)"]
+        return ["assert int(self.arr[:4].sum()) == 6",
+                "assert not self.arr.flags.writeable"]
+
+    def check(self, run, graph):
+        import numpy as np
+
+        end_task = run["end"].task
+        a = end_task.data.arr
+        assert isinstance(a, np.ndarray) and a.shape == (4096,)
+        assert int(a.sum()) == 4095 * 4096 // 2
+        start_sha = run["start"].task["arr"].sha
+        assert end_task["arr"].sha == start_sha, \
+            "numpy artifact re-serialized along the way"

@@ -7,6 +7,7 @@ from .matrix.harness import (
     ArtifactFlow,
     CurrentInfoFlow,
     ForeachContextFlow,
+    NumpyArtifactFlow,
     ResumeFlow,
     StepCounterFlow,
     run_matrix_case,
@@ -18,6 +19,7 @@ BEHAVIORS = {
     "current": CurrentInfoFlow,
     "foreach_ctx": ForeachContextFlow,
     "resume": ResumeFlow,
+    "numpy": NumpyArtifactFlow,
 }
 
 CASES = [(g, b) for g in GRAPHS for b in BEHAVIORS]
