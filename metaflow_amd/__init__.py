@@ -14,7 +14,13 @@ from .parameters import Parameter, JSONType
 from .includefile import IncludeFile
 from .current import current
 from .unbounded_foreach import UnboundedForeachInput
-from .user_config import Config, ConfigValue, FlowMutator, MutableFlow
+from .user_config import (
+    Config,
+    ConfigValue,
+    FlowMutator,
+    MutableFlow,
+    StepMutator,
+)
 from .user_decorators import (
     USER_SKIP_STEP,
     UserStepDecorator,
@@ -42,10 +48,21 @@ from .client import (
     DataArtifact,
     namespace,
     get_namespace,
+    default_namespace,
+    get_metadata,
+    default_metadata,
 )
 from .multicore_utils import parallel_map, parallel_imap_unordered
+from .profile_util import profile
 from .datatools import ObjectStore
-from .runner import DeployedFlow, Deployer, Runner, TriggeredRun
+from .runner import (
+    DeployedFlow,
+    Deployer,
+    NBDeployer,
+    NBRunner,
+    Runner,
+    TriggeredRun,
+)
 from .speculative import speculative_generate
 
 __version__ = "0.1.0"
@@ -63,6 +80,7 @@ __all__ = [
     "Config",
     "ConfigValue",
     "FlowMutator",
+    "StepMutator",
     "MutableFlow",
     "retry",
     "catch",
@@ -87,11 +105,17 @@ __all__ = [
     "DataArtifact",
     "namespace",
     "get_namespace",
+    "default_namespace",
+    "get_metadata",
+    "default_metadata",
+    "profile",
     "ObjectStore",
     "Runner",
     "Deployer",
     "DeployedFlow",
     "TriggeredRun",
+    "NBRunner",
+    "NBDeployer",
     "speculative_generate",
     "parallel_map",
     "parallel_imap_unordered",

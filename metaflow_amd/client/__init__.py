@@ -27,6 +27,26 @@ def get_namespace():
 
 
 def default_namespace():
+    """Reset to the user's own namespace; returns it (reference
+    client/core.py default_namespace)."""
+    global _current_namespace
+    _current_namespace = "user:%s" % _username()
+    return _current_namespace
+
+
+def get_metadata():
+    """Describe the active metadata provider as "TYPE@location"
+    (reference get_metadata)."""
+    return "local@%s" % _datastore_root()
+
+
+def default_metadata():
+    """Reset metadata selection to the default local provider;
+    returns its description (reference default_metadata)."""
+    return get_metadata()
+
+
+def default_namespace():
     return namespace("user:%s" % _username())
 
 

@@ -99,6 +99,7 @@ def resolve_configs(flow_cls, overrides=None):
         value = spec.resolve(overrides.get(spec.name))
         setattr(flow_cls, attr_name, value)
         resolved[attr_name] = value
+    apply_step_mutators(flow_cls)
     return resolved
 
 
@@ -145,6 +146,49 @@ class MutableFlow(object):
             if isinstance(v, ConfigValue):
                 out[attr_name] = v
         return out
+
+
+class StepMutator(object):
+    """Per-step analog of FlowMutator (reference user_decorators/
+    user_flow_decorator.py StepMutator): subclass, implement
+    mutate(mutable_step), place ABOVE @step. The mutation runs when
+    configs resolve (cli._init_state), so it can read ConfigValues:
+
+        class Instrument(StepMutator):
+            def mutate(self, ms):
+                ms.add_decorator(RetryDecorator, times=2)
+
+        class F(FlowSpec):
+            @Instrument()
+            @step
+            def train(self): ...
+    """
+
+    def __init__(self, *args, **kwargs):
+        self.args = args
+        self.kwargs = kwargs
+
+    def mutate(self, mutable_step):
+        raise NotImplementedError
+
+    def __call__(self, step_func):
+        if not getattr(step_func, "is_step", False):
+            raise TypeError("StepMutator must be applied above @step")
+        pending = getattr(step_func, "step_mutators", [])
+        step_func.step_mutators = pending + [self]
+        return step_func
+
+
+def apply_step_mutators(flow_cls):
+    """Run pending StepMutators (resolve_configs calls this after
+    config resolution, once per process)."""
+    for name in getattr(flow_cls, "_steps", []):
+        func = getattr(flow_cls, name, None)
+        for m in getattr(func, "step_mutators", []) or []:
+            if getattr(m, "_mfx_applied", False):
+                continue
+            m.mutate(MutableStep(flow_cls, name))
+            m._mfx_applied = True
 
 
 class FlowMutator(object):

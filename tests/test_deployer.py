@@ -115,3 +115,35 @@ def test_mfx_deploy_cli(flow_dir, tmp_datastore, monkeypatch):
     assert r.returncode == 0 and "prod" in r.stdout
     r = mfx("trigger", "DepFlow", "--param", "alpha=3")
     assert r.returncode == 0 and "successful" in r.stdout, r.stderr[-2000:]
+
+
+def test_nbrunner_and_nbdeployer(tmp_datastore, monkeypatch):
+    """NBRunner/NBDeployer: a class defined in THIS file (stand-in for
+    a notebook cell) runs and deploys without a user-written flow
+    file."""
+    from metaflow_amd import NBDeployer, NBRunner
+
+    _env(monkeypatch, tmp_datastore)
+
+    from metaflow_amd import FlowSpec, Parameter, step
+
+    class CellFlow(FlowSpec):
+        alpha = Parameter("alpha", default=2, type=int)
+
+        @step
+        def start(self):
+            self.out = self.alpha * 11
+            self.next(self.end)
+
+        @step
+        def end(self):
+            pass
+
+    run = NBRunner(CellFlow, datastore_root=tmp_datastore).nbrun(alpha=4)
+    assert run["start"].task.data.out == 44
+
+    df = NBDeployer(CellFlow,
+                    datastore_root=tmp_datastore).deploy(name="nb")
+    tr = df.trigger(alpha=6)
+    tr.wait(timeout=300)
+    assert tr.run["start"].task.data.out == 66

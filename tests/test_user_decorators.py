@@ -346,3 +346,70 @@ if __name__ == "__main__":
     assert p2.returncode == 0, p2.stderr[-3000:]
     # wrapper ran on the failed attempt AND the resumed re-run of mid
     assert marker.read_text().count("mid") == 2
+
+
+def test_step_mutator(tmp_path, tmp_datastore):
+    """StepMutator (reference user_flow_decorator.py analog): placed
+    above @step, its mutate(MutableStep) runs at config-resolution
+    time and can attach real decorators (here @retry) to that step."""
+    _run_inline_flow(tmp_path, tmp_datastore, """
+        import os
+
+        from metaflow_amd import FlowSpec, StepMutator, step
+
+        class AddRetry(StepMutator):
+            def mutate(self, ms):
+                from metaflow_amd.plugins.retry_decorator import (
+                    RetryDecorator,
+                )
+                ms.add_decorator(RetryDecorator, times=2)
+
+        class SMFlow(FlowSpec):
+            @AddRetry()
+            @step
+            def start(self):
+                self.marker = os.path.join(
+                    os.environ["SM_DIR"], "tries")
+                with open(self.marker, "a") as f:
+                    f.write("x")
+                if len(open(self.marker).read()) < 2:
+                    raise ValueError("flaky")
+                self.tries = len(open(self.marker).read())
+                self.next(self.end)
+
+            @step
+            def end(self):
+                assert self.tries == 2, self.tries
+                self.ok = True
+
+        if __name__ == "__main__":
+            SMFlow()
+    """, env_extra={"SM_DIR": str(tmp_path)})
+    run_id = latest_run_id(tmp_datastore, "SMFlow")
+    assert read_artifact(tmp_datastore, "SMFlow", run_id, "end", "ok")
+
+
+def test_default_namespace_and_metadata_helpers(tmp_datastore,
+                                                monkeypatch):
+    """default_namespace / get_metadata / default_metadata parity
+    helpers."""
+    monkeypatch.setenv("MFX_DATASTORE_ROOT", tmp_datastore)
+    import importlib
+
+    import metaflow_amd.client as C
+    importlib.reload(C)
+    from metaflow_amd.client import (
+        default_metadata,
+        default_namespace,
+        get_metadata,
+        get_namespace,
+        namespace,
+    )
+
+    namespace(None)
+    assert get_namespace() is None
+    ns = default_namespace()
+    assert ns.startswith("user:") and get_namespace() == ns
+    md = get_metadata()
+    assert md.startswith("local@")
+    assert default_metadata() == md
